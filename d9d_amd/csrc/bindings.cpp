@@ -1,0 +1,27 @@
+// Python bindings for the d9d_amd CDNA4 kernel extension.
+#include <torch/extension.h>
+
+// rms_norm.hip
+std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w, double eps, bool zero_centered);
+std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor x, torch::Tensor w, torch::Tensor dy, torch::Tensor inv_rms, bool zero_centered);
+
+// silu_mul.hip
+torch::Tensor silu_mul_fwd(torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor a, torch::Tensor b, torch::Tensor g);
+
+// stochastic.hip
+void copy_fp32_to_bf16_stochastic_(torch::Tensor dst, torch::Tensor src, int64_t seed);
+void adamw_stochastic_bf16_(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
+                            double lr, double beta1, double beta2, double eps, double weight_decay,
+                            int64_t step, int64_t seed);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm_fwd", &rms_norm_fwd, "RMSNorm forward (bf16, CDNA4)");
+  m.def("rms_norm_bwd", &rms_norm_bwd, "RMSNorm backward (bf16, CDNA4)");
+  m.def("silu_mul_fwd", &silu_mul_fwd, "fused silu(a)*b forward");
+  m.def("silu_mul_bwd", &silu_mul_bwd, "fused silu(a)*b backward");
+  m.def("copy_fp32_to_bf16_stochastic_", &copy_fp32_to_bf16_stochastic_,
+        "stochastic-rounding fp32->bf16 copy");
+  m.def("adamw_stochastic_bf16_", &adamw_stochastic_bf16_,
+        "fused AdamW step with SR bf16 param writes");
+}

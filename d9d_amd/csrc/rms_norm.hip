@@ -1,0 +1,225 @@
+// RMSNorm forward/backward for CDNA4 (gfx950).
+//
+// MI355X-native design (replaces the reference's Triton kernels,
+// d9d/kernel/normalization/rms/op.py): memory-bound row kernels, bf16 I/O
+// vectorized 16 B/lane, fp32 math, whole row staged in LDS so HBM traffic is
+// one read + one write. Backward uses per-block fp32 dw accumulation in LDS
+// followed by global fp32 atomics (fast on CDNA4).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace d9d {
+
+// One row per block. BLOCK threads stage the raw bf16 row in LDS, compute
+// sum(x^2) in fp32, then normalize out of LDS.
+template <int BLOCK>
+__global__ void rms_norm_fwd_kernel(
+    const ushort* __restrict__ x,  // (M, N) bf16 bits
+    const ushort* __restrict__ w,  // (N,) bf16 bits
+    ushort* __restrict__ y,        // (M, N)
+    float* __restrict__ inv_rms,   // (M,)
+    int64_t M, int64_t N, float eps, float w_offset) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  ushort* row_lds = reinterpret_cast<ushort*>(smem_raw);          // N bf16
+  float* red_lds = reinterpret_cast<float*>(row_lds + ((N + 7) & ~7ll));
+
+  const int64_t n_vec = N / 8;
+  for (int64_t row = blockIdx.x; row < M; row += gridDim.x) {
+    const ushort* xrow = x + row * N;
+    float sumsq = 0.f;
+    for (int64_t v = threadIdx.x; v < n_vec; v += BLOCK) {
+      Bf16x8 pack;
+      pack.u = *reinterpret_cast<const ushort8v*>(xrow + v * 8);
+      *reinterpret_cast<ushort8v*>(row_lds + v * 8) = pack.u;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_bits_to_f32(pack.s[j]);
+        sumsq += f * f;
+      }
+    }
+    // scalar tail
+    for (int64_t i = n_vec * 8 + threadIdx.x; i < N; i += BLOCK) {
+      ushort bits = xrow[i];
+      row_lds[i] = bits;
+      float f = bf16_bits_to_f32(bits);
+      sumsq += f * f;
+    }
+    __syncthreads();
+    sumsq = block_reduce_sum<BLOCK>(sumsq, red_lds);
+    const float inv = rsqrtf(sumsq / static_cast<float>(N) + eps);
+    if (threadIdx.x == 0) inv_rms[row] = inv;
+
+    ushort* yrow = y + row * N;
+    for (int64_t v = threadIdx.x; v < n_vec; v += BLOCK) {
+      Bf16x8 xin, win, out;
+      xin.u = *reinterpret_cast<const ushort8v*>(row_lds + v * 8);
+      win.u = *reinterpret_cast<const ushort8v*>(w + v * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = bf16_bits_to_f32(xin.s[j]);
+        float wf = bf16_bits_to_f32(win.s[j]) + w_offset;
+        out.s[j] = f32_to_bf16_rne(xf * inv * wf);
+      }
+      *reinterpret_cast<ushort8v*>(yrow + v * 8) = out.u;
+    }
+    for (int64_t i = n_vec * 8 + threadIdx.x; i < N; i += BLOCK) {
+      float xf = bf16_bits_to_f32(row_lds[i]);
+      float wf = bf16_bits_to_f32(w[i]) + w_offset;
+      yrow[i] = f32_to_bf16_rne(xf * inv * wf);
+    }
+    __syncthreads();
+  }
+}
+
+// Persistent backward: each block walks rows with stride gridDim, keeps a
+// per-block fp32 dw accumulator in LDS, and atomically adds it to the global
+// fp32 dw buffer once at the end.
+template <int BLOCK>
+__global__ void rms_norm_bwd_kernel(
+    const ushort* __restrict__ x,   // (M, N)
+    const ushort* __restrict__ w,   // (N,)
+    const ushort* __restrict__ dy,  // (M, N)
+    const float* __restrict__ inv_rms,  // (M,)
+    ushort* __restrict__ dx,        // (M, N)
+    float* __restrict__ dw,         // (N,) fp32, pre-zeroed
+    int64_t M, int64_t N, float w_offset) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* dw_lds = reinterpret_cast<float*>(smem_raw);  // N fp32
+  float* red_lds = dw_lds + N;                         // 16 fp32
+
+  for (int64_t i = threadIdx.x; i < N; i += BLOCK) dw_lds[i] = 0.f;
+  __syncthreads();
+
+  const int64_t n_vec = N / 8;
+  const float inv_n = 1.f / static_cast<float>(N);
+  for (int64_t row = blockIdx.x; row < M; row += gridDim.x) {
+    const ushort* xrow = x + row * N;
+    const ushort* grow = dy + row * N;
+    const float inv = inv_rms[row];
+
+    // Pass 1: s = sum(dy * w * x_hat); accumulate dw += dy * x_hat.
+    float s = 0.f;
+    for (int64_t v = threadIdx.x; v < n_vec; v += BLOCK) {
+      Bf16x8 xin, win, gin;
+      xin.u = *reinterpret_cast<const ushort8v*>(xrow + v * 8);
+      win.u = *reinterpret_cast<const ushort8v*>(w + v * 8);
+      gin.u = *reinterpret_cast<const ushort8v*>(grow + v * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_bits_to_f32(xin.s[j]) * inv;
+        float g = bf16_bits_to_f32(gin.s[j]);
+        float wf = bf16_bits_to_f32(win.s[j]) + w_offset;
+        s += g * wf * xh;
+        dw_lds[v * 8 + j] += g * xh;  // block-private; no atomics needed yet
+      }
+    }
+    for (int64_t i = n_vec * 8 + threadIdx.x; i < N; i += BLOCK) {
+      float xh = bf16_bits_to_f32(xrow[i]) * inv;
+      float g = bf16_bits_to_f32(grow[i]);
+      float wf = bf16_bits_to_f32(w[i]) + w_offset;
+      s += g * wf * xh;
+      dw_lds[i] += g * xh;
+    }
+    __syncthreads();
+    s = block_reduce_sum<BLOCK>(s, red_lds) * inv_n;
+
+    // Pass 2: dx = inv * (dy * w - x_hat * s).
+    ushort* dxrow = dx + row * N;
+    for (int64_t v = threadIdx.x; v < n_vec; v += BLOCK) {
+      Bf16x8 xin, win, gin, out;
+      xin.u = *reinterpret_cast<const ushort8v*>(xrow + v * 8);
+      win.u = *reinterpret_cast<const ushort8v*>(w + v * 8);
+      gin.u = *reinterpret_cast<const ushort8v*>(grow + v * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_bits_to_f32(xin.s[j]) * inv;
+        float g = bf16_bits_to_f32(gin.s[j]);
+        float wf = bf16_bits_to_f32(win.s[j]) + w_offset;
+        out.s[j] = f32_to_bf16_rne(inv * (g * wf - xh * s));
+      }
+      *reinterpret_cast<ushort8v*>(dxrow + v * 8) = out.u;
+    }
+    for (int64_t i = n_vec * 8 + threadIdx.x; i < N; i += BLOCK) {
+      float xh = bf16_bits_to_f32(xrow[i]) * inv;
+      float g = bf16_bits_to_f32(grow[i]);
+      float wf = bf16_bits_to_f32(w[i]) + w_offset;
+      dxrow[i] = f32_to_bf16_rne(inv * (g * wf - xh * s));
+    }
+    __syncthreads();
+  }
+
+  // Flush the block-private dw accumulator.
+  for (int64_t i = threadIdx.x; i < N; i += BLOCK) {
+    atomicAdd(dw + i, dw_lds[i]);
+  }
+}
+
+// NOTE on dx formula: dx_j = inv * (dy_j * w_j) - x_hat_j * inv * mean_k(dy_k w_k x_hat_k)
+//                         = inv * (dy_j w_j - x_hat_j * s)   with s = mean(dy*w*x_hat).
+
+}  // namespace d9d
+
+// ---- host wrappers ----------------------------------------------------------
+
+static void check_bf16_2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+std::vector<torch::Tensor> rms_norm_fwd(
+    torch::Tensor x, torch::Tensor w, double eps, bool zero_centered) {
+  check_bf16_2d(x, "x");
+  check_bf16_2d(w, "w");
+  const int64_t N = x.size(-1);
+  const int64_t M = x.numel() / N;
+  TORCH_CHECK(w.numel() == N, "weight size mismatch");
+
+  auto y = torch::empty_like(x);
+  auto inv_rms = torch::empty({M}, x.options().dtype(torch::kFloat32));
+
+  constexpr int kBlock = 256;
+  const int grid = static_cast<int>(std::min<int64_t>(M, 2048));
+  const size_t smem = ((N + 7) & ~7ll) * sizeof(ushort) + 16 * sizeof(float);
+  TORCH_CHECK(smem <= 160 * 1024, "rms_norm: N too large for LDS staging: ", N);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      (d9d::rms_norm_fwd_kernel<kBlock>), dim3(grid), dim3(kBlock), smem, stream,
+      reinterpret_cast<const ushort*>(x.data_ptr()),
+      reinterpret_cast<const ushort*>(w.data_ptr()),
+      reinterpret_cast<ushort*>(y.data_ptr()),
+      inv_rms.data_ptr<float>(),
+      M, N, static_cast<float>(eps), zero_centered ? 1.0f : 0.0f);
+  return {y, inv_rms};
+}
+
+std::vector<torch::Tensor> rms_norm_bwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy, torch::Tensor inv_rms,
+    bool zero_centered) {
+  check_bf16_2d(x, "x");
+  check_bf16_2d(w, "w");
+  check_bf16_2d(dy, "dy");
+  const int64_t N = x.size(-1);
+  const int64_t M = x.numel() / N;
+
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({N}, x.options().dtype(torch::kFloat32));
+
+  constexpr int kBlock = 256;
+  const int grid = static_cast<int>(std::min<int64_t>(M, 1024));
+  const size_t smem = N * sizeof(float) + 16 * sizeof(float);
+  TORCH_CHECK(smem <= 160 * 1024, "rms_norm bwd: N too large for LDS dw: ", N);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      (d9d::rms_norm_bwd_kernel<kBlock>), dim3(grid), dim3(kBlock), smem, stream,
+      reinterpret_cast<const ushort*>(x.data_ptr()),
+      reinterpret_cast<const ushort*>(w.data_ptr()),
+      reinterpret_cast<const ushort*>(dy.data_ptr()),
+      inv_rms.data_ptr<float>(),
+      reinterpret_cast<ushort*>(dx.data_ptr()),
+      dw.data_ptr<float>(),
+      M, N, zero_centered ? 1.0f : 0.0f);
+  return {dx, dw};
+}

@@ -1,0 +1,139 @@
+// Stochastic-rounding kernels: fp32->bf16 SR copy and fused AdamW step with
+// SR bf16 parameter writes. Replaces the reference's Triton kernels
+// (d9d/kernel/stochastic/copy.py, d9d/kernel/stochastic/adamw_step.py).
+//
+// RNG is counter-based (splitmix64 of seed ^ element index): stateless,
+// reproducible given the host-provided seed, no curand/rocrand state.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace d9d {
+
+template <int BLOCK>
+__global__ void copy_fp32_to_bf16_sr_kernel(
+    const float* __restrict__ src,
+    ushort* __restrict__ dst,
+    int64_t n, uint64_t seed) {
+  const int64_t n_vec = n / 4;
+  const int64_t tid = blockIdx.x * static_cast<int64_t>(BLOCK) + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
+  for (int64_t v = tid; v < n_vec; v += stride) {
+    const float4v in = *reinterpret_cast<const float4v*>(src + v * 4);
+    const uint64_t r = splitmix64(seed ^ static_cast<uint64_t>(v));
+    ushort4v out;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      out[j] = f32_to_bf16_stochastic(in[j], static_cast<uint32_t>(r >> (16 * j)));
+    }
+    *reinterpret_cast<ushort4v*>(dst + v * 4) = out;
+  }
+  for (int64_t i = n_vec * 4 + tid; i < n; i += stride) {
+    const uint64_t r = splitmix64(seed ^ (0x8000000000000000ull | static_cast<uint64_t>(i)));
+    dst[i] = f32_to_bf16_stochastic(src[i], static_cast<uint32_t>(r));
+  }
+}
+
+// Fused AdamW: bf16 params+grads, fp32 moments, fp32 math, SR bf16 writes.
+template <int BLOCK>
+__global__ void adamw_sr_bf16_kernel(
+    ushort* __restrict__ p,         // (n,) bf16 param
+    const ushort* __restrict__ g,   // (n,) bf16 grad
+    float* __restrict__ m,          // (n,) fp32 exp_avg
+    float* __restrict__ v,          // (n,) fp32 exp_avg_sq
+    int64_t n,
+    float lr, float beta1, float beta2, float eps, float weight_decay,
+    float bias_corr1, float bias_corr2,  // 1 - beta^t, precomputed on host
+    uint64_t seed) {
+  const int64_t n_vec = n / 4;
+  const int64_t tid = blockIdx.x * static_cast<int64_t>(BLOCK) + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
+  const float decay_mul = 1.f - lr * weight_decay;
+  const float inv_bc1 = 1.f / bias_corr1;
+  const float inv_sqrt_bc2 = rsqrtf(bias_corr2);
+
+  for (int64_t vi = tid; vi < n_vec; vi += stride) {
+    ushort4v pv = *reinterpret_cast<const ushort4v*>(p + vi * 4);
+    const ushort4v gv = *reinterpret_cast<const ushort4v*>(g + vi * 4);
+    float4v mv = *reinterpret_cast<const float4v*>(m + vi * 4);
+    float4v vv = *reinterpret_cast<const float4v*>(v + vi * 4);
+    const uint64_t r = splitmix64(seed ^ static_cast<uint64_t>(vi));
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float pf = bf16_bits_to_f32(pv[j]) * decay_mul;
+      const float gf = bf16_bits_to_f32(gv[j]);
+      float mf = beta1 * mv[j] + (1.f - beta1) * gf;
+      float vf = beta2 * vv[j] + (1.f - beta2) * gf * gf;
+      const float m_hat = mf * inv_bc1;
+      const float denom = sqrtf(vf) * inv_sqrt_bc2 + eps;
+      pf -= lr * m_hat / denom;
+      mv[j] = mf;
+      vv[j] = vf;
+      pv[j] = f32_to_bf16_stochastic(pf, static_cast<uint32_t>(r >> (16 * j)));
+    }
+    *reinterpret_cast<ushort4v*>(p + vi * 4) = pv;
+    *reinterpret_cast<float4v*>(m + vi * 4) = mv;
+    *reinterpret_cast<float4v*>(v + vi * 4) = vv;
+  }
+  for (int64_t i = n_vec * 4 + tid; i < n; i += stride) {
+    float pf = bf16_bits_to_f32(p[i]) * decay_mul;
+    const float gf = bf16_bits_to_f32(g[i]);
+    float mf = beta1 * m[i] + (1.f - beta1) * gf;
+    float vf = beta2 * v[i] + (1.f - beta2) * gf * gf;
+    pf -= lr * (mf * inv_bc1) / (sqrtf(vf) * inv_sqrt_bc2 + eps);
+    m[i] = mf;
+    v[i] = vf;
+    const uint64_t r = splitmix64(seed ^ (0x8000000000000000ull | static_cast<uint64_t>(i)));
+    p[i] = f32_to_bf16_stochastic(pf, static_cast<uint32_t>(r));
+  }
+}
+
+}  // namespace d9d
+
+void copy_fp32_to_bf16_stochastic_(
+    torch::Tensor dst, torch::Tensor src, int64_t seed) {
+  TORCH_CHECK(src.is_cuda() && src.scalar_type() == torch::kFloat32 && src.is_contiguous());
+  TORCH_CHECK(dst.scalar_type() == torch::kBFloat16 && dst.is_contiguous());
+  TORCH_CHECK(dst.numel() == src.numel());
+  const int64_t n = src.numel();
+  if (n == 0) return;
+  constexpr int kBlock = 256;
+  const int grid = static_cast<int>(
+      std::min<int64_t>(d9d::ceil_div(n / 4 + 1, kBlock), 2048));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      (d9d::copy_fp32_to_bf16_sr_kernel<kBlock>), dim3(grid), dim3(kBlock), 0,
+      stream, src.data_ptr<float>(),
+      reinterpret_cast<ushort*>(dst.data_ptr()), n,
+      static_cast<uint64_t>(seed));
+}
+
+void adamw_stochastic_bf16_(
+    torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
+    double lr, double beta1, double beta2, double eps, double weight_decay,
+    int64_t step, int64_t seed) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kBFloat16 && p.is_contiguous());
+  TORCH_CHECK(g.scalar_type() == torch::kBFloat16 && g.is_contiguous());
+  TORCH_CHECK(m.scalar_type() == torch::kFloat32 && m.is_contiguous());
+  TORCH_CHECK(v.scalar_type() == torch::kFloat32 && v.is_contiguous());
+  const int64_t n = p.numel();
+  TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n);
+  if (n == 0) return;
+
+  const float bc1 = 1.f - powf(static_cast<float>(beta1), static_cast<float>(step));
+  const float bc2 = 1.f - powf(static_cast<float>(beta2), static_cast<float>(step));
+
+  constexpr int kBlock = 256;
+  const int grid = static_cast<int>(
+      std::min<int64_t>(d9d::ceil_div(n / 4 + 1, kBlock), 2048));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      (d9d::adamw_sr_bf16_kernel<kBlock>), dim3(grid), dim3(kBlock), 0, stream,
+      reinterpret_cast<ushort*>(p.data_ptr()),
+      reinterpret_cast<const ushort*>(g.data_ptr()),
+      m.data_ptr<float>(), v.data_ptr<float>(), n,
+      static_cast<float>(lr), static_cast<float>(beta1),
+      static_cast<float>(beta2), static_cast<float>(eps),
+      static_cast<float>(weight_decay), bc1, bc2, static_cast<uint64_t>(seed));
+}
