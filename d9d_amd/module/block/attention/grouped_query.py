@@ -1,0 +1,110 @@
+"""Grouped-query attention block (reference: d9d/module/block/attention/grouped_query.py).
+
+q/k/v/o projections, optional QK-RMSNorm (optionally zero-centered), partial
+RoPE (`rope_dim` <= head_dim), optional sigmoid output gate (Qwen3.5 style),
+SDPA = the framework's CDNA4 flash-attention op.
+"""
+
+import math
+
+import torch
+from torch import nn
+
+from ....ops import flash_attn_func
+from ..normalization import RMSNorm
+from ..positional import RopeLayout, apply_rotary_emb
+
+
+class GroupedQueryAttention(nn.Module):
+    def __init__(
+        self,
+        hidden_size: int,
+        num_attention_heads: int,
+        num_key_value_heads: int,
+        head_dim: int,
+        rope_dim: int | None = None,
+        rope_layout: RopeLayout = RopeLayout.HALF,
+        use_qk_norm: bool = True,
+        qk_norm_zero_centered: bool = False,
+        rms_norm_eps: float = 1e-6,
+        use_output_gate: bool = False,
+        sliding_window: int | None = None,
+        use_sinks: bool = False,
+        device=None,
+        dtype=None,
+    ) -> None:
+        super().__init__()
+        assert num_attention_heads % num_key_value_heads == 0
+        self.num_heads = num_attention_heads
+        self.num_kv_heads = num_key_value_heads
+        self.head_dim = head_dim
+        self.rope_dim = rope_dim if rope_dim is not None else head_dim
+        self.rope_layout = rope_layout
+        self.sliding_window = sliding_window
+
+        kw = {"device": device, "dtype": dtype, "bias": False}
+        q_out = num_attention_heads * head_dim
+        self.q_proj = nn.Linear(hidden_size, q_out * (2 if use_output_gate else 1), **kw)
+        self.k_proj = nn.Linear(hidden_size, num_key_value_heads * head_dim, **kw)
+        self.v_proj = nn.Linear(hidden_size, num_key_value_heads * head_dim, **kw)
+        self.o_proj = nn.Linear(q_out, hidden_size, **kw)
+        self.use_output_gate = use_output_gate
+
+        if use_qk_norm:
+            self.q_norm = RMSNorm(head_dim, eps=rms_norm_eps,
+                                  zero_centered=qk_norm_zero_centered,
+                                  device=device, dtype=dtype)
+            self.k_norm = RMSNorm(head_dim, eps=rms_norm_eps,
+                                  zero_centered=qk_norm_zero_centered,
+                                  device=device, dtype=dtype)
+        else:
+            self.q_norm = None
+            self.k_norm = None
+
+        if use_sinks:
+            self.sinks = nn.Parameter(
+                torch.empty(num_attention_heads, device=device, dtype=torch.float32)
+            )
+        else:
+            self.sinks = None
+
+    def reset_parameters(self) -> None:
+        for lin in (self.q_proj, self.k_proj, self.v_proj, self.o_proj):
+            nn.init.normal_(lin.weight, mean=0.0, std=0.02 / math.sqrt(2))
+        if self.q_norm is not None:
+            self.q_norm.reset_parameters()
+            self.k_norm.reset_parameters()
+        if self.sinks is not None:
+            nn.init.zeros_(self.sinks)
+
+    def forward(
+        self,
+        hidden_states: torch.Tensor,  # (B, S, H)
+        rotary_cos_sin: tuple[torch.Tensor, torch.Tensor],
+    ) -> torch.Tensor:
+        B, S, _ = hidden_states.shape
+        q = self.q_proj(hidden_states)
+        if self.use_output_gate:
+            q, gate = q.chunk(2, dim=-1)
+        k = self.k_proj(hidden_states).view(B, S, self.num_kv_heads, self.head_dim)
+        v = self.v_proj(hidden_states).view(B, S, self.num_kv_heads, self.head_dim)
+        q = q.view(B, S, self.num_heads, self.head_dim)
+
+        if self.q_norm is not None:
+            q = self.q_norm(q)
+            k = self.k_norm(k)
+
+        cos, sin = rotary_cos_sin
+        cos = cos[..., : self.rope_dim]
+        sin = sin[..., : self.rope_dim]
+        q = apply_rotary_emb(q, cos, sin, self.rope_layout)
+        k = apply_rotary_emb(k, cos, sin, self.rope_layout)
+
+        window = (-1, -1) if self.sliding_window is None else (self.sliding_window, -1)
+        attn = flash_attn_func(
+            q, k, v, causal=True, window_size=window, sinks=self.sinks
+        )
+        attn = attn.reshape(B, S, self.num_heads * self.head_dim)
+        if self.use_output_gate:
+            attn = attn * torch.sigmoid(gate)
+        return self.o_proj(attn)

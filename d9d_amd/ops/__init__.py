@@ -1,10 +1,21 @@
 from .rms_norm import rms_norm
 from .swiglu import silu_mul
 from .stochastic import copy_fp32_to_bf16_stochastic_, adamw_stochastic_bf16_
+from .attention import flash_attn_func
+from .cce import linear_cross_entropy, LM_IGNORE_INDEX, VocabParallelOptions
+from .gmm import gmm
+from .moe_permute import moe_permute, moe_unpermute
 
 __all__ = [
     "rms_norm",
     "silu_mul",
     "copy_fp32_to_bf16_stochastic_",
     "adamw_stochastic_bf16_",
+    "flash_attn_func",
+    "linear_cross_entropy",
+    "LM_IGNORE_INDEX",
+    "VocabParallelOptions",
+    "gmm",
+    "moe_permute",
+    "moe_unpermute",
 ]

@@ -1,0 +1,118 @@
+"""Flash attention op — the framework's single SDPA implementation.
+
+GPU: hand-written CDNA4 flash-attention kernel (csrc/attention.hip) with
+online softmax, causal masking, GQA head packing, optional sliding window,
+optional learnable attention sinks, and LSE output (needed for context-
+parallel ring merging). CPU: eager fp32 oracle with identical semantics
+(reference wrapper: d9d/kernel/flash_attn/function.py).
+
+Layout: q (B, S, Hq, D), k/v (B, S, Hkv, D), out (B, S, Hq, D),
+lse (B, Hq, S) natural-log-sum-exp of the scores row.
+"""
+
+import math
+
+import torch
+
+from ._ext import get_ext
+
+
+def _eager_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool,
+    softmax_scale: float,
+    window_size: tuple[int, int],
+    sinks: torch.Tensor | None,
+):
+    """fp32 eager oracle. Returns (out, lse)."""
+    B, S, Hq, D = q.shape
+    Hkv = k.shape[2]
+    rep = Hq // Hkv
+
+    q32 = q.float().permute(0, 2, 1, 3)  # (B, Hq, S, D)
+    k32 = k.float().permute(0, 2, 1, 3)
+    v32 = v.float().permute(0, 2, 1, 3)
+    if rep > 1:
+        k32 = k32.repeat_interleave(rep, dim=1)
+        v32 = v32.repeat_interleave(rep, dim=1)
+
+    scores = torch.matmul(q32, k32.transpose(-1, -2)) * softmax_scale  # (B,Hq,S,Skv)
+    Skv = scores.shape[-1]
+    q_pos = torch.arange(S, device=q.device).unsqueeze(1)
+    kv_pos = torch.arange(Skv, device=q.device).unsqueeze(0)
+    mask = torch.zeros(S, Skv, dtype=torch.bool, device=q.device)
+    if causal:
+        mask |= kv_pos > q_pos
+    left, right = window_size
+    if left >= 0:
+        mask |= kv_pos < q_pos - left
+    if right >= 0 and not causal:
+        mask |= kv_pos > q_pos + right
+    scores = scores.masked_fill(mask, float("-inf"))
+
+    if sinks is not None:
+        # One virtual "sink" column per head: its logit joins the softmax
+        # denominator but contributes no value (reference: flash4 sinks).
+        sink_col = sinks.float().view(1, Hq, 1, 1).expand(B, Hq, S, 1)
+        full = torch.cat([scores, sink_col], dim=-1)
+        lse = torch.logsumexp(full, dim=-1)  # (B,Hq,S)
+        p = torch.exp(scores - lse.unsqueeze(-1))
+    else:
+        lse = torch.logsumexp(scores, dim=-1)
+        p = torch.exp(scores - lse.unsqueeze(-1))
+
+    out = torch.matmul(p, v32)  # (B,Hq,S,D)
+    return out.permute(0, 2, 1, 3).to(q.dtype), lse
+
+
+class _FlashAttnFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, sinks, causal, softmax_scale, window_left):
+        ext = get_ext()
+        out, lse = ext.flash_attn_fwd(
+            q.contiguous(), k.contiguous(), v.contiguous(),
+            causal, softmax_scale, window_left,
+        )
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.causal = causal
+        ctx.softmax_scale = softmax_scale
+        ctx.window_left = window_left
+        ctx.has_sinks = sinks is not None
+        if ctx.has_sinks:
+            raise NotImplementedError("sinks on the HIP path land with the flash kernel v2")
+        return out, lse
+
+    @staticmethod
+    def backward(ctx, dout, dlse):
+        q, k, v, out, lse = ctx.saved_tensors
+        ext = get_ext()
+        dq, dk, dv = ext.flash_attn_bwd(
+            dout.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(),
+            out.contiguous(), lse, ctx.causal, ctx.softmax_scale, ctx.window_left,
+        )
+        return dq, dk, dv, None, None, None, None
+
+
+def flash_attn_func(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    *,
+    causal: bool = True,
+    softmax_scale: float | None = None,
+    window_size: tuple[int, int] = (-1, -1),
+    sinks: torch.Tensor | None = None,
+    return_lse: bool = False,
+):
+    """Scaled-dot-product attention. q (B,S,Hq,D); k/v (B,S,Hkv,D)."""
+    if softmax_scale is None:
+        softmax_scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda and sinks is None and window_size == (-1, -1):
+        out, lse = _FlashAttnFunction.apply(q, k, v, None, causal, softmax_scale, -1)
+    else:
+        out, lse = _eager_attention(q, k, v, causal, softmax_scale, window_size, sinks)
+    if return_lse:
+        return out, lse
+    return out

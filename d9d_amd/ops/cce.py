@@ -1,0 +1,116 @@
+"""Fused linear + cross-entropy (CCE) — never materializes the (T, V) logits.
+
+Replaces the reference's vendored cut-cross-entropy
+(d9d/kernel/cce/cce.py:47-216): per-token loss = lse(e @ c^T) - logit[target],
+with the logits recomputed in vocab/row chunks in backward. Matmuls run on
+MFMA via rocBLAS (library GEMM); a fully fused HIP kernel is the follow-up.
+
+Supports vocab-parallel reduction (VocabParallelOptions analog): pass a
+process group and each rank's local vocab shard; lse is all-reduced via
+logsumexp-merge and the target logit summed (exactly one rank owns a target).
+"""
+
+from dataclasses import dataclass
+
+import torch
+import torch.distributed as dist
+
+LM_IGNORE_INDEX = -100
+_ROW_CHUNK = 2048
+
+
+@dataclass
+class VocabParallelOptions:
+    group: object  # ProcessGroup
+    vocab_start: int
+    vocab_end: int
+
+
+def _chunk_fwd(e32, c, targets, vocab_start):
+    """Return (lse, target_logit) for a row chunk; target_logit=0 for ignored/out-of-shard."""
+    logits = torch.matmul(e32, c.t().to(e32.dtype))  # (Tc, V_local)
+    lse = torch.logsumexp(logits.float(), dim=-1)
+    local_targets = targets - vocab_start
+    in_shard = (local_targets >= 0) & (local_targets < logits.shape[-1]) & (
+        targets != LM_IGNORE_INDEX
+    )
+    safe = local_targets.clamp(0, logits.shape[-1] - 1)
+    tgt_logit = logits.float().gather(1, safe.unsqueeze(1)).squeeze(1)
+    tgt_logit = torch.where(in_shard, tgt_logit, torch.zeros_like(tgt_logit))
+    return lse, tgt_logit
+
+
+class _LinearCrossEntropyFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, e, c, targets, vp_group, vocab_start, vocab_end):
+        T = e.shape[0]
+        lse_parts = []
+        tgt_parts = []
+        for s in range(0, T, _ROW_CHUNK):
+            sl = slice(s, min(s + _ROW_CHUNK, T))
+            lse_c, tgt_c = _chunk_fwd(e[sl], c, targets[sl], vocab_start)
+            lse_parts.append(lse_c)
+            tgt_parts.append(tgt_c)
+        lse = torch.cat(lse_parts)
+        tgt_logit = torch.cat(tgt_parts)
+
+        if vp_group is not None:
+            # Merge lse across vocab shards: lse_full = log sum_r exp(lse_r).
+            world = dist.get_world_size(vp_group)
+            all_lse = torch.empty((world,) + lse.shape, dtype=lse.dtype, device=lse.device)
+            dist.all_gather_into_tensor(all_lse, lse, group=vp_group)
+            lse = torch.logsumexp(all_lse, dim=0)
+            dist.all_reduce(tgt_logit, group=vp_group)
+
+        loss = lse - tgt_logit  # = -log p(target)
+        ignored = targets == LM_IGNORE_INDEX
+        loss = torch.where(ignored, torch.zeros_like(loss), loss)
+
+        ctx.save_for_backward(e, c, targets, lse)
+        ctx.vp = (vp_group, vocab_start, vocab_end)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        e, c, targets, lse = ctx.saved_tensors
+        vp_group, vocab_start, _ = ctx.vp
+        T, H = e.shape
+        V = c.shape[0]
+        de = torch.empty_like(e)
+        dc = torch.zeros(c.shape, dtype=torch.float32, device=c.device)
+        ignored = targets == LM_IGNORE_INDEX
+        dl = torch.where(ignored, torch.zeros_like(dloss), dloss).float()
+
+        for s in range(0, T, _ROW_CHUNK):
+            sl = slice(s, min(s + _ROW_CHUNK, T))
+            e32 = e[sl]
+            logits = torch.matmul(e32, c.t().to(e32.dtype)).float()  # (Tc, V)
+            p = torch.exp(logits - lse[sl].unsqueeze(1))
+            local_targets = targets[sl] - vocab_start
+            in_shard = (local_targets >= 0) & (local_targets < V) & (~ignored[sl])
+            safe = local_targets.clamp(0, V - 1)
+            p.scatter_add_(
+                1, safe.unsqueeze(1),
+                torch.where(in_shard, -torch.ones_like(safe, dtype=p.dtype), torch.zeros_like(safe, dtype=p.dtype)).unsqueeze(1),
+            )
+            p *= dl[sl].unsqueeze(1)  # (Tc, V) = d logits
+            pb = p.to(c.dtype)
+            de[sl] = torch.matmul(pb, c).to(e.dtype)
+            dc += torch.matmul(pb.t(), e32.to(c.dtype)).float()
+
+        return de, dc.to(c.dtype), None, None, None, None
+
+
+def linear_cross_entropy(
+    embeddings: torch.Tensor,  # (T, H)
+    classifier: torch.Tensor,  # (V_local, H)
+    targets: torch.Tensor,  # (T,) int64; LM_IGNORE_INDEX skipped
+    vocab_parallel: VocabParallelOptions | None = None,
+) -> torch.Tensor:
+    """Per-token negative log-likelihood, zeros at ignored positions."""
+    if vocab_parallel is None:
+        return _LinearCrossEntropyFunction.apply(embeddings, classifier, targets, None, 0, classifier.shape[0])
+    return _LinearCrossEntropyFunction.apply(
+        embeddings, classifier, targets,
+        vocab_parallel.group, vocab_parallel.vocab_start, vocab_parallel.vocab_end,
+    )

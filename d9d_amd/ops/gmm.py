@@ -1,0 +1,71 @@
+"""Grouped GEMM for MoE experts (reference: d9d/kernel/gmm/function.py).
+
+out[rows_e] = a[rows_e] @ b[e]  for each expert e, rows partitioned by
+`batch_sizes` (CPU int64, per-expert row counts, in order).
+
+GPU: hand-written CDNA4 MFMA grouped-GEMM kernel (csrc/gmm.hip) with
+tile-list scheduling over ragged group sizes. Until the kernel lands the
+GPU path falls back to per-expert rocBLAS GEMMs (library GEMM).
+Backward is direction-gated via GLOBAL_GRAD_CONTEXT (da = g @ b^T grouped;
+db[e] = a[rows_e]^T @ g[rows_e]).
+"""
+
+import torch
+
+from ..core.autograd import GLOBAL_GRAD_CONTEXT, GradDirection
+from ._ext import get_ext, has_ext
+
+
+def _gmm_loop(a: torch.Tensor, b: torch.Tensor, batch_sizes: torch.Tensor) -> torch.Tensor:
+    out = a.new_empty((a.shape[0], b.shape[2]))
+    start = 0
+    for e, n in enumerate(batch_sizes.tolist()):
+        if n:
+            out[start : start + n] = a[start : start + n] @ b[e]
+        start += n
+    return out
+
+
+def _gmm_accum_db(a: torch.Tensor, g: torch.Tensor, batch_sizes: torch.Tensor, E: int) -> torch.Tensor:
+    db = g.new_empty((E, a.shape[1], g.shape[1]))
+    start = 0
+    for e, n in enumerate(batch_sizes.tolist()):
+        if n:
+            db[e] = a[start : start + n].t() @ g[start : start + n]
+        else:
+            db[e].zero_()
+        start += n
+    return db
+
+
+def _gmm_forward_impl(a, b, batch_sizes):
+    if a.is_cuda and has_ext() and hasattr(get_ext(), "gmm"):
+        return get_ext().gmm(a.contiguous(), b.contiguous(), batch_sizes)
+    return _gmm_loop(a, b, batch_sizes)
+
+
+class _GroupedGemmFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b, batch_sizes):
+        ctx.save_for_backward(a, b, batch_sizes)
+        return _gmm_forward_impl(a, b, batch_sizes)
+
+    @staticmethod
+    def backward(ctx, g):
+        a, b, batch_sizes, = ctx.saved_tensors
+        g = g.contiguous()
+        da = db = None
+        if ctx.needs_input_grad[0] and GLOBAL_GRAD_CONTEXT.computes(GradDirection.INPUTS):
+            da = _gmm_forward_impl(g, b.transpose(1, 2).contiguous(), batch_sizes)
+        if ctx.needs_input_grad[1] and GLOBAL_GRAD_CONTEXT.computes(GradDirection.WEIGHTS):
+            if a.is_cuda and has_ext() and hasattr(get_ext(), "gmm_db"):
+                db = get_ext().gmm_db(a.contiguous(), g, batch_sizes, b.shape[0])
+            else:
+                db = _gmm_accum_db(a, g, batch_sizes, b.shape[0])
+        return da, db, None
+
+
+def gmm(a: torch.Tensor, b: torch.Tensor, batch_sizes: torch.Tensor) -> torch.Tensor:
+    """a (sum_T, K); b (E, K, N); batch_sizes (E,) int64 on CPU."""
+    assert batch_sizes.device.type == "cpu"
+    return _GroupedGemmFunction.apply(a, b, batch_sizes)

@@ -1,0 +1,160 @@
+"""Kernel numerics: HIP kernels vs plain-torch fp32 references.
+
+CPU tests exercise the autograd wrappers' reference paths; @gpu tests compare
+the CDNA4 kernels against the same fp32 references on an MI355X.
+"""
+
+import pytest
+import torch
+
+from d9d_amd.ops import (
+    adamw_stochastic_bf16_,
+    copy_fp32_to_bf16_stochastic_,
+    rms_norm,
+    silu_mul,
+)
+
+
+def _rms_ref(x32, w32, eps, zero_centered):
+    inv = torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + eps)
+    return x32 * inv * (w32 + (1.0 if zero_centered else 0.0))
+
+
+# ---- CPU wrapper correctness (vs torch autograd in fp32) ---------------------
+
+
+@pytest.mark.parametrize("zero_centered", [False, True])
+def test_rms_norm_cpu_matches_autograd(zero_centered):
+    x = torch.randn(6, 96, dtype=torch.float32, requires_grad=True)
+    w = torch.randn(96, dtype=torch.float32, requires_grad=True)
+    y = rms_norm(x, w, eps=1e-6, zero_centered=zero_centered)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    y_ref = _rms_ref(x2, w2, 1e-6, zero_centered)
+    y_ref.backward(g)
+
+    torch.testing.assert_close(y, y_ref, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(w.grad, w2.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_silu_mul_cpu_matches_autograd():
+    a = torch.randn(5, 33, requires_grad=True)
+    b = torch.randn(5, 33, requires_grad=True)
+    out = silu_mul(a, b)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    a2 = a.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.silu(a2) * b2
+    ref.backward(g)
+
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(a.grad, a2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(b.grad, b2.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_sr_copy_cpu_expectation():
+    src = torch.full((20000,), 1.0 + 1 / 512, dtype=torch.float32)  # between bf16 grid points
+    dst = torch.empty_like(src, dtype=torch.bfloat16)
+    copy_fp32_to_bf16_stochastic_(dst, src, seed=7)
+    mean = dst.float().mean().item()
+    assert abs(mean - (1.0 + 1 / 512)) < 2e-3  # SR is unbiased in expectation
+    assert dst.float().unique().numel() == 2  # rounds to the two neighbours
+
+
+# ---- GPU kernel parity -------------------------------------------------------
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [(128, 768), (64, 1024), (33, 7168), (17, 100)])
+@pytest.mark.parametrize("zero_centered", [False, True])
+def test_rms_norm_gpu_parity(shape, zero_centered):
+    device = torch.device("cuda")
+    x = torch.randn(shape, dtype=torch.bfloat16, device=device, requires_grad=True)
+    w = torch.randn(shape[-1], dtype=torch.bfloat16, device=device, requires_grad=True)
+    y = rms_norm(x, w, eps=1e-6, zero_centered=zero_centered)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    x32 = x.detach().float().requires_grad_(True)
+    w32 = w.detach().float().requires_grad_(True)
+    y_ref = _rms_ref(x32, w32, 1e-6, zero_centered)
+    y_ref.backward(g.float())
+
+    torch.testing.assert_close(y.float(), y_ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(x.grad.float(), x32.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(
+        w.grad.float(), w32.grad, rtol=3e-2, atol=3e-1
+    )  # dw sums over M rows in bf16 inputs
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("n", [4096 * 576, 1000003])
+def test_silu_mul_gpu_parity(n):
+    device = torch.device("cuda")
+    a = torch.randn(n, dtype=torch.bfloat16, device=device, requires_grad=True)
+    b = torch.randn(n, dtype=torch.bfloat16, device=device, requires_grad=True)
+    out = silu_mul(a, b)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    a32 = a.detach().float().requires_grad_(True)
+    b32 = b.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.silu(a32) * b32
+    ref.backward(g.float())
+
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(a.grad.float(), a32.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(b.grad.float(), b32.grad, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.gpu
+def test_sr_copy_gpu_unbiased():
+    device = torch.device("cuda")
+    val = 1.0 + 1 / 512
+    src = torch.full((1 << 20,), val, dtype=torch.float32, device=device)
+    dst = torch.empty_like(src, dtype=torch.bfloat16)
+    copy_fp32_to_bf16_stochastic_(dst, src, seed=123)
+    mean = dst.float().mean().item()
+    assert abs(mean - val) < 5e-4
+    assert dst.float().unique().numel() == 2
+    # different seeds give different rounding patterns
+    dst2 = torch.empty_like(dst)
+    copy_fp32_to_bf16_stochastic_(dst2, src, seed=124)
+    assert not torch.equal(dst, dst2)
+
+
+@pytest.mark.gpu
+def test_adamw_gpu_matches_fp32_reference():
+    device = torch.device("cuda")
+    n = 4097
+    p32 = torch.randn(n, dtype=torch.float32, device=device)
+    p = p32.to(torch.bfloat16)
+    p32 = p.float()  # start exactly at the bf16 values
+    g = torch.randn(n, dtype=torch.bfloat16, device=device)
+    m = torch.randn(n, dtype=torch.float32, device=device).abs() * 0.1
+    v = torch.randn(n, dtype=torch.float32, device=device).abs() * 0.01
+    m_ref, v_ref = m.clone(), v.clone()
+
+    lr, b1, b2, eps, wd, step = 1e-2, 0.9, 0.95, 1e-8, 0.1, 3
+    adamw_stochastic_bf16_(
+        p, g, m, v, lr=lr, beta1=b1, beta2=b2, eps=eps, weight_decay=wd,
+        step=step, seed=42,
+    )
+
+    g32 = g.float()
+    p_ref = p32 * (1 - lr * wd)
+    m_ref.mul_(b1).add_(g32, alpha=1 - b1)
+    v_ref.mul_(b2).addcmul_(g32, g32, value=1 - b2)
+    denom = (v_ref / (1 - b2**step)).sqrt().add_(eps)
+    p_ref.addcdiv_(m_ref / (1 - b1**step), denom, value=-lr)
+
+    torch.testing.assert_close(m, m_ref, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(v, v_ref, rtol=1e-5, atol=1e-7)
+    # SR write is within one ulp of the fp32 result
+    torch.testing.assert_close(p.float(), p_ref, rtol=1e-2, atol=1e-2)
