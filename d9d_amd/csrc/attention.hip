@@ -1,0 +1,789 @@
+// CDNA4 (gfx950) flash attention: forward + backward, bf16, causal, GQA, LSE.
+//
+// Replaces the reference's external flash-attn wheels
+// (d9d/kernel/flash_attn/function.py). MI355X-first design:
+//   * MFMA v_mfma_f32_16x16x32_bf16 tiles, wave64.
+//   * Forward: workgroup = 4 waves x 16 q-rows (QBLK 64), KV tiles of 64
+//     staged in LDS — K row-major with a ((row&15)<<4) XOR swizzle so the
+//     B-fragment ds_read_b128s are bank-conflict-free; V staged transposed
+//     for contiguous PV B-fragments. Online softmax in fp32 with running
+//     (m, l) per row; LSE written for CP ring merging.
+//   * Backward: FA2-style; workgroup owns a KV tile, loops q tiles;
+//     dK/dV accumulate in AGPRs; dQ via fp32 global atomics (fast on CDNA4);
+//     delta = rowsum(dO*O) precomputed by a small kernel.
+//
+// Layouts: q (B,Sq,Hq,D), k/v (B,Skv,Hkv,D), out (B,Sq,Hq,D), lse (B,Hq,Sq).
+// D (head_dim) templated in {32, 64, 96, 128}; host pads other sizes.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace d9d {
+
+typedef __bf16 bf16_t;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+D9D_DEVICE f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// Fragment maps for v_mfma_f32_16x16x32_bf16 (M=N=16, K=32):
+//   A[m][k]: lane l holds m = l&15, k = (l>>4)*8 + j   (j = 0..7)
+//   B[k][n]: lane l holds n = l&15, k = (l>>4)*8 + j
+//   C[m][n]: lane l, reg r: m = (l>>4)*4 + r, n = l&15
+// (verified on-device by mfma_selfcheck below).
+
+constexpr int kQBlk = 64;   // q rows per workgroup (16 per wave)
+constexpr int kKvBlk = 64;  // kv rows per tile
+constexpr float kLog2e = 1.44269504088896340736f;
+
+// XOR swizzles keep ds_read_b128 bank-conflict-free (guide G4). The XOR must
+// stay inside the LDS row: row-major [64][D] tiles (D*2-byte rows) use mask
+// kSwzRM(D); transposed [D][64] tiles (128-byte rows) use mask 7.
+template <int D>
+constexpr int swz_rm_mask() {
+  return D == 128 ? 15 : (D >= 64 ? 7 : 3);
+}
+
+// ---------------------------------------------------------------------------
+// Forward
+// ---------------------------------------------------------------------------
+
+template <int D>
+__global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
+    const bf16_t* __restrict__ q,   // (B,Sq,Hq,D)
+    const bf16_t* __restrict__ k,   // (B,Skv,Hkv,D)
+    const bf16_t* __restrict__ v,   // (B,Skv,Hkv,D)
+    bf16_t* __restrict__ out,       // (B,Sq,Hq,D)
+    float* __restrict__ lse,        // (B,Hq,Sq)
+    int B, int Sq, int Skv, int Hq, int Hkv,
+    float scale, int causal, int window_left) {
+  constexpr int kNT = D / 16;   // n-tiles over head dim
+  constexpr int kKS = D / 32;   // k-steps over head dim
+  constexpr int kRowBytes = D * 2;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* k_lds = reinterpret_cast<bf16_t*>(smem);            // [64][D] swizzled
+  bf16_t* vt_lds = k_lds + kKvBlk * D;                        // [D][64] transposed
+  // per-wave P scratch: [4][16][kKvBlk+8]
+  bf16_t* p_lds = vt_lds + D * kKvBlk;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q_tile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int h = bh % Hq;
+  const int hkv = h / (Hq / Hkv);
+
+  const int64_t q_base = ((int64_t)b * Sq * Hq + h) * D;
+  const int64_t kv_base = ((int64_t)b * Skv * Hkv + hkv) * D;
+  const int64_t q_row_stride = (int64_t)Hq * D;
+  const int64_t kv_row_stride = (int64_t)Hkv * D;
+
+  // ---- load this wave's 16 q rows into A fragments -------------------------
+  const int q_row_local = lane & 15;           // m
+  const int q_row_global = q_tile * kQBlk + wave * 16 + q_row_local;
+  bf16x8 q_frag[kKS];
+  {
+    const int safe_row = min(q_row_global, Sq - 1);
+    const bf16_t* qp = q + q_base + (int64_t)safe_row * q_row_stride;
+#pragma unroll
+    for (int ks = 0; ks < kKS; ++ks) {
+      const int d0 = ks * 32 + (lane >> 4) * 8;
+      q_frag[ks] = *reinterpret_cast<const bf16x8*>(qp + d0);
+    }
+  }
+
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -1e30f;
+    l_run[r] = 0.f;
+  }
+  f32x4 o_acc[kNT];
+#pragma unroll
+  for (int nt = 0; nt < kNT; ++nt) o_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int q_tile_last_row = min(q_tile * kQBlk + kQBlk - 1, Sq - 1);
+  int kv_end = Skv;
+  if (causal) kv_end = min(Skv, q_tile_last_row + 1);
+  const int num_kv_tiles = (kv_end + kKvBlk - 1) / kKvBlk;
+
+  for (int kt = 0; kt < num_kv_tiles; ++kt) {
+    const int kv0 = kt * kKvBlk;
+    // ---- stage K (row-major + swizzle) and V^T ------------------------------
+    {
+      // K: 64 rows x D cols; 256 threads, 8 bf16 (16B) per thread per pass.
+      constexpr int elems = kKvBlk * D;
+      for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
+        const int row = idx / D;
+        const int col = idx % D;
+        const int g_row = min(kv0 + row, Skv - 1);
+        const bf16x8 val = *reinterpret_cast<const bf16x8*>(
+            k + kv_base + (int64_t)g_row * kv_row_stride + col);
+        const int byte = (col * 2) ^ ((row & swz_rm_mask<D>()) << 4);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(k_lds) + row * kRowBytes + byte) = val;
+      }
+      // V^T: read V row-major, write transposed (scalar writes).
+      for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
+        const int row = idx / D;  // kv row
+        const int col = idx % D;
+        const int g_row = min(kv0 + row, Skv - 1);
+        const bf16x8 val = *reinterpret_cast<const bf16x8*>(
+            v + kv_base + (int64_t)g_row * kv_row_stride + col);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = col + j;
+          const int byte = (row * 2) ^ ((d & 7) << 4);
+          *reinterpret_cast<bf16_t*>(
+              reinterpret_cast<char*>(vt_lds) + d * (kKvBlk * 2) + byte) = val[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S = Q @ K^T (per wave: 16 x 64) -----------------------------------
+    f32x4 s_acc[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) s_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int kv_row = nt * 16 + (lane & 15);
+#pragma unroll
+      for (int ks = 0; ks < kKS; ++ks) {
+        const int d0 = ks * 32 + (lane >> 4) * 8;
+        const int byte = (d0 * 2) ^ ((kv_row & swz_rm_mask<D>()) << 4);
+        const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<char*>(k_lds) + kv_row * kRowBytes + byte);
+        s_acc[nt] = mfma16(q_frag[ks], kb, s_acc[nt]);
+      }
+    }
+
+    // ---- mask + scale + online softmax -------------------------------------
+    const int my_q_row = q_tile * kQBlk + wave * 16;  // + (lane>>4)*4 + r
+    float p_val[4][4];  // [nt][r]
+    float m_new[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) m_new[r] = m_run[r];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int col = kv0 + nt * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = my_q_row + (lane >> 4) * 4 + r;
+        float s = s_acc[nt][r] * scale;
+        bool masked = col >= Skv;
+        if (causal) masked |= col > row;
+        if (window_left >= 0) masked |= col < row - window_left;
+        p_val[nt][r] = masked ? -1e30f : s;
+      }
+    }
+    // row max: reduce over 16 lanes (same lane>>4 group) and the 4 nt slots.
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = fmaxf(fmaxf(p_val[0][r], p_val[1][r]),
+                       fmaxf(p_val[2][r], p_val[3][r]));
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) {
+        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+      }
+      m_new[r] = fmaxf(m_new[r], mx);
+    }
+    // P = exp(s - m_new); row sum.
+    float l_add[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float acc = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const float p = __builtin_amdgcn_exp2f((p_val[nt][r] - m_new[r]) * kLog2e);
+        p_val[nt][r] = p;
+        acc += p;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) {
+        acc += __shfl_xor(acc, off, 64);
+      }
+      l_add[r] = acc;
+    }
+    // rescale running state + O.
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float alpha = __builtin_amdgcn_exp2f((m_run[r] - m_new[r]) * kLog2e);
+      l_run[r] = l_run[r] * alpha + l_add[r];
+      m_run[r] = m_new[r];
+#pragma unroll
+      for (int nt = 0; nt < kNT; ++nt) o_acc[nt][r] *= alpha;
+    }
+
+    // ---- write P to this wave's LDS scratch, C-layout -> row-major ---------
+    // p_scratch row = q_local (0..15), col = kv col (0..63), row pitch 64+8.
+    {
+      bf16_t* pw = p_lds + wave * 16 * (kKvBlk + 8);
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = (lane >> 4) * 4 + r;
+          const int col = nt * 16 + (lane & 15);
+          pw[row * (kKvBlk + 8) + col] = (bf16_t)p_val[nt][r];
+        }
+      }
+    }
+
+    // ---- O += P @ V (A = P from scratch, B = V^T tile) ---------------------
+    {
+      const bf16_t* pr = p_lds + wave * 16 * (kKvBlk + 8);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {  // K = 64 kv -> 2 steps of 32
+        const int kv_off = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+            pr + (lane & 15) * (kKvBlk + 8) + kv_off);
+#pragma unroll
+        for (int nt = 0; nt < kNT; ++nt) {
+          const int d = nt * 16 + (lane & 15);
+          const int byte = (kv_off * 2) ^ ((d & 7) << 4);
+          const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(vt_lds) + d * (kKvBlk * 2) + byte);
+          o_acc[nt] = mfma16(pa, vb, o_acc[nt]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: O /= l; stage O in LDS row-major; coalesced store ---------
+  {
+    bf16_t* o_lds = k_lds;  // reuse
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+      const int row = wave * 16 + (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int nt = 0; nt < kNT; ++nt) {
+        o_lds[row * D + nt * 16 + (lane & 15)] = (bf16_t)(o_acc[nt][r] * inv_l);
+      }
+    }
+    // LSE: one lane per row.
+    if ((lane & 15) == 0) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int rg = q_tile * kQBlk + wave * 16 + (lane >> 4) * 4 + r;
+        if (rg < Sq) {
+          lse[((int64_t)b * Hq + h) * Sq + rg] =
+              m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+        }
+      }
+    }
+    __syncthreads();
+    constexpr int elems = kQBlk * D;
+    for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
+      const int row = idx / D;
+      const int col = idx % D;
+      const int rg = q_tile * kQBlk + row;
+      if (rg < Sq) {
+        *reinterpret_cast<bf16x8*>(out + q_base + (int64_t)rg * q_row_stride + col) =
+            *reinterpret_cast<const bf16x8*>(o_lds + row * D + col);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward
+// ---------------------------------------------------------------------------
+
+// delta[b,h,q] = sum_d dO * O  (fp32)
+__global__ void attn_delta_kernel(
+    const bf16_t* __restrict__ dout,  // (B,Sq,Hq,D)
+    const bf16_t* __restrict__ out,
+    float* __restrict__ delta,        // (B,Hq,Sq)
+    int B, int Sq, int Hq, int D) {
+  const int64_t row = blockIdx.x;  // over B*Sq*Hq
+  if (row >= (int64_t)B * Sq * Hq) return;
+  const int h = row % Hq;
+  const int s = (row / Hq) % Sq;
+  const int b = row / ((int64_t)Hq * Sq);
+  const bf16_t* dp = dout + row * D;
+  const bf16_t* op = out + row * D;
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < D; i += 64) {
+    acc += (float)dp[i] * (float)op[i];
+  }
+  acc = wave_reduce_sum(acc);
+  if (threadIdx.x == 0) {
+    delta[((int64_t)b * Hq + h) * Sq + s] = acc;
+  }
+}
+
+template <int D>
+__global__ __launch_bounds__(256, 1) void flash_bwd_kernel(
+    const bf16_t* __restrict__ q,
+    const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v,
+    const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse,    // (B,Hq,Sq)
+    const float* __restrict__ delta,  // (B,Hq,Sq)
+    float* __restrict__ dq,           // (B,Sq,Hq,D) fp32 accum
+    float* __restrict__ dk,           // (B,Skv,Hkv,D) fp32 accum
+    float* __restrict__ dv,           // (B,Skv,Hkv,D) fp32 accum
+    int B, int Sq, int Skv, int Hq, int Hkv,
+    float scale, int causal, int window_left) {
+  constexpr int kNT = D / 16;
+  constexpr int kKS = D / 32;
+  constexpr int kRowBytes = D * 2;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);   // [64][D] swizzled rows
+  bf16_t* qt_lds = q_lds + kQBlk * D;                // [D][64] swizzled
+  bf16_t* do_lds = qt_lds + D * kQBlk;               // [64][D]
+  bf16_t* dot_lds = do_lds + kQBlk * D;              // [D][64]
+  bf16_t* kt_lds = dot_lds + D * kQBlk;              // [D][64]
+  bf16_t* x_lds = kt_lds + D * kKvBlk;               // [64][64+8] shared scratch
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int kv_tile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int h = bh % Hq;
+  const int hkv = h / (Hq / Hkv);
+
+  const int64_t q_base = ((int64_t)b * Sq * Hq + h) * D;
+  const int64_t kv_base = ((int64_t)b * Skv * Hkv + hkv) * D;
+  const int64_t q_row_stride = (int64_t)Hq * D;
+  const int64_t kv_row_stride = (int64_t)Hkv * D;
+  const float* lse_row = lse + ((int64_t)b * Hq + h) * Sq;
+  const float* delta_row = delta + ((int64_t)b * Hq + h) * Sq;
+
+  const int kv0 = kv_tile * kKvBlk;
+
+  // This wave's 16 kv rows: K and V A-fragments in registers.
+  const int kv_row_local = lane & 15;
+  const int kv_row_global = min(kv0 + wave * 16 + kv_row_local, Skv - 1);
+  bf16x8 k_frag[kKS], v_frag[kKS];
+  {
+    const bf16_t* kp = k + kv_base + (int64_t)kv_row_global * kv_row_stride;
+    const bf16_t* vp = v + kv_base + (int64_t)kv_row_global * kv_row_stride;
+#pragma unroll
+    for (int ks = 0; ks < kKS; ++ks) {
+      const int d0 = ks * 32 + (lane >> 4) * 8;
+      k_frag[ks] = *reinterpret_cast<const bf16x8*>(kp + d0);
+      v_frag[ks] = *reinterpret_cast<const bf16x8*>(vp + d0);
+    }
+  }
+
+  // Stage K^T once (for dQ's B fragments).
+  {
+    constexpr int elems = kKvBlk * D;
+    for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
+      const int row = idx / D;
+      const int col = idx % D;
+      const int g_row = min(kv0 + row, Skv - 1);
+      const bf16x8 val = *reinterpret_cast<const bf16x8*>(
+          k + kv_base + (int64_t)g_row * kv_row_stride + col);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d = col + j;
+        const int byte = (row * 2) ^ ((d & 7) << 4);
+        *reinterpret_cast<bf16_t*>(
+            reinterpret_cast<char*>(kt_lds) + d * (kKvBlk * 2) + byte) = val[j];
+      }
+    }
+  }
+
+  f32x4 dk_acc[kNT], dv_acc[kNT];
+#pragma unroll
+  for (int nt = 0; nt < kNT; ++nt) {
+    dk_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+    dv_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  int q_start = 0;
+  if (causal) q_start = (kv0 / kQBlk) * kQBlk;  // first q tile that can see kv0
+  if (window_left >= 0) q_start = max(q_start, 0);
+
+  for (int qt = q_start; qt < Sq; qt += kQBlk) {
+    // ---- stage Q, Q^T, dO, dO^T -------------------------------------------
+    __syncthreads();
+    {
+      constexpr int elems = kQBlk * D;
+      for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
+        const int row = idx / D;
+        const int col = idx % D;
+        const int g_row = min(qt + row, Sq - 1);
+        const bf16x8 qv = *reinterpret_cast<const bf16x8*>(
+            q + q_base + (int64_t)g_row * q_row_stride + col);
+        const bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(
+            dout + q_base + (int64_t)g_row * q_row_stride + col);
+        const int byte = (col * 2) ^ ((row & swz_rm_mask<D>()) << 4);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(q_lds) + row * kRowBytes + byte) = qv;
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(do_lds) + row * kRowBytes + byte) = dv8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = col + j;
+          const int tbyte = (row * 2) ^ ((d & 7) << 4);
+          *reinterpret_cast<bf16_t*>(
+              reinterpret_cast<char*>(qt_lds) + d * (kQBlk * 2) + tbyte) = qv[j];
+          *reinterpret_cast<bf16_t*>(
+              reinterpret_cast<char*>(dot_lds) + d * (kQBlk * 2) + tbyte) = dv8[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K @ Q^T (per wave: 16 kv x 64 q) ----------------------------
+    f32x4 st_acc[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) st_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int q_row = nt * 16 + (lane & 15);
+#pragma unroll
+      for (int ks = 0; ks < kKS; ++ks) {
+        const int d0 = ks * 32 + (lane >> 4) * 8;
+        const int byte = (d0 * 2) ^ ((q_row & swz_rm_mask<D>()) << 4);
+        const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<char*>(q_lds) + q_row * kRowBytes + byte);
+        st_acc[nt] = mfma16(k_frag[ks], qb, st_acc[nt]);
+      }
+    }
+
+    // ---- P^T = exp(S^T * scale - lse[q]); masked entries 0 -----------------
+    float pt_val[4][4];  // [nt][r]
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int q_glob = qt + nt * 16 + (lane & 15);
+      const float l = (q_glob < Sq) ? lse_row[min(q_glob, Sq - 1)] : 1e30f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kv_glob = kv0 + wave * 16 + (lane >> 4) * 4 + r;
+        float s = st_acc[nt][r] * scale;
+        bool masked = (kv_glob >= Skv) || (q_glob >= Sq);
+        if (causal) masked |= kv_glob > q_glob;
+        if (window_left >= 0) masked |= kv_glob < q_glob - window_left;
+        pt_val[nt][r] =
+            masked ? 0.f : __builtin_amdgcn_exp2f((s - l) * kLog2e);
+      }
+    }
+
+    // ---- dP^T = V @ dO^T ---------------------------------------------------
+    f32x4 dpt_acc[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) dpt_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int q_row = nt * 16 + (lane & 15);
+#pragma unroll
+      for (int ks = 0; ks < kKS; ++ks) {
+        const int d0 = ks * 32 + (lane >> 4) * 8;
+        const int byte = (d0 * 2) ^ ((q_row & swz_rm_mask<D>()) << 4);
+        const bf16x8 db = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<char*>(do_lds) + q_row * kRowBytes + byte);
+        dpt_acc[nt] = mfma16(v_frag[ks], db, dpt_acc[nt]);
+      }
+    }
+
+    // ---- dV += P^T @ dO  (A = P^T via x_lds, B = dO^T) --------------------
+    {
+      // write this wave's P^T rows (kv-local) into x_lds [kv 64][q 64+8]
+      bf16_t* xw = x_lds;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
+          const int q_l = nt * 16 + (lane & 15);
+          xw[kv_l * (kQBlk + 8) + q_l] = (bf16_t)pt_val[nt][r];
+        }
+      }
+      __syncthreads();
+      const bf16_t* xr = x_lds + (wave * 16) * (kQBlk + 8);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int q_off = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+            xr + (lane & 15) * (kQBlk + 8) + q_off);
+#pragma unroll
+        for (int nt = 0; nt < kNT; ++nt) {
+          const int d = nt * 16 + (lane & 15);
+          const int byte = (q_off * 2) ^ ((d & 7) << 4);
+          const bf16x8 dob = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(dot_lds) + d * (kQBlk * 2) + byte);
+          dv_acc[nt] = mfma16(pa, dob, dv_acc[nt]);
+        }
+      }
+    }
+
+    // ---- dS^T = P^T * (dP^T - delta[q]) * scale ---------------------------
+    float dst_val[4][4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int q_glob = qt + nt * 16 + (lane & 15);
+      const float dlt = (q_glob < Sq) ? delta_row[min(q_glob, Sq - 1)] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        dst_val[nt][r] = pt_val[nt][r] * (dpt_acc[nt][r] - dlt) * scale;
+      }
+    }
+
+    // ---- dK += dS^T @ Q  (A = dS^T via x_lds, B = Q^T) --------------------
+    {
+      __syncthreads();  // x_lds reuse
+      bf16_t* xw = x_lds;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
+          const int q_l = nt * 16 + (lane & 15);
+          xw[kv_l * (kQBlk + 8) + q_l] = (bf16_t)dst_val[nt][r];
+        }
+      }
+      __syncthreads();
+      const bf16_t* xr = x_lds + (wave * 16) * (kQBlk + 8);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int q_off = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 da = *reinterpret_cast<const bf16x8*>(
+            xr + (lane & 15) * (kQBlk + 8) + q_off);
+#pragma unroll
+        for (int nt = 0; nt < kNT; ++nt) {
+          const int d = nt * 16 + (lane & 15);
+          const int byte = (q_off * 2) ^ ((d & 7) << 4);
+          const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(qt_lds) + d * (kQBlk * 2) + byte);
+          dk_acc[nt] = mfma16(da, qb, dk_acc[nt]);
+        }
+      }
+    }
+
+    // ---- dQ partial = dS @ K; atomicAdd (A = dS q-major via x_lds) --------
+    {
+      __syncthreads();
+      bf16_t* xw = x_lds;  // now [q 64][kv 64+8]
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
+          const int q_l = nt * 16 + (lane & 15);
+          xw[q_l * (kKvBlk + 8) + kv_l] = (bf16_t)dst_val[nt][r];
+        }
+      }
+      __syncthreads();
+      // each wave: 16 q rows (wave*16 ..), K = 64 kv, N = D
+      f32x4 dq_acc[kNT];
+#pragma unroll
+      for (int nt = 0; nt < kNT; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      const bf16_t* xr = x_lds + (wave * 16) * (kKvBlk + 8);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int kv_off = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 da = *reinterpret_cast<const bf16x8*>(
+            xr + (lane & 15) * (kKvBlk + 8) + kv_off);
+#pragma unroll
+        for (int nt = 0; nt < kNT; ++nt) {
+          const int d = nt * 16 + (lane & 15);
+          const int byte = (kv_off * 2) ^ ((d & 7) << 4);
+          const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(kt_lds) + d * (kKvBlk * 2) + byte);
+          dq_acc[nt] = mfma16(da, kb, dq_acc[nt]);
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q_glob = qt + wave * 16 + (lane >> 4) * 4 + r;
+        if (q_glob < Sq) {
+#pragma unroll
+          for (int nt = 0; nt < kNT; ++nt) {
+            atomicAdd(dq + q_base + (int64_t)q_glob * q_row_stride + nt * 16 + (lane & 15),
+                      dq_acc[nt][r]);
+          }
+        }
+      }
+    }
+  }
+
+  // ---- flush dK, dV (atomicAdd: GQA groups and padded tiles overlap) -------
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kv_glob = kv0 + wave * 16 + (lane >> 4) * 4 + r;
+    if (kv_glob < Skv) {
+#pragma unroll
+      for (int nt = 0; nt < kNT; ++nt) {
+        const int d = nt * 16 + (lane & 15);
+        atomicAdd(dk + kv_base + (int64_t)kv_glob * kv_row_stride + d, dk_acc[nt][r]);
+        atomicAdd(dv + kv_base + (int64_t)kv_glob * kv_row_stride + d, dv_acc[nt][r]);
+      }
+    }
+  }
+}
+
+// MFMA fragment-map self-check: C = A@B for one 16x32 @ 32x16 tile.
+__global__ void mfma_selfcheck_kernel(
+    const bf16_t* __restrict__ a,  // (16, 32) row-major
+    const bf16_t* __restrict__ b,  // (32, 16) row-major
+    float* __restrict__ c) {       // (16, 16) row-major
+  const int lane = threadIdx.x & 63;
+  bf16x8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = a[(lane & 15) * 32 + (lane >> 4) * 8 + j];
+    bf[j] = b[((lane >> 4) * 8 + j) * 16 + (lane & 15)];
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = mfma16(af, bf, acc);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    c[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+  }
+}
+
+}  // namespace d9d
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+namespace {
+
+int padded_head_dim(int d) {
+  for (int cand : {32, 64, 96, 128}) {
+    if (d <= cand) return cand;
+  }
+  TORCH_CHECK(false, "head_dim too large: ", d);
+  return -1;
+}
+
+torch::Tensor maybe_pad_d(torch::Tensor t, int D_pad) {
+  const int D = t.size(-1);
+  if (D == D_pad) return t.contiguous();
+  auto padded = torch::zeros(
+      {t.size(0), t.size(1), t.size(2), (int64_t)D_pad}, t.options());
+  padded.narrow(-1, 0, D).copy_(t);
+  return padded;
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> flash_attn_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    bool causal, double softmax_scale, int64_t window_left) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1), Hkv = k.size(2);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA head mismatch");
+
+  const int D_pad = padded_head_dim(D);
+  auto qp = maybe_pad_d(q, D_pad);
+  auto kp = maybe_pad_d(k, D_pad);
+  auto vp = maybe_pad_d(v, D_pad);
+
+  auto out = torch::empty_like(qp);
+  auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+
+  const dim3 grid((Sq + d9d::kQBlk - 1) / d9d::kQBlk, B * Hq);
+  const size_t smem =
+      (size_t)(d9d::kKvBlk * D_pad + D_pad * d9d::kKvBlk + 4 * 16 * (d9d::kKvBlk + 8)) *
+      sizeof(__bf16);
+  auto stream = at::hip::getCurrentHIPStream();
+
+#define LAUNCH_FWD(DP)                                                        \
+  hipLaunchKernelGGL((d9d::flash_fwd_kernel<DP>), grid, dim3(256), smem,      \
+                     stream,                                                  \
+                     reinterpret_cast<const __bf16*>(qp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(kp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(vp.data_ptr()),          \
+                     reinterpret_cast<__bf16*>(out.data_ptr()),               \
+                     lse.data_ptr<float>(), B, Sq, Skv, Hq, Hkv,              \
+                     (float)softmax_scale, causal ? 1 : 0, (int)window_left)
+  switch (D_pad) {
+    case 32: LAUNCH_FWD(32); break;
+    case 64: LAUNCH_FWD(64); break;
+    case 96: LAUNCH_FWD(96); break;
+    case 128: LAUNCH_FWD(128); break;
+  }
+#undef LAUNCH_FWD
+
+  if (D_pad != D) out = out.narrow(-1, 0, D).contiguous();
+  return {out, lse};
+}
+
+std::vector<torch::Tensor> flash_attn_bwd(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor lse,
+    bool causal, double softmax_scale, int64_t window_left) {
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1), Hkv = k.size(2);
+  const int D_pad = padded_head_dim(D);
+
+  auto qp = maybe_pad_d(q, D_pad);
+  auto kp = maybe_pad_d(k, D_pad);
+  auto vp = maybe_pad_d(v, D_pad);
+  auto dop = maybe_pad_d(dout, D_pad);
+  auto op = maybe_pad_d(out, D_pad);
+
+  auto f32opt = q.options().dtype(torch::kFloat32);
+  auto delta = torch::empty({B, Hq, Sq}, f32opt);
+  auto dq32 = torch::zeros({B, Sq, Hq, D_pad}, f32opt);
+  auto dk32 = torch::zeros({B, Skv, Hkv, D_pad}, f32opt);
+  auto dv32 = torch::zeros({B, Skv, Hkv, D_pad}, f32opt);
+
+  auto stream = at::hip::getCurrentHIPStream();
+  {
+    const int64_t rows = (int64_t)B * Sq * Hq;
+    hipLaunchKernelGGL(d9d::attn_delta_kernel, dim3(rows), dim3(64), 0, stream,
+                       reinterpret_cast<const __bf16*>(dop.data_ptr()),
+                       reinterpret_cast<const __bf16*>(op.data_ptr()),
+                       delta.data_ptr<float>(), B, Sq, Hq, D_pad);
+  }
+
+  const dim3 grid((Skv + d9d::kKvBlk - 1) / d9d::kKvBlk, B * Hq);
+  const size_t smem =
+      (size_t)(4 * d9d::kQBlk * D_pad + D_pad * d9d::kKvBlk + 64 * (64 + 8)) *
+      sizeof(__bf16);
+
+#define LAUNCH_BWD(DP)                                                        \
+  hipLaunchKernelGGL((d9d::flash_bwd_kernel<DP>), grid, dim3(256), smem,      \
+                     stream,                                                  \
+                     reinterpret_cast<const __bf16*>(qp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(kp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(vp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(dop.data_ptr()),         \
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),          \
+                     dq32.data_ptr<float>(), dk32.data_ptr<float>(),          \
+                     dv32.data_ptr<float>(), B, Sq, Skv, Hq, Hkv,             \
+                     (float)softmax_scale, causal ? 1 : 0, (int)window_left)
+  switch (D_pad) {
+    case 32: LAUNCH_BWD(32); break;
+    case 64: LAUNCH_BWD(64); break;
+    case 96: LAUNCH_BWD(96); break;
+    case 128: LAUNCH_BWD(128); break;
+  }
+#undef LAUNCH_BWD
+
+  auto dq = dq32.narrow(-1, 0, D).to(torch::kBFloat16);
+  auto dk = dk32.narrow(-1, 0, D).to(torch::kBFloat16);
+  auto dv = dv32.narrow(-1, 0, D).to(torch::kBFloat16);
+  return {dq.contiguous(), dk.contiguous(), dv.contiguous()};
+}
+
+torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.sizes() == torch::IntArrayRef({16, 32}));
+  TORCH_CHECK(b.sizes() == torch::IntArrayRef({32, 16}));
+  auto c = torch::empty({16, 16}, a.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::mfma_selfcheck_kernel, dim3(1), dim3(64), 0, stream,
+                     reinterpret_cast<const __bf16*>(a.contiguous().data_ptr()),
+                     reinterpret_cast<const __bf16*>(b.contiguous().data_ptr()),
+                     c.data_ptr<float>());
+  return c;
+}

@@ -15,7 +15,18 @@ void adamw_stochastic_bf16_(torch::Tensor p, torch::Tensor g, torch::Tensor m, t
                             double lr, double beta1, double beta2, double eps, double weight_decay,
                             int64_t step, int64_t seed);
 
+// attention.hip
+std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                          bool causal, double softmax_scale, int64_t window_left);
+std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                          torch::Tensor v, torch::Tensor out, torch::Tensor lse,
+                                          bool causal, double softmax_scale, int64_t window_left);
+torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("flash_attn_fwd", &flash_attn_fwd, "CDNA4 flash attention forward");
+  m.def("flash_attn_bwd", &flash_attn_bwd, "CDNA4 flash attention backward");
+  m.def("mfma_selfcheck", &mfma_selfcheck, "MFMA fragment-map self check");
   m.def("rms_norm_fwd", &rms_norm_fwd, "RMSNorm forward (bf16, CDNA4)");
   m.def("rms_norm_bwd", &rms_norm_bwd, "RMSNorm backward (bf16, CDNA4)");
   m.def("silu_mul_fwd", &silu_mul_fwd, "fused silu(a)*b forward");
