@@ -69,13 +69,14 @@ class GroupedQueryAttention(nn.Module):
             self.sinks = None
 
     def reset_parameters(self) -> None:
-        for lin in (self.q_proj, self.k_proj, self.v_proj, self.o_proj):
-            nn.init.normal_(lin.weight, mean=0.0, std=0.02 / math.sqrt(2))
-        if self.q_norm is not None:
-            self.q_norm.reset_parameters()
-            self.k_norm.reset_parameters()
-        if self.sinks is not None:
-            nn.init.zeros_(self.sinks)
+        with torch.no_grad():
+            for lin in (self.q_proj, self.k_proj, self.v_proj, self.o_proj):
+                nn.init.normal_(lin.weight, mean=0.0, std=0.02 / math.sqrt(2))
+            if self.q_norm is not None:
+                self.q_norm.reset_parameters()
+                self.k_norm.reset_parameters()
+            if self.sinks is not None:
+                nn.init.zeros_(self.sinks)
 
     def forward(
         self,

@@ -37,8 +37,9 @@ class SplitTokenEmbeddings(nn.Module):
         self.vocab_size = off
 
     def reset_parameters(self) -> None:
-        for emb in self.embeddings.values():
-            nn.init.normal_(emb.weight, mean=0.0, std=0.02)
+        with torch.no_grad():
+            for emb in self.embeddings.values():
+                nn.init.normal_(emb.weight, mean=0.0, std=0.02)
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         out = None

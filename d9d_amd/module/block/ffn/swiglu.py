@@ -25,8 +25,9 @@ class SwiGLU(nn.Module):
         self.down_proj = nn.Linear(intermediate_size, hidden_size, **kw)
 
     def reset_parameters(self) -> None:
-        for lin in (self.gate_proj, self.up_proj, self.down_proj):
-            nn.init.normal_(lin.weight, mean=0.0, std=0.02 / math.sqrt(2))
+        with torch.no_grad():
+            for lin in (self.gate_proj, self.up_proj, self.down_proj):
+                nn.init.normal_(lin.weight, mean=0.0, std=0.02 / math.sqrt(2))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.down_proj(silu_mul(self.gate_proj(x), self.up_proj(x)))

@@ -18,7 +18,8 @@ class ClassificationHead(nn.Module):
         self.proj = nn.Linear(hidden_size, num_classes, bias=False, device=device, dtype=dtype)
 
     def reset_parameters(self) -> None:
-        nn.init.normal_(self.proj.weight, mean=0.0, std=0.02)
+        with torch.no_grad():
+            nn.init.normal_(self.proj.weight, mean=0.0, std=0.02)
 
     def forward(
         self,

@@ -20,8 +20,9 @@ class EmbeddingHead(nn.Module):
             self.proj = None
 
     def reset_parameters(self) -> None:
-        if self.proj is not None:
-            nn.init.normal_(self.proj.weight, mean=0.0, std=0.02)
+        with torch.no_grad():
+            if self.proj is not None:
+                nn.init.normal_(self.proj.weight, mean=0.0, std=0.02)
 
     def forward(
         self,

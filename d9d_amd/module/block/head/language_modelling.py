@@ -34,8 +34,9 @@ class SplitLanguageModellingHead(nn.Module):
         self.vocab_size = sum(split_sizes.values())
 
     def reset_parameters(self) -> None:
-        for w in self.weights.values():
-            nn.init.normal_(w, mean=0.0, std=0.02)
+        with torch.no_grad():
+            for w in self.weights.values():
+                nn.init.normal_(w, mean=0.0, std=0.02)
 
     def full_weight(self) -> torch.Tensor:
         return torch.cat([self.weights[name] for name in self.order], dim=0)

@@ -146,6 +146,7 @@ class RcclAllToAllCommunicationHandler(MoECommunicationHandler):
                 "order": order,
                 "out_splits": out_splits,
                 "in_splits": in_splits,
+                "local_expert_offset": dist.get_rank(self.group) * self.experts_per_rank,
             },
         )
         return expert_rows, batch_sizes, ctx

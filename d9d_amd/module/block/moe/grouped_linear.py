@@ -30,7 +30,8 @@ class GroupedLinear(nn.Module):
         )
 
     def reset_parameters(self) -> None:
-        nn.init.normal_(self.weight, mean=0.0, std=0.02 / math.sqrt(2))
+        with torch.no_grad():
+            nn.init.normal_(self.weight, mean=0.0, std=0.02 / math.sqrt(2))
 
     def _local_weight(self) -> torch.Tensor:
         w = self.weight

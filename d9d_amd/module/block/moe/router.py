@@ -34,9 +34,10 @@ class TopKRouter(nn.Module):
             self.expert_bias = None
 
     def reset_parameters(self) -> None:
-        nn.init.normal_(self.gate.weight, mean=0.0, std=0.02)
-        if self.expert_bias is not None:
-            nn.init.zeros_(self.expert_bias)
+        with torch.no_grad():
+            nn.init.normal_(self.gate.weight, mean=0.0, std=0.02)
+            if self.expert_bias is not None:
+                nn.init.zeros_(self.expert_bias)
 
     def forward(self, x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
         """x (T, H) -> (probs (T, k) fp32, indices (T, k) int64)."""

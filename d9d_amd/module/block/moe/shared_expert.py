@@ -27,8 +27,9 @@ class SharedSwiGLU(nn.Module):
             self.output_gate = nn.Linear(hidden_size, hidden_size, **kw)
 
     def reset_parameters(self) -> None:
-        for name, lin in self.named_children():
-            nn.init.normal_(lin.weight, mean=0.0, std=0.02 / math.sqrt(2))
+        with torch.no_grad():
+            for name, lin in self.named_children():
+                nn.init.normal_(lin.weight, mean=0.0, std=0.02 / math.sqrt(2))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         out = self.down_proj(silu_mul(self.gate_proj(x), self.up_proj(x)))
