@@ -23,7 +23,13 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q, t
                                           bool causal, double softmax_scale, int64_t window_left);
 torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b);
 
+// gmm.hip
+torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes);
+torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes, int64_t num_experts);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
+  m.def("gmm_db", &gmm_db, "CDNA4 grouped GEMM weight-grad");
   m.def("flash_attn_fwd", &flash_attn_fwd, "CDNA4 flash attention forward");
   m.def("flash_attn_bwd", &flash_attn_bwd, "CDNA4 flash attention backward");
   m.def("mfma_selfcheck", &mfma_selfcheck, "MFMA fragment-map self check");

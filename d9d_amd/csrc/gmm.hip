@@ -1,0 +1,366 @@
+// CDNA4 grouped GEMM for MoE expert compute.
+//
+// Replaces the reference's nv-grouped-gemm wheel (d9d/kernel/gmm/function.py).
+// out[rows_e] = a[rows_e] @ b[e] for each expert e; row counts are ragged.
+// Host builds tiny per-expert offset tables (row_offsets, m-tile prefix) from
+// the CPU batch_sizes; ONE kernel launch covers every expert — each block
+// binary-searches its m-tile, so there is no per-expert launch overhead
+// (the eager fallback was E x 3 rocBLAS launches per MoE layer).
+//
+// Geometry: 128x64 tiles, 4 waves (2x2), BK=64, double-buffered LDS with the
+// ((row&15)<<4 style) XOR swizzle for conflict-free ds_read_b128,
+// v_mfma_f32_16x16x32_bf16.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace d9d {
+
+typedef __bf16 bf16_t;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+D9D_DEVICE f32x4 mfma16g(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+constexpr int kBM = 128;
+constexpr int kBN = 64;
+constexpr int kBK = 64;
+
+// A LDS tile: [kBM][kBK] row-major, rows kBK*2=128 bytes -> swizzle mask 7.
+// B LDS tile: [kBN][kBK] row-major (B^T layout: we need B[k][n] fragments with
+//   k contiguous... see below), also 128-byte rows.
+//
+// Fragment needs:
+//   A[m][k]: lane m=l&15, k=(l>>4)*8+j  -> A tile row-major [m][k], contiguous in k. OK.
+//   B[k][n]: lane n=l&15, k=(l>>4)*8+j  -> need k contiguous at fixed n ->
+//   stage B TRANSPOSED: bt[n][k] (64 x 64).
+//
+// b (E,K,N) is row-major in N, so reading b[k][n0..n0+7] is contiguous: we
+// read 8 consecutive n at fixed k and scatter into bt[n][k] (8 scalar LDS
+// writes) — done once per K-tile per block.
+
+__global__ __launch_bounds__(256, 2) void gmm_kernel(
+    const bf16_t* __restrict__ a,    // (T, K)
+    const bf16_t* __restrict__ b,    // (E, K, N)
+    bf16_t* __restrict__ out,        // (T, N)
+    const int* __restrict__ row_off,      // (E+1,)
+    const int* __restrict__ mtile_pref,   // (E+1,) prefix of ceil(rows_e/kBM)
+    int E, int K, int N) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* a_lds = reinterpret_cast<bf16_t*>(smem);        // 2 x [kBM][kBK]
+  bf16_t* bt_lds = a_lds + 2 * kBM * kBK;                 // 2 x [kBN][kBK]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 1;   // 0..1: wave row (64 rows each)
+  const int wn = wave & 1;    // 0..1: wave col (32 cols each)
+
+  // ---- map blockIdx.x -> (expert, m_tile) via binary search ----------------
+  const int mt_global = blockIdx.x;
+  // largest e with mtile_pref[e] <= mt_global (expert may own several m-tiles)
+  int lo = 0, hi = E - 1;
+  while (lo < hi) {
+    const int mid = (lo + hi + 1) >> 1;
+    if (mtile_pref[mid] <= mt_global) lo = mid; else hi = mid - 1;
+  }
+  const int e = lo;
+  const int m_tile = mt_global - mtile_pref[e];
+  const int row0 = row_off[e] + m_tile * kBM;
+  const int row_end = row_off[e + 1];
+  const int n0 = blockIdx.y * kBN;
+
+  const bf16_t* b_e = b + (int64_t)e * K * N;
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int n_ktiles = (K + kBK - 1) / kBK;
+
+  auto stage = [&](int kt, int buf) {
+    const int k0 = kt * kBK;
+    // A: kBM x kBK, 16B per thread -> 128*64*2/16 = 1024 ops / 256 threads = 4
+    bf16_t* al = a_lds + buf * kBM * kBK;
+#pragma unroll
+    for (int it = 0; it < (kBM * kBK) / (256 * 8); ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int row = idx / kBK;
+      const int col = idx % kBK;
+      const int g_row = row0 + row;
+      bf16x8 val = {};
+      if (g_row < row_end && k0 + col < K) {
+        val = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + col);
+      }
+      const int byte = (col * 2) ^ ((row & 7) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(al) + row * (kBK * 2) + byte) = val;
+    }
+    // B^T: stage b[k0..k0+63][n0..n0+63] transposed into bt[n][k].
+    bf16_t* bl = bt_lds + buf * kBN * kBK;
+#pragma unroll
+    for (int it = 0; it < (kBN * kBK) / (256 * 8); ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int k = idx / kBN;       // 0..63 within tile
+      const int n = idx % kBN;       // 0..56 step 8
+      bf16x8 val = {};
+      if (k0 + k < K) {
+        if (n0 + n + 7 < N) {
+          val = *reinterpret_cast<const bf16x8*>(b_e + (int64_t)(k0 + k) * N + n0 + n);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            val[j] = (n0 + n + j < N) ? b_e[(int64_t)(k0 + k) * N + n0 + n + j]
+                                      : (bf16_t)0.f;
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int nn = n + j;
+        const int byte = (k * 2) ^ ((nn & 7) << 4);
+        *reinterpret_cast<bf16_t*>(
+            reinterpret_cast<char*>(bl) + nn * (kBK * 2) + byte) = val[j];
+      }
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  for (int kt = 0; kt < n_ktiles; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < n_ktiles) stage(kt + 1, buf ^ 1);
+
+    const bf16_t* al = a_lds + buf * kBM * kBK;
+    const bf16_t* bl = bt_lds + buf * kBN * kBK;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {  // kBK=64 -> 2 MFMA k-steps
+      const int kk = ks * 32 + (lane >> 4) * 8;
+      bf16x8 a_frag[4], b_frag[2];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = wm * 64 + i * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((row & 7) << 4);
+        a_frag[i] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(al) + row * (kBK * 2) + byte);
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int col = wn * 32 + j * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((col & 7) << 4);
+        b_frag[j] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(bl) + col * (kBK * 2) + byte);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = mfma16g(a_frag[i], b_frag[j], acc[i][j]);
+    }
+    __syncthreads();
+  }
+
+  // ---- store ---------------------------------------------------------------
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + wm * 64 + i * 16 + (lane >> 4) * 4 + r;
+        const int col = n0 + wn * 32 + j * 16 + (lane & 15);
+        if (row < row_end && col < N) {
+          out[(int64_t)row * N + col] = (bf16_t)acc[i][j][r];
+        }
+      }
+    }
+  }
+}
+
+// db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N). Grid (ceil(K/64), ceil(N/64), E).
+__global__ __launch_bounds__(256, 2) void gmm_db_kernel(
+    const bf16_t* __restrict__ a,   // (T, K)
+    const bf16_t* __restrict__ g,   // (T, N)
+    bf16_t* __restrict__ db,        // (E, K, N)
+    const int* __restrict__ row_off,
+    int E, int K, int N) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // at: [64 k][64 rows] (transposed a tile), gl: [64 rows][64 n] -> need
+  // A[m=k][kk=row] k-contig rows: at[k][row]; B[kk=row][n]: need row contig at
+  // fixed n -> stage g transposed too: gt[n][row].
+  bf16_t* at_lds = reinterpret_cast<bf16_t*>(smem);   // [64][64+pad? swizzled]
+  bf16_t* gt_lds = at_lds + 64 * 64;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 1;  // 2x2 waves over (k 64 x n 64): each 32x32
+  const int wn = wave & 1;
+
+  const int e = blockIdx.z;
+  const int k0 = blockIdx.x * 64;
+  const int n0 = blockIdx.y * 64;
+  const int r_start = row_off[e];
+  const int r_end = row_off[e + 1];
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int rt = r_start; rt < r_end; rt += 64) {
+    __syncthreads();
+    // stage a^T: a[row][k0..] -> at[k][row]; 64x64 tile
+    for (int idx = threadIdx.x * 8; idx < 64 * 64; idx += 256 * 8) {
+      const int row = idx / 64;  // row within tile
+      const int kk = idx % 64;   // k within tile (contiguous 8)
+      const int g_row = rt + row;
+      bf16x8 va = {}, vg = {};
+      if (g_row < r_end) {
+        if (k0 + kk + 7 < K) {
+          va = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + kk);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            va[j] = (k0 + kk + j < K) ? a[(int64_t)g_row * K + k0 + kk + j] : (bf16_t)0.f;
+        }
+        if (n0 + kk + 7 < N) {
+          vg = *reinterpret_cast<const bf16x8*>(g + (int64_t)g_row * N + n0 + kk);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            vg[j] = (n0 + kk + j < N) ? g[(int64_t)g_row * N + n0 + kk + j] : (bf16_t)0.f;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int kcol = kk + j;
+        const int byte_a = (row * 2) ^ ((kcol & 7) << 4);
+        *reinterpret_cast<bf16_t*>(
+            reinterpret_cast<char*>(at_lds) + kcol * (64 * 2) + byte_a) = va[j];
+        const int byte_g = (row * 2) ^ ((kcol & 7) << 4);
+        *reinterpret_cast<bf16_t*>(
+            reinterpret_cast<char*>(gt_lds) + kcol * (64 * 2) + byte_g) = vg[j];
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {  // 64 rows -> 2 k-steps
+      const int rr = ks * 32 + (lane >> 4) * 8;
+      bf16x8 a_frag[2], g_frag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int krow = wm * 32 + i * 16 + (lane & 15);
+        const int byte = (rr * 2) ^ ((krow & 7) << 4);
+        a_frag[i] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(at_lds) + krow * (64 * 2) + byte);
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int ncol = wn * 32 + j * 16 + (lane & 15);
+        const int byte = (rr * 2) ^ ((ncol & 7) << 4);
+        g_frag[j] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(gt_lds) + ncol * (64 * 2) + byte);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = mfma16g(a_frag[i], g_frag[j], acc[i][j]);
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kk = k0 + wm * 32 + i * 16 + (lane >> 4) * 4 + r;
+        const int nn = n0 + wn * 32 + j * 16 + (lane & 15);
+        if (kk < K && nn < N) {
+          db[((int64_t)e * K + kk) * N + nn] = (bf16_t)acc[i][j][r];
+        }
+      }
+    }
+  }
+}
+
+}  // namespace d9d
+
+// ---------------------------------------------------------------------------
+
+static std::pair<torch::Tensor, torch::Tensor> build_offsets(
+    torch::Tensor batch_sizes, torch::Device device, int tile_m) {
+  const int E = batch_sizes.numel();
+  auto row_off = torch::empty({E + 1}, torch::dtype(torch::kInt32));
+  auto mtile_pref = torch::empty({E + 1}, torch::dtype(torch::kInt32));
+  auto bs = batch_sizes.to(torch::kInt64);
+  const int64_t* p = bs.data_ptr<int64_t>();
+  int32_t* ro = row_off.data_ptr<int32_t>();
+  int32_t* mp = mtile_pref.data_ptr<int32_t>();
+  int rows = 0, tiles = 0;
+  for (int e = 0; e < E; ++e) {
+    ro[e] = rows;
+    mp[e] = tiles;
+    rows += (int)p[e];
+    tiles += (int)((p[e] + tile_m - 1) / tile_m);
+  }
+  ro[E] = rows;
+  mp[E] = tiles;
+  return {row_off.to(device, /*non_blocking=*/true),
+          mtile_pref.to(device, /*non_blocking=*/true)};
+}
+
+torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
+  TORCH_CHECK(b.is_cuda() && b.scalar_type() == torch::kBFloat16 && b.is_contiguous());
+  TORCH_CHECK(batch_sizes.device().is_cpu());
+  const int T = a.size(0), K = a.size(1);
+  const int E = b.size(0), N = b.size(2);
+  TORCH_CHECK(b.size(1) == K, "gmm K mismatch");
+
+  auto out = torch::empty({(int64_t)T, (int64_t)N}, a.options());
+  if (T == 0) return out;
+  auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
+  const int total_mtiles = ((int64_t)T + d9d::kBM * E) > 0
+      ? mtile_pref[E].item<int>() : 0;
+  if (total_mtiles == 0) return out;
+
+  const dim3 grid(total_mtiles, (N + d9d::kBN - 1) / d9d::kBN);
+  const size_t smem = (2 * d9d::kBM * d9d::kBK + 2 * d9d::kBN * d9d::kBK) * sizeof(__bf16);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::gmm_kernel, grid, dim3(256), smem, stream,
+                     reinterpret_cast<const __bf16*>(a.data_ptr()),
+                     reinterpret_cast<const __bf16*>(b.data_ptr()),
+                     reinterpret_cast<__bf16*>(out.data_ptr()),
+                     row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
+                     E, K, N);
+  return out;
+}
+
+torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes,
+                     int64_t num_experts) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
+  TORCH_CHECK(g.is_cuda() && g.scalar_type() == torch::kBFloat16 && g.is_contiguous());
+  const int K = a.size(1), N = g.size(1);
+  const int E = (int)num_experts;
+
+  auto db = torch::zeros({(int64_t)E, (int64_t)K, (int64_t)N}, a.options());
+  if (a.size(0) == 0) return db;
+  auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
+
+  const dim3 grid((K + 63) / 64, (N + 63) / 64, E);
+  const size_t smem = (2 * 64 * 64) * sizeof(__bf16);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::gmm_db_kernel, grid, dim3(256), smem, stream,
+                     reinterpret_cast<const __bf16*>(a.data_ptr()),
+                     reinterpret_cast<const __bf16*>(g.data_ptr()),
+                     reinterpret_cast<__bf16*>(db.data_ptr()),
+                     row_off.data_ptr<int>(), E, K, N);
+  return db;
+}

@@ -73,6 +73,99 @@ __global__ void rms_norm_fwd_kernel(
   }
 }
 
+// Small-N specialization (e.g. per-head QK norm, N <= 512): one WAVE per row,
+// row in registers, no LDS round-trip. 4 rows in flight per block.
+template <int BLOCK, int VPL>  // VPL = values per lane (N <= 64*VPL)
+__global__ void rms_norm_fwd_smalln_kernel(
+    const ushort* __restrict__ x,
+    const ushort* __restrict__ w,
+    ushort* __restrict__ y,
+    float* __restrict__ inv_rms,
+    int64_t M, int64_t N, float eps, float w_offset) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t waves_total = (int64_t)gridDim.x * (BLOCK / 64);
+  float wvals[VPL];
+#pragma unroll
+  for (int j = 0; j < VPL; ++j) {
+    const int64_t i = (int64_t)j * 64 + lane;
+    wvals[j] = (i < N) ? bf16_bits_to_f32(w[i]) + w_offset : 0.f;
+  }
+  for (int64_t row = blockIdx.x * (BLOCK / 64) + wave; row < M; row += waves_total) {
+    const ushort* xrow = x + row * N;
+    float vals[VPL];
+    float sumsq = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) {
+      const int64_t i = (int64_t)j * 64 + lane;
+      vals[j] = (i < N) ? bf16_bits_to_f32(xrow[i]) : 0.f;
+      sumsq += vals[j] * vals[j];
+    }
+    sumsq = wave_reduce_sum(sumsq);
+    const float inv = rsqrtf(sumsq / (float)N + eps);
+    if (lane == 0) inv_rms[row] = inv;
+    ushort* yrow = y + row * N;
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) {
+      const int64_t i = (int64_t)j * 64 + lane;
+      if (i < N) yrow[i] = f32_to_bf16_rne(vals[j] * inv * wvals[j]);
+    }
+  }
+}
+
+template <int BLOCK, int VPL>
+__global__ void rms_norm_bwd_smalln_kernel(
+    const ushort* __restrict__ x,
+    const ushort* __restrict__ w,
+    const ushort* __restrict__ dy,
+    const float* __restrict__ inv_rms,
+    ushort* __restrict__ dx,
+    float* __restrict__ dw,
+    int64_t M, int64_t N, float w_offset) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* dw_lds = reinterpret_cast<float*>(smem_raw);  // N fp32, block-shared
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t waves_total = (int64_t)gridDim.x * (BLOCK / 64);
+  for (int64_t i = threadIdx.x; i < N; i += BLOCK) dw_lds[i] = 0.f;
+  __syncthreads();
+
+  float wvals[VPL];
+#pragma unroll
+  for (int j = 0; j < VPL; ++j) {
+    const int64_t i = (int64_t)j * 64 + lane;
+    wvals[j] = (i < N) ? bf16_bits_to_f32(w[i]) + w_offset : 0.f;
+  }
+  const float inv_n = 1.f / (float)N;
+  for (int64_t row = blockIdx.x * (BLOCK / 64) + wave; row < M; row += waves_total) {
+    const ushort* xrow = x + row * N;
+    const ushort* grow = dy + row * N;
+    const float inv = inv_rms[row];
+    float xh[VPL], g[VPL];
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) {
+      const int64_t i = (int64_t)j * 64 + lane;
+      xh[j] = (i < N) ? bf16_bits_to_f32(xrow[i]) * inv : 0.f;
+      g[j] = (i < N) ? bf16_bits_to_f32(grow[i]) : 0.f;
+      s += g[j] * wvals[j] * xh[j];
+    }
+    s = wave_reduce_sum(s) * inv_n;
+    ushort* dxrow = dx + row * N;
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) {
+      const int64_t i = (int64_t)j * 64 + lane;
+      if (i < N) {
+        dxrow[i] = f32_to_bf16_rne(inv * (g[j] * wvals[j] - xh[j] * s));
+        atomicAdd(&dw_lds[i], g[j] * xh[j]);  // LDS atomics: cheap on CDNA4
+      }
+    }
+  }
+  __syncthreads();
+  for (int64_t i = threadIdx.x; i < N; i += BLOCK) atomicAdd(dw + i, dw_lds[i]);
+}
+
 // Persistent backward: each block walks rows with stride gridDim, keeps a
 // per-block fp32 dw accumulator in LDS, and atomically adds it to the global
 // fp32 dw buffer once at the end.
@@ -181,10 +274,27 @@ std::vector<torch::Tensor> rms_norm_fwd(
   auto inv_rms = torch::empty({M}, x.options().dtype(torch::kFloat32));
 
   constexpr int kBlock = 256;
+  auto stream = at::hip::getCurrentHIPStream();
+  if (N <= 512) {
+    const int grid = static_cast<int>(std::min<int64_t>((M + 3) / 4, 2048));
+#define SMALL_FWD(VPL)                                                         \
+  hipLaunchKernelGGL((d9d::rms_norm_fwd_smalln_kernel<kBlock, VPL>),           \
+                     dim3(grid), dim3(kBlock), 0, stream,                      \
+                     reinterpret_cast<const ushort*>(x.data_ptr()),            \
+                     reinterpret_cast<const ushort*>(w.data_ptr()),            \
+                     reinterpret_cast<ushort*>(y.data_ptr()),                  \
+                     inv_rms.data_ptr<float>(), M, N,                          \
+                     static_cast<float>(eps), zero_centered ? 1.0f : 0.0f)
+    if (N <= 64) SMALL_FWD(1);
+    else if (N <= 128) SMALL_FWD(2);
+    else if (N <= 256) SMALL_FWD(4);
+    else SMALL_FWD(8);
+#undef SMALL_FWD
+    return {y, inv_rms};
+  }
   const int grid = static_cast<int>(std::min<int64_t>(M, 2048));
   const size_t smem = ((N + 7) & ~7ll) * sizeof(ushort) + 16 * sizeof(float);
   TORCH_CHECK(smem <= 160 * 1024, "rms_norm: N too large for LDS staging: ", N);
-  auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(
       (d9d::rms_norm_fwd_kernel<kBlock>), dim3(grid), dim3(kBlock), smem, stream,
       reinterpret_cast<const ushort*>(x.data_ptr()),
@@ -208,10 +318,29 @@ std::vector<torch::Tensor> rms_norm_bwd(
   auto dw = torch::zeros({N}, x.options().dtype(torch::kFloat32));
 
   constexpr int kBlock = 256;
+  auto stream = at::hip::getCurrentHIPStream();
+  if (N <= 512) {
+    const int sgrid = static_cast<int>(std::min<int64_t>((M + 3) / 4, 1024));
+    const size_t ssmem = N * sizeof(float);
+#define SMALL_BWD(VPL)                                                         \
+  hipLaunchKernelGGL((d9d::rms_norm_bwd_smalln_kernel<kBlock, VPL>),           \
+                     dim3(sgrid), dim3(kBlock), ssmem, stream,                 \
+                     reinterpret_cast<const ushort*>(x.data_ptr()),            \
+                     reinterpret_cast<const ushort*>(w.data_ptr()),            \
+                     reinterpret_cast<const ushort*>(dy.data_ptr()),           \
+                     inv_rms.data_ptr<float>(),                                \
+                     reinterpret_cast<ushort*>(dx.data_ptr()),                 \
+                     dw.data_ptr<float>(), M, N, zero_centered ? 1.0f : 0.0f)
+    if (N <= 64) SMALL_BWD(1);
+    else if (N <= 128) SMALL_BWD(2);
+    else if (N <= 256) SMALL_BWD(4);
+    else SMALL_BWD(8);
+#undef SMALL_BWD
+    return {dx, dw};
+  }
   const int grid = static_cast<int>(std::min<int64_t>(M, 1024));
   const size_t smem = N * sizeof(float) + 16 * sizeof(float);
   TORCH_CHECK(smem <= 160 * 1024, "rms_norm bwd: N too large for LDS dw: ", N);
-  auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(
       (d9d::rms_norm_bwd_kernel<kBlock>), dim3(grid), dim3(kBlock), smem, stream,
       reinterpret_cast<const ushort*>(x.data_ptr()),
