@@ -321,7 +321,7 @@ __global__ void attn_delta_kernel(
 }
 
 template <int D>
-__global__ __launch_bounds__(256, 1) void flash_bwd_kernel(
+__global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     const bf16_t* __restrict__ q,
     const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v,
@@ -339,10 +339,9 @@ __global__ __launch_bounds__(256, 1) void flash_bwd_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);   // [64][D] swizzled rows
-  bf16_t* qt_lds = q_lds + kQBlk * D;                // [D][64] swizzled
-  bf16_t* do_lds = qt_lds + D * kQBlk;               // [64][D]
-  bf16_t* dot_lds = do_lds + kQBlk * D;              // [D][64]
-  bf16_t* kt_lds = dot_lds + D * kQBlk;              // [D][64]
+  bf16_t* do_lds = q_lds + kQBlk * D;                // [64][D]
+  bf16_t* t_lds = do_lds + kQBlk * D;                // [D][64]: dO^T then Q^T
+  bf16_t* kt_lds = t_lds + D * kQBlk;                // [D][64]
   bf16_t* x_lds = kt_lds + D * kKvBlk;               // [64][64+8] shared scratch
 
   const int lane = threadIdx.x & 63;
@@ -430,9 +429,7 @@ __global__ __launch_bounds__(256, 1) void flash_bwd_kernel(
           const int d = col + j;
           const int tbyte = (row * 2) ^ ((d & 7) << 4);
           *reinterpret_cast<bf16_t*>(
-              reinterpret_cast<char*>(qt_lds) + d * (kQBlk * 2) + tbyte) = qv[j];
-          *reinterpret_cast<bf16_t*>(
-              reinterpret_cast<char*>(dot_lds) + d * (kQBlk * 2) + tbyte) = dv8[j];
+              reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + tbyte) = dv8[j];
         }
       }
     }
@@ -515,7 +512,7 @@ __global__ __launch_bounds__(256, 1) void flash_bwd_kernel(
           const int d = nt * 16 + (lane & 15);
           const int byte = (q_off * 2) ^ ((d & 7) << 4);
           const bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(dot_lds) + d * (kQBlk * 2) + byte);
+              reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + byte);
           dv_acc[nt] = mfma16(pa, dob, dv_acc[nt]);
         }
       }
@@ -533,9 +530,26 @@ __global__ __launch_bounds__(256, 1) void flash_bwd_kernel(
       }
     }
 
-    // ---- dK += dS^T @ Q  (A = dS^T via x_lds, B = Q^T) --------------------
+    // ---- dK += dS^T @ Q  (A = dS^T via x_lds, B = Q^T re-staged) ----------
     {
-      __syncthreads();  // x_lds reuse
+      __syncthreads();  // x_lds + t_lds reuse
+      {
+        constexpr int elems = kQBlk * D;
+        for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
+          const int row = idx / D;
+          const int col = idx % D;
+          const int g_row = min(qt + row, Sq - 1);
+          const bf16x8 qv = *reinterpret_cast<const bf16x8*>(
+              q + q_base + (int64_t)g_row * q_row_stride + col);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int d = col + j;
+            const int tbyte = (row * 2) ^ ((d & 7) << 4);
+            *reinterpret_cast<bf16_t*>(
+                reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + tbyte) = qv[j];
+          }
+        }
+      }
       bf16_t* xw = x_lds;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -558,7 +572,7 @@ __global__ __launch_bounds__(256, 1) void flash_bwd_kernel(
           const int d = nt * 16 + (lane & 15);
           const int byte = (q_off * 2) ^ ((d & 7) << 4);
           const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(qt_lds) + d * (kQBlk * 2) + byte);
+              reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + byte);
           dk_acc[nt] = mfma16(da, qb, dk_acc[nt]);
         }
       }
@@ -748,7 +762,7 @@ std::vector<torch::Tensor> flash_attn_bwd(
 
   const dim3 grid((Skv + d9d::kKvBlk - 1) / d9d::kKvBlk, B * Hq);
   const size_t smem =
-      (size_t)(4 * d9d::kQBlk * D_pad + D_pad * d9d::kKvBlk + 64 * (64 + 8)) *
+      (size_t)(3 * d9d::kQBlk * D_pad + D_pad * d9d::kKvBlk + 64 * (64 + 8)) *
       sizeof(__bf16);
 
 #define LAUNCH_BWD(DP)                                                        \

@@ -168,6 +168,14 @@ class _DenseWithHead(nn.Module):
     def stage_info(self) -> PipelineStageInfo:
         return self.model.stage_info
 
+    _last_stage_extra: frozenset[str] = frozenset()
+
+    def pipeline_input_names(self) -> set[str]:
+        names = self.model.pipeline_input_names()
+        if self.model.stage_info.is_last_stage:
+            names |= self._last_stage_extra
+        return names
+
     def init_weights(self) -> None:
         self.reset_parameters()
 
@@ -179,6 +187,8 @@ class _DenseWithHead(nn.Module):
 
 
 class Qwen3DenseForCausalLM(_DenseWithHead):
+    _last_stage_extra = frozenset({"labels"})
+
     def __init__(self, params, stage_info=None, device=None, dtype=None) -> None:
         super().__init__(params, stage_info, device=device, dtype=dtype)
         self.lm_head = (
@@ -218,6 +228,8 @@ class Qwen3DenseForCausalLM(_DenseWithHead):
 
 
 class Qwen3DenseForClassification(_DenseWithHead):
+    _last_stage_extra = frozenset({"pooling_mask"})
+
     def __init__(self, params, stage_info=None, device=None, dtype=None) -> None:
         super().__init__(params, stage_info, device=device, dtype=dtype)
         self.head = (
@@ -249,6 +261,8 @@ class Qwen3DenseForClassification(_DenseWithHead):
 
 
 class Qwen3DenseForEmbedding(_DenseWithHead):
+    _last_stage_extra = frozenset({"pooling_mask"})
+
     def __init__(self, params, stage_info=None, device=None, dtype=None) -> None:
         super().__init__(params, stage_info, device=device, dtype=dtype)
         self.head = (

@@ -150,6 +150,12 @@ class Qwen3MoEModel(nn.Module):
             hidden_states = self.norm(hidden_states)
         return {"hidden_states": hidden_states}
 
+    def pipeline_input_names(self) -> set[str]:
+        names = {"position_ids"}
+        if self.stage_info.is_first_stage:
+            names.add("input_ids")
+        return names
+
     # -- pipelining shape inference (ModuleSupportsPipelining) ----------------
 
     def infer_stage_inputs_from_pipeline_inputs(
@@ -214,6 +220,12 @@ class Qwen3MoEForCausalLM(nn.Module):
     @property
     def stage_info(self) -> PipelineStageInfo:
         return self.model.stage_info
+
+    def pipeline_input_names(self) -> set[str]:
+        names = self.model.pipeline_input_names()
+        if self.model.stage_info.is_last_stage:
+            names.add("labels")
+        return names
 
     def forward(
         self,

@@ -1,0 +1,120 @@
+"""Schedule program builders (reference: d9d/pipelining/infra/schedule/program/).
+
+Each builder returns, for one pp rank, a compute-only `Program`; the factory
+wraps it with communication actions. Stage→rank topology is the LOOP style
+(global stage g lives on rank g % pp, local index g // pp); the V style
+(zig-zag) is used by ZBV/DualPipeV.
+"""
+
+from .actions import Action, ActionKind, Program
+
+
+def loop_stage_to_rank(num_stages: int, pp: int) -> list[int]:
+    return [g % pp for g in range(num_stages)]
+
+
+def v_stage_to_rank(num_stages: int, pp: int) -> list[int]:
+    """Zig-zag: ranks 0..pp-1 then pp-1..0 (requires num_stages = 2k*pp)."""
+    out = []
+    direction = 1
+    r = 0
+    for _ in range(num_stages):
+        out.append(r)
+        nr = r + direction
+        if nr == pp:
+            direction = -1
+            nr = pp - 1
+        elif nr < 0:
+            direction = 1
+            nr = 0
+        r = nr
+    return out
+
+
+def local_stages(stage_to_rank: list[int], rank: int) -> list[int]:
+    """Global stage ids owned by `rank`, in global order."""
+    return [g for g, r in enumerate(stage_to_rank) if r == rank]
+
+
+def build_gpipe(rank: int, pp: int, num_stages: int, num_microbatches: int,
+                forward_only: bool = False) -> Program:
+    """All forwards (per local stage in global order), then all backwards."""
+    owned = local_stages(loop_stage_to_rank(num_stages, pp), rank)
+    prog: Program = []
+    for li, _ in enumerate(owned):
+        for mb in range(num_microbatches):
+            prog.append(Action(ActionKind.FORWARD_COMPUTE, li, mb))
+    if not forward_only:
+        for li in reversed(range(len(owned))):
+            for mb in range(num_microbatches):
+                prog.append(Action(ActionKind.BACKWARD_COMPUTE, li, mb))
+    return prog
+
+
+def build_looped_bfs(rank: int, pp: int, num_stages: int, num_microbatches: int,
+                     forward_only: bool = False) -> Program:
+    """BFS over stages: same as gpipe at program level (interleaving emerges
+    from cross-rank overlap); reference: bfs.py:14."""
+    return build_gpipe(rank, pp, num_stages, num_microbatches, forward_only)
+
+
+def build_1f1b(rank: int, pp: int, num_stages: int, num_microbatches: int,
+               zero_bubble: bool = False) -> Program:
+    """Non-interleaved 1F1B (num_stages == pp). Warmup = pp-1-rank forwards,
+    then steady 1F1B, then drain. zero_bubble: split B into B_in at B's slot
+    and B_w deferred to the tail (ZB1P, reference: interleaved.py)."""
+    assert num_stages == pp, "1F1B requires one stage per rank (use looped_bfs otherwise)"
+    warmup = min(pp - 1 - rank, num_microbatches)
+    prog: Program = []
+    b_kind = ActionKind.BACKWARD_INPUT if zero_bubble else ActionKind.BACKWARD_COMPUTE
+    fwd_mb = 0
+    bwd_mb = 0
+    for _ in range(warmup):
+        prog.append(Action(ActionKind.FORWARD_COMPUTE, 0, fwd_mb))
+        fwd_mb += 1
+    # steady: 1F then 1B
+    while fwd_mb < num_microbatches:
+        prog.append(Action(ActionKind.FORWARD_COMPUTE, 0, fwd_mb))
+        fwd_mb += 1
+        prog.append(Action(b_kind, 0, bwd_mb))
+        if zero_bubble:
+            prog.append(Action(ActionKind.BACKWARD_WEIGHT, 0, bwd_mb))
+        bwd_mb += 1
+    while bwd_mb < num_microbatches:
+        prog.append(Action(b_kind, 0, bwd_mb))
+        if zero_bubble:
+            prog.append(Action(ActionKind.BACKWARD_WEIGHT, 0, bwd_mb))
+        bwd_mb += 1
+    return prog
+
+
+def build_zb1p(rank: int, pp: int, num_stages: int, num_microbatches: int) -> Program:
+    """ZB1P: 1F1B with input/weight-split backward; weight grads deferred so
+    d(input) propagates to the previous stage as early as possible."""
+    assert num_stages == pp
+    warmup = min(pp - 1 - rank, num_microbatches)
+    prog: Program = []
+    fwd_mb = 0
+    bwd_mb = 0
+    w_queue: list[int] = []
+    for _ in range(warmup):
+        prog.append(Action(ActionKind.FORWARD_COMPUTE, 0, fwd_mb))
+        fwd_mb += 1
+    while fwd_mb < num_microbatches:
+        prog.append(Action(ActionKind.FORWARD_COMPUTE, 0, fwd_mb))
+        fwd_mb += 1
+        prog.append(Action(ActionKind.BACKWARD_INPUT, 0, bwd_mb))
+        w_queue.append(bwd_mb)
+        bwd_mb += 1
+        # fill the bubble on late ranks: flush one deferred W per steady slot
+        if rank >= pp - 1 and w_queue:
+            prog.append(Action(ActionKind.BACKWARD_WEIGHT, 0, w_queue.pop(0)))
+    while bwd_mb < num_microbatches:
+        prog.append(Action(ActionKind.BACKWARD_INPUT, 0, bwd_mb))
+        w_queue.append(bwd_mb)
+        bwd_mb += 1
+        if w_queue:
+            prog.append(Action(ActionKind.BACKWARD_WEIGHT, 0, w_queue.pop(0)))
+    for mb in w_queue:
+        prog.append(Action(ActionKind.BACKWARD_WEIGHT, 0, mb))
+    return prog
