@@ -1,0 +1,121 @@
+"""model_state mapper DAG + streaming safetensors IO tests."""
+
+import torch
+from torch import nn
+
+from d9d_amd.model_state import (
+    ChunkTensors,
+    ConcatenateTensors,
+    Identity,
+    Parallel,
+    PrefixScope,
+    Rename,
+    Sequential,
+    Shard,
+    StackTensors,
+    Transpose,
+    UnstackTensors,
+    identity_mapper_from_module,
+    load_model_state,
+    read_model_state,
+    save_module_state,
+    write_model_state,
+)
+
+
+def test_leaf_mappers():
+    t = torch.randn(3, 4)
+    assert torch.equal(Rename("a", "b").apply({"a": t})["b"], t)
+    assert torch.equal(
+        Transpose("a").apply({"a": t})["a"], t.t().contiguous()
+    )
+    stack = StackTensors(["x0", "x1"], "x", dim=0)
+    out = stack.apply({"x0": t, "x1": t + 1})
+    assert out["x"].shape == (2, 3, 4)
+    unstack = UnstackTensors("x", ["x0", "x1"], dim=0)
+    back = unstack.apply(out)
+    assert torch.equal(back["x0"], t) and torch.equal(back["x1"], t + 1)
+
+
+def test_sequential_chains_groups():
+    # rename then stack: net group {a.0, a.1} -> {stacked}
+    m = Sequential(
+        Parallel(Rename("a.0", "b.0"), Rename("a.1", "b.1")),
+        StackTensors(["b.0", "b.1"], "stacked"),
+    )
+    groups = m.state_dependency_groups()
+    assert len(groups) == 1
+    g = groups[0]
+    assert g.inputs == frozenset({"a.0", "a.1"})
+    assert g.outputs == frozenset({"stacked"})
+    t0, t1 = torch.randn(2), torch.randn(2)
+    out = m.apply({"a.0": t0, "a.1": t1})
+    assert torch.equal(out["stacked"], torch.stack([t0, t1]))
+
+
+def test_sequential_gap_fill_passthrough():
+    m = Sequential(
+        Parallel(Identity("x"), Identity("y")),
+        Rename("x", "z"),  # y passes through untouched
+    )
+    out = m.apply({"x": torch.ones(1), "y": torch.zeros(1)})
+    assert set(out) == {"z", "y"}
+
+
+def test_prefix_scope_and_shard():
+    inner = Parallel(Rename("w", "weight"), Rename("b", "bias"))
+    scoped = PrefixScope("layer0.", inner)
+    out = scoped.apply({"layer0.w": torch.ones(1), "layer0.b": torch.zeros(1)})
+    assert set(out) == {"layer0.weight", "layer0.bias"}
+
+    sharded0 = Shard(inner, rank=0, world_size=2)
+    sharded1 = Shard(inner, rank=1, world_size=2)
+    g0 = sharded0.state_dependency_groups()
+    g1 = sharded1.state_dependency_groups()
+    assert len(g0) == 1 and len(g1) == 1 and g0 != g1
+
+
+def test_write_read_roundtrip(tmp_path):
+    state = {f"k{i}": torch.randn(16, 16) for i in range(5)}
+    mapper = Parallel(*[Identity(k) for k in state])
+    write_model_state(mapper, dict(state), tmp_path, shard_size_gb=1e-6)  # tiny shards
+    files = list(tmp_path.glob("*.safetensors"))
+    assert len(files) >= 2  # forced multi-shard
+    loaded = dict(read_model_state(tmp_path))
+    assert set(loaded) == set(state)
+    for k in state:
+        torch.testing.assert_close(loaded[k], state[k])
+
+
+def test_module_save_load_roundtrip(tmp_path):
+    torch.manual_seed(0)
+    m1 = nn.Sequential(nn.Linear(4, 8), nn.Linear(8, 2))
+    save_module_state(m1, tmp_path)
+    m2 = nn.Sequential(nn.Linear(4, 8), nn.Linear(8, 2))
+    loaded = load_model_state(m2, tmp_path, strict=True)
+    assert len(loaded) == 4
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        torch.testing.assert_close(p1, p2)
+
+
+def test_transform_on_write(tmp_path):
+    # experts stored separately -> stacked on export (HF MODULE_LIST format flavor)
+    experts = {f"expert.{i}.w": torch.randn(4, 4) for i in range(3)}
+    mapper = StackTensors([f"expert.{i}.w" for i in range(3)], "experts.w", dim=0)
+    write_model_state(mapper, dict(experts), tmp_path)
+    loaded = dict(read_model_state(tmp_path))
+    assert loaded["experts.w"].shape == (3, 4, 4)
+    # and the reverse import path
+    back = ChunkTensors("experts.w", [f"expert.{i}.w" for i in range(3)], dim=0)
+    state = back.apply(loaded)
+    for i in range(3):
+        torch.testing.assert_close(
+            state[f"expert.{i}.w"].squeeze(0), experts[f"expert.{i}.w"]
+        )
+
+
+def test_identity_mapper_from_module():
+    m = nn.Linear(3, 3)
+    mapper = identity_mapper_from_module(m)
+    groups = mapper.state_dependency_groups()
+    assert len(groups) == 2
