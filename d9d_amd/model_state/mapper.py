@@ -154,6 +154,26 @@ class ChunkTensors(ModelStateMapper):
         return {d: p.contiguous() for d, p in zip(self.dsts, parts)}
 
 
+class SliceRows(ModelStateMapper):
+    """Split one tensor into named variable-size row slices (e.g. the
+    regular/special vocab split of SplitTokenEmbeddings)."""
+
+    def __init__(self, src: str, dsts: list[tuple[str, int]], dim: int = 0) -> None:
+        self.src, self.dsts, self.dim = src, list(dsts), dim
+
+    def state_dependency_groups(self):
+        return [StateGroup.of([self.src], [d for d, _ in self.dsts])]
+
+    def apply_group(self, group, tensors):
+        t = tensors[self.src]
+        out = {}
+        off = 0
+        for name, size in self.dsts:
+            out[name] = t.narrow(self.dim, off, size).contiguous()
+            off += size
+        return out
+
+
 class Distribute(_SingleTensor):
     """Full local tensor -> DTensor, sliced locally (no communication;
     reference: leaf/dtensor.py Distribute with src_data_rank=None)."""

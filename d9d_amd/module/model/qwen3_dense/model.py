@@ -129,6 +129,12 @@ class Qwen3DenseModel(nn.Module):
             hidden_states = self.norm(hidden_states)
         return {"hidden_states": hidden_states}
 
+    def pipeline_input_names(self) -> set[str]:
+        names = {"position_ids"}
+        if self.stage_info.is_first_stage:
+            names.add("input_ids")
+        return names
+
     def infer_stage_inputs_from_pipeline_inputs(
         self, pipeline_inputs: dict[str, Any], num_microbatches: int
     ) -> dict[str, torch.Tensor]:

@@ -48,7 +48,9 @@ class PipelineScheduleExecutor:
         pp_rank: int,
         group=None,
         input_spec=None,
+        forward_only: bool = False,
     ) -> None:
+        self.forward_only = forward_only
         self.stages = stages
         self.global_stage_ids = global_stage_ids
         self.rank_of_stage = rank_of_stage
@@ -167,6 +169,9 @@ class PipelineScheduleExecutor:
                 if loss is not None:
                     loss_cache[(s, mb)] = loss
                     losses.append(loss.detach())
+            if self.forward_only:
+                stage._input_cache.pop(mb, None)
+                stage._output_cache.pop(mb, None)
 
         elif kind is ActionKind.FORWARD_SEND:
             nxt = self._next_info(s)
@@ -219,12 +224,13 @@ class OfflinePipelineExecutor:
     """Single-process fallback: runs all stages sequentially (reference: offline.py)."""
 
     def __init__(self, stages: list[PipelineStage], num_microbatches: int,
-                 input_spec=None) -> None:
+                 input_spec=None, forward_only: bool = False) -> None:
         self.stages = stages
         self.num_microbatches = num_microbatches
         self.input_spec = input_spec or SpecShard(dim=0)
         self.has_first_stage = True
         self.has_last_stage = True
+        self.forward_only = forward_only
 
     def configure_buffers(self, pipeline_inputs: dict[str, Any]) -> None:
         for stage in self.stages:
@@ -256,6 +262,11 @@ class OfflinePipelineExecutor:
             loss = loss_fn(mb, outs, microbatches[mb]) if loss_fn else None
             if loss is not None:
                 losses.append(loss.detach())
+            if self.forward_only:
+                for stage in self.stages:
+                    stage._input_cache.pop(mb, None)
+                    stage._output_cache.pop(mb, None)
+                continue
             # backward through all stages via autograd chain
             grads = None
             for si in reversed(range(len(self.stages))):

@@ -96,12 +96,14 @@ def build_schedule(
         modules.append(module)
         stages.append(PipelineStage(module, g, num_stages, device))
 
+    forward_only = config.schedule == "inference"
     if pp_size == 1:
-        schedule = OfflinePipelineExecutor(stages, num_microbatches, input_spec)
+        schedule = OfflinePipelineExecutor(
+            stages, num_microbatches, input_spec, forward_only=forward_only
+        )
         return PipelineScheduleInfo(schedule, stages, modules, True, True)
 
     name = config.schedule
-    forward_only = name == "inference"
     if name in ("gpipe", "looped_bfs", "inference"):
         prog = build_looped_bfs(pp_rank, pp_size, num_stages, num_microbatches, forward_only)
     elif name == "1f1b":
@@ -131,6 +133,7 @@ def build_schedule(
         pp_rank=pp_rank,
         group=pp_group,
         input_spec=input_spec,
+        forward_only=forward_only,
     )
     return PipelineScheduleInfo(
         schedule, stages, modules,

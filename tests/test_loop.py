@@ -1,0 +1,157 @@
+"""End-to-end Trainer tests on CPU: local mode + gloo ws=2 DP replicate."""
+
+import pytest
+import torch
+from torch.utils.data import Dataset
+
+from d9d_amd.core.dist_context import DeviceMeshParameters
+from d9d_amd.loop import TrainingConfigurator, TrainerConfig
+from d9d_amd.loop.auto import (
+    AutoLRSchedulerProvider,
+    AutoOptimizerProvider,
+    LRSchedulerConfig,
+    OptimizerConfig,
+)
+from d9d_amd.loop.config import BatchingConfig, CheckpointingConfig
+from d9d_amd.loop.control import DatasetProvider, ModelProvider, TrainTask
+from d9d_amd.metric import WeightedMeanMetric
+from d9d_amd.module.model.qwen3_dense import (
+    Qwen3DenseForCausalLM,
+    Qwen3DenseModelParameters,
+)
+from tests.helpers import run_distributed
+
+
+class _SyntheticLM(Dataset):
+    def __init__(self, vocab, seq, n=64, seed=0):
+        g = torch.Generator().manual_seed(seed)
+        self.data = torch.randint(0, vocab, (n, seq + 1), generator=g)
+
+    def __len__(self):
+        return len(self.data)
+
+    def __getitem__(self, i):
+        return self.data[i]
+
+
+class _LMDatasetProvider(DatasetProvider):
+    def __init__(self, params):
+        self.params = params
+
+    def build_dataset(self, ctx):
+        return _SyntheticLM(self.params.vocab_size, 32)
+
+
+class _LMModelProvider(ModelProvider):
+    def __init__(self, params, parallelize=None):
+        self.params = params
+        self.parallelize = parallelize
+
+    def initialize_model_stage(self, stage_info):
+        return Qwen3DenseForCausalLM(self.params, stage_info)
+
+    def parallelize_model_stage(self, module, ctx):
+        if self.parallelize:
+            return self.parallelize(module, ctx)
+        return module
+
+
+class _LMTask(TrainTask):
+    def build_forward_inputs(self, batch):
+        return {"input_ids": batch[:, :-1], "labels": batch[:, 1:]}
+
+    def compute_loss(self, outputs, mb_inputs):
+        return outputs["loss"].mean(), 1.0
+
+    def create_metrics(self):
+        return {"train_loss": WeightedMeanMetric()}
+
+    def update_metrics(self, metrics, outputs, mb_inputs):
+        metrics["train_loss"].update(outputs["loss"].detach().mean(), 1.0)
+
+
+def _make_config(total_steps=3, save_dir=None):
+    return TrainerConfig(
+        batching=BatchingConfig(global_batch_size=8, microbatch_size=4),
+        total_steps=total_steps,
+        checkpointing=CheckpointingConfig(
+            save_dir=save_dir, period_steps=2, num_to_keep=2
+        ),
+    )
+
+
+def _build_trainer(total_steps=3, save_dir=None, parallelize=None, mesh=None):
+    params = Qwen3DenseModelParameters.tiny()
+    return TrainingConfigurator(
+        _make_config(total_steps, save_dir),
+        mesh or DeviceMeshParameters(),
+        _LMModelProvider(params, parallelize),
+        _LMDatasetProvider(params),
+        AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+        AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+        _LMTask(),
+    ).configure(device_type="cpu")
+
+
+def test_trainer_local_runs_and_loss_decreases():
+    trainer = _build_trainer(total_steps=5)
+    trainer.train()
+    assert trainer.stepper.step == 5
+    assert trainer.last_losses  # last-stage rank collects losses
+
+
+def test_trainer_checkpoint_resume(tmp_path):
+    trainer = _build_trainer(total_steps=2, save_dir=str(tmp_path))
+    trainer.train()
+    assert trainer.checkpointer.existing_checkpoints() == [2]
+
+    trainer2 = _build_trainer(total_steps=4, save_dir=str(tmp_path))
+    trainer2.train()
+    assert trainer2.stepper.step == 4
+    # resumed from 2 then trained 2 more, checkpointed at 4
+    assert 4 in trainer2.checkpointer.existing_checkpoints()
+
+
+def test_trainer_sleep_wake():
+    trainer = _build_trainer(total_steps=1)
+    trainer.train()
+    trainer.grad_manager.install()
+    trainer.sleep()
+    assert trainer.is_sleeping
+    trainer.wake()
+    assert not trainer.is_sleeping
+    trainer.grad_manager.uninstall()
+
+
+def test_trainer_export(tmp_path):
+    trainer = _build_trainer(total_steps=1)
+    trainer.train()
+    trainer.export(str(tmp_path / "export"))
+    from d9d_amd.model_state import read_model_state
+
+    keys = dict(read_model_state(tmp_path / "export"))
+    assert any("lm_head" in k for k in keys)
+
+
+def _dp2_trainer(rank, world_size):
+    from d9d_amd.parallel import parallelize_replicate
+
+    def parallelize(module, ctx):
+        return parallelize_replicate(module, ctx.mesh_for("dense"))
+
+    mesh = DeviceMeshParameters(data_parallel_replicate=2)
+    trainer = _build_trainer(total_steps=2, parallelize=parallelize, mesh=mesh)
+    trainer.train()
+    # replicated params must stay identical across ranks after optimizer steps
+    p = next(iter(trainer.modules_by_key.values())).lm_head.weights["regular"]
+    from torch.distributed.tensor import DTensor
+
+    local = p.to_local() if isinstance(p, DTensor) else p
+    checksum = local.float().sum().item()
+    return round(checksum, 4)
+
+
+@pytest.mark.distributed
+def test_trainer_dp2_replicate_consistent():
+    results = run_distributed(_dp2_trainer, world_size=2, timeout=300)
+    assert results[0] == results[1]
