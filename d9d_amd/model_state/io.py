@@ -81,8 +81,9 @@ class _StreamingApplier:
                     g, {k: self.pending[k] for k in g.inputs}
                 )
                 yield outs
-        # evict inputs nothing else needs
-        still_needed = {k for rem_g in self.remaining.values() for k in rem_g}
+        # evict inputs no UNFIRED group needs (arrived inputs of pending
+        # groups must stay resident until the group fires)
+        still_needed = {k for g in self.remaining for k in g.inputs}
         for k in list(self.pending):
             if k not in still_needed:
                 del self.pending[k]
