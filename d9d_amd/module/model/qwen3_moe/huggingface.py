@@ -1,21 +1,20 @@
 """HF <-> d9d_amd state mappers for Qwen3-MoE (reference: qwen3_moe/huggingface.py).
 
-HF layout (transformers Qwen3MoeForCausalLM):
-  model.embed_tokens.weight (V, H)
-  model.layers.N.self_attn.{q,k,v,o}_proj.weight, {q,k}_norm.weight
-  model.layers.N.{input,post_attention}_layernorm.weight
-  model.layers.N.mlp.gate.weight (E, H)
-  model.layers.N.mlp.experts.M.{gate,up,down}_proj.weight  (MODULE_LIST format)
-  model.norm.weight, lm_head.weight (V, H)
+Two HF expert formats are supported (reference: huggingface.py MODULE_LIST and
+FUSED formats):
+  * "module_list": model.layers.N.mlp.experts.M.{gate,up,down}_proj.weight,
+    each (out, in) -> transpose + stack into our (E, in, out).
+  * "fused" (transformers >= 5.x packed experts):
+    experts.gate_up_proj (E, 2I, H) and experts.down_proj (E, H, I)
+    -> chunk gate/up on dim 1, transpose (1, 2).
 
-d9d_amd layout differences:
+Common differences from HF:
   * split vocab: embed_tokens.embeddings.{seg}.weight / lm_head.weights.{seg}
-  * stacked experts: mlp.experts.{gate,up,down}_proj.weight (E, in, out)
-    (HF per-expert weights are (out, in) -> transpose then stack)
-  * router: mlp.router.gate.weight
+  * router: mlp.router.gate.weight (HF: mlp.gate.weight)
 """
 
 from ....model_state.mapper import (
+    StateGroup,
     ConcatenateTensors,
     Identity,
     ModelStateMapper,
@@ -30,11 +29,77 @@ from ....model_state.mapper import (
 from .params import Qwen3MoEModelParameters
 
 
+class _FusedToStacked(ModelStateMapper):
+    """HF packed experts -> our stacked (E, in, out) weights for one layer."""
+
+    def __init__(self, prefix: str, intermediate: int) -> None:
+        self.prefix = prefix
+        self.intermediate = intermediate
+
+    def state_dependency_groups(self):
+        pre = self.prefix
+        return [
+            StateGroup.of(
+                [pre + "mlp.experts.gate_up_proj"],
+                [pre + "mlp.experts.gate_proj.weight", pre + "mlp.experts.up_proj.weight"],
+            ),
+            StateGroup.of(
+                [pre + "mlp.experts.down_proj"],
+                [pre + "mlp.experts.down_proj.weight"],
+            ),
+        ]
+
+    def apply_group(self, group, tensors):
+        pre = self.prefix
+        if pre + "mlp.experts.gate_up_proj" in tensors:
+            gu = tensors[pre + "mlp.experts.gate_up_proj"]  # (E, 2I, H)
+            gate = gu[:, : self.intermediate, :].transpose(1, 2).contiguous()
+            up = gu[:, self.intermediate :, :].transpose(1, 2).contiguous()
+            return {
+                pre + "mlp.experts.gate_proj.weight": gate,
+                pre + "mlp.experts.up_proj.weight": up,
+            }
+        down = tensors[pre + "mlp.experts.down_proj"]  # (E, H, I)
+        return {pre + "mlp.experts.down_proj.weight": down.transpose(1, 2).contiguous()}
+
+
+class _StackedToFused(ModelStateMapper):
+    def __init__(self, prefix: str, intermediate: int) -> None:
+        self.prefix = prefix
+        self.intermediate = intermediate
+
+    def state_dependency_groups(self):
+        pre = self.prefix
+        return [
+            StateGroup.of(
+                [pre + "mlp.experts.gate_proj.weight", pre + "mlp.experts.up_proj.weight"],
+                [pre + "mlp.experts.gate_up_proj"],
+            ),
+            StateGroup.of(
+                [pre + "mlp.experts.down_proj.weight"],
+                [pre + "mlp.experts.down_proj"],
+            ),
+        ]
+
+    def apply_group(self, group, tensors):
+        import torch
+
+        pre = self.prefix
+        if pre + "mlp.experts.gate_proj.weight" in tensors:
+            gate = tensors[pre + "mlp.experts.gate_proj.weight"].transpose(1, 2)
+            up = tensors[pre + "mlp.experts.up_proj.weight"].transpose(1, 2)
+            return {pre + "mlp.experts.gate_up_proj": torch.cat([gate, up], dim=1).contiguous()}
+        down = tensors[pre + "mlp.experts.down_proj.weight"]
+        return {pre + "mlp.experts.down_proj": down.transpose(1, 2).contiguous()}
+
+
 def _vocab_splits(p: Qwen3MoEModelParameters) -> list[tuple[str, int]]:
     return [(name, p.split_vocab_size[name]) for name in p.split_vocab_order]
 
 
-def hf_to_d9d_mapper(p: Qwen3MoEModelParameters) -> ModelStateMapper:
+def hf_to_d9d_mapper(
+    p: Qwen3MoEModelParameters, expert_format: str = "fused"
+) -> ModelStateMapper:
     mappers: list[ModelStateMapper] = []
 
     mappers.append(
@@ -68,29 +133,34 @@ def hf_to_d9d_mapper(p: Qwen3MoEModelParameters) -> ModelStateMapper:
         ):
             mappers.append(Identity(pre + key))
         mappers.append(Rename(pre + "mlp.gate.weight", pre + "mlp.router.gate.weight"))
-        for proj in ("gate_proj", "up_proj", "down_proj"):
-            # HF expert weight (out, in) -> transpose (in, out) -> stack (E, in, out)
-            per_expert = [
-                Transpose(
-                    f"{pre}mlp.experts.{e}.{proj}.weight",
-                    f"{pre}mlp.experts._t{e}.{proj}",
+        if expert_format == "fused":
+            mappers.append(_FusedToStacked(pre, p.intermediate_size))
+        else:
+            for proj in ("gate_proj", "up_proj", "down_proj"):
+                # HF expert weight (out, in) -> transpose (in, out) -> stack
+                per_expert = [
+                    Transpose(
+                        f"{pre}mlp.experts.{e}.{proj}.weight",
+                        f"{pre}mlp.experts._t{e}.{proj}",
+                    )
+                    for e in range(p.num_experts)
+                ]
+                mappers.append(
+                    Sequential(
+                        Parallel(*per_expert),
+                        StackTensors(
+                            [f"{pre}mlp.experts._t{e}.{proj}" for e in range(p.num_experts)],
+                            f"{pre}mlp.experts.{proj}.weight",
+                            dim=0,
+                        ),
+                    )
                 )
-                for e in range(p.num_experts)
-            ]
-            mappers.append(
-                Sequential(
-                    Parallel(*per_expert),
-                    StackTensors(
-                        [f"{pre}mlp.experts._t{e}.{proj}" for e in range(p.num_experts)],
-                        f"{pre}mlp.experts.{proj}.weight",
-                        dim=0,
-                    ),
-                )
-            )
     return Parallel(*mappers)
 
 
-def d9d_to_hf_mapper(p: Qwen3MoEModelParameters) -> ModelStateMapper:
+def d9d_to_hf_mapper(
+    p: Qwen3MoEModelParameters, expert_format: str = "fused"
+) -> ModelStateMapper:
     mappers: list[ModelStateMapper] = []
 
     mappers.append(
@@ -124,18 +194,21 @@ def d9d_to_hf_mapper(p: Qwen3MoEModelParameters) -> ModelStateMapper:
         ):
             mappers.append(Identity(pre + key))
         mappers.append(Rename(pre + "mlp.router.gate.weight", pre + "mlp.gate.weight"))
-        for proj in ("gate_proj", "up_proj", "down_proj"):
-            unstack = UnstackTensors(
-                f"{pre}mlp.experts.{proj}.weight",
-                [f"{pre}mlp.experts._t{e}.{proj}" for e in range(p.num_experts)],
-                dim=0,
-            )
-            per_expert = [
-                Transpose(
-                    f"{pre}mlp.experts._t{e}.{proj}",
-                    f"{pre}mlp.experts.{e}.{proj}.weight",
+        if expert_format == "fused":
+            mappers.append(_StackedToFused(pre, p.intermediate_size))
+        else:
+            for proj in ("gate_proj", "up_proj", "down_proj"):
+                unstack = UnstackTensors(
+                    f"{pre}mlp.experts.{proj}.weight",
+                    [f"{pre}mlp.experts._t{e}.{proj}" for e in range(p.num_experts)],
+                    dim=0,
                 )
-                for e in range(p.num_experts)
-            ]
-            mappers.append(Sequential(unstack, Parallel(*per_expert)))
+                per_expert = [
+                    Transpose(
+                        f"{pre}mlp.experts._t{e}.{proj}",
+                        f"{pre}mlp.experts.{e}.{proj}.weight",
+                    )
+                    for e in range(p.num_experts)
+                ]
+                mappers.append(Sequential(unstack, Parallel(*per_expert)))
     return Parallel(*mappers)
