@@ -59,7 +59,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     bf16_t* __restrict__ out,       // (B,Sq,Hq,D)
     float* __restrict__ lse,        // (B,Hq,Sq)
     int B, int Sq, int Skv, int Hq, int Hkv,
-    float scale, int causal, int window_left) {
+    float scale, int causal, int window_left, int q_offset) {
   constexpr int kNT = D / 16;   // n-tiles over head dim
   constexpr int kKS = D / 32;   // k-steps over head dim
   constexpr int kRowBytes = D * 2;
@@ -109,7 +109,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
 
   const int q_tile_last_row = min(q_tile * kQBlk + kQBlk - 1, Sq - 1);
   int kv_end = Skv;
-  if (causal) kv_end = min(Skv, q_tile_last_row + 1);
+  if (causal) kv_end = min(Skv, q_tile_last_row + q_offset + 1);
   const int num_kv_tiles = (kv_end + kKvBlk - 1) / kKvBlk;
 
   for (int kt = 0; kt < num_kv_tiles; ++kt) {
@@ -174,7 +174,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       const int col = kv0 + nt * 16 + (lane & 15);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = my_q_row + (lane >> 4) * 4 + r;
+        const int row = my_q_row + (lane >> 4) * 4 + r + q_offset;
         float s = s_acc[nt][r] * scale;
         bool masked = col >= Skv;
         if (causal) masked |= col > row;
@@ -332,7 +332,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     float* __restrict__ dk,           // (B,Skv,Hkv,D) fp32 accum
     float* __restrict__ dv,           // (B,Skv,Hkv,D) fp32 accum
     int B, int Sq, int Skv, int Hq, int Hkv,
-    float scale, int causal, int window_left) {
+    float scale, int causal, int window_left, int q_offset) {
   constexpr int kNT = D / 16;
   constexpr int kKS = D / 32;
   constexpr int kRowBytes = D * 2;
@@ -403,7 +403,10 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   }
 
   int q_start = 0;
-  if (causal) q_start = (kv0 / kQBlk) * kQBlk;  // first q tile that can see kv0
+  if (causal) {
+    // first q tile whose last GLOBAL position reaches kv0
+    q_start = (max(kv0 - q_offset, 0) / kQBlk) * kQBlk;
+  }
   if (window_left >= 0) q_start = max(q_start, 0);
 
   for (int qt = q_start; qt < Sq; qt += kQBlk) {
@@ -463,8 +466,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
         const int kv_glob = kv0 + wave * 16 + (lane >> 4) * 4 + r;
         float s = st_acc[nt][r] * scale;
         bool masked = (kv_glob >= Skv) || (q_glob >= Sq);
-        if (causal) masked |= kv_glob > q_glob;
-        if (window_left >= 0) masked |= kv_glob < q_glob - window_left;
+        if (causal) masked |= kv_glob > q_glob + q_offset;
+        if (window_left >= 0) masked |= kv_glob < q_glob + q_offset - window_left;
         pt_val[nt][r] =
             masked ? 0.f : __builtin_amdgcn_exp2f((s - l) * kLog2e);
       }
@@ -689,7 +692,7 @@ torch::Tensor maybe_pad_d(torch::Tensor t, int D_pad) {
 
 std::vector<torch::Tensor> flash_attn_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
-    bool causal, double softmax_scale, int64_t window_left) {
+    bool causal, double softmax_scale, int64_t window_left, int64_t q_offset) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
   const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
@@ -718,7 +721,8 @@ std::vector<torch::Tensor> flash_attn_fwd(
                      reinterpret_cast<const __bf16*>(vp.data_ptr()),          \
                      reinterpret_cast<__bf16*>(out.data_ptr()),               \
                      lse.data_ptr<float>(), B, Sq, Skv, Hq, Hkv,              \
-                     (float)softmax_scale, causal ? 1 : 0, (int)window_left)
+                     (float)softmax_scale, causal ? 1 : 0, (int)window_left,   \
+                     (int)q_offset)
   switch (D_pad) {
     case 32: LAUNCH_FWD(32); break;
     case 64: LAUNCH_FWD(64); break;
@@ -734,7 +738,7 @@ std::vector<torch::Tensor> flash_attn_fwd(
 std::vector<torch::Tensor> flash_attn_bwd(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor out, torch::Tensor lse,
-    bool causal, double softmax_scale, int64_t window_left) {
+    bool causal, double softmax_scale, int64_t window_left, int64_t q_offset) {
   const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Skv = k.size(1), Hkv = k.size(2);
   const int D_pad = padded_head_dim(D);
@@ -775,7 +779,8 @@ std::vector<torch::Tensor> flash_attn_bwd(
                      lse.data_ptr<float>(), delta.data_ptr<float>(),          \
                      dq32.data_ptr<float>(), dk32.data_ptr<float>(),          \
                      dv32.data_ptr<float>(), B, Sq, Skv, Hq, Hkv,             \
-                     (float)softmax_scale, causal ? 1 : 0, (int)window_left)
+                     (float)softmax_scale, causal ? 1 : 0, (int)window_left,   \
+                     (int)q_offset)
   switch (D_pad) {
     case 32: LAUNCH_BWD(32); break;
     case 64: LAUNCH_BWD(64); break;

@@ -68,6 +68,13 @@ class GroupedQueryAttention(nn.Module):
         else:
             self.sinks = None
 
+        # Context parallelism: installed by parallelize_context_parallel.
+        # When set, K/V are all-gathered along the sequence over the cp group
+        # and the local Q block attends with its global position offset.
+        self._cp_group = None
+        self._cp_rank = 0
+        self._cp_size = 1
+
     def reset_parameters(self) -> None:
         with torch.no_grad():
             for lin in (self.q_proj, self.k_proj, self.v_proj, self.o_proj):
@@ -102,8 +109,16 @@ class GroupedQueryAttention(nn.Module):
         k = apply_rotary_emb(k, cos, sin, self.rope_layout)
 
         window = (-1, -1) if self.sliding_window is None else (self.sliding_window, -1)
+        q_offset = 0
+        if self._cp_group is not None and self._cp_size > 1:
+            from ....parallel.tensor import _AllGatherSeq
+
+            k = _AllGatherSeq.apply(k, self._cp_group, 1)
+            v = _AllGatherSeq.apply(v, self._cp_group, 1)
+            q_offset = self._cp_rank * S
         attn = flash_attn_func(
-            q, k, v, causal=True, window_size=window, sinks=self.sinks
+            q, k, v, causal=True, window_size=window, sinks=self.sinks,
+            q_offset=q_offset,
         )
         attn = attn.reshape(B, S, self.num_heads * self.head_dim)
         if self.use_output_gate:

@@ -25,8 +25,10 @@ def _eager_attention(
     softmax_scale: float,
     window_size: tuple[int, int],
     sinks: torch.Tensor | None,
+    q_offset: int = 0,
 ):
-    """fp32 eager oracle. Returns (out, lse)."""
+    """fp32 eager oracle. Returns (out, lse). `q_offset` is the global
+    position of q row 0 (context parallel: local q block vs gathered KV)."""
     B, S, Hq, D = q.shape
     Hkv = k.shape[2]
     rep = Hq // Hkv
@@ -40,7 +42,7 @@ def _eager_attention(
 
     scores = torch.matmul(q32, k32.transpose(-1, -2)) * softmax_scale  # (B,Hq,S,Skv)
     Skv = scores.shape[-1]
-    q_pos = torch.arange(S, device=q.device).unsqueeze(1)
+    q_pos = torch.arange(S, device=q.device).unsqueeze(1) + q_offset
     kv_pos = torch.arange(Skv, device=q.device).unsqueeze(0)
     mask = torch.zeros(S, Skv, dtype=torch.bool, device=q.device)
     if causal:
@@ -69,16 +71,17 @@ def _eager_attention(
 
 class _FlashAttnFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, sinks, causal, softmax_scale, window_left):
+    def forward(ctx, q, k, v, sinks, causal, softmax_scale, window_left, q_offset):
         ext = get_ext()
         out, lse = ext.flash_attn_fwd(
             q.contiguous(), k.contiguous(), v.contiguous(),
-            causal, softmax_scale, window_left,
+            causal, softmax_scale, window_left, q_offset,
         )
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.causal = causal
         ctx.softmax_scale = softmax_scale
         ctx.window_left = window_left
+        ctx.q_offset = q_offset
         ctx.has_sinks = sinks is not None
         if ctx.has_sinks:
             raise NotImplementedError("sinks on the HIP path land with the flash kernel v2")
@@ -91,8 +94,9 @@ class _FlashAttnFunction(torch.autograd.Function):
         dq, dk, dv = ext.flash_attn_bwd(
             dout.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(),
             out.contiguous(), lse, ctx.causal, ctx.softmax_scale, ctx.window_left,
+            ctx.q_offset,
         )
-        return dq, dk, dv, None, None, None, None
+        return dq, dk, dv, None, None, None, None, None
 
 
 def flash_attn_func(
@@ -105,14 +109,19 @@ def flash_attn_func(
     window_size: tuple[int, int] = (-1, -1),
     sinks: torch.Tensor | None = None,
     return_lse: bool = False,
+    q_offset: int = 0,
 ):
     """Scaled-dot-product attention. q (B,S,Hq,D); k/v (B,S,Hkv,D)."""
     if softmax_scale is None:
         softmax_scale = 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda and sinks is None and window_size == (-1, -1):
-        out, lse = _FlashAttnFunction.apply(q, k, v, None, causal, softmax_scale, -1)
+        out, lse = _FlashAttnFunction.apply(
+            q, k, v, None, causal, softmax_scale, -1, q_offset
+        )
     else:
-        out, lse = _eager_attention(q, k, v, causal, softmax_scale, window_size, sinks)
+        out, lse = _eager_attention(
+            q, k, v, causal, softmax_scale, window_size, sinks, q_offset
+        )
     if return_lse:
         return out, lse
     return out
