@@ -91,3 +91,23 @@ def test_flash_attn_long_seq_smoke():
     torch.cuda.synchronize()
     assert torch.isfinite(out.float()).all()
     assert torch.isfinite(q.grad.float()).all()
+
+
+@pytest.mark.gpu
+def test_flash_attn_q_offset_cp_parity():
+    """CP decomposition: local q block + full KV + q_offset == full attention."""
+    device = torch.device("cuda")
+    B, S, Hq, Hkv, D = 2, 256, 8, 2, 128
+    q = torch.randn(B, S, Hq, D, dtype=torch.bfloat16, device=device)
+    k = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device=device)
+    v = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device=device)
+    full = flash_attn_func(q, k, v, causal=True)
+    half = S // 2
+    for r in range(2):
+        part = flash_attn_func(
+            q[:, r * half : (r + 1) * half], k, v, causal=True, q_offset=r * half
+        )
+        torch.testing.assert_close(
+            part.float(), full[:, r * half : (r + 1) * half].float(),
+            rtol=2e-2, atol=2e-2,
+        )
