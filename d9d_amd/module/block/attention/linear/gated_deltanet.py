@@ -1,0 +1,176 @@
+"""Gated DeltaNet linear attention (reference: d9d/module/block/attention/linear/gated_deltanet.py).
+
+QKV projections + causal short depthwise conv, log-sigmoid decay gate,
+beta = sigmoid, GQA head expansion, chunked gated delta rule recurrence,
+per-head RMSNorm and silu-mul output gate.
+
+The chunked delta-rule recurrence (replacing fla-core's CUDA/Triton
+`chunk_gated_delta_rule`) is implemented as a torch chunked scan — matmuls
+land on MFMA via rocBLAS; a fused HIP kernel is the planned follow-up.
+"""
+
+import math
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from .....ops import silu_mul
+from ...normalization import RMSNorm
+
+
+class CausalShortDepthwiseConv1d(nn.Module):
+    """Per-channel causal conv over the sequence (reference uses
+    fla.modules.conv.causal_conv1d; this is the MI355X-native equivalent)."""
+
+    def __init__(self, channels: int, kernel_size: int = 4, device=None, dtype=None):
+        super().__init__()
+        self.kernel_size = kernel_size
+        self.weight = nn.Parameter(
+            torch.empty(channels, kernel_size, device=device, dtype=dtype)
+        )
+
+    def reset_parameters(self) -> None:
+        with torch.no_grad():
+            nn.init.normal_(self.weight, std=1.0 / math.sqrt(self.kernel_size))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:  # (B, S, C)
+        B, S, C = x.shape
+        xt = x.transpose(1, 2)  # (B, C, S)
+        xt = F.pad(xt, (self.kernel_size - 1, 0))
+        out = F.conv1d(xt, self.weight.unsqueeze(1), groups=C)
+        return F.silu(out.transpose(1, 2))
+
+
+def chunk_gated_delta_rule(
+    q: torch.Tensor,  # (B, H, S, Dk)
+    k: torch.Tensor,  # (B, H, S, Dk)
+    v: torch.Tensor,  # (B, H, S, Dv)
+    beta: torch.Tensor,  # (B, H, S) write strength in [0, 1]
+    decay_log: torch.Tensor,  # (B, H, S) log decay (<= 0)
+    chunk_size: int = 64,
+) -> torch.Tensor:
+    """Gated delta rule: S_t = S_{t-1} * exp(g_t) * (I - beta_t k_t k_t^T) +
+    beta_t k_t v_t^T; o_t = q_t @ S_t.
+
+    Sequential over chunks, exact recurrence within each chunk (the
+    correctness oracle the HIP kernel will be tested against).
+    """
+    B, H, S, Dk = q.shape
+    Dv = v.shape[-1]
+    out = torch.zeros(B, H, S, Dv, dtype=torch.float32, device=q.device)
+    state = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+    q32, k32, v32 = q.float(), k.float(), v.float()
+    beta32 = beta.float()
+    g32 = decay_log.float()
+    for s0 in range(0, S, chunk_size):
+        s1 = min(s0 + chunk_size, S)
+        for t in range(s0, s1):
+            kt = k32[:, :, t]  # (B,H,Dk)
+            vt = v32[:, :, t]  # (B,H,Dv)
+            bt = beta32[:, :, t].unsqueeze(-1)  # (B,H,1)
+            gt = g32[:, :, t].exp().unsqueeze(-1).unsqueeze(-1)
+            state = state * gt
+            # delta update: remove old association along k_t, write new one
+            pred = torch.einsum("bhk,bhkv->bhv", kt, state)
+            state = state + torch.einsum(
+                "bhk,bhv->bhkv", kt, bt * (vt - pred)
+            )
+            out[:, :, t] = torch.einsum("bhk,bhkv->bhv", q32[:, :, t], state)
+    return out.to(v.dtype)
+
+
+class LogSigmoidDecayGate(nn.Module):
+    """a_t = -softplus(-(w x + b)) -> log decay in (-inf, 0)
+    (reference: LogSigmoidDecayGate with fused_kda_gate)."""
+
+    def __init__(self, hidden_size: int, num_heads: int, device=None, dtype=None):
+        super().__init__()
+        self.proj = nn.Linear(hidden_size, num_heads, device=device, dtype=dtype)
+
+    def reset_parameters(self) -> None:
+        with torch.no_grad():
+            nn.init.normal_(self.proj.weight, std=0.02)
+            nn.init.zeros_(self.proj.bias)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:  # (B,S,H) log decay
+        return F.logsigmoid(self.proj(x).float())
+
+
+class GatedDeltaNet(nn.Module):
+    def __init__(
+        self,
+        hidden_size: int,
+        num_heads: int = 4,
+        num_kv_heads: int | None = None,
+        head_k_dim: int = 64,
+        head_v_dim: int = 64,
+        conv_kernel_size: int = 4,
+        rms_norm_eps: float = 1e-6,
+        device=None,
+        dtype=None,
+    ) -> None:
+        super().__init__()
+        kw = {"device": device, "dtype": dtype, "bias": False}
+        self.num_heads = num_heads
+        self.num_kv_heads = num_kv_heads or num_heads
+        self.head_k_dim = head_k_dim
+        self.head_v_dim = head_v_dim
+
+        self.q_proj = nn.Linear(hidden_size, self.num_kv_heads * head_k_dim, **kw)
+        self.k_proj = nn.Linear(hidden_size, self.num_kv_heads * head_k_dim, **kw)
+        self.v_proj = nn.Linear(hidden_size, num_heads * head_v_dim, **kw)
+        self.q_conv = CausalShortDepthwiseConv1d(
+            self.num_kv_heads * head_k_dim, conv_kernel_size, device=device, dtype=dtype)
+        self.k_conv = CausalShortDepthwiseConv1d(
+            self.num_kv_heads * head_k_dim, conv_kernel_size, device=device, dtype=dtype)
+        self.v_conv = CausalShortDepthwiseConv1d(
+            num_heads * head_v_dim, conv_kernel_size, device=device, dtype=dtype)
+
+        self.beta_proj = nn.Linear(hidden_size, num_heads, **kw)
+        self.decay_gate = LogSigmoidDecayGate(hidden_size, num_heads, device=device, dtype=dtype)
+
+        self.out_norm = RMSNorm(head_v_dim, eps=rms_norm_eps, device=device, dtype=dtype)
+        self.out_gate = nn.Linear(hidden_size, num_heads * head_v_dim, **kw)
+        self.o_proj = nn.Linear(num_heads * head_v_dim, hidden_size, **kw)
+
+    def reset_parameters(self) -> None:
+        with torch.no_grad():
+            for m in (self.q_proj, self.k_proj, self.v_proj, self.beta_proj,
+                      self.out_gate, self.o_proj):
+                nn.init.normal_(m.weight, std=0.02 / math.sqrt(2))
+        for m in (self.q_conv, self.k_conv, self.v_conv, self.decay_gate, self.out_norm):
+            m.reset_parameters()
+
+    def forward(self, hidden_states: torch.Tensor, rotary_cos_sin=None) -> torch.Tensor:
+        B, S, _ = hidden_states.shape
+        H, Hkv = self.num_heads, self.num_kv_heads
+
+        q = self.q_conv(self.q_proj(hidden_states)).view(B, S, Hkv, self.head_k_dim)
+        k = self.k_conv(self.k_proj(hidden_states)).view(B, S, Hkv, self.head_k_dim)
+        v = self.v_conv(self.v_proj(hidden_states)).view(B, S, H, self.head_v_dim)
+
+        if H != Hkv:  # GQA expansion
+            rep = H // Hkv
+            q = q.repeat_interleave(rep, dim=2)
+            k = k.repeat_interleave(rep, dim=2)
+
+        # L2-normalize q/k per head (delta-rule convention)
+        q = F.normalize(q.float(), dim=-1)
+        k = F.normalize(k.float(), dim=-1)
+
+        beta = torch.sigmoid(self.beta_proj(hidden_states).float())  # (B,S,H)
+        decay = self.decay_gate(hidden_states)  # (B,S,H)
+
+        out = chunk_gated_delta_rule(
+            q.permute(0, 2, 1, 3),
+            k.permute(0, 2, 1, 3),
+            v.permute(0, 2, 1, 3).to(v.dtype),
+            beta.permute(0, 2, 1),
+            decay.permute(0, 2, 1),
+        ).permute(0, 2, 1, 3)  # (B,S,H,Dv)
+
+        out = self.out_norm(out)
+        gate = self.out_gate(hidden_states).view(B, S, H, self.head_v_dim)
+        out = silu_mul(gate.contiguous(), out.contiguous().to(gate.dtype))
+        return self.o_proj(out.reshape(B, S, H * self.head_v_dim))

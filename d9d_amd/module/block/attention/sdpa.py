@@ -1,0 +1,82 @@
+"""SDPA backend family (reference: d9d/module/block/attention/sdpa/, DEP-0008).
+
+The MI355X framework has exactly ONE production backend — the CDNA4 flash
+kernel — plus the eager fp32 oracle used as the numerics reference in tests.
+The selection machinery (protocol, discriminated config, env override) is
+kept so user configs stay source-compatible with the reference:
+resolution order = explicit config > D9D_BACKEND_AUTO_SDPA env > auto.
+"""
+
+import os
+from typing import Annotated, Literal, Protocol, Union, runtime_checkable
+
+import torch
+from pydantic import BaseModel, Field
+
+from ....ops.attention import _eager_attention, flash_attn_func
+
+
+@runtime_checkable
+class SdpaBackend(Protocol):
+    def __call__(
+        self,
+        q: torch.Tensor,
+        k: torch.Tensor,
+        v: torch.Tensor,
+        *,
+        causal: bool = True,
+        softmax_scale: float | None = None,
+        window_size: tuple[int, int] = (-1, -1),
+        sinks: torch.Tensor | None = None,
+        q_offset: int = 0,
+    ) -> torch.Tensor: ...
+
+
+class Cdna4FlashSdpaConfig(BaseModel):
+    backend: Literal["cdna4_flash"] = "cdna4_flash"
+
+
+class EagerSdpaConfig(BaseModel):
+    backend: Literal["eager"] = "eager"
+
+
+class AutoSdpaConfig(BaseModel):
+    backend: Literal["auto"] = "auto"
+
+
+SdpaBackendConfig = Annotated[
+    Union[Cdna4FlashSdpaConfig, EagerSdpaConfig, AutoSdpaConfig],
+    Field(discriminator="backend"),
+]
+
+SDPA_ENV_VAR = "D9D_BACKEND_AUTO_SDPA"
+
+
+def _cdna4_flash(q, k, v, *, causal=True, softmax_scale=None,
+                 window_size=(-1, -1), sinks=None, q_offset=0):
+    return flash_attn_func(
+        q, k, v, causal=causal, softmax_scale=softmax_scale,
+        window_size=window_size, sinks=sinks, q_offset=q_offset,
+    )
+
+
+def _eager(q, k, v, *, causal=True, softmax_scale=None,
+           window_size=(-1, -1), sinks=None, q_offset=0):
+    import math
+
+    scale = softmax_scale or 1.0 / math.sqrt(q.shape[-1])
+    out, _ = _eager_attention(q, k, v, causal, scale, window_size, sinks, q_offset)
+    return out
+
+
+def build_sdpa_backend(config: SdpaBackendConfig | None = None) -> SdpaBackend:
+    name = None
+    if config is not None and config.backend != "auto":
+        name = config.backend
+    elif SDPA_ENV_VAR in os.environ:
+        name = os.environ[SDPA_ENV_VAR]
+    if name == "eager":
+        return _eager
+    # auto and cdna4_flash both resolve to the flash op (which itself keeps
+    # the eager path for CPU and for features the kernel lacks).
+    return _cdna4_flash

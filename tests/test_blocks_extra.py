@@ -1,0 +1,98 @@
+"""Tests for MLA, GatedDeltaNet, SDPA family, hidden-state aggregator."""
+
+import torch
+
+from d9d_amd.module.block.attention import (
+    GatedDeltaNet,
+    MultiHeadLatentAttention,
+    build_sdpa_backend,
+    chunk_gated_delta_rule,
+)
+from d9d_amd.module.block.attention.sdpa import EagerSdpaConfig
+from d9d_amd.module.block.hidden_states_aggregator import HiddenStatesAggregator
+from d9d_amd.module.block.positional import RotaryEmbeddingProvider
+
+
+def test_mla_forward_backward_shapes():
+    mla = MultiHeadLatentAttention(
+        64, 4, qk_nope_head_dim=16, qk_rope_head_dim=8, v_head_dim=16,
+        kv_lora_rank=32, q_lora_rank=24,
+    )
+    mla.reset_parameters()
+    prov = RotaryEmbeddingProvider(rope_dim=8)
+    pos = torch.arange(10).unsqueeze(0).expand(2, 10)
+    x = torch.randn(2, 10, 64, requires_grad=True)
+    out = mla(x, prov(pos))
+    assert out.shape == (2, 10, 64)
+    out.sum().backward()
+    assert x.grad is not None
+    assert mla.kv_up.weight.grad is not None
+
+
+def test_delta_rule_is_causal_and_decays():
+    torch.manual_seed(0)
+    B, H, S, D = 1, 2, 8, 4
+    q = torch.randn(B, H, S, D)
+    k = torch.randn(B, H, S, D)
+    v = torch.randn(B, H, S, D)
+    beta = torch.sigmoid(torch.randn(B, H, S))
+    g = torch.zeros(B, H, S)  # no decay
+
+    out = chunk_gated_delta_rule(q, k, v, beta, g)
+    # causality: changing future inputs must not change earlier outputs
+    v2 = v.clone()
+    v2[:, :, -1] += 100
+    out2 = chunk_gated_delta_rule(q, k, v2, beta, g)
+    torch.testing.assert_close(out[:, :, :-1], out2[:, :, :-1])
+    assert not torch.allclose(out[:, :, -1], out2[:, :, -1])
+
+    # full decay wipes the state: each output only sees its own write
+    g_full = torch.full((B, H, S), -50.0)
+    out3 = chunk_gated_delta_rule(q, k, v, beta, g_full)
+    t = 3
+    expected = torch.einsum(
+        "bhk,bhkv->bhv",
+        q[:, :, t],
+        torch.einsum("bhk,bhv->bhkv", k[:, :, t], beta[:, :, t, None] * v[:, :, t]),
+    )
+    torch.testing.assert_close(out3[:, :, t], expected.to(out3.dtype), rtol=1e-4, atol=1e-5)
+
+
+def test_gdn_module_runs():
+    gdn = GatedDeltaNet(32, num_heads=4, num_kv_heads=2, head_k_dim=8, head_v_dim=8)
+    gdn.reset_parameters()
+    x = torch.randn(2, 12, 32, requires_grad=True)
+    out = gdn(x)
+    out.sum().backward()
+    assert out.shape == x.shape
+    assert x.grad is not None
+
+
+def test_sdpa_factory_env_and_config(monkeypatch):
+    import math
+
+    backend = build_sdpa_backend(EagerSdpaConfig())
+    q = torch.randn(1, 8, 2, 16)
+    k = torch.randn(1, 8, 2, 16)
+    v = torch.randn(1, 8, 2, 16)
+    out = backend(q, k, v, causal=True)
+    from d9d_amd.ops.attention import _eager_attention
+
+    ref, _ = _eager_attention(q, k, v, True, 1 / math.sqrt(16), (-1, -1), None)
+    torch.testing.assert_close(out, ref)
+
+    monkeypatch.setenv("D9D_BACKEND_AUTO_SDPA", "eager")
+    assert build_sdpa_backend(None).__name__ == "_eager"
+    monkeypatch.delenv("D9D_BACKEND_AUTO_SDPA")
+    assert build_sdpa_backend(None).__name__ == "_cdna4_flash"
+
+
+def test_hidden_states_aggregator():
+    agg = HiddenStatesAggregator("mean")
+    h = torch.randn(2, 6, 8)
+    a = agg.initial(h)
+    a = agg.append(a, h)
+    a = agg.append(a, h * 2)
+    assert a.shape == (2, 2, 8)
+    torch.testing.assert_close(a[0], h.mean(1))
+    assert HiddenStatesAggregator("none").append(None, h) is None
