@@ -29,6 +29,39 @@ D9D_DEVICE f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+union ushort2_t {  // 4 bf16 lanes packed for one 8-byte LDS store
+  uint64_t u;
+  ushort s[4];
+};
+
+// Stage a (64 rows x D cols) global tile TRANSPOSED into a [D][64] LDS image
+// (128-byte rows, ((d&7)<<4) XOR swizzle) with vectorized 8B LDS writes:
+// each op covers a (4 rows x 2 cols) block.
+template <int D>
+D9D_DEVICE void stage_transposed_tile(
+    bf16_t* dst, const bf16_t* src, int64_t row_stride,
+    int row0, int row_clamp, int tid) {
+  for (int idx = tid; idx < (64 / 4) * (D / 2); idx += 256) {
+    const int d0 = (idx % (D / 2)) * 2;
+    const int rb = (idx / (D / 2)) * 4;
+    ushort2_t c0, c1;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int g_row = min(row0 + rb + i, row_clamp);
+      const uint32_t pair = *reinterpret_cast<const uint32_t*>(
+          src + (int64_t)g_row * row_stride + d0);
+      c0.s[i] = (ushort)(pair & 0xffffu);
+      c1.s[i] = (ushort)(pair >> 16);
+    }
+    const int byte0 = (rb * 2) ^ ((d0 & 7) << 4);
+    const int byte1 = (rb * 2) ^ (((d0 + 1) & 7) << 4);
+    *reinterpret_cast<uint64_t*>(
+        reinterpret_cast<char*>(dst) + d0 * 128 + byte0) = c0.u;
+    *reinterpret_cast<uint64_t*>(
+        reinterpret_cast<char*>(dst) + (d0 + 1) * 128 + byte1) = c1.u;
+  }
+}
+
 // Fragment maps for v_mfma_f32_16x16x32_bf16 (M=N=16, K=32):
 //   A[m][k]: lane l holds m = l&15, k = (l>>4)*8 + j   (j = 0..7)
 //   B[k][n]: lane l holds n = l&15, k = (l>>4)*8 + j
@@ -63,11 +96,12 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
   constexpr int kNT = D / 16;   // n-tiles over head dim
   constexpr int kKS = D / 32;   // k-steps over head dim
   constexpr int kRowBytes = D * 2;
+  constexpr int kQB = 128;      // q rows per workgroup: 32 per wave (2 m-tiles)
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* k_lds = reinterpret_cast<bf16_t*>(smem);            // [64][D] swizzled
   bf16_t* vt_lds = k_lds + kKvBlk * D;                        // [D][64] transposed
-  // per-wave P scratch: [4][16][kKvBlk+8]
+  // per-wave P scratch: [4][32][kKvBlk+8]
   bf16_t* p_lds = vt_lds + D * kKvBlk;
 
   const int lane = threadIdx.x & 63;
@@ -83,40 +117,42 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
   const int64_t q_row_stride = (int64_t)Hq * D;
   const int64_t kv_row_stride = (int64_t)Hkv * D;
 
-  // ---- load this wave's 16 q rows into A fragments -------------------------
-  const int q_row_local = lane & 15;           // m
-  const int q_row_global = q_tile * kQBlk + wave * 16 + q_row_local;
-  bf16x8 q_frag[kKS];
-  {
+  // ---- load this wave's 32 q rows (2 m-tiles) into A fragments -------------
+  bf16x8 q_frag[2][kKS];
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+    const int q_row_global = q_tile * kQB + wave * 32 + m * 16 + (lane & 15);
     const int safe_row = min(q_row_global, Sq - 1);
     const bf16_t* qp = q + q_base + (int64_t)safe_row * q_row_stride;
 #pragma unroll
     for (int ks = 0; ks < kKS; ++ks) {
-      const int d0 = ks * 32 + (lane >> 4) * 8;
-      q_frag[ks] = *reinterpret_cast<const bf16x8*>(qp + d0);
+      q_frag[m][ks] = *reinterpret_cast<const bf16x8*>(qp + ks * 32 + (lane >> 4) * 8);
     }
   }
 
-  float m_run[4], l_run[4];
+  float m_run[2][4], l_run[2][4];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m_run[r] = -1e30f;
-    l_run[r] = 0.f;
-  }
-  f32x4 o_acc[kNT];
+  for (int m = 0; m < 2; ++m)
 #pragma unroll
-  for (int nt = 0; nt < kNT; ++nt) o_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+    for (int r = 0; r < 4; ++r) {
+      m_run[m][r] = -1e30f;
+      l_run[m][r] = 0.f;
+    }
+  f32x4 o_acc[2][kNT];
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int nt = 0; nt < kNT; ++nt) o_acc[m][nt] = {0.f, 0.f, 0.f, 0.f};
 
-  const int q_tile_last_row = min(q_tile * kQBlk + kQBlk - 1, Sq - 1);
+  const int q_tile_last_row = min(q_tile * kQB + kQB - 1, Sq - 1);
   int kv_end = Skv;
   if (causal) kv_end = min(Skv, q_tile_last_row + q_offset + 1);
   const int num_kv_tiles = (kv_end + kKvBlk - 1) / kKvBlk;
 
   for (int kt = 0; kt < num_kv_tiles; ++kt) {
     const int kv0 = kt * kKvBlk;
-    // ---- stage K (row-major + swizzle) and V^T ------------------------------
+    // ---- stage K (row-major + swizzle, 16B ops) ----------------------------
     {
-      // K: 64 rows x D cols; 256 threads, 8 bf16 (16B) per thread per pass.
       constexpr int elems = kKvBlk * D;
       for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
         const int row = idx / D;
@@ -128,166 +164,180 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
         *reinterpret_cast<bf16x8*>(
             reinterpret_cast<char*>(k_lds) + row * kRowBytes + byte) = val;
       }
-      // V^T: read V row-major, write transposed (scalar writes).
-      for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
-        const int row = idx / D;  // kv row
-        const int col = idx % D;
-        const int g_row = min(kv0 + row, Skv - 1);
-        const bf16x8 val = *reinterpret_cast<const bf16x8*>(
-            v + kv_base + (int64_t)g_row * kv_row_stride + col);
+      // V^T staging, vectorized: each op covers a (4 kv) x (2 d) block —
+      // two 4B global reads per kv-pair... per-thread: 2 d at fixed kv row
+      // read as one 4B load x 4 rows; write two 8B ds_writes (d, kv..kv+3).
+      for (int idx = threadIdx.x; idx < (kKvBlk / 4) * (D / 2); idx += 256) {
+        const int d0 = (idx % (D / 2)) * 2;
+        const int kvb = (idx / (D / 2)) * 4;
+        ushort2_t col0, col1;  // two d-columns x 4 kv rows
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = col + j;
-          const int byte = (row * 2) ^ ((d & 7) << 4);
-          *reinterpret_cast<bf16_t*>(
-              reinterpret_cast<char*>(vt_lds) + d * (kKvBlk * 2) + byte) = val[j];
+        for (int i = 0; i < 4; ++i) {
+          const int g_row = min(kv0 + kvb + i, Skv - 1);
+          const uint32_t pair = *reinterpret_cast<const uint32_t*>(
+              v + kv_base + (int64_t)g_row * kv_row_stride + d0);
+          col0.s[i] = (ushort)(pair & 0xffffu);
+          col1.s[i] = (ushort)(pair >> 16);
         }
+        const int byte0 = (kvb * 2) ^ ((d0 & 7) << 4);
+        const int byte1 = (kvb * 2) ^ (((d0 + 1) & 7) << 4);
+        *reinterpret_cast<uint64_t*>(
+            reinterpret_cast<char*>(vt_lds) + d0 * (kKvBlk * 2) + byte0) = col0.u;
+        *reinterpret_cast<uint64_t*>(
+            reinterpret_cast<char*>(vt_lds) + (d0 + 1) * (kKvBlk * 2) + byte1) = col1.u;
       }
     }
     __syncthreads();
 
-    // ---- S = Q @ K^T (per wave: 16 x 64) -----------------------------------
-    f32x4 s_acc[4];
+    // per m-tile: QK^T -> softmax -> P -> PV
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) s_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+    for (int m = 0; m < 2; ++m) {
+      f32x4 s_acc[4];
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
-      const int kv_row = nt * 16 + (lane & 15);
-#pragma unroll
-      for (int ks = 0; ks < kKS; ++ks) {
-        const int d0 = ks * 32 + (lane >> 4) * 8;
-        const int byte = (d0 * 2) ^ ((kv_row & swz_rm_mask<D>()) << 4);
-        const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<char*>(k_lds) + kv_row * kRowBytes + byte);
-        s_acc[nt] = mfma16(q_frag[ks], kb, s_acc[nt]);
-      }
-    }
-
-    // ---- mask + scale + online softmax -------------------------------------
-    const int my_q_row = q_tile * kQBlk + wave * 16;  // + (lane>>4)*4 + r
-    float p_val[4][4];  // [nt][r]
-    float m_new[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) m_new[r] = m_run[r];
-#pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
-      const int col = kv0 + nt * 16 + (lane & 15);
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = my_q_row + (lane >> 4) * 4 + r + q_offset;
-        float s = s_acc[nt][r] * scale;
-        bool masked = col >= Skv;
-        if (causal) masked |= col > row;
-        if (window_left >= 0) masked |= col < row - window_left;
-        p_val[nt][r] = masked ? -1e30f : s;
-      }
-    }
-    // row max: reduce over 16 lanes (same lane>>4 group) and the 4 nt slots.
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float mx = fmaxf(fmaxf(p_val[0][r], p_val[1][r]),
-                       fmaxf(p_val[2][r], p_val[3][r]));
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1) {
-        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
-      }
-      m_new[r] = fmaxf(m_new[r], mx);
-    }
-    // P = exp(s - m_new); row sum.
-    float l_add[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float acc = 0.f;
+      for (int nt = 0; nt < 4; ++nt) s_acc[nt] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        const float p = __builtin_amdgcn_exp2f((p_val[nt][r] - m_new[r]) * kLog2e);
-        p_val[nt][r] = p;
-        acc += p;
-      }
+        const int kv_row = nt * 16 + (lane & 15);
 #pragma unroll
-      for (int off = 1; off < 16; off <<= 1) {
-        acc += __shfl_xor(acc, off, 64);
-      }
-      l_add[r] = acc;
-    }
-    // rescale running state + O.
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const float alpha = __builtin_amdgcn_exp2f((m_run[r] - m_new[r]) * kLog2e);
-      l_run[r] = l_run[r] * alpha + l_add[r];
-      m_run[r] = m_new[r];
-#pragma unroll
-      for (int nt = 0; nt < kNT; ++nt) o_acc[nt][r] *= alpha;
-    }
-
-    // ---- write P to this wave's LDS scratch, C-layout -> row-major ---------
-    // p_scratch row = q_local (0..15), col = kv col (0..63), row pitch 64+8.
-    {
-      bf16_t* pw = p_lds + wave * 16 * (kKvBlk + 8);
-#pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = (lane >> 4) * 4 + r;
-          const int col = nt * 16 + (lane & 15);
-          pw[row * (kKvBlk + 8) + col] = (bf16_t)p_val[nt][r];
+        for (int ks = 0; ks < kKS; ++ks) {
+          const int d0 = ks * 32 + (lane >> 4) * 8;
+          const int byte = (d0 * 2) ^ ((kv_row & swz_rm_mask<D>()) << 4);
+          const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(k_lds) + kv_row * kRowBytes + byte);
+          s_acc[nt] = mfma16(q_frag[m][ks], kb, s_acc[nt]);
         }
       }
-    }
 
-    // ---- O += P @ V (A = P from scratch, B = V^T tile) ---------------------
-    {
-      const bf16_t* pr = p_lds + wave * 16 * (kKvBlk + 8);
+      const int my_q_row = q_tile * kQB + wave * 32 + m * 16;
+      float p_val[4][4];  // [nt][r]
+      float m_new[4];
 #pragma unroll
-      for (int ks2 = 0; ks2 < 2; ++ks2) {  // K = 64 kv -> 2 steps of 32
-        const int kv_off = ks2 * 32 + (lane >> 4) * 8;
-        const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-            pr + (lane & 15) * (kKvBlk + 8) + kv_off);
+      for (int r = 0; r < 4; ++r) m_new[r] = m_run[m][r];
 #pragma unroll
-        for (int nt = 0; nt < kNT; ++nt) {
-          const int d = nt * 16 + (lane & 15);
-          const int byte = (kv_off * 2) ^ ((d & 7) << 4);
-          const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(vt_lds) + d * (kKvBlk * 2) + byte);
-          o_acc[nt] = mfma16(pa, vb, o_acc[nt]);
+      for (int nt = 0; nt < 4; ++nt) {
+        const int col = kv0 + nt * 16 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = my_q_row + (lane >> 4) * 4 + r + q_offset;
+          float s = s_acc[nt][r] * scale;
+          bool masked = col >= Skv;
+          if (causal) masked |= col > row;
+          if (window_left >= 0) masked |= col < row - window_left;
+          p_val[nt][r] = masked ? -1e30f : s;
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float mx = fmaxf(fmaxf(p_val[0][r], p_val[1][r]),
+                         fmaxf(p_val[2][r], p_val[3][r]));
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+          mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+        }
+        m_new[r] = fmaxf(m_new[r], mx);
+      }
+      float l_add[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float acc = 0.f;
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const float p = __builtin_amdgcn_exp2f((p_val[nt][r] - m_new[r]) * kLog2e);
+          p_val[nt][r] = p;
+          acc += p;
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+          acc += __shfl_xor(acc, off, 64);
+        }
+        l_add[r] = acc;
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float alpha = __builtin_amdgcn_exp2f((m_run[m][r] - m_new[r]) * kLog2e);
+        l_run[m][r] = l_run[m][r] * alpha + l_add[r];
+        m_run[m][r] = m_new[r];
+#pragma unroll
+        for (int nt = 0; nt < kNT; ++nt) o_acc[m][nt][r] *= alpha;
+      }
+
+      // P C-layout -> row-major LDS scratch (own-wave slice) -> A fragments
+      {
+        bf16_t* pw = p_lds + (wave * 32 + m * 16) * (kKvBlk + 8);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = (lane >> 4) * 4 + r;
+            const int col = nt * 16 + (lane & 15);
+            pw[row * (kKvBlk + 8) + col] = (bf16_t)p_val[nt][r];
+          }
+        }
+        const bf16_t* pr = pw;
+#pragma unroll
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
+          const int kv_off = ks2 * 32 + (lane >> 4) * 8;
+          const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+              pr + (lane & 15) * (kKvBlk + 8) + kv_off);
+#pragma unroll
+          for (int nt = 0; nt < kNT; ++nt) {
+            const int d = nt * 16 + (lane & 15);
+            const int byte = (kv_off * 2) ^ ((d & 7) << 4);
+            const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<char*>(vt_lds) + d * (kKvBlk * 2) + byte);
+            o_acc[m][nt] = mfma16(pa, vb, o_acc[m][nt]);
+          }
         }
       }
     }
     __syncthreads();
   }
 
-  // ---- epilogue: O /= l; stage O in LDS row-major; coalesced store ---------
+  // ---- epilogue: O /= l; stage in LDS; coalesced 16B stores; LSE -----------
   {
-    bf16_t* o_lds = k_lds;  // reuse
+    bf16_t* o_lds = k_lds;  // reuse: [64][D] staged twice (two 64-row halves)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
-      const int row = wave * 16 + (lane >> 4) * 4 + r;
+    for (int half = 0; half < 2; ++half) {
+      __syncthreads();
+      // waves 0,1 hold q rows 0..63 of the tile; waves 2,3 rows 64..127
+      if ((wave >> 1) == half) {
 #pragma unroll
-      for (int nt = 0; nt < kNT; ++nt) {
-        o_lds[row * D + nt * 16 + (lane & 15)] = (bf16_t)(o_acc[nt][r] * inv_l);
+        for (int m = 0; m < 2; ++m) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float inv_l = (l_run[m][r] > 0.f) ? 1.f / l_run[m][r] : 0.f;
+            const int row = (wave & 1) * 32 + m * 16 + (lane >> 4) * 4 + r;
+#pragma unroll
+            for (int nt = 0; nt < kNT; ++nt) {
+              o_lds[row * D + nt * 16 + (lane & 15)] =
+                  (bf16_t)(o_acc[m][nt][r] * inv_l);
+            }
+          }
+        }
       }
-    }
-    // LSE: one lane per row.
-    if ((lane & 15) == 0) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int rg = q_tile * kQBlk + wave * 16 + (lane >> 4) * 4 + r;
+      __syncthreads();
+      constexpr int elems = 64 * D;
+      for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
+        const int row = idx / D;
+        const int col = idx % D;
+        const int rg = q_tile * kQB + half * 64 + row;
         if (rg < Sq) {
-          lse[((int64_t)b * Hq + h) * Sq + rg] =
-              m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+          *reinterpret_cast<bf16x8*>(out + q_base + (int64_t)rg * q_row_stride + col) =
+              *reinterpret_cast<const bf16x8*>(o_lds + row * D + col);
         }
       }
     }
-    __syncthreads();
-    constexpr int elems = kQBlk * D;
-    for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
-      const int row = idx / D;
-      const int col = idx % D;
-      const int rg = q_tile * kQBlk + row;
-      if (rg < Sq) {
-        *reinterpret_cast<bf16x8*>(out + q_base + (int64_t)rg * q_row_stride + col) =
-            *reinterpret_cast<const bf16x8*>(o_lds + row * D + col);
+    if ((lane & 15) == 0) {
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int rg = q_tile * kQB + wave * 32 + m * 16 + (lane >> 4) * 4 + r;
+          if (rg < Sq) {
+            lse[((int64_t)b * Hq + h) * Sq + rg] =
+                m_run[m][r] + __logf(fmaxf(l_run[m][r], 1e-30f));
+          }
+        }
       }
     }
   }
@@ -377,23 +427,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   }
 
   // Stage K^T once (for dQ's B fragments).
-  {
-    constexpr int elems = kKvBlk * D;
-    for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
-      const int row = idx / D;
-      const int col = idx % D;
-      const int g_row = min(kv0 + row, Skv - 1);
-      const bf16x8 val = *reinterpret_cast<const bf16x8*>(
-          k + kv_base + (int64_t)g_row * kv_row_stride + col);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d = col + j;
-        const int byte = (row * 2) ^ ((d & 7) << 4);
-        *reinterpret_cast<bf16_t*>(
-            reinterpret_cast<char*>(kt_lds) + d * (kKvBlk * 2) + byte) = val[j];
-      }
-    }
-  }
+  stage_transposed_tile<D>(
+      kt_lds, k + kv_base, kv_row_stride, kv0, Skv - 1, threadIdx.x);
 
   f32x4 dk_acc[kNT], dv_acc[kNT];
 #pragma unroll
@@ -427,14 +462,10 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
             reinterpret_cast<char*>(q_lds) + row * kRowBytes + byte) = qv;
         *reinterpret_cast<bf16x8*>(
             reinterpret_cast<char*>(do_lds) + row * kRowBytes + byte) = dv8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = col + j;
-          const int tbyte = (row * 2) ^ ((d & 7) << 4);
-          *reinterpret_cast<bf16_t*>(
-              reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + tbyte) = dv8[j];
-        }
       }
+      // dO^T (for dV's B fragments), vectorized transpose
+      stage_transposed_tile<D>(
+          t_lds, dout + q_base, q_row_stride, qt, Sq - 1, threadIdx.x);
     }
     __syncthreads();
 
@@ -536,23 +567,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     // ---- dK += dS^T @ Q  (A = dS^T via x_lds, B = Q^T re-staged) ----------
     {
       __syncthreads();  // x_lds + t_lds reuse
-      {
-        constexpr int elems = kQBlk * D;
-        for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
-          const int row = idx / D;
-          const int col = idx % D;
-          const int g_row = min(qt + row, Sq - 1);
-          const bf16x8 qv = *reinterpret_cast<const bf16x8*>(
-              q + q_base + (int64_t)g_row * q_row_stride + col);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int d = col + j;
-            const int tbyte = (row * 2) ^ ((d & 7) << 4);
-            *reinterpret_cast<bf16_t*>(
-                reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + tbyte) = qv[j];
-          }
-        }
-      }
+      stage_transposed_tile<D>(
+          t_lds, q + q_base, q_row_stride, qt, Sq - 1, threadIdx.x);
       bf16_t* xw = x_lds;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -707,9 +723,9 @@ std::vector<torch::Tensor> flash_attn_fwd(
   auto out = torch::empty_like(qp);
   auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
 
-  const dim3 grid((Sq + d9d::kQBlk - 1) / d9d::kQBlk, B * Hq);
+  const dim3 grid((Sq + 127) / 128, B * Hq);
   const size_t smem =
-      (size_t)(d9d::kKvBlk * D_pad + D_pad * d9d::kKvBlk + 4 * 16 * (d9d::kKvBlk + 8)) *
+      (size_t)(d9d::kKvBlk * D_pad + D_pad * d9d::kKvBlk + 128 * (d9d::kKvBlk + 8)) *
       sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
 

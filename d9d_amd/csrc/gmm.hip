@@ -58,8 +58,10 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
   const int wm = wave >> 1;   // 0..1: wave row (64 rows each)
   const int wn = wave & 1;    // 0..1: wave col (32 cols each)
 
-  // ---- map blockIdx.x -> (expert, m_tile) via binary search ----------------
-  const int mt_global = blockIdx.x;
+  // ---- map blockIdx.y -> (expert, m_tile) via binary search ----------------
+  // grid.x is the (small) n-tile index: consecutive blocks share the same A
+  // rows, so A re-reads hit L2 instead of HBM.
+  const int mt_global = blockIdx.y;
   // largest e with mtile_pref[e] <= mt_global (expert may own several m-tiles)
   int lo = 0, hi = E - 1;
   while (lo < hi) {
@@ -70,7 +72,7 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
   const int m_tile = mt_global - mtile_pref[e];
   const int row0 = row_off[e] + m_tile * kBM;
   const int row_end = row_off[e + 1];
-  const int n0 = blockIdx.y * kBN;
+  const int n0 = blockIdx.x * kBN;
 
   const bf16_t* b_e = b + (int64_t)e * K * N;
 
@@ -181,7 +183,11 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
   }
 }
 
-// db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N). Grid (ceil(K/64), ceil(N/64), E).
+// db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N).
+// Grid (ceil(N/128), ceil(K/128), E): 128x128 tiles (4 waves as 2x2, 64x64
+// each) looping the expert's rows in chunks of 64; both operands staged
+// transposed ([dim][row], 128-byte rows, XOR swizzle) for k-contiguous
+// fragments.
 __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
     const bf16_t* __restrict__ a,   // (T, K)
     const bf16_t* __restrict__ g,   // (T, N)
@@ -189,62 +195,58 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
     const int* __restrict__ row_off,
     int E, int K, int N) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // at: [64 k][64 rows] (transposed a tile), gl: [64 rows][64 n] -> need
-  // A[m=k][kk=row] k-contig rows: at[k][row]; B[kk=row][n]: need row contig at
-  // fixed n -> stage g transposed too: gt[n][row].
-  bf16_t* at_lds = reinterpret_cast<bf16_t*>(smem);   // [64][64+pad? swizzled]
-  bf16_t* gt_lds = at_lds + 64 * 64;
+  bf16_t* at_lds = reinterpret_cast<bf16_t*>(smem);   // [128 k][64 rows]
+  bf16_t* gt_lds = at_lds + 128 * 64;                 // [128 n][64 rows]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = wave >> 1;  // 2x2 waves over (k 64 x n 64): each 32x32
-  const int wn = wave & 1;
+  const int wm = wave >> 1;  // k-half
+  const int wn = wave & 1;   // n-half
 
   const int e = blockIdx.z;
-  const int k0 = blockIdx.x * 64;
-  const int n0 = blockIdx.y * 64;
+  const int k0 = blockIdx.y * 128;
+  const int n0 = blockIdx.x * 128;
   const int r_start = row_off[e];
   const int r_end = row_off[e + 1];
 
-  f32x4 acc[2][2];
+  f32x4 acc[4][4];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   for (int rt = r_start; rt < r_end; rt += 64) {
     __syncthreads();
-    // stage a^T: a[row][k0..] -> at[k][row]; 64x64 tile
-    for (int idx = threadIdx.x * 8; idx < 64 * 64; idx += 256 * 8) {
-      const int row = idx / 64;  // row within tile
-      const int kk = idx % 64;   // k within tile (contiguous 8)
+    // stage a^T and g^T: [dim 128][row 64]; 128*64 elems each, 8 per thread op
+    for (int idx = threadIdx.x * 8; idx < 128 * 64; idx += 256 * 8) {
+      const int row = idx / 128;   // row within 64-chunk
+      const int d0 = idx % 128;    // dim offset (8 contiguous)
       const int g_row = rt + row;
       bf16x8 va = {}, vg = {};
       if (g_row < r_end) {
-        if (k0 + kk + 7 < K) {
-          va = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + kk);
+        if (k0 + d0 + 7 < K) {
+          va = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + d0);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            va[j] = (k0 + kk + j < K) ? a[(int64_t)g_row * K + k0 + kk + j] : (bf16_t)0.f;
+            va[j] = (k0 + d0 + j < K) ? a[(int64_t)g_row * K + k0 + d0 + j] : (bf16_t)0.f;
         }
-        if (n0 + kk + 7 < N) {
-          vg = *reinterpret_cast<const bf16x8*>(g + (int64_t)g_row * N + n0 + kk);
+        if (n0 + d0 + 7 < N) {
+          vg = *reinterpret_cast<const bf16x8*>(g + (int64_t)g_row * N + n0 + d0);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            vg[j] = (n0 + kk + j < N) ? g[(int64_t)g_row * N + n0 + kk + j] : (bf16_t)0.f;
+            vg[j] = (n0 + d0 + j < N) ? g[(int64_t)g_row * N + n0 + d0 + j] : (bf16_t)0.f;
         }
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const int kcol = kk + j;
-        const int byte_a = (row * 2) ^ ((kcol & 7) << 4);
+        const int d = d0 + j;
+        const int byte = (row * 2) ^ ((d & 7) << 4);
         *reinterpret_cast<bf16_t*>(
-            reinterpret_cast<char*>(at_lds) + kcol * (64 * 2) + byte_a) = va[j];
-        const int byte_g = (row * 2) ^ ((kcol & 7) << 4);
+            reinterpret_cast<char*>(at_lds) + d * (64 * 2) + byte) = va[j];
         *reinterpret_cast<bf16_t*>(
-            reinterpret_cast<char*>(gt_lds) + kcol * (64 * 2) + byte_g) = vg[j];
+            reinterpret_cast<char*>(gt_lds) + d * (64 * 2) + byte) = vg[j];
       }
     }
     __syncthreads();
@@ -252,36 +254,36 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {  // 64 rows -> 2 k-steps
       const int rr = ks * 32 + (lane >> 4) * 8;
-      bf16x8 a_frag[2], g_frag[2];
+      bf16x8 a_frag[4], g_frag[4];
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        const int krow = wm * 32 + i * 16 + (lane & 15);
+      for (int i = 0; i < 4; ++i) {
+        const int krow = wm * 64 + i * 16 + (lane & 15);
         const int byte = (rr * 2) ^ ((krow & 7) << 4);
         a_frag[i] = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const char*>(at_lds) + krow * (64 * 2) + byte);
       }
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
-        const int ncol = wn * 32 + j * 16 + (lane & 15);
+      for (int j = 0; j < 4; ++j) {
+        const int ncol = wn * 64 + j * 16 + (lane & 15);
         const int byte = (rr * 2) ^ ((ncol & 7) << 4);
         g_frag[j] = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const char*>(gt_lds) + ncol * (64 * 2) + byte);
       }
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
+      for (int i = 0; i < 4; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j) acc[i][j] = mfma16g(a_frag[i], g_frag[j], acc[i][j]);
+        for (int j = 0; j < 4; ++j) acc[i][j] = mfma16g(a_frag[i], g_frag[j], acc[i][j]);
     }
   }
 
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int i = 0; i < 4; ++i) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < 4; ++j) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int kk = k0 + wm * 32 + i * 16 + (lane >> 4) * 4 + r;
-        const int nn = n0 + wn * 32 + j * 16 + (lane & 15);
+        const int kk = k0 + wm * 64 + i * 16 + (lane >> 4) * 4 + r;
+        const int nn = n0 + wn * 64 + j * 16 + (lane & 15);
         if (kk < K && nn < N) {
           db[((int64_t)e * K + kk) * N + nn] = (bf16_t)acc[i][j][r];
         }
@@ -331,7 +333,7 @@ torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
       ? mtile_pref[E].item<int>() : 0;
   if (total_mtiles == 0) return out;
 
-  const dim3 grid(total_mtiles, (N + d9d::kBN - 1) / d9d::kBN);
+  const dim3 grid((N + d9d::kBN - 1) / d9d::kBN, total_mtiles);
   const size_t smem = (2 * d9d::kBM * d9d::kBK + 2 * d9d::kBN * d9d::kBK) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(d9d::gmm_kernel, grid, dim3(256), smem, stream,
@@ -354,8 +356,8 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
   if (a.size(0) == 0) return db;
   auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
 
-  const dim3 grid((K + 63) / 64, (N + 63) / 64, E);
-  const size_t smem = (2 * 64 * 64) * sizeof(__bf16);
+  const dim3 grid((N + 127) / 128, (K + 127) / 128, E);
+  const size_t smem = (2 * 128 * 64) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(d9d::gmm_db_kernel, grid, dim3(256), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
