@@ -30,6 +30,11 @@ def parse_args():
     p.add_argument("--microbatch", type=int, default=8)
     p.add_argument("--grad-accum", type=int, default=16)
     p.add_argument("--model", type=str, default="qwen3_moe")
+    p.add_argument(
+        "--parallelism", type=str, default="dp", choices=["dp", "ep"],
+        help="multi-GPU strategy: dp = replicate + bucketed grad all-reduce; "
+             "ep = expert parallelism (experts sharded, RCCL all-to-all dispatch)",
+    )
     return p.parse_args()
 
 
@@ -79,6 +84,36 @@ def main():
     model.init_weights()
     model.train()
 
+    sync = None
+    if distributed:
+        from d9d_amd.core.dist_context import DeviceMeshParameters
+        from d9d_amd.internals.grad_sync import GradientSynchronizer
+
+        if args.parallelism == "ep":
+            mesh_params = DeviceMeshParameters(
+                data_parallel_replicate=world, expert_parallel=world
+            )
+            ctx = mesh_params.build()
+            from d9d_amd.parallel import (
+                parallelize_expert_parallel,
+                parallelize_replicate,
+            )
+
+            parallelize_expert_parallel(model, ctx.mesh_for("expert"))
+            # everything that is NOT expert-sharded replicates over dp
+            parallelize_replicate(model, ctx.mesh_for("dense"))
+        else:
+            mesh_params = DeviceMeshParameters(data_parallel_replicate=world)
+            ctx = mesh_params.build()
+            from d9d_amd.parallel import parallelize_replicate
+
+            parallelize_replicate(model, ctx.mesh_for("dense"))
+        sync = GradientSynchronizer(
+            list(model.named_parameters()),
+            accumulation_steps=args.grad_accum,
+            bucket_bytes=128 * 1024 * 1024,  # xGMI ring: few big buckets
+        )
+
     opt = StochasticAdamW(model.parameters(), lr=3e-4, weight_decay=0.1, seed=rank)
 
     B, S = args.microbatch, args.seq_len
@@ -91,43 +126,30 @@ def main():
         torch.randint(0, vocab, (B, S + 1), device=device) for _ in range(2)
     ]
 
-    grads_synced = [p for p in model.parameters() if p.requires_grad]
+    inv_world = 1.0 / world
 
     def one_step():
-        opt.zero_grad(set_to_none=True)
+        if sync is not None:
+            sync.zero_grad()
+        opt.zero_grad(set_to_none=False)
         for mb in range(n_micro):
             data = batches[mb % len(batches)]
             input_ids = data[:, :-1]
             labels = data[:, 1:]
             out = model(input_ids=input_ids, labels=labels)
             (out["loss"].mean() / n_micro).backward()
-        if distributed:
-            # v1 DP sync: flat bucketed all-reduce (overlapped sync lands with
-            # the GradientSynchronizer integration).
-            bucket: list[torch.Tensor] = []
-            size = 0
-            for p in grads_synced:
-                if p.grad is None:
-                    continue
-                bucket.append(p.grad)
-                size += p.grad.numel()
-                if size >= 64 * 1024 * 1024 // 2:
-                    flat = torch.cat([g.reshape(-1) for g in bucket])
-                    dist.all_reduce(flat)
-                    flat /= world
-                    off = 0
-                    for g in bucket:
-                        g.copy_(flat[off : off + g.numel()].view_as(g))
-                        off += g.numel()
-                    bucket, size = [], 0
-            if bucket:
-                flat = torch.cat([g.reshape(-1) for g in bucket])
-                dist.all_reduce(flat)
-                flat /= world
-                off = 0
-                for g in bucket:
-                    g.copy_(flat[off : off + g.numel()].view_as(g))
-                    off += g.numel()
+        if sync is not None:
+            # bucketed all-reduce launched by post-accumulate hooks on a side
+            # stream, overlapped with backward; join + average here
+            sync.wait()
+            from torch.distributed.tensor import DTensor
+
+            grads = [
+                (p.grad.to_local() if isinstance(p.grad, DTensor) else p.grad)
+                for p in model.parameters()
+                if p.grad is not None
+            ]
+            torch._foreach_mul_(grads, inv_world)
         opt.step()
 
     # Warmup
