@@ -149,45 +149,71 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
   if (causal) kv_end = min(Skv, q_tile_last_row + q_offset + 1);
   const int num_kv_tiles = (kv_end + kKvBlk - 1) / kKvBlk;
 
-  for (int kt = 0; kt < num_kv_tiles; ++kt) {
+  // T14 split staging: next tile's K rows and V (4x2 transpose blocks) are
+  // loaded into registers while the current tile's MFMAs run; LDS writes and
+  // their vmcnt waits land after the compute barrier.
+  bf16x8 k_reg[4];
+  ushort2_t v_c0[4], v_c1[4];
+
+  auto load_regs = [&](int kt) {
     const int kv0 = kt * kKvBlk;
-    // ---- stage K (row-major + swizzle, 16B ops) ----------------------------
-    {
-      constexpr int elems = kKvBlk * D;
-      for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
-        const int row = idx / D;
-        const int col = idx % D;
-        const int g_row = min(kv0 + row, Skv - 1);
-        const bf16x8 val = *reinterpret_cast<const bf16x8*>(
-            k + kv_base + (int64_t)g_row * kv_row_stride + col);
-        const int byte = (col * 2) ^ ((row & swz_rm_mask<D>()) << 4);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(k_lds) + row * kRowBytes + byte) = val;
-      }
-      // V^T staging, vectorized: each op covers a (4 kv) x (2 d) block —
-      // two 4B global reads per kv-pair... per-thread: 2 d at fixed kv row
-      // read as one 4B load x 4 rows; write two 8B ds_writes (d, kv..kv+3).
-      for (int idx = threadIdx.x; idx < (kKvBlk / 4) * (D / 2); idx += 256) {
-        const int d0 = (idx % (D / 2)) * 2;
-        const int kvb = (idx / (D / 2)) * 4;
-        ushort2_t col0, col1;  // two d-columns x 4 kv rows
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int g_row = min(kv0 + kvb + i, Skv - 1);
-          const uint32_t pair = *reinterpret_cast<const uint32_t*>(
-              v + kv_base + (int64_t)g_row * kv_row_stride + d0);
-          col0.s[i] = (ushort)(pair & 0xffffu);
-          col1.s[i] = (ushort)(pair >> 16);
-        }
-        const int byte0 = (kvb * 2) ^ ((d0 & 7) << 4);
-        const int byte1 = (kvb * 2) ^ (((d0 + 1) & 7) << 4);
-        *reinterpret_cast<uint64_t*>(
-            reinterpret_cast<char*>(vt_lds) + d0 * (kKvBlk * 2) + byte0) = col0.u;
-        *reinterpret_cast<uint64_t*>(
-            reinterpret_cast<char*>(vt_lds) + (d0 + 1) * (kKvBlk * 2) + byte1) = col1.u;
+    for (int it = 0; it < (kKvBlk * D) / (256 * 8); ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int g_row = min(kv0 + row, Skv - 1);
+      k_reg[it] = *reinterpret_cast<const bf16x8*>(
+          k + kv_base + (int64_t)g_row * kv_row_stride + col);
+    }
+#pragma unroll
+    for (int it = 0; it < (kKvBlk / 4) * (D / 2) / 256; ++it) {
+      const int idx = threadIdx.x + it * 256;
+      const int d0 = (idx % (D / 2)) * 2;
+      const int kvb = (idx / (D / 2)) * 4;
+      v_c0[it].u = v_c1[it].u = 0;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int g_row = min(kv0 + kvb + i, Skv - 1);
+        const uint32_t pair = *reinterpret_cast<const uint32_t*>(
+            v + kv_base + (int64_t)g_row * kv_row_stride + d0);
+        v_c0[it].s[i] = (ushort)(pair & 0xffffu);
+        v_c1[it].s[i] = (ushort)(pair >> 16);
       }
     }
-    __syncthreads();
+  };
+
+  auto store_lds = [&]() {
+#pragma unroll
+    for (int it = 0; it < (kKvBlk * D) / (256 * 8); ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int byte = (col * 2) ^ ((row & swz_rm_mask<D>()) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(k_lds) + row * kRowBytes + byte) = k_reg[it];
+    }
+#pragma unroll
+    for (int it = 0; it < (kKvBlk / 4) * (D / 2) / 256; ++it) {
+      const int idx = threadIdx.x + it * 256;
+      const int d0 = (idx % (D / 2)) * 2;
+      const int kvb = (idx / (D / 2)) * 4;
+      const int byte0 = (kvb * 2) ^ ((d0 & 7) << 4);
+      const int byte1 = (kvb * 2) ^ (((d0 + 1) & 7) << 4);
+      *reinterpret_cast<uint64_t*>(
+          reinterpret_cast<char*>(vt_lds) + d0 * (kKvBlk * 2) + byte0) = v_c0[it].u;
+      *reinterpret_cast<uint64_t*>(
+          reinterpret_cast<char*>(vt_lds) + (d0 + 1) * (kKvBlk * 2) + byte1) = v_c1[it].u;
+    }
+  };
+
+  load_regs(0);
+  store_lds();
+  __syncthreads();
+
+  for (int kt = 0; kt < num_kv_tiles; ++kt) {
+    const int kv0 = kt * kKvBlk;
+    if (kt + 1 < num_kv_tiles) load_regs(kt + 1);  // issue early
 
     // per m-tile: QK^T -> softmax -> P -> PV
 #pragma unroll
@@ -291,6 +317,10 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       }
     }
     __syncthreads();
+    if (kt + 1 < num_kv_tiles) {
+      store_lds();
+      __syncthreads();
+    }
   }
 
   // ---- epilogue: O /= l; stage in LDS; coalesced 16B stores; LSE -----------

@@ -84,59 +84,78 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
 
   const int n_ktiles = (K + kBK - 1) / kBK;
 
-  auto stage = [&](int kt, int buf) {
+  // T14 split staging: per k-tile, 4x bf16x8 of A and 2x bf16x8 of B per
+  // thread are loaded into registers EARLY (issue overlaps the previous
+  // tile's MFMAs) and written to the alternate LDS buffer late.
+  bf16x8 a_reg[4], b_reg[2];
+
+  auto load_regs = [&](int kt) {
     const int k0 = kt * kBK;
-    // A: kBM x kBK, 16B per thread -> 128*64*2/16 = 1024 ops / 256 threads = 4
-    bf16_t* al = a_lds + buf * kBM * kBK;
 #pragma unroll
-    for (int it = 0; it < (kBM * kBK) / (256 * 8); ++it) {
+    for (int it = 0; it < 4; ++it) {
       const int idx = (threadIdx.x + it * 256) * 8;
       const int row = idx / kBK;
       const int col = idx % kBK;
       const int g_row = row0 + row;
-      bf16x8 val = {};
+      a_reg[it] = bf16x8{};
       if (g_row < row_end && k0 + col < K) {
-        val = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + col);
+        a_reg[it] = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + col);
       }
-      const int byte = (col * 2) ^ ((row & 7) << 4);
-      *reinterpret_cast<bf16x8*>(
-          reinterpret_cast<char*>(al) + row * (kBK * 2) + byte) = val;
     }
-    // B^T: stage b[k0..k0+63][n0..n0+63] transposed into bt[n][k].
-    bf16_t* bl = bt_lds + buf * kBN * kBK;
 #pragma unroll
-    for (int it = 0; it < (kBN * kBK) / (256 * 8); ++it) {
+    for (int it = 0; it < 2; ++it) {
       const int idx = (threadIdx.x + it * 256) * 8;
-      const int k = idx / kBN;       // 0..63 within tile
-      const int n = idx % kBN;       // 0..56 step 8
-      bf16x8 val = {};
-      if (k0 + k < K) {
+      const int kk = idx / kBN;
+      const int n = idx % kBN;
+      b_reg[it] = bf16x8{};
+      if (k0 + kk < K) {
         if (n0 + n + 7 < N) {
-          val = *reinterpret_cast<const bf16x8*>(b_e + (int64_t)(k0 + k) * N + n0 + n);
+          b_reg[it] = *reinterpret_cast<const bf16x8*>(b_e + (int64_t)(k0 + kk) * N + n0 + n);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            val[j] = (n0 + n + j < N) ? b_e[(int64_t)(k0 + k) * N + n0 + n + j]
-                                      : (bf16_t)0.f;
+            b_reg[it][j] = (n0 + n + j < N) ? b_e[(int64_t)(k0 + kk) * N + n0 + n + j]
+                                            : (bf16_t)0.f;
           }
         }
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int nn = n + j;
-        const int byte = (k * 2) ^ ((nn & 7) << 4);
-        *reinterpret_cast<bf16_t*>(
-            reinterpret_cast<char*>(bl) + nn * (kBK * 2) + byte) = val[j];
       }
     }
   };
 
-  stage(0, 0);
+  auto store_lds = [&](int buf) {
+    bf16_t* al = a_lds + buf * kBM * kBK;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int row = idx / kBK;
+      const int col = idx % kBK;
+      const int byte = (col * 2) ^ ((row & 7) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(al) + row * (kBK * 2) + byte) = a_reg[it];
+    }
+    bf16_t* bl = bt_lds + buf * kBN * kBK;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int kk = idx / kBN;
+      const int n = idx % kBN;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int nn = n + j;
+        const int byte = (kk * 2) ^ ((nn & 7) << 4);
+        *reinterpret_cast<bf16_t*>(
+            reinterpret_cast<char*>(bl) + nn * (kBK * 2) + byte) = b_reg[it][j];
+      }
+    }
+  };
+
+  load_regs(0);
+  store_lds(0);
   __syncthreads();
 
   for (int kt = 0; kt < n_ktiles; ++kt) {
     const int buf = kt & 1;
-    if (kt + 1 < n_ktiles) stage(kt + 1, buf ^ 1);
+    if (kt + 1 < n_ktiles) load_regs(kt + 1);  // issue early, use late
 
     const bf16_t* al = a_lds + buf * kBM * kBK;
     const bf16_t* bl = bt_lds + buf * kBN * kBK;
@@ -162,6 +181,9 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
       for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = mfma16g(a_frag[i], b_frag[j], acc[i][j]);
+    }
+    if (kt + 1 < n_ktiles) {
+      store_lds(buf ^ 1);  // writes into the idle buffer; vmcnt waits land here
     }
     __syncthreads();
   }
@@ -215,41 +237,75 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  for (int rt = r_start; rt < r_end; rt += 64) {
-    __syncthreads();
-    // stage a^T and g^T: [dim 128][row 64]; 128*64 elems each, 8 per thread op
-    for (int idx = threadIdx.x * 8; idx < 128 * 64; idx += 256 * 8) {
-      const int row = idx / 128;   // row within 64-chunk
-      const int d0 = idx % 128;    // dim offset (8 contiguous)
-      const int g_row = rt + row;
-      bf16x8 va = {}, vg = {};
-      if (g_row < r_end) {
-        if (k0 + d0 + 7 < K) {
-          va = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + d0);
-        } else {
+  // T14 staging: 4 (row x 2-col) blocks per thread per tensor, loaded into
+  // registers early (overlapping the previous chunk's MFMAs), transposed into
+  // LDS with 8-byte writes late. Rows past r_end are zero-filled (they enter
+  // the row-sum).
+  union u64u {
+    uint64_t u;
+    ushort s[4];
+  };
+  u64u a_c0[4], a_c1[4], g_c0[4], g_c1[4];
+
+  auto load_regs = [&](int rt) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            va[j] = (k0 + d0 + j < K) ? a[(int64_t)g_row * K + k0 + d0 + j] : (bf16_t)0.f;
+    for (int it = 0; it < 4; ++it) {
+      const int idx = threadIdx.x + it * 256;
+      const int d0 = (idx % 64) * 2;
+      const int rb = (idx / 64) * 4;
+      a_c0[it].u = a_c1[it].u = g_c0[it].u = g_c1[it].u = 0;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int g_row = rt + rb + i;
+        if (g_row < r_end) {
+          if (k0 + d0 + 1 < K) {
+            const uint32_t pa = *reinterpret_cast<const uint32_t*>(
+                a + (int64_t)g_row * K + k0 + d0);
+            a_c0[it].s[i] = (ushort)(pa & 0xffffu);
+            a_c1[it].s[i] = (ushort)(pa >> 16);
+          } else if (k0 + d0 < K) {
+            a_c0[it].s[i] = *reinterpret_cast<const ushort*>(
+                a + (int64_t)g_row * K + k0 + d0);
+          }
+          if (n0 + d0 + 1 < N) {
+            const uint32_t pg = *reinterpret_cast<const uint32_t*>(
+                g + (int64_t)g_row * N + n0 + d0);
+            g_c0[it].s[i] = (ushort)(pg & 0xffffu);
+            g_c1[it].s[i] = (ushort)(pg >> 16);
+          } else if (n0 + d0 < N) {
+            g_c0[it].s[i] = *reinterpret_cast<const ushort*>(
+                g + (int64_t)g_row * N + n0 + d0);
+          }
         }
-        if (n0 + d0 + 7 < N) {
-          vg = *reinterpret_cast<const bf16x8*>(g + (int64_t)g_row * N + n0 + d0);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vg[j] = (n0 + d0 + j < N) ? g[(int64_t)g_row * N + n0 + d0 + j] : (bf16_t)0.f;
-        }
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d = d0 + j;
-        const int byte = (row * 2) ^ ((d & 7) << 4);
-        *reinterpret_cast<bf16_t*>(
-            reinterpret_cast<char*>(at_lds) + d * (64 * 2) + byte) = va[j];
-        *reinterpret_cast<bf16_t*>(
-            reinterpret_cast<char*>(gt_lds) + d * (64 * 2) + byte) = vg[j];
       }
     }
-    __syncthreads();
+  };
+
+  auto store_lds = [&]() {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = threadIdx.x + it * 256;
+      const int d0 = (idx % 64) * 2;
+      const int rb = (idx / 64) * 4;
+      const int byte0 = (rb * 2) ^ ((d0 & 7) << 4);
+      const int byte1 = (rb * 2) ^ (((d0 + 1) & 7) << 4);
+      *reinterpret_cast<uint64_t*>(
+          reinterpret_cast<char*>(at_lds) + d0 * 128 + byte0) = a_c0[it].u;
+      *reinterpret_cast<uint64_t*>(
+          reinterpret_cast<char*>(at_lds) + (d0 + 1) * 128 + byte1) = a_c1[it].u;
+      *reinterpret_cast<uint64_t*>(
+          reinterpret_cast<char*>(gt_lds) + d0 * 128 + byte0) = g_c0[it].u;
+      *reinterpret_cast<uint64_t*>(
+          reinterpret_cast<char*>(gt_lds) + (d0 + 1) * 128 + byte1) = g_c1[it].u;
+    }
+  };
+
+  load_regs(r_start);
+  store_lds();
+  __syncthreads();
+
+  for (int rt = r_start; rt < r_end; rt += 64) {
+    if (rt + 64 < r_end) load_regs(rt + 64);  // issue early
 
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {  // 64 rows -> 2 k-steps
@@ -273,6 +329,11 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
       for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[i][j] = mfma16g(a_frag[i], g_frag[j], acc[i][j]);
+    }
+    __syncthreads();
+    if (rt + 64 < r_end) {
+      store_lds();
+      __syncthreads();
     }
   }
 
