@@ -89,6 +89,9 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
   // tile's MFMAs) and written to the alternate LDS buffer late.
   bf16x8 a_reg[4], b_reg[2];
 
+  // Branchless staging: clamp addresses into range, select-zero after the
+  // load (a per-element branch around a load makes hipcc serialize every
+  // load behind a vmcnt(0) drain).
   auto load_regs = [&](int kt) {
     const int k0 = kt * kBK;
 #pragma unroll
@@ -97,27 +100,33 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
       const int row = idx / kBK;
       const int col = idx % kBK;
       const int g_row = row0 + row;
-      a_reg[it] = bf16x8{};
-      if (g_row < row_end && k0 + col < K) {
-        a_reg[it] = *reinterpret_cast<const bf16x8*>(a + (int64_t)g_row * K + k0 + col);
-      }
+      const bool ok = g_row < row_end && k0 + col + 7 < K;
+      const int64_t sr = min(g_row, row_end - 1);
+      const int64_t sc = min(k0 + col, max(K - 8, 0));
+      const bf16x8 val = *reinterpret_cast<const bf16x8*>(a + sr * K + sc);
+      a_reg[it] = ok ? val : bf16x8{};
     }
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       const int idx = (threadIdx.x + it * 256) * 8;
       const int kk = idx / kBN;
       const int n = idx % kBN;
-      b_reg[it] = bf16x8{};
-      if (k0 + kk < K) {
-        if (n0 + n + 7 < N) {
-          b_reg[it] = *reinterpret_cast<const bf16x8*>(b_e + (int64_t)(k0 + kk) * N + n0 + n);
-        } else {
+      const bool ok = k0 + kk < K && n0 + n + 7 < N;
+      const int64_t sk = min(k0 + kk, K - 1);
+      const int64_t sn = min(n0 + n, max(N - 8, 0));
+      const bf16x8 val = *reinterpret_cast<const bf16x8*>(b_e + sk * N + sn);
+      if (ok) {
+        b_reg[it] = val;
+      } else {
+        // slow edge path (rare): per-element fill
+        bf16x8 ev = {};
+        if (k0 + kk < K) {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            b_reg[it][j] = (n0 + n + j < N) ? b_e[(int64_t)(k0 + kk) * N + n0 + n + j]
-                                            : (bf16_t)0.f;
+            if (n0 + n + j < N) ev[j] = b_e[(int64_t)(k0 + kk) * N + n0 + n + j];
           }
         }
+        b_reg[it] = ev;
       }
     }
   };
@@ -253,30 +262,23 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
       const int idx = threadIdx.x + it * 256;
       const int d0 = (idx % 64) * 2;
       const int rb = (idx / 64) * 4;
-      a_c0[it].u = a_c1[it].u = g_c0[it].u = g_c1[it].u = 0;
+      const bool k_ok = k0 + d0 + 1 < K;
+      const bool n_ok = n0 + d0 + 1 < N;
+      const int64_t sk = min((int64_t)(k0 + d0), (int64_t)max(K - 2, 0));
+      const int64_t sn = min((int64_t)(n0 + d0), (int64_t)max(N - 2, 0));
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int g_row = rt + rb + i;
-        if (g_row < r_end) {
-          if (k0 + d0 + 1 < K) {
-            const uint32_t pa = *reinterpret_cast<const uint32_t*>(
-                a + (int64_t)g_row * K + k0 + d0);
-            a_c0[it].s[i] = (ushort)(pa & 0xffffu);
-            a_c1[it].s[i] = (ushort)(pa >> 16);
-          } else if (k0 + d0 < K) {
-            a_c0[it].s[i] = *reinterpret_cast<const ushort*>(
-                a + (int64_t)g_row * K + k0 + d0);
-          }
-          if (n0 + d0 + 1 < N) {
-            const uint32_t pg = *reinterpret_cast<const uint32_t*>(
-                g + (int64_t)g_row * N + n0 + d0);
-            g_c0[it].s[i] = (ushort)(pg & 0xffffu);
-            g_c1[it].s[i] = (ushort)(pg >> 16);
-          } else if (n0 + d0 < N) {
-            g_c0[it].s[i] = *reinterpret_cast<const ushort*>(
-                g + (int64_t)g_row * N + n0 + d0);
-          }
-        }
+        const bool row_ok = g_row < r_end;
+        const int64_t sr = min((int64_t)g_row, (int64_t)(r_end - 1));
+        const uint32_t pa = *reinterpret_cast<const uint32_t*>(a + sr * K + sk);
+        const uint32_t pg = *reinterpret_cast<const uint32_t*>(g + sr * N + sn);
+        const uint32_t va = (row_ok && k_ok) ? pa : 0u;
+        const uint32_t vg = (row_ok && n_ok) ? pg : 0u;
+        a_c0[it].s[i] = (ushort)(va & 0xffffu);
+        a_c1[it].s[i] = (ushort)(va >> 16);
+        g_c0[it].s[i] = (ushort)(vg & 0xffffu);
+        g_c1[it].s[i] = (ushort)(vg >> 16);
       }
     }
   };

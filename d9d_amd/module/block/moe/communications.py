@@ -20,7 +20,7 @@ from ....ops import moe_permute, moe_unpermute
 @dataclass
 class DispatchContext:
     num_tokens: int
-    row_to_token: torch.Tensor
+    permute_ctx: tuple
     permuted_probs: torch.Tensor
     extra: dict[str, Any] = field(default_factory=dict)
 
@@ -48,20 +48,18 @@ class NoCommunicationHandler(MoECommunicationHandler):
         self.num_experts = num_experts
 
     def dispatch(self, tokens, probs, indices):
-        permuted, permuted_probs, row_to_token, tokens_per_expert = moe_permute(
+        permuted, permuted_probs, permute_ctx, tokens_per_expert = moe_permute(
             tokens, indices, probs, self.num_experts
         )
         ctx = DispatchContext(
             num_tokens=tokens.shape[0],
-            row_to_token=row_to_token,
+            permute_ctx=permute_ctx,
             permuted_probs=permuted_probs,
         )
         return permuted, tokens_per_expert.cpu(), ctx
 
     def combine(self, expert_out, ctx):
-        return moe_unpermute(
-            expert_out, ctx.permuted_probs, ctx.row_to_token, ctx.num_tokens
-        )
+        return moe_unpermute(expert_out, ctx.permuted_probs, ctx.permute_ctx)
 
 
 class _AllToAllSingle(torch.autograd.Function):
@@ -111,7 +109,7 @@ class RcclAllToAllCommunicationHandler(MoECommunicationHandler):
 
     def dispatch(self, tokens, probs, indices):
         # Local sort by global expert id == by (dest rank, dest local expert).
-        permuted, permuted_probs, row_to_token, tokens_per_expert = moe_permute(
+        permuted, permuted_probs, permute_ctx, tokens_per_expert = moe_permute(
             tokens, indices, probs, self.num_experts
         )
         # Split sizes: rows per destination rank.
@@ -140,7 +138,7 @@ class RcclAllToAllCommunicationHandler(MoECommunicationHandler):
 
         ctx = DispatchContext(
             num_tokens=tokens.shape[0],
-            row_to_token=row_to_token,
+            permute_ctx=permute_ctx,
             permuted_probs=permuted_probs,
             extra={
                 "order": order,
@@ -160,6 +158,4 @@ class RcclAllToAllCommunicationHandler(MoECommunicationHandler):
         back = _all_to_all(
             by_src, ctx.extra["in_splits"], ctx.extra["out_splits"], self.group
         )
-        return moe_unpermute(
-            back, ctx.permuted_probs, ctx.row_to_token, ctx.num_tokens
-        )
+        return moe_unpermute(back, ctx.permuted_probs, ctx.permute_ctx)
