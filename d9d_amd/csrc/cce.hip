@@ -1,0 +1,234 @@
+// Fused linear + cross-entropy FORWARD for CDNA4 (gfx950).
+//
+// Replaces the cut-cross-entropy forward (reference: d9d/kernel/cce/cce.py
+// cce_lse_forward_kernel): per token, lse over the full vocab and the target
+// logit, WITHOUT materializing the (T, V) logits. The (T,V) GEMM runs on
+// MFMA inside the kernel with an online logsumexp.
+//
+// Geometry: one block = 64 token rows (4 waves x 16-row m-tile). The token
+// block's embeddings (64 x K bf16) live in LDS for the whole sweep; the
+// classifier streams as (64 vocab x 64 k) tiles, T14-staged (loads issued a
+// tile early into registers, written to the alternate LDS buffer after the
+// MFMAs). All resident blocks sweep the vocab in the same order, so the
+// classifier is served from L2/L3 after the first wavefront of blocks.
+//
+// K (hidden) is a runtime dim; K <= 1024 fits the 160 KiB LDS (the host
+// falls back to the chunked rocBLAS path above that). Backward stays the
+// chunked-GEMM path (cce.py) — it is GEMM-shaped and rocBLAS-efficient.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace d9d {
+
+typedef __bf16 bf16_t;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+D9D_DEVICE f32x4 mfma16c(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+constexpr float kLog2eC = 1.44269504088896340736f;
+
+__global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
+    const bf16_t* __restrict__ e,   // (T, K)
+    const bf16_t* __restrict__ c,   // (V, K)
+    const int64_t* __restrict__ targets,  // (T,)
+    float* __restrict__ lse_out,          // (T,)
+    float* __restrict__ tgt_out,          // (T,) target logit or -inf
+    int T, int V, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* e_lds = reinterpret_cast<bf16_t*>(smem);       // [64][K] swizzled
+  bf16_t* c_lds = e_lds + 64 * K;                        // 2 x [64][64]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int t0 = blockIdx.x * 64;
+
+  // ---- stage e block (row-major [64][K], ((row&15)<<4) swizzle) ------------
+  for (int idx = threadIdx.x * 8; idx < 64 * K; idx += 256 * 8) {
+    const int row = idx / K;
+    const int col = idx % K;
+    const int g_row = min(t0 + row, T - 1);
+    const bf16x8 val = *reinterpret_cast<const bf16x8*>(e + (int64_t)g_row * K + col);
+    const int byte = (col * 2) ^ ((row & 15) << 4);
+    *reinterpret_cast<bf16x8*>(
+        reinterpret_cast<char*>(e_lds) + row * (K * 2) + byte) = val;
+  }
+  __syncthreads();
+
+  const int my_row_local = wave * 16 + (lane & 15);
+
+  float m_run[4], l_run[4], tgt[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -1e30f;
+    l_run[r] = 0.f;
+    tgt[r] = -1e30f;
+  }
+
+  const int n_vtiles = (V + 63) / 64;
+  const int n_ktiles = K / 64;  // K % 64 == 0 enforced by the host
+
+  // T14 c staging: per (vtile, ktile) iteration each thread owns 2 bf16x8.
+  bf16x8 c_reg[2];
+  auto load_c = [&](int it_lin) {
+    const int vt = it_lin / n_ktiles;
+    const int kt = it_lin % n_ktiles;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int idx = (threadIdx.x + i * 256) * 8;
+      const int vrow = idx / 64;  // vocab row within tile
+      const int col = idx % 64;
+      const int g_v = min(vt * 64 + vrow, V - 1);
+      c_reg[i] = *reinterpret_cast<const bf16x8*>(
+          c + (int64_t)g_v * K + kt * 64 + col);
+    }
+  };
+  auto store_c = [&](int buf) {
+    bf16_t* cl = c_lds + buf * 64 * 64;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int idx = (threadIdx.x + i * 256) * 8;
+      const int vrow = idx / 64;
+      const int col = idx % 64;
+      const int byte = (col * 2) ^ ((vrow & 7) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(cl) + vrow * 128 + byte) = c_reg[i];
+    }
+  };
+
+  const int total_iters = n_vtiles * n_ktiles;
+  load_c(0);
+  store_c(0);
+  __syncthreads();
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int it = 0; it < total_iters; ++it) {
+    const int vt = it / n_ktiles;
+    const int kt = it % n_ktiles;
+    const int buf = it & 1;
+    if (it + 1 < total_iters) load_c(it + 1);
+
+    const bf16_t* cl = c_lds + buf * 64 * 64;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kk = kt * 64 + ks * 32 + (lane >> 4) * 8;
+      const int ebyte = (kk * 2) ^ ((my_row_local & 15) << 4);
+      const bf16x8 ea = *reinterpret_cast<const bf16x8*>(
+          reinterpret_cast<char*>(e_lds) + my_row_local * (K * 2) + ebyte);
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int vrow = nt * 16 + (lane & 15);
+        const int cbyte = ((ks * 32 + (lane >> 4) * 8) * 2) ^ ((vrow & 7) << 4);
+        const bf16x8 cb = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(cl) + vrow * 128 + cbyte);
+        acc[nt] = mfma16c(ea, cb, acc[nt]);
+      }
+    }
+
+    if (kt == n_ktiles - 1) {
+      // logits tile complete: online lse + target capture, reset acc.
+      float m_new[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) m_new[r] = m_run[r];
+      float vals[4][4];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int col = vt * 64 + nt * 16 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          vals[nt][r] = (col < V) ? acc[nt][r] : -1e30f;
+        }
+      }
+      // target capture: row = t0 + wave*16 + (lane>>4)*4 + r
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row_g = t0 + wave * 16 + (lane >> 4) * 4 + r;
+        const int64_t tg = targets[min(row_g, T - 1)];
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int col = vt * 64 + nt * 16 + (lane & 15);
+          float hit = (col == (int)tg) ? vals[nt][r] : -1e30f;
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1) {
+            hit = fmaxf(hit, __shfl_xor(hit, off, 64));
+          }
+          tgt[r] = fmaxf(tgt[r], hit);
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float mx = fmaxf(fmaxf(vals[0][r], vals[1][r]),
+                         fmaxf(vals[2][r], vals[3][r]));
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+          mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+        }
+        m_new[r] = fmaxf(m_new[r], mx);
+        float add = 0.f;
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          add += __builtin_amdgcn_exp2f((vals[nt][r] - m_new[r]) * kLog2eC);
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+          add += __shfl_xor(add, off, 64);
+        }
+        const float alpha = __builtin_amdgcn_exp2f((m_run[r] - m_new[r]) * kLog2eC);
+        l_run[r] = l_run[r] * alpha + add;
+        m_run[r] = m_new[r];
+      }
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+    }
+
+    __syncthreads();
+    if (it + 1 < total_iters) {
+      store_c(buf ^ 1);
+    }
+    __syncthreads();
+  }
+
+  // ---- write lse + target logit -------------------------------------------
+  if ((lane & 15) == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row_g = t0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (row_g < T) {
+        lse_out[row_g] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+        tgt_out[row_g] = tgt[r];
+      }
+    }
+  }
+}
+
+}  // namespace d9d
+
+std::vector<torch::Tensor> cce_fwd(
+    torch::Tensor e, torch::Tensor c, torch::Tensor targets) {
+  TORCH_CHECK(e.is_cuda() && e.scalar_type() == torch::kBFloat16 && e.is_contiguous());
+  TORCH_CHECK(c.scalar_type() == torch::kBFloat16 && c.is_contiguous());
+  TORCH_CHECK(targets.scalar_type() == torch::kInt64);
+  const int T = e.size(0), K = e.size(1), V = c.size(0);
+  TORCH_CHECK(K % 64 == 0, "cce_fwd requires hidden % 64 == 0");
+  const size_t smem = (size_t)(64 * K + 2 * 64 * 64) * sizeof(__bf16);
+  TORCH_CHECK(smem <= 160 * 1024, "cce_fwd: hidden too large for LDS: ", K);
+
+  auto lse = torch::empty({T}, e.options().dtype(torch::kFloat32));
+  auto tgt = torch::empty({T}, e.options().dtype(torch::kFloat32));
+  if (T == 0) return {lse, tgt};
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      d9d::cce_fwd_kernel, dim3((T + 63) / 64), dim3(256), smem, stream,
+      reinterpret_cast<const __bf16*>(e.data_ptr()),
+      reinterpret_cast<const __bf16*>(c.data_ptr()),
+      targets.contiguous().data_ptr<int64_t>(),
+      lse.data_ptr<float>(), tgt.data_ptr<float>(), T, V, K);
+  return {lse, tgt};
+}

@@ -40,19 +40,48 @@ def _chunk_fwd(e32, c, targets, vocab_start):
     return lse, tgt_logit
 
 
+def _kernel_forward(e, c, targets, vocab_start):
+    """Fused CDNA4 forward: one kernel computes per-token lse + target logit."""
+    from ._ext import get_ext
+
+    local_targets = (targets - vocab_start).clamp(min=-1)  # out-of-shard -> -1
+    lse, tgt = get_ext().cce_fwd(e.contiguous(), c.contiguous(), local_targets)
+    in_shard = (targets - vocab_start >= 0) & (targets - vocab_start < c.shape[0]) & (
+        targets != LM_IGNORE_INDEX
+    )
+    tgt_logit = torch.where(in_shard, tgt, torch.zeros_like(tgt))
+    return lse, tgt_logit
+
+
+def _can_use_kernel(e, c):
+    from ._ext import has_ext
+
+    return (
+        e.is_cuda
+        and has_ext()
+        and e.dtype == torch.bfloat16
+        and c.dtype == torch.bfloat16
+        and e.shape[1] % 64 == 0
+        and (64 * e.shape[1] + 2 * 64 * 64) * 2 <= 160 * 1024
+    )
+
+
 class _LinearCrossEntropyFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, e, c, targets, vp_group, vocab_start, vocab_end):
         T = e.shape[0]
-        lse_parts = []
-        tgt_parts = []
-        for s in range(0, T, _ROW_CHUNK):
-            sl = slice(s, min(s + _ROW_CHUNK, T))
-            lse_c, tgt_c = _chunk_fwd(e[sl], c, targets[sl], vocab_start)
-            lse_parts.append(lse_c)
-            tgt_parts.append(tgt_c)
-        lse = torch.cat(lse_parts)
-        tgt_logit = torch.cat(tgt_parts)
+        if _can_use_kernel(e, c):
+            lse, tgt_logit = _kernel_forward(e, c, targets, vocab_start)
+        else:
+            lse_parts = []
+            tgt_parts = []
+            for s in range(0, T, _ROW_CHUNK):
+                sl = slice(s, min(s + _ROW_CHUNK, T))
+                lse_c, tgt_c = _chunk_fwd(e[sl], c, targets[sl], vocab_start)
+                lse_parts.append(lse_c)
+                tgt_parts.append(tgt_c)
+            lse = torch.cat(lse_parts)
+            tgt_logit = torch.cat(tgt_parts)
 
         if vp_group is not None:
             # Merge lse across vocab shards: lse_full = log sum_r exp(lse_r).

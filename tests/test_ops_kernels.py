@@ -158,3 +158,47 @@ def test_adamw_gpu_matches_fp32_reference():
     torch.testing.assert_close(v, v_ref, rtol=1e-5, atol=1e-7)
     # SR write is within one ulp of the fp32 result
     torch.testing.assert_close(p.float(), p_ref, rtol=1e-2, atol=1e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("T,V,K", [(256, 5000, 768), (130, 1000, 128), (512, 151669, 768)])
+def test_cce_fwd_kernel_parity(T, V, K):
+    from d9d_amd.ops.cce import _chunk_fwd, _can_use_kernel, _kernel_forward
+
+    device = torch.device("cuda")
+    e = torch.randn(T, K, dtype=torch.bfloat16, device=device) * 0.5
+    c = torch.randn(V, K, dtype=torch.bfloat16, device=device) * 0.02
+    targets = torch.randint(0, V, (T,), device=device)
+    targets[::17] = -100  # ignore index
+
+    assert _can_use_kernel(e, c)
+    lse, tgt = _kernel_forward(e, c, targets, 0)
+    ref_lse, ref_tgt = _chunk_fwd(e, c, targets, 0)
+    torch.testing.assert_close(lse, ref_lse, rtol=2e-3, atol=2e-3)
+    torch.testing.assert_close(tgt, ref_tgt, rtol=2e-3, atol=2e-3)
+
+
+@pytest.mark.gpu
+def test_moe_permute_kernels_parity():
+    from d9d_amd.ops import moe_permute, moe_unpermute
+
+    device = torch.device("cuda")
+    T, H, E, K = 1024, 768, 16, 4
+    tokens = torch.randn(T, H, dtype=torch.bfloat16, device=device, requires_grad=True)
+    indices = torch.stack([torch.randperm(E, device=device)[:K] for _ in range(T)])
+    probs = torch.rand(T, K, device=device)
+    probs = (probs / probs.sum(-1, keepdim=True)).requires_grad_(True)
+
+    perm, pprobs, ctx, counts = moe_permute(tokens, indices, probs, E)
+    out = moe_unpermute(perm * 2.0, pprobs, ctx)
+    out.sum().backward()
+
+    t2 = tokens.detach().float().requires_grad_(True)
+    p2 = probs.detach().float().requires_grad_(True)
+    # reference: out[t] = sum_k p[t,k] * 2 * tokens[t]
+    ref = (p2.sum(-1, keepdim=True) * 2 * t2)
+    ref.sum().backward()
+
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(tokens.grad.float(), t2.grad, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(probs.grad.float(), p2.grad, rtol=2e-2, atol=1e-1)
