@@ -239,6 +239,7 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
   const int n0 = blockIdx.x * 128;
   const int r_start = row_off[e];
   const int r_end = row_off[e + 1];
+  if (r_start >= r_end) return;  // empty expert: db stays zero (host zeros it)
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -270,7 +271,7 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
       for (int i = 0; i < 4; ++i) {
         const int g_row = rt + rb + i;
         const bool row_ok = g_row < r_end;
-        const int64_t sr = min((int64_t)g_row, (int64_t)(r_end - 1));
+        const int64_t sr = min((int64_t)g_row, (int64_t)max(r_end - 1, 0));
         const uint32_t pa = *reinterpret_cast<const uint32_t*>(a + sr * K + sk);
         const uint32_t pg = *reinterpret_cast<const uint32_t*>(g + sr * N + sn);
         const uint32_t va = (row_ok && k_ok) ? pa : 0u;
