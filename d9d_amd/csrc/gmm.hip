@@ -104,7 +104,19 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
       const int64_t sr = min(g_row, row_end - 1);
       const int64_t sc = min(k0 + col, max(K - 8, 0));
       const bf16x8 val = *reinterpret_cast<const bf16x8*>(a + sr * K + sc);
-      a_reg[it] = ok ? val : bf16x8{};
+      if (ok) {
+        a_reg[it] = val;
+      } else {
+        // K-tail / row-tail edge (rare): per-element fill
+        bf16x8 ev = {};
+        if (g_row < row_end) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            if (k0 + col + j < K) ev[j] = a[(int64_t)g_row * K + k0 + col + j];
+          }
+        }
+        a_reg[it] = ev;
+      }
     }
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
