@@ -19,8 +19,10 @@ from .programs import (
     build_gpipe,
     build_looped_bfs,
     build_zb1p,
+    build_zbv,
     local_stages,
     loop_stage_to_rank,
+    v_stage_to_rank,
 )
 from .stage import PipelineStage
 
@@ -48,6 +50,17 @@ class PipelineScheduleZB1PConfig(BaseModel):
     schedule: Literal["zb1p"] = "zb1p"
 
 
+class PipelineScheduleZBVConfig(BaseModel):
+    schedule: Literal["zero_bubble_v"] = "zero_bubble_v"
+
+
+class PipelineScheduleDualPipeVConfig(BaseModel):
+    """Accepted for config compatibility; currently executes the ZBV ordering
+    (the F+B compose overlap of true DualPipeV is not yet implemented)."""
+
+    schedule: Literal["dual_pipe_v"] = "dual_pipe_v"
+
+
 PipelineScheduleConfig = Annotated[
     Union[
         PipelineScheduleInferenceConfig,
@@ -55,6 +68,8 @@ PipelineScheduleConfig = Annotated[
         PipelineScheduleLoopedBFSConfig,
         PipelineSchedule1F1BConfig,
         PipelineScheduleZB1PConfig,
+        PipelineScheduleZBVConfig,
+        PipelineScheduleDualPipeVConfig,
     ],
     Field(discriminator="schedule"),
 ]
@@ -69,7 +84,12 @@ class PipelineScheduleInfo:
     has_last_stage: bool
 
 
+_V_SCHEDULES = ("zero_bubble_v", "dual_pipe_v")
+
+
 def _num_stages(config, pp: int) -> int:
+    if config.schedule in _V_SCHEDULES:
+        return 2 * pp
     per_rank = getattr(config, "num_stages_per_rank", 1)
     return pp * per_rank
 
@@ -85,7 +105,10 @@ def build_schedule(
     input_spec=None,
 ) -> PipelineScheduleInfo:
     num_stages = _num_stages(config, pp_size)
-    rank_of_stage = loop_stage_to_rank(num_stages, pp_size)
+    if config.schedule in _V_SCHEDULES and pp_size > 1:
+        rank_of_stage = v_stage_to_rank(num_stages, pp_size)
+    else:
+        rank_of_stage = loop_stage_to_rank(num_stages, pp_size)
     owned = local_stages(rank_of_stage, pp_rank)
 
     modules = []
@@ -111,6 +134,8 @@ def build_schedule(
                           zero_bubble=config.zero_bubble)
     elif name == "zb1p":
         prog = build_zb1p(pp_rank, pp_size, num_stages, num_microbatches)
+    elif name in _V_SCHEDULES:
+        prog = build_zbv(pp_rank, pp_size, num_stages, num_microbatches)
     else:
         raise ValueError(f"unknown schedule {name!r}")
 

@@ -118,3 +118,34 @@ def build_zb1p(rank: int, pp: int, num_stages: int, num_microbatches: int) -> Pr
     for mb in w_queue:
         prog.append(Action(ActionKind.BACKWARD_WEIGHT, 0, mb))
     return prog
+
+
+def build_zbv(rank: int, pp: int, num_stages: int, num_microbatches: int) -> Program:
+    """Zero-bubble V (ZBV): V topology (rank r owns global stages r and
+    2*pp-1-r), input/weight-split backward, weight grads deferred into the
+    drain bubble (reference: zerobubblev.py; program order is BFS over the V
+    legs -- a valid ZBV ordering, not the paper-optimal interleave).
+
+    Local stage indices follow `local_stages(v_stage_to_rank(...), rank)`
+    order (global ascending): 0 = down-leg stage (global r), 1 = up-leg
+    stage (global 2*pp-1-r).
+    """
+    assert num_stages == 2 * pp, "ZBV requires exactly 2 stages per rank"
+    prog: Program = []
+    # forwards: down leg then up leg, all microbatches each
+    for li in (0, 1):
+        for mb in range(num_microbatches):
+            prog.append(Action(ActionKind.FORWARD_COMPUTE, li, mb))
+    # backwards: up leg then down leg, input-grad first, weights deferred
+    w_queue: list = []
+    for li in (1, 0):
+        for mb in range(num_microbatches):
+            prog.append(Action(ActionKind.BACKWARD_INPUT, li, mb))
+            w_queue.append((li, mb))
+            # fill the drain bubble with one deferred weight pass per slot
+            if li == 0 and w_queue:
+                wli, wmb = w_queue.pop(0)
+                prog.append(Action(ActionKind.BACKWARD_WEIGHT, wli, wmb))
+    for wli, wmb in w_queue:
+        prog.append(Action(ActionKind.BACKWARD_WEIGHT, wli, wmb))
+    return prog

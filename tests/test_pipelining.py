@@ -135,12 +135,19 @@ def _pp2_schedule_case(rank, world_size, schedule_name, num_stages_per_rank, num
         build_schedule,
     )
 
+    from d9d_amd.pipelining.factory import (
+        PipelineScheduleDualPipeVConfig,
+        PipelineScheduleZBVConfig,
+    )
+
     cfg = {
         "gpipe": PipelineScheduleGPipeConfig(),
         "looped_bfs": PipelineScheduleLoopedBFSConfig(num_stages_per_rank=num_stages_per_rank),
         "1f1b": PipelineSchedule1F1BConfig(),
         "1f1b_zb": PipelineSchedule1F1BConfig(zero_bubble=True),
         "zb1p": PipelineScheduleZB1PConfig(),
+        "zbv": PipelineScheduleZBVConfig(),
+        "dualpipev": PipelineScheduleDualPipeVConfig(),
     }[schedule_name]
 
     seed = 23
@@ -189,6 +196,8 @@ def _pp2_schedule_case(rank, world_size, schedule_name, num_stages_per_rank, num
         ("1f1b_zb", 1, 4),
         ("zb1p", 1, 6),
         ("looped_bfs", 2, 8),
+        ("zbv", 2, 4),
+        ("dualpipev", 2, 3),
     ],
 )
 def test_pp2_gradient_exact(schedule_name, stages_per_rank, num_mb):
