@@ -29,10 +29,44 @@ D9D_DEVICE f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+// XOR swizzles keep ds_read_b128 bank-conflict-free (guide G4). The XOR must
+// stay inside the LDS row: row-major [64][D] tiles (D*2-byte rows) use mask
+// kSwzRM(D); transposed [D][64] tiles (128-byte rows) use mask 7.
+template <int D>
+constexpr int swz_rm_mask() {
+  return D == 128 ? 15 : (D >= 64 ? 7 : 3);
+}
+
 union ushort2_t {  // 4 bf16 lanes packed for one 8-byte LDS store
   uint64_t u;
   ushort s[4];
 };
+
+// Transpose a swizzled row-major [64][D] LDS image into a [D][64] LDS image
+// (128-byte rows, ((d&7)<<4) swizzle) using 4B LDS reads + 8B LDS writes.
+template <int D>
+D9D_DEVICE void transpose_lds_tile(bf16_t* dst, const bf16_t* src_rm, int tid) {
+  for (int idx = tid; idx < (64 / 4) * (D / 2); idx += 256) {
+    const int d0 = (idx % (D / 2)) * 2;
+    const int rb = (idx / (D / 2)) * 4;
+    ushort2_t c0, c1;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = rb + i;
+      const int byte = (d0 * 2) ^ ((row & swz_rm_mask<D>()) << 4);
+      const uint32_t pair = *reinterpret_cast<const uint32_t*>(
+          reinterpret_cast<const char*>(src_rm) + row * (D * 2) + byte);
+      c0.s[i] = (ushort)(pair & 0xffffu);
+      c1.s[i] = (ushort)(pair >> 16);
+    }
+    const int byte0 = (rb * 2) ^ ((d0 & 7) << 4);
+    const int byte1 = (rb * 2) ^ (((d0 + 1) & 7) << 4);
+    *reinterpret_cast<uint64_t*>(
+        reinterpret_cast<char*>(dst) + d0 * 128 + byte0) = c0.u;
+    *reinterpret_cast<uint64_t*>(
+        reinterpret_cast<char*>(dst) + (d0 + 1) * 128 + byte1) = c1.u;
+  }
+}
 
 // Stage a (64 rows x D cols) global tile TRANSPOSED into a [D][64] LDS image
 // (128-byte rows, ((d&7)<<4) XOR swizzle) with vectorized 8B LDS writes:
@@ -71,14 +105,6 @@ D9D_DEVICE void stage_transposed_tile(
 constexpr int kQBlk = 64;   // q rows per workgroup (16 per wave)
 constexpr int kKvBlk = 64;  // kv rows per tile
 constexpr float kLog2e = 1.44269504088896340736f;
-
-// XOR swizzles keep ds_read_b128 bank-conflict-free (guide G4). The XOR must
-// stay inside the LDS row: row-major [64][D] tiles (D*2-byte rows) use mask
-// kSwzRM(D); transposed [D][64] tiles (128-byte rows) use mask 7.
-template <int D>
-constexpr int swz_rm_mask() {
-  return D == 128 ? 15 : (D >= 64 ? 7 : 3);
-}
 
 // ---------------------------------------------------------------------------
 // Forward
@@ -474,29 +500,46 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   }
   if (window_left >= 0) q_start = max(q_start, 0);
 
-  for (int qt = q_start; qt < Sq; qt += kQBlk) {
-    // ---- stage Q, Q^T, dO, dO^T -------------------------------------------
-    __syncthreads();
-    {
-      constexpr int elems = kQBlk * D;
-      for (int idx = threadIdx.x * 8; idx < elems; idx += 256 * 8) {
-        const int row = idx / D;
-        const int col = idx % D;
-        const int g_row = min(qt + row, Sq - 1);
-        const bf16x8 qv = *reinterpret_cast<const bf16x8*>(
-            q + q_base + (int64_t)g_row * q_row_stride + col);
-        const bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(
-            dout + q_base + (int64_t)g_row * q_row_stride + col);
-        const int byte = (col * 2) ^ ((row & swz_rm_mask<D>()) << 4);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(q_lds) + row * kRowBytes + byte) = qv;
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(do_lds) + row * kRowBytes + byte) = dv8;
-      }
-      // dO^T (for dV's B fragments), vectorized transpose
-      stage_transposed_tile<D>(
-          t_lds, dout + q_base, q_row_stride, qt, Sq - 1, threadIdx.x);
+  // T14: each q-tile's Q/dO rows are prefetched into registers while the
+  // previous tile's MFMAs run; the transposed images are derived from the
+  // row-major LDS copies (cheap ds_reads) instead of re-reading global.
+  bf16x8 q_reg[(kQBlk * D) / (256 * 8)], do_reg[(kQBlk * D) / (256 * 8)];
+
+  auto load_qdo_regs = [&](int qt_next) {
+#pragma unroll
+    for (int it = 0; it < (kQBlk * D) / (256 * 8); ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int g_row = min(qt_next + row, Sq - 1);
+      q_reg[it] = *reinterpret_cast<const bf16x8*>(
+          q + q_base + (int64_t)g_row * q_row_stride + col);
+      do_reg[it] = *reinterpret_cast<const bf16x8*>(
+          dout + q_base + (int64_t)g_row * q_row_stride + col);
     }
+  };
+  auto store_qdo_lds = [&]() {
+#pragma unroll
+    for (int it = 0; it < (kQBlk * D) / (256 * 8); ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int byte = (col * 2) ^ ((row & swz_rm_mask<D>()) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(q_lds) + row * kRowBytes + byte) = q_reg[it];
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(do_lds) + row * kRowBytes + byte) = do_reg[it];
+    }
+  };
+
+  load_qdo_regs(q_start);
+  store_qdo_lds();
+  __syncthreads();
+
+  for (int qt = q_start; qt < Sq; qt += kQBlk) {
+    if (qt + kQBlk < Sq) load_qdo_regs(qt + kQBlk);  // issue early
+    // dO^T for dV's B fragments, from the row-major LDS image
+    transpose_lds_tile<D>(t_lds, do_lds, threadIdx.x);
     __syncthreads();
 
     // ---- S^T = K @ Q^T (per wave: 16 kv x 64 q) ----------------------------
@@ -597,8 +640,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     // ---- dK += dS^T @ Q  (A = dS^T via x_lds, B = Q^T re-staged) ----------
     {
       __syncthreads();  // x_lds + t_lds reuse
-      stage_transposed_tile<D>(
-          t_lds, q + q_base, q_row_stride, qt, Sq - 1, threadIdx.x);
+      transpose_lds_tile<D>(t_lds, q_lds, threadIdx.x);
       bf16_t* xw = x_lds;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -672,6 +714,11 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
         }
       }
     }
+    __syncthreads();
+    if (qt + kQBlk < Sq) {
+      store_qdo_lds();  // prefetched next Q/dO land after all phases read LDS
+    }
+    __syncthreads();
   }
 
   // ---- flush dK, dV (atomicAdd: GQA groups and padded tiles overlap) -------
