@@ -32,7 +32,7 @@ D9D_DEVICE f32x4 mfma16c(bf16x8 a, bf16x8 b, f32x4 c) {
 
 constexpr float kLog2eC = 1.44269504088896340736f;
 
-__global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
+__global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
     const bf16_t* __restrict__ e,   // (T, K)
     const bf16_t* __restrict__ c,   // (V, K)
     const int64_t* __restrict__ targets,  // (T,)
@@ -41,14 +41,16 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
     int T, int V, int K) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* e_lds = reinterpret_cast<bf16_t*>(smem);       // [64][K] swizzled
-  bf16_t* c_lds = e_lds + 64 * K;                        // 2 x [64][64]
+  bf16_t* c_lds = e_lds + 64 * K;                        // 2 x [128][64]
 
   const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
+  const int wave = threadIdx.x >> 6;   // 8 waves: (vocab half) x (4 m-tiles)
+  const int m_tile = wave & 3;
+  const int vhalf = wave >> 2;
   const int t0 = blockIdx.x * 64;
 
   // ---- stage e block (row-major [64][K], ((row&15)<<4) swizzle) ------------
-  for (int idx = threadIdx.x * 8; idx < 64 * K; idx += 256 * 8) {
+  for (int idx = threadIdx.x * 8; idx < 64 * K; idx += 512 * 8) {
     const int row = idx / K;
     const int col = idx % K;
     const int g_row = min(t0 + row, T - 1);
@@ -59,7 +61,7 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
   }
   __syncthreads();
 
-  const int my_row_local = wave * 16 + (lane & 15);
+  const int my_row_local = m_tile * 16 + (lane & 15);
 
   float m_run[4], l_run[4], tgt[4];
 #pragma unroll
@@ -69,7 +71,7 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
     tgt[r] = -1e30f;
   }
 
-  const int n_vtiles = (V + 63) / 64;
+  const int n_vtiles = (V + 127) / 128;  // 128-wide vocab tiles (2 x 64 halves)
   const int n_ktiles = K / 64;  // K % 64 == 0 enforced by the host
 
   // T14 c staging: per (vtile, ktile) iteration each thread owns 2 bf16x8.
@@ -79,19 +81,19 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
     const int kt = it_lin % n_ktiles;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int idx = (threadIdx.x + i * 256) * 8;
-      const int vrow = idx / 64;  // vocab row within tile
+      const int idx = (threadIdx.x + i * 512) * 8;
+      const int vrow = idx / 64;  // vocab row within the 128-wide tile
       const int col = idx % 64;
-      const int g_v = min(vt * 64 + vrow, V - 1);
+      const int g_v = min(vt * 128 + vrow, V - 1);
       c_reg[i] = *reinterpret_cast<const bf16x8*>(
           c + (int64_t)g_v * K + kt * 64 + col);
     }
   };
   auto store_c = [&](int buf) {
-    bf16_t* cl = c_lds + buf * 64 * 64;
+    bf16_t* cl = c_lds + buf * 128 * 64;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int idx = (threadIdx.x + i * 256) * 8;
+      const int idx = (threadIdx.x + i * 512) * 8;
       const int vrow = idx / 64;
       const int col = idx % 64;
       const int byte = (col * 2) ^ ((vrow & 7) << 4);
@@ -115,7 +117,7 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
     const int buf = it & 1;
     if (it + 1 < total_iters) load_c(it + 1);
 
-    const bf16_t* cl = c_lds + buf * 64 * 64;
+    const bf16_t* cl = c_lds + buf * 128 * 64;
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const int kk = kt * 64 + ks * 32 + (lane >> 4) * 8;
@@ -124,7 +126,7 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
           reinterpret_cast<char*>(e_lds) + my_row_local * (K * 2) + ebyte);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        const int vrow = nt * 16 + (lane & 15);
+        const int vrow = vhalf * 64 + nt * 16 + (lane & 15);
         const int cbyte = ((ks * 32 + (lane >> 4) * 8) * 2) ^ ((vrow & 7) << 4);
         const bf16x8 cb = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const char*>(cl) + vrow * 128 + cbyte);
@@ -140,20 +142,20 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
       float vals[4][4];
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        const int col = vt * 64 + nt * 16 + (lane & 15);
+        const int col = vt * 128 + vhalf * 64 + nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           vals[nt][r] = (col < V) ? acc[nt][r] : -1e30f;
         }
       }
-      // target capture: row = t0 + wave*16 + (lane>>4)*4 + r
+      // target capture: row = t0 + m_tile*16 + (lane>>4)*4 + r
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row_g = t0 + wave * 16 + (lane >> 4) * 4 + r;
+        const int row_g = t0 + m_tile * 16 + (lane >> 4) * 4 + r;
         const int64_t tg = targets[min(row_g, T - 1)];
 #pragma unroll
         for (int nt = 0; nt < 4; ++nt) {
-          const int col = vt * 64 + nt * 16 + (lane & 15);
+          const int col = vt * 128 + vhalf * 64 + nt * 16 + (lane & 15);
           float hit = (col == (int)tg) ? vals[nt][r] : -1e30f;
 #pragma unroll
           for (int off = 1; off < 16; off <<= 1) {
@@ -195,14 +197,33 @@ __global__ __launch_bounds__(256, 1) void cce_fwd_kernel(
     __syncthreads();
   }
 
-  // ---- write lse + target logit -------------------------------------------
+  // ---- merge the two vocab halves, write lse + target logit ---------------
+  // scratch (reuses c_lds): per half, per row: lse and tgt.
+  float* half_lse = reinterpret_cast<float*>(c_lds);          // [2][64]
+  float* half_tgt = half_lse + 2 * 64;                        // [2][64]
+  __syncthreads();
   if ((lane & 15) == 0) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row_g = t0 + wave * 16 + (lane >> 4) * 4 + r;
+      const int row_l = m_tile * 16 + (lane >> 4) * 4 + r;
+      half_lse[vhalf * 64 + row_l] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+      half_tgt[vhalf * 64 + row_l] = tgt[r];
+    }
+  }
+  __syncthreads();
+  if (vhalf == 0 && (lane & 15) == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row_l = m_tile * 16 + (lane >> 4) * 4 + r;
+      const int row_g = t0 + row_l;
       if (row_g < T) {
-        lse_out[row_g] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
-        tgt_out[row_g] = tgt[r];
+        const float a = half_lse[row_l];
+        const float b = half_lse[64 + row_l];
+        const float mx = fmaxf(a, b);
+        lse_out[row_g] = mx + __logf(
+            __builtin_amdgcn_exp2f((a - mx) * kLog2eC) +
+            __builtin_amdgcn_exp2f((b - mx) * kLog2eC));
+        tgt_out[row_g] = fmaxf(half_tgt[row_l], half_tgt[64 + row_l]);
       }
     }
   }
@@ -217,7 +238,7 @@ std::vector<torch::Tensor> cce_fwd(
   TORCH_CHECK(targets.scalar_type() == torch::kInt64);
   const int T = e.size(0), K = e.size(1), V = c.size(0);
   TORCH_CHECK(K % 64 == 0, "cce_fwd requires hidden % 64 == 0");
-  const size_t smem = (size_t)(64 * K + 2 * 64 * 64) * sizeof(__bf16);
+  const size_t smem = (size_t)(64 * K + 2 * 128 * 64) * sizeof(__bf16);
   TORCH_CHECK(smem <= 160 * 1024, "cce_fwd: hidden too large for LDS: ", K);
 
   auto lse = torch::empty({T}, e.options().dtype(torch::kFloat32));
@@ -225,7 +246,7 @@ std::vector<torch::Tensor> cce_fwd(
   if (T == 0) return {lse, tgt};
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(
-      d9d::cce_fwd_kernel, dim3((T + 63) / 64), dim3(256), smem, stream,
+      d9d::cce_fwd_kernel, dim3((T + 63) / 64), dim3(512), smem, stream,
       reinterpret_cast<const __bf16*>(e.data_ptr()),
       reinterpret_cast<const __bf16*>(c.data_ptr()),
       targets.contiguous().data_ptr<int64_t>(),

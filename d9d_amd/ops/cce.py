@@ -62,7 +62,7 @@ def _can_use_kernel(e, c):
         and e.dtype == torch.bfloat16
         and c.dtype == torch.bfloat16
         and e.shape[1] % 64 == 0
-        and (64 * e.shape[1] + 2 * 64 * 64) * 2 <= 160 * 1024
+        and (64 * e.shape[1] + 2 * 128 * 64) * 2 <= 160 * 1024
     )
 
 
