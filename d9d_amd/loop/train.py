@@ -128,13 +128,21 @@ class TrainingConfigurator:
             batch_mesh = ctx.mesh_for("batch")
             dp_rank = batch_mesh.get_local_rank("dp")
             dataset = ShardedDataset(dataset, dp_rank, dp)
-        data_loader = DataLoader(
-            dataset,
-            batch_size=maths.data_loader_batch_size,
-            num_workers=cfg.data_loading.num_workers,
-            pin_memory=cfg.data_loading.pin_memory,
-            collate_fn=self.dataset_provider.collate,
-            drop_last=True,
+        from .data import StatefulDataLoaderLite
+
+        dp_rank = 0
+        if ctx.is_distributed and dp > 1:
+            dp_rank = ctx.mesh_for("batch").get_local_rank("dp")
+        data_loader = StatefulDataLoaderLite(
+            DataLoader(
+                dataset,
+                batch_size=maths.data_loader_batch_size,
+                num_workers=cfg.data_loading.num_workers,
+                pin_memory=cfg.data_loading.pin_memory,
+                collate_fn=self.dataset_provider.collate,
+                drop_last=True,
+            ),
+            dp_rank=dp_rank,
         )
 
         # -- model stages through the schedule factory ------------------------
@@ -278,6 +286,7 @@ class Trainer:
         return _StatefulDict(
             stepper=self.stepper,
             tracked_modules=_ModuleStates(self.modules_by_key),
+            data_loader=self.data_loader,
             optimizer=self.optimizer,
             lr_scheduler=self.lr_scheduler,
             metrics=self.collector,
