@@ -247,6 +247,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       f32x4 s_acc[4];
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) s_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         const int kv_row = nt * 16 + (lane & 15);
@@ -259,6 +260,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
           s_acc[nt] = mfma16(q_frag[m][ks], kb, s_acc[nt]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 
       const int my_q_row = q_tile * kQB + wave * 32 + m * 16;
       float p_val[4][4];  // [nt][r]

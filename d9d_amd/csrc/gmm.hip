@@ -180,6 +180,7 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
 
     const bf16_t* al = a_lds + buf * kBM * kBK;
     const bf16_t* bl = bt_lds + buf * kBN * kBK;
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {  // kBK=64 -> 2 MFMA k-steps
       const int kk = ks * 32 + (lane >> 4) * 8;
@@ -203,6 +204,7 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = mfma16g(a_frag[i], b_frag[j], acc[i][j]);
     }
+    __builtin_amdgcn_s_setprio(0);
     if (kt + 1 < n_ktiles) {
       store_lds(buf ^ 1);  // writes into the idle buffer; vmcnt waits land here
     }
