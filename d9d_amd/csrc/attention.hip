@@ -328,6 +328,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
           }
         }
         const bf16_t* pr = pw;
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks2 = 0; ks2 < 2; ++ks2) {
           const int kv_off = ks2 * 32 + (lane >> 4) * 8;
@@ -342,6 +343,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
             o_acc[m][nt] = mfma16(pa, vb, o_acc[m][nt]);
           }
         }
+        __builtin_amdgcn_s_setprio(0);
       }
     }
     __syncthreads();
@@ -548,6 +550,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     f32x4 st_acc[4];
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) st_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
       const int q_row = nt * 16 + (lane & 15);
@@ -560,6 +563,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
         st_acc[nt] = mfma16(k_frag[ks], qb, st_acc[nt]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- P^T = exp(S^T * scale - lse[q]); masked entries 0 -----------------
     float pt_val[4][4];  // [nt][r]
@@ -583,6 +587,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     f32x4 dpt_acc[4];
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) dpt_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
       const int q_row = nt * 16 + (lane & 15);
@@ -595,6 +600,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
         dpt_acc[nt] = mfma16(v_frag[ks], db, dpt_acc[nt]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- dV += P^T @ dO  (A = P^T via x_lds, B = dO^T) --------------------
     {
@@ -611,6 +617,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
       }
       __syncthreads();
       const bf16_t* xr = x_lds + (wave * 16) * (kQBlk + 8);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         const int q_off = ks2 * 32 + (lane >> 4) * 8;
@@ -625,6 +632,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
           dv_acc[nt] = mfma16(pa, dob, dv_acc[nt]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     // ---- dS^T = P^T * (dP^T - delta[q]) * scale ---------------------------
@@ -655,6 +663,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
       }
       __syncthreads();
       const bf16_t* xr = x_lds + (wave * 16) * (kQBlk + 8);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         const int q_off = ks2 * 32 + (lane >> 4) * 8;
@@ -669,6 +678,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
           dk_acc[nt] = mfma16(da, qb, dk_acc[nt]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     // ---- dQ partial = dS @ K; atomicAdd (A = dS q-major via x_lds) --------
@@ -690,6 +700,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
 #pragma unroll
       for (int nt = 0; nt < kNT; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
       const bf16_t* xr = x_lds + (wave * 16) * (kKvBlk + 8);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         const int kv_off = ks2 * 32 + (lane >> 4) * 8;
@@ -704,6 +715,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
           dq_acc[nt] = mfma16(da, kb, dq_acc[nt]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int q_glob = qt + wave * 16 + (lane >> 4) * 4 + r;

@@ -122,20 +122,16 @@ __global__ void rms_norm_bwd_smalln_kernel(
     ushort* __restrict__ dx,
     float* __restrict__ dw,
     int64_t M, int64_t N, float w_offset) {
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* dw_lds = reinterpret_cast<float*>(smem_raw);  // N fp32, block-shared
-
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int64_t waves_total = (int64_t)gridDim.x * (BLOCK / 64);
-  for (int64_t i = threadIdx.x; i < N; i += BLOCK) dw_lds[i] = 0.f;
-  __syncthreads();
 
-  float wvals[VPL];
+  float wvals[VPL], dw_acc[VPL];
 #pragma unroll
   for (int j = 0; j < VPL; ++j) {
     const int64_t i = (int64_t)j * 64 + lane;
     wvals[j] = (i < N) ? bf16_bits_to_f32(w[i]) + w_offset : 0.f;
+    dw_acc[j] = 0.f;
   }
   const float inv_n = 1.f / (float)N;
   for (int64_t row = blockIdx.x * (BLOCK / 64) + wave; row < M; row += waves_total) {
@@ -158,12 +154,16 @@ __global__ void rms_norm_bwd_smalln_kernel(
       const int64_t i = (int64_t)j * 64 + lane;
       if (i < N) {
         dxrow[i] = f32_to_bf16_rne(inv * (g[j] * wvals[j] - xh[j] * s));
-        atomicAdd(&dw_lds[i], g[j] * xh[j]);  // LDS atomics: cheap on CDNA4
+        dw_acc[j] += g[j] * xh[j];  // register accumulation per wave
       }
     }
   }
-  __syncthreads();
-  for (int64_t i = threadIdx.x; i < N; i += BLOCK) atomicAdd(dw + i, dw_lds[i]);
+  // one global atomic per (wave, element)
+#pragma unroll
+  for (int j = 0; j < VPL; ++j) {
+    const int64_t i = (int64_t)j * 64 + lane;
+    if (i < N) atomicAdd(dw + i, dw_acc[j]);
+  }
 }
 
 // Persistent backward: each block walks rows with stride gridDim, keeps a
@@ -321,7 +321,7 @@ std::vector<torch::Tensor> rms_norm_bwd(
   auto stream = at::hip::getCurrentHIPStream();
   if (N <= 512) {
     const int sgrid = static_cast<int>(std::min<int64_t>((M + 3) / 4, 1024));
-    const size_t ssmem = N * sizeof(float);
+    const size_t ssmem = 0;
 #define SMALL_BWD(VPL)                                                         \
   hipLaunchKernelGGL((d9d::rms_norm_bwd_smalln_kernel<kBlock, VPL>),           \
                      dim3(sgrid), dim3(kBlock), ssmem, stream,                 \

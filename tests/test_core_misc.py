@@ -27,3 +27,50 @@ def test_offload_tensor_identity_preserved_cpu():
     assert id(t) == original_id
     onload_tensor(t, torch.device("cpu"))
     assert t.device.type == "cpu"
+
+
+def test_pipelined_optimizer_aggregate():
+    import torch
+    from torch import nn
+
+    from d9d_amd.pipelining.training import PipelinedLRScheduler, PipelinedOptimizer
+    from d9d_amd.lr_scheduler import PiecewiseLRScheduler, Phase, LinearCurve
+
+    m1, m2 = nn.Linear(4, 4), nn.Linear(4, 4)
+    o1 = torch.optim.SGD(m1.parameters(), lr=0.1)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.2)
+    popt = PipelinedOptimizer([o1, o2])
+    assert len(popt.param_groups) == 2
+    sched = PipelinedLRScheduler([
+        PiecewiseLRScheduler(o1, [Phase(10, LinearCurve(1.0, 0.0))]),
+        PiecewiseLRScheduler(o2, [Phase(10, LinearCurve(1.0, 0.0))]),
+    ])
+    (m1(torch.randn(2, 4)).sum() + m2(torch.randn(2, 4)).sum()).backward()
+    popt.step()
+    sched.step()
+    assert len(sched.get_last_lr()) == 2
+    sd = popt.state_dict()
+    popt.load_state_dict(sd)
+    popt.zero_grad()
+
+
+def test_main_process_stateful():
+    from d9d_amd.internals.state import MainProcessStateful
+
+    class Obj:
+        def __init__(self):
+            self.v = 1
+
+        def state_dict(self):
+            return {"v": self.v}
+
+        def load_state_dict(self, sd):
+            self.v = sd["v"]
+
+    o = Obj()
+    wrapped = MainProcessStateful(o)
+    sd = wrapped.state_dict()
+    assert sd == {"v": 1}
+    o.v = 5
+    wrapped.load_state_dict(sd)
+    assert o.v == 1
