@@ -86,9 +86,11 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
         if vp_group is not None:
             # Merge lse across vocab shards: lse_full = log sum_r exp(lse_r).
             world = dist.get_world_size(vp_group)
-            all_lse = torch.empty((world,) + lse.shape, dtype=lse.dtype, device=lse.device)
-            dist.all_gather_into_tensor(all_lse, lse, group=vp_group)
-            lse = torch.logsumexp(all_lse, dim=0)
+            all_lse = torch.empty(
+                world * lse.numel(), dtype=lse.dtype, device=lse.device
+            )
+            dist.all_gather_into_tensor(all_lse, lse.contiguous(), group=vp_group)
+            lse = torch.logsumexp(all_lse.view(world, -1), dim=0)
             dist.all_reduce(tgt_logit, group=vp_group)
 
         loss = lse - tgt_logit  # = -log p(target)

@@ -182,3 +182,46 @@ def _grad_norm_clip(rank, world_size):
 @pytest.mark.distributed
 def test_grad_norm_clip_ws2():
     assert all(run_distributed(_grad_norm_clip, world_size=2))
+
+
+def _vocab_parallel_cce(rank, world_size):
+    import torch.distributed as dist
+
+    from d9d_amd.ops.cce import VocabParallelOptions, linear_cross_entropy
+
+    torch.manual_seed(2)
+    T, H, V = 12, 16, 40
+    e = torch.randn(T, H)
+    c = torch.randn(V, H) * 0.1
+    targets = torch.randint(0, V, (T,))
+    targets[0] = -100
+
+    # reference: full-vocab loss
+    e_ref = e.clone().requires_grad_(True)
+    c_ref = c.clone().requires_grad_(True)
+    ref = linear_cross_entropy(e_ref, c_ref, targets)
+    ref.sum().backward()
+
+    # vocab-parallel: this rank owns rows [rank*20, rank*20+20)
+    shard = 20
+    e_vp = e.clone().requires_grad_(True)
+    c_local = c[rank * shard : (rank + 1) * shard].clone().requires_grad_(True)
+    vp = VocabParallelOptions(dist.group.WORLD, rank * shard, (rank + 1) * shard)
+    loss = linear_cross_entropy(e_vp, c_local, targets, vocab_parallel=vp)
+    loss.sum().backward()
+
+    torch.testing.assert_close(loss, ref, rtol=1e-4, atol=1e-5)
+    # e grads: each rank computes the full-e grad for its shard's logits;
+    # summed across ranks == reference
+    eg = e_vp.grad.clone()
+    dist.all_reduce(eg)
+    torch.testing.assert_close(eg, e_ref.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(
+        c_local.grad, c_ref.grad[rank * shard : (rank + 1) * shard], rtol=1e-4, atol=1e-5
+    )
+    return True
+
+
+@pytest.mark.distributed
+def test_vocab_parallel_cce_ws2():
+    assert all(run_distributed(_vocab_parallel_cce, world_size=2))
