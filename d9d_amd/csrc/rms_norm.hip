@@ -74,7 +74,8 @@ __global__ void rms_norm_fwd_kernel(
 }
 
 // Small-N specialization (e.g. per-head QK norm, N <= 512): one WAVE per row,
-// row in registers, no LDS round-trip. 4 rows in flight per block.
+// row in registers, no LDS round-trip. Per-lane elements are CONTIGUOUS
+// (lane*VPL + j) so loads/stores vectorize to 4-16 B per lane.
 template <int BLOCK, int VPL>  // VPL = values per lane (N <= 64*VPL)
 __global__ void rms_norm_fwd_smalln_kernel(
     const ushort* __restrict__ x,
@@ -85,30 +86,54 @@ __global__ void rms_norm_fwd_smalln_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int64_t waves_total = (int64_t)gridDim.x * (BLOCK / 64);
+  const int64_t base = (int64_t)lane * VPL;
+  const bool full = (VPL >= 2) && (base + VPL <= N);
+
   float wvals[VPL];
 #pragma unroll
   for (int j = 0; j < VPL; ++j) {
-    const int64_t i = (int64_t)j * 64 + lane;
-    wvals[j] = (i < N) ? bf16_bits_to_f32(w[i]) + w_offset : 0.f;
+    wvals[j] = (base + j < N) ? bf16_bits_to_f32(w[base + j]) + w_offset : 0.f;
   }
   for (int64_t row = blockIdx.x * (BLOCK / 64) + wave; row < M; row += waves_total) {
     const ushort* xrow = x + row * N;
+    ushort bits[VPL];
+    if (full) {
+#pragma unroll
+      for (int j = 0; j + 1 < VPL; j += 2) {
+        const uint32_t p = *reinterpret_cast<const uint32_t*>(xrow + base + j);
+        bits[j] = (ushort)(p & 0xffffu);
+        bits[j + 1] = (ushort)(p >> 16);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) {
+        bits[j] = (base + j < N) ? xrow[base + j] : (ushort)0;
+      }
+    }
     float vals[VPL];
     float sumsq = 0.f;
 #pragma unroll
     for (int j = 0; j < VPL; ++j) {
-      const int64_t i = (int64_t)j * 64 + lane;
-      vals[j] = (i < N) ? bf16_bits_to_f32(xrow[i]) : 0.f;
+      vals[j] = bf16_bits_to_f32(bits[j]);
       sumsq += vals[j] * vals[j];
     }
     sumsq = wave_reduce_sum(sumsq);
     const float inv = rsqrtf(sumsq / (float)N + eps);
     if (lane == 0) inv_rms[row] = inv;
     ushort* yrow = y + row * N;
+    if (full) {
 #pragma unroll
-    for (int j = 0; j < VPL; ++j) {
-      const int64_t i = (int64_t)j * 64 + lane;
-      if (i < N) yrow[i] = f32_to_bf16_rne(vals[j] * inv * wvals[j]);
+      for (int j = 0; j + 1 < VPL; j += 2) {
+        const uint32_t p =
+            (uint32_t)f32_to_bf16_rne(vals[j] * inv * wvals[j]) |
+            ((uint32_t)f32_to_bf16_rne(vals[j + 1] * inv * wvals[j + 1]) << 16);
+        *reinterpret_cast<uint32_t*>(yrow + base + j) = p;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) {
+        if (base + j < N) yrow[base + j] = f32_to_bf16_rne(vals[j] * inv * wvals[j]);
+      }
     }
   }
 }
@@ -125,12 +150,13 @@ __global__ void rms_norm_bwd_smalln_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int64_t waves_total = (int64_t)gridDim.x * (BLOCK / 64);
+  const int64_t base = (int64_t)lane * VPL;
+  const bool full = (VPL >= 2) && (base + VPL <= N);
 
   float wvals[VPL], dw_acc[VPL];
 #pragma unroll
   for (int j = 0; j < VPL; ++j) {
-    const int64_t i = (int64_t)j * 64 + lane;
-    wvals[j] = (i < N) ? bf16_bits_to_f32(w[i]) + w_offset : 0.f;
+    wvals[j] = (base + j < N) ? bf16_bits_to_f32(w[base + j]) + w_offset : 0.f;
     dw_acc[j] = 0.f;
   }
   const float inv_n = 1.f / (float)N;
@@ -138,31 +164,55 @@ __global__ void rms_norm_bwd_smalln_kernel(
     const ushort* xrow = x + row * N;
     const ushort* grow = dy + row * N;
     const float inv = inv_rms[row];
+    ushort xb[VPL], gb[VPL];
+    if (full) {
+#pragma unroll
+      for (int j = 0; j + 1 < VPL; j += 2) {
+        const uint32_t px = *reinterpret_cast<const uint32_t*>(xrow + base + j);
+        const uint32_t pg = *reinterpret_cast<const uint32_t*>(grow + base + j);
+        xb[j] = (ushort)(px & 0xffffu); xb[j + 1] = (ushort)(px >> 16);
+        gb[j] = (ushort)(pg & 0xffffu); gb[j + 1] = (ushort)(pg >> 16);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) {
+        xb[j] = (base + j < N) ? xrow[base + j] : (ushort)0;
+        gb[j] = (base + j < N) ? grow[base + j] : (ushort)0;
+      }
+    }
     float xh[VPL], g[VPL];
     float s = 0.f;
 #pragma unroll
     for (int j = 0; j < VPL; ++j) {
-      const int64_t i = (int64_t)j * 64 + lane;
-      xh[j] = (i < N) ? bf16_bits_to_f32(xrow[i]) * inv : 0.f;
-      g[j] = (i < N) ? bf16_bits_to_f32(grow[i]) : 0.f;
+      xh[j] = bf16_bits_to_f32(xb[j]) * inv;
+      g[j] = bf16_bits_to_f32(gb[j]);
       s += g[j] * wvals[j] * xh[j];
     }
     s = wave_reduce_sum(s) * inv_n;
     ushort* dxrow = dx + row * N;
+    if (full) {
 #pragma unroll
-    for (int j = 0; j < VPL; ++j) {
-      const int64_t i = (int64_t)j * 64 + lane;
-      if (i < N) {
-        dxrow[i] = f32_to_bf16_rne(inv * (g[j] * wvals[j] - xh[j] * s));
-        dw_acc[j] += g[j] * xh[j];  // register accumulation per wave
+      for (int j = 0; j + 1 < VPL; j += 2) {
+        const uint32_t p =
+            (uint32_t)f32_to_bf16_rne(inv * (g[j] * wvals[j] - xh[j] * s)) |
+            ((uint32_t)f32_to_bf16_rne(inv * (g[j + 1] * wvals[j + 1] - xh[j + 1] * s)) << 16);
+        *reinterpret_cast<uint32_t*>(dxrow + base + j) = p;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) {
+        if (base + j < N) {
+          dxrow[base + j] = f32_to_bf16_rne(inv * (g[j] * wvals[j] - xh[j] * s));
+        }
       }
     }
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) dw_acc[j] += g[j] * xh[j];
   }
   // one global atomic per (wave, element)
 #pragma unroll
   for (int j = 0; j < VPL; ++j) {
-    const int64_t i = (int64_t)j * 64 + lane;
-    if (i < N) atomicAdd(dw + i, dw_acc[j]);
+    if (base + j < N) atomicAdd(dw + base + j, dw_acc[j]);
   }
 }
 
