@@ -29,6 +29,10 @@ torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b);
 torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes);
 torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes, int64_t num_experts);
 
+// rope.hip
+std::vector<torch::Tensor> rope_qk(torch::Tensor q, torch::Tensor k,
+                                   torch::Tensor cos_t, torch::Tensor sin_t, double sin_sign);
+
 // cce.hip
 std::vector<torch::Tensor> cce_fwd(torch::Tensor e, torch::Tensor c, torch::Tensor targets);
 
@@ -41,6 +45,7 @@ torch::Tensor moe_row_dot(torch::Tensor grad_out, torch::Tensor expert_out, torc
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
   m.def("gmm_db", &gmm_db, "CDNA4 grouped GEMM weight-grad");
+  m.def("rope_qk", &rope_qk, "fused q/k rotary embedding");
   m.def("cce_fwd", &cce_fwd, "fused linear cross-entropy forward (lse + target logit)");
   m.def("moe_gather_rows", &moe_gather_rows, "MoE row gather (optional scale)");
   m.def("moe_csr_combine", &moe_csr_combine, "MoE weighted replica combine");

@@ -238,6 +238,7 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
     const bf16_t* __restrict__ g,   // (T, N)
     bf16_t* __restrict__ db,        // (E, K, N)
     const int* __restrict__ row_off,
+    const int* __restrict__ expert_order,  // experts sorted by row count desc
     int E, int K, int N) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* at_lds = reinterpret_cast<bf16_t*>(smem);   // [128 k][64 rows]
@@ -248,7 +249,8 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
   const int wm = wave >> 1;  // k-half
   const int wn = wave & 1;   // n-half
 
-  const int e = blockIdx.z;
+  // Heavy experts dispatch first so the tail wave holds the small tiles.
+  const int e = expert_order[blockIdx.z];
   const int k0 = blockIdx.y * 128;
   const int n0 = blockIdx.x * 128;
   const int r_start = row_off[e];
@@ -433,6 +435,9 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
   auto db = torch::zeros({(int64_t)E, (int64_t)K, (int64_t)N}, a.options());
   if (a.size(0) == 0) return db;
   auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
+  auto order_cpu = torch::argsort(batch_sizes.to(torch::kInt64), /*dim=*/0,
+                                  /*descending=*/true).to(torch::kInt32);
+  auto expert_order = order_cpu.to(a.device(), /*non_blocking=*/true);
 
   const dim3 grid((N + 127) / 128, (K + 127) / 128, E);
   const size_t smem = (2 * 128 * 64) * sizeof(__bf16);
@@ -441,6 +446,7 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
                      reinterpret_cast<__bf16*>(db.data_ptr()),
-                     row_off.data_ptr<int>(), E, K, N);
+                     row_off.data_ptr<int>(), expert_order.data_ptr<int>(),
+                     E, K, N);
   return db;
 }

@@ -105,8 +105,13 @@ class GroupedQueryAttention(nn.Module):
         cos, sin = rotary_cos_sin
         cos = cos[..., : self.rope_dim]
         sin = sin[..., : self.rope_dim]
-        q = apply_rotary_emb(q, cos, sin, self.rope_layout)
-        k = apply_rotary_emb(k, cos, sin, self.rope_layout)
+        from ....ops.rope import rope_qk, rope_qk_available
+
+        if self.rope_layout is RopeLayout.HALF and rope_qk_available(q, self.rope_dim):
+            q, k = rope_qk(q, k, cos, sin)
+        else:
+            q = apply_rotary_emb(q, cos, sin, self.rope_layout)
+            k = apply_rotary_emb(k, cos, sin, self.rope_layout)
 
         window = (-1, -1) if self.sliding_window is None else (self.sliding_window, -1)
         q_offset = 0

@@ -202,3 +202,36 @@ def test_moe_permute_kernels_parity():
     torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(tokens.grad.float(), t2.grad, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(probs.grad.float(), p2.grad, rtol=2e-2, atol=1e-1)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("rope_dim", [128, 64])
+def test_rope_qk_kernel_parity(rope_dim):
+    from d9d_amd.module.block.positional import RotaryEmbeddingProvider, apply_rotary_emb
+    from d9d_amd.ops.rope import rope_qk
+
+    device = torch.device("cuda")
+    B, S, Hq, Hkv, D = 2, 64, 4, 2, 128
+    prov = RotaryEmbeddingProvider(rope_dim=rope_dim, device=device)
+    pos = torch.arange(S, device=device).unsqueeze(0).expand(B, S)
+    cos, sin = prov(pos)
+    q = torch.randn(B, S, Hq, D, dtype=torch.bfloat16, device=device, requires_grad=True)
+    k = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device=device, requires_grad=True)
+
+    q_out, k_out = rope_qk(q, k, cos, sin)
+    g_q = torch.randn_like(q_out)
+    g_k = torch.randn_like(k_out)
+    (q_out * g_q).sum().backward()
+    (k_out * g_k).sum().backward()
+
+    q2 = q.detach().clone().requires_grad_(True)
+    k2 = k.detach().clone().requires_grad_(True)
+    q_ref = apply_rotary_emb(q2, cos, sin)
+    k_ref = apply_rotary_emb(k2, cos, sin)
+    (q_ref * g_q).sum().backward()
+    (k_ref * g_k).sum().backward()
+
+    torch.testing.assert_close(q_out.float(), q_ref.float(), rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(k_out.float(), k_ref.float(), rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(q.grad.float(), q2.grad.float(), rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(k.grad.float(), k2.grad.float(), rtol=2e-2, atol=2e-2)
