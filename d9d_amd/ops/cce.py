@@ -112,22 +112,41 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
         ignored = targets == LM_IGNORE_INDEX
         dl = torch.where(ignored, torch.zeros_like(dloss), dloss).float()
 
+        bf16_fast = e.dtype == torch.bfloat16 and c.dtype == torch.bfloat16
         for s in range(0, T, _ROW_CHUNK):
             sl = slice(s, min(s + _ROW_CHUNK, T))
-            e32 = e[sl]
-            logits = torch.matmul(e32, c.t().to(e32.dtype)).float()  # (Tc, V)
-            p = torch.exp(logits - lse[sl].unsqueeze(1))
+            e_chunk = e[sl]
+            logits = torch.matmul(e_chunk, c.t().to(e_chunk.dtype))  # (Tc, V)
             local_targets = targets[sl] - vocab_start
             in_shard = (local_targets >= 0) & (local_targets < V) & (~ignored[sl])
             safe = local_targets.clamp(0, V - 1)
-            p.scatter_add_(
-                1, safe.unsqueeze(1),
-                torch.where(in_shard, -torch.ones_like(safe, dtype=p.dtype), torch.zeros_like(safe, dtype=p.dtype)).unsqueeze(1),
-            )
-            p *= dl[sl].unsqueeze(1)  # (Tc, V) = d logits
-            pb = p.to(c.dtype)
+            if bf16_fast:
+                # softmax in bf16 (exp in (0, 1], ~3 digits: grads are bf16
+                # downstream anyway) — avoids two fp32 copies of (Tc, V).
+                logits -= lse[sl].unsqueeze(1).to(logits.dtype)
+                p = logits.exp_()
+                p.scatter_add_(
+                    1, safe.unsqueeze(1),
+                    torch.where(
+                        in_shard, -torch.ones_like(safe, dtype=p.dtype),
+                        torch.zeros_like(safe, dtype=p.dtype),
+                    ).unsqueeze(1),
+                )
+                p *= dl[sl].unsqueeze(1).to(p.dtype)
+                pb = p
+            else:
+                p = torch.exp(logits.float() - lse[sl].unsqueeze(1))
+                p.scatter_add_(
+                    1, safe.unsqueeze(1),
+                    torch.where(
+                        in_shard, -torch.ones_like(safe, dtype=p.dtype),
+                        torch.zeros_like(safe, dtype=p.dtype),
+                    ).unsqueeze(1),
+                )
+                p *= dl[sl].unsqueeze(1)
+                pb = p.to(c.dtype)
             de[sl] = torch.matmul(pb, c).to(e.dtype)
-            dc += torch.matmul(pb.t(), e32.to(c.dtype)).float()
+            dc += torch.matmul(pb.t(), e_chunk.to(c.dtype)).float()
 
         return de, dc.to(c.dtype), None, None, None, None
 
