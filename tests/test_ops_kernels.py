@@ -221,15 +221,13 @@ def test_rope_qk_kernel_parity(rope_dim):
     q_out, k_out = rope_qk(q, k, cos, sin)
     g_q = torch.randn_like(q_out)
     g_k = torch.randn_like(k_out)
-    (q_out * g_q).sum().backward()
-    (k_out * g_k).sum().backward()
+    ((q_out * g_q).sum() + (k_out * g_k).sum()).backward()
 
     q2 = q.detach().clone().requires_grad_(True)
     k2 = k.detach().clone().requires_grad_(True)
     q_ref = apply_rotary_emb(q2, cos, sin)
     k_ref = apply_rotary_emb(k2, cos, sin)
-    (q_ref * g_q).sum().backward()
-    (k_ref * g_k).sum().backward()
+    ((q_ref * g_q).sum() + (k_ref * g_k).sum()).backward()
 
     torch.testing.assert_close(q_out.float(), q_ref.float(), rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(k_out.float(), k_ref.float(), rtol=2e-2, atol=2e-2)
