@@ -44,9 +44,9 @@ union ushort2_t {  // 4 bf16 lanes packed for one 8-byte LDS store
 
 // Transpose a swizzled row-major [64][D] LDS image into a [D][64] LDS image
 // (128-byte rows, ((d&7)<<4) swizzle) using 4B LDS reads + 8B LDS writes.
-template <int D>
+template <int D, int BLOCK = 256>
 D9D_DEVICE void transpose_lds_tile(bf16_t* dst, const bf16_t* src_rm, int tid) {
-  for (int idx = tid; idx < (64 / 4) * (D / 2); idx += 256) {
+  for (int idx = tid; idx < (64 / 4) * (D / 2); idx += BLOCK) {
     const int d0 = (idx % (D / 2)) * 2;
     const int rb = (idx / (D / 2)) * 4;
     ushort2_t c0, c1;
@@ -71,11 +71,11 @@ D9D_DEVICE void transpose_lds_tile(bf16_t* dst, const bf16_t* src_rm, int tid) {
 // Stage a (64 rows x D cols) global tile TRANSPOSED into a [D][64] LDS image
 // (128-byte rows, ((d&7)<<4) XOR swizzle) with vectorized 8B LDS writes:
 // each op covers a (4 rows x 2 cols) block.
-template <int D>
+template <int D, int ROWS = 64, int BLOCK = 256>
 D9D_DEVICE void stage_transposed_tile(
     bf16_t* dst, const bf16_t* src, int64_t row_stride,
     int row0, int row_clamp, int tid) {
-  for (int idx = tid; idx < (64 / 4) * (D / 2); idx += 256) {
+  for (int idx = tid; idx < (ROWS / 4) * (D / 2); idx += BLOCK) {
     const int d0 = (idx % (D / 2)) * 2;
     const int rb = (idx / (D / 2)) * 4;
     ushort2_t c0, c1;
@@ -90,9 +90,9 @@ D9D_DEVICE void stage_transposed_tile(
     const int byte0 = (rb * 2) ^ ((d0 & 7) << 4);
     const int byte1 = (rb * 2) ^ (((d0 + 1) & 7) << 4);
     *reinterpret_cast<uint64_t*>(
-        reinterpret_cast<char*>(dst) + d0 * 128 + byte0) = c0.u;
+        reinterpret_cast<char*>(dst) + d0 * (ROWS * 2) + byte0) = c0.u;
     *reinterpret_cast<uint64_t*>(
-        reinterpret_cast<char*>(dst) + (d0 + 1) * 128 + byte1) = c1.u;
+        reinterpret_cast<char*>(dst) + (d0 + 1) * (ROWS * 2) + byte1) = c1.u;
   }
 }
 
@@ -430,8 +430,11 @@ __global__ void attn_delta_kernel(
   }
 }
 
+constexpr int kBwdKv = 128;      // kv rows per block (8 waves x 16)
+constexpr int kBwdThreads = 512;
+
 template <int D>
-__global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
+__global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
     const bf16_t* __restrict__ q,
     const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v,
@@ -451,11 +454,11 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);   // [64][D] swizzled rows
   bf16_t* do_lds = q_lds + kQBlk * D;                // [64][D]
   bf16_t* t_lds = do_lds + kQBlk * D;                // [D][64]: dO^T then Q^T
-  bf16_t* kt_lds = t_lds + D * kQBlk;                // [D][64]
-  bf16_t* x_lds = kt_lds + D * kKvBlk;               // [64][64+8] shared scratch
+  bf16_t* kt_lds = t_lds + D * kQBlk;                // [D][kBwdKv]
+  bf16_t* x_lds = kt_lds + D * kBwdKv;               // [kBwdKv][64+8] scratch
 
   const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
+  const int wave = threadIdx.x >> 6;   // 8 waves x 16 kv rows = 128 kv/block
   const int kv_tile = blockIdx.x;
   const int bh = blockIdx.y;
   const int b = bh / Hq;
@@ -469,7 +472,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   const float* lse_row = lse + ((int64_t)b * Hq + h) * Sq;
   const float* delta_row = delta + ((int64_t)b * Hq + h) * Sq;
 
-  const int kv0 = kv_tile * kKvBlk;
+  const int kv0 = kv_tile * kBwdKv;
 
   // This wave's 16 kv rows: K and V A-fragments in registers.
   const int kv_row_local = lane & 15;
@@ -487,7 +490,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   }
 
   // Stage K^T once (for dQ's B fragments).
-  stage_transposed_tile<D>(
+  stage_transposed_tile<D, kBwdKv, kBwdThreads>(
       kt_lds, k + kv_base, kv_row_stride, kv0, Skv - 1, threadIdx.x);
 
   f32x4 dk_acc[kNT], dv_acc[kNT];
@@ -507,12 +510,15 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   // T14: each q-tile's Q/dO rows are prefetched into registers while the
   // previous tile's MFMAs run; the transposed images are derived from the
   // row-major LDS copies (cheap ds_reads) instead of re-reading global.
-  bf16x8 q_reg[(kQBlk * D) / (256 * 8)], do_reg[(kQBlk * D) / (256 * 8)];
+  constexpr int kQdoRegs =
+      (kQBlk * D + kBwdThreads * 8 - 1) / (kBwdThreads * 8);  // >= 1 (D <= 128)
+  bf16x8 q_reg[kQdoRegs], do_reg[kQdoRegs];
 
   auto load_qdo_regs = [&](int qt_next) {
 #pragma unroll
-    for (int it = 0; it < (kQBlk * D) / (256 * 8); ++it) {
-      const int idx = (threadIdx.x + it * 256) * 8;
+    for (int it = 0; it < kQdoRegs; ++it) {
+      const int idx = (threadIdx.x + it * kBwdThreads) * 8;
+      if (idx >= kQBlk * D) break;
       const int row = idx / D;
       const int col = idx % D;
       const int g_row = min(qt_next + row, Sq - 1);
@@ -524,8 +530,9 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   };
   auto store_qdo_lds = [&]() {
 #pragma unroll
-    for (int it = 0; it < (kQBlk * D) / (256 * 8); ++it) {
-      const int idx = (threadIdx.x + it * 256) * 8;
+    for (int it = 0; it < kQdoRegs; ++it) {
+      const int idx = (threadIdx.x + it * kBwdThreads) * 8;
+      if (idx >= kQBlk * D) break;
       const int row = idx / D;
       const int col = idx % D;
       const int byte = (col * 2) ^ ((row & swz_rm_mask<D>()) << 4);
@@ -543,7 +550,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
   for (int qt = q_start; qt < Sq; qt += kQBlk) {
     if (qt + kQBlk < Sq) load_qdo_regs(qt + kQBlk);  // issue early
     // dO^T for dV's B fragments, from the row-major LDS image
-    transpose_lds_tile<D>(t_lds, do_lds, threadIdx.x);
+    transpose_lds_tile<D, kBwdThreads>(t_lds, do_lds, threadIdx.x);
     __syncthreads();
 
     // ---- S^T = K @ Q^T (per wave: 16 kv x 64 q) ----------------------------
@@ -650,7 +657,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     // ---- dK += dS^T @ Q  (A = dS^T via x_lds, B = Q^T re-staged) ----------
     {
       __syncthreads();  // x_lds + t_lds reuse
-      transpose_lds_tile<D>(t_lds, q_lds, threadIdx.x);
+      transpose_lds_tile<D, kBwdThreads>(t_lds, q_lds, threadIdx.x);
       bf16_t* xw = x_lds;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -684,46 +691,48 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kernel(
     // ---- dQ partial = dS @ K; atomicAdd (A = dS q-major via x_lds) --------
     {
       __syncthreads();
-      bf16_t* xw = x_lds;  // now [q 64][kv 64+8]
+      bf16_t* xw = x_lds;  // now [q 64][kv kBwdKv+8]
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
           const int q_l = nt * 16 + (lane & 15);
-          xw[q_l * (kKvBlk + 8) + kv_l] = (bf16_t)dst_val[nt][r];
+          xw[q_l * (kBwdKv + 8) + kv_l] = (bf16_t)dst_val[nt][r];
         }
       }
       __syncthreads();
-      // each wave: 16 q rows (wave*16 ..), K = 64 kv, N = D
-      f32x4 dq_acc[kNT];
+      // waves 0-3: one 16-row q m-tile each, K = kBwdKv kv, N = D
+      if (wave < 4) {
+        f32x4 dq_acc[kNT];
 #pragma unroll
-      for (int nt = 0; nt < kNT; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
-      const bf16_t* xr = x_lds + (wave * 16) * (kKvBlk + 8);
-      __builtin_amdgcn_s_setprio(1);
+        for (int nt = 0; nt < kNT; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+        const bf16_t* xr = x_lds + (wave * 16) * (kBwdKv + 8);
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int ks2 = 0; ks2 < 2; ++ks2) {
-        const int kv_off = ks2 * 32 + (lane >> 4) * 8;
-        const bf16x8 da = *reinterpret_cast<const bf16x8*>(
-            xr + (lane & 15) * (kKvBlk + 8) + kv_off);
-#pragma unroll
-        for (int nt = 0; nt < kNT; ++nt) {
-          const int d = nt * 16 + (lane & 15);
-          const int byte = (kv_off * 2) ^ ((d & 7) << 4);
-          const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(kt_lds) + d * (kKvBlk * 2) + byte);
-          dq_acc[nt] = mfma16(da, kb, dq_acc[nt]);
-        }
-      }
-      __builtin_amdgcn_s_setprio(0);
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int q_glob = qt + wave * 16 + (lane >> 4) * 4 + r;
-        if (q_glob < Sq) {
+        for (int ks2 = 0; ks2 < kBwdKv / 32; ++ks2) {
+          const int kv_off = ks2 * 32 + (lane >> 4) * 8;
+          const bf16x8 da = *reinterpret_cast<const bf16x8*>(
+              xr + (lane & 15) * (kBwdKv + 8) + kv_off);
 #pragma unroll
           for (int nt = 0; nt < kNT; ++nt) {
-            atomicAdd(dq + q_base + (int64_t)q_glob * q_row_stride + nt * 16 + (lane & 15),
-                      dq_acc[nt][r]);
+            const int d = nt * 16 + (lane & 15);
+            const int byte = (kv_off * 2) ^ ((d & 7) << 4);
+            const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<char*>(kt_lds) + d * (kBwdKv * 2) + byte);
+            dq_acc[nt] = mfma16(da, kb, dq_acc[nt]);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int q_glob = qt + wave * 16 + (lane >> 4) * 4 + r;
+          if (q_glob < Sq) {
+#pragma unroll
+            for (int nt = 0; nt < kNT; ++nt) {
+              atomicAdd(dq + q_base + (int64_t)q_glob * q_row_stride + nt * 16 + (lane & 15),
+                        dq_acc[nt][r]);
+            }
           }
         }
       }
@@ -871,13 +880,14 @@ std::vector<torch::Tensor> flash_attn_bwd(
                        delta.data_ptr<float>(), B, Sq, Hq, D_pad);
   }
 
-  const dim3 grid((Skv + d9d::kKvBlk - 1) / d9d::kKvBlk, B * Hq);
+  const dim3 grid((Skv + d9d::kBwdKv - 1) / d9d::kBwdKv, B * Hq);
   const size_t smem =
-      (size_t)(3 * d9d::kQBlk * D_pad + D_pad * d9d::kKvBlk + 64 * (64 + 8)) *
+      (size_t)(3 * d9d::kQBlk * D_pad + D_pad * d9d::kBwdKv +
+               d9d::kBwdKv * (64 + 8)) *
       sizeof(__bf16);
 
 #define LAUNCH_BWD(DP)                                                        \
-  hipLaunchKernelGGL((d9d::flash_bwd_kernel<DP>), grid, dim3(256), smem,      \
+  hipLaunchKernelGGL((d9d::flash_bwd_kernel<DP>), grid, dim3(512), smem,      \
                      stream,                                                  \
                      reinterpret_cast<const __bf16*>(qp.data_ptr()),          \
                      reinterpret_cast<const __bf16*>(kp.data_ptr()),          \
