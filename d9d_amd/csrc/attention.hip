@@ -3,14 +3,16 @@
 // Replaces the reference's external flash-attn wheels
 // (d9d/kernel/flash_attn/function.py). MI355X-first design:
 //   * MFMA v_mfma_f32_16x16x32_bf16 tiles, wave64.
-//   * Forward: workgroup = 4 waves x 16 q-rows (QBLK 64), KV tiles of 64
-//     staged in LDS — K row-major with a ((row&15)<<4) XOR swizzle so the
-//     B-fragment ds_read_b128s are bank-conflict-free; V staged transposed
-//     for contiguous PV B-fragments. Online softmax in fp32 with running
-//     (m, l) per row; LSE written for CP ring merging.
-//   * Backward: FA2-style; workgroup owns a KV tile, loops q tiles;
-//     dK/dV accumulate in AGPRs; dQ via fp32 global atomics (fast on CDNA4);
-//     delta = rowsum(dO*O) precomputed by a small kernel.
+//   * Forward: workgroup = 4 waves x 32 q-rows (two 16-row m-tiles per wave,
+//     q block 128), KV tiles of 64 staged with the T14 issue-early/write-late
+//     split — K row-major with an XOR swizzle (conflict-free ds_read_b128
+//     B-fragments), V transposed with vectorized 8-byte LDS writes. Online
+//     softmax in fp32; s_setprio(1) around MFMA clusters; LSE written for
+//     context-parallel merging.
+//   * Backward: FA2 decomposition; workgroup = 8 waves owning a 128-row KV
+//     tile, looping q tiles (Q/dO prefetched into registers under the MFMAs,
+//     transposed images derived LDS-to-LDS); dK/dV accumulate in registers,
+//     dQ via fp32 global atomics; delta = rowsum(dO*O) precomputed.
 //
 // Layouts: q (B,Sq,Hq,D), k/v (B,Skv,Hkv,D), out (B,Sq,Hq,D), lse (B,Hq,Sq).
 // D (head_dim) templated in {32, 64, 96, 128}; host pads other sizes.
