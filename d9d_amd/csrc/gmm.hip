@@ -229,11 +229,13 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
 }
 
 // db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N).
-// Grid (ceil(N/128), ceil(K/128), E): 128x128 tiles (4 waves as 2x2, 64x64
-// each) looping the expert's rows in chunks of 64; both operands staged
-// transposed ([dim][row], 128-byte rows, XOR swizzle) for k-contiguous
-// fragments.
-__global__ __launch_bounds__(256, 2) void gmm_db_kernel(
+// Grid (ceil(N/256), ceil(K/256), E): 256x256 tiles per workgroup -- 8 waves
+// as 2 (k-halves of 128) x 4 (n-quarters of 64) -- looping the expert's rows
+// in chunks of 64. Both operands are staged transposed ([dim][row], 128-byte
+// LDS rows, XOR swizzle) for k-contiguous fragments. The big tile is what
+// matters here: this kernel is HBM-bound on re-reads (a is read once per
+// n-tile, g once per k-tile), and 256x256 halves that traffic vs 128x128.
+__global__ __launch_bounds__(512, 1) void gmm_db_kernel(
     const bf16_t* __restrict__ a,   // (T, K)
     const bf16_t* __restrict__ g,   // (T, N)
     bf16_t* __restrict__ db,        // (E, K, N)
@@ -241,32 +243,32 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
     const int* __restrict__ expert_order,  // experts sorted by row count desc
     int E, int K, int N) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* at_lds = reinterpret_cast<bf16_t*>(smem);   // [128 k][64 rows]
-  bf16_t* gt_lds = at_lds + 128 * 64;                 // [128 n][64 rows]
+  bf16_t* at_lds = reinterpret_cast<bf16_t*>(smem);   // [256 k][64 rows]
+  bf16_t* gt_lds = at_lds + 256 * 64;                 // [256 n][64 rows]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = wave >> 1;  // k-half
-  const int wn = wave & 1;   // n-half
+  const int wm = wave >> 2;  // k-half (128 rows of db)
+  const int wn = wave & 3;   // n-quarter (64 cols)
 
   // Heavy experts dispatch first so the tail wave holds the small tiles.
   const int e = expert_order[blockIdx.z];
-  const int k0 = blockIdx.y * 128;
-  const int n0 = blockIdx.x * 128;
+  const int k0 = blockIdx.y * 256;
+  const int n0 = blockIdx.x * 256;
   const int r_start = row_off[e];
   const int r_end = row_off[e + 1];
   if (r_start >= r_end) return;  // empty expert: db stays zero (host zeros it)
 
-  f32x4 acc[4][4];
+  f32x4 acc[8][4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 8; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // T14 staging: 4 (row x 2-col) blocks per thread per tensor, loaded into
-  // registers early (overlapping the previous chunk's MFMAs), transposed into
-  // LDS with 8-byte writes late. Rows past r_end are zero-filled (they enter
-  // the row-sum).
+  // T14 staging: each thread owns 4 (4-row x 2-col) blocks per tensor, loaded
+  // into registers early (overlapping the previous chunk's MFMAs) with
+  // branchless clamped addresses, transposed into LDS with 8-byte writes
+  // late. Rows past r_end are zero-filled (they enter the row-sum).
   union u64u {
     uint64_t u;
     ushort s[4];
@@ -276,9 +278,9 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
   auto load_regs = [&](int rt) {
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
-      const int idx = threadIdx.x + it * 256;
-      const int d0 = (idx % 64) * 2;
-      const int rb = (idx / 64) * 4;
+      const int idx = threadIdx.x + it * 512;
+      const int d0 = (idx & 127) * 2;
+      const int rb = (idx >> 7) * 4;
       const bool k_ok = k0 + d0 + 1 < K;
       const bool n_ok = n0 + d0 + 1 < N;
       const int64_t sk = min((int64_t)(k0 + d0), (int64_t)max(K - 2, 0));
@@ -303,9 +305,9 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
   auto store_lds = [&]() {
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
-      const int idx = threadIdx.x + it * 256;
-      const int d0 = (idx % 64) * 2;
-      const int rb = (idx / 64) * 4;
+      const int idx = threadIdx.x + it * 512;
+      const int d0 = (idx & 127) * 2;
+      const int rb = (idx >> 7) * 4;
       const int byte0 = (rb * 2) ^ ((d0 & 7) << 4);
       const int byte1 = (rb * 2) ^ (((d0 + 1) & 7) << 4);
       *reinterpret_cast<uint64_t*>(
@@ -326,13 +328,14 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
   for (int rt = r_start; rt < r_end; rt += 64) {
     if (rt + 64 < r_end) load_regs(rt + 64);  // issue early
 
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {  // 64 rows -> 2 k-steps
       const int rr = ks * 32 + (lane >> 4) * 8;
-      bf16x8 a_frag[4], g_frag[4];
+      bf16x8 a_frag[8], g_frag[4];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int krow = wm * 64 + i * 16 + (lane & 15);
+      for (int i = 0; i < 8; ++i) {
+        const int krow = wm * 128 + i * 16 + (lane & 15);
         const int byte = (rr * 2) ^ ((krow & 7) << 4);
         a_frag[i] = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const char*>(at_lds) + krow * (64 * 2) + byte);
@@ -345,10 +348,11 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
             reinterpret_cast<const char*>(gt_lds) + ncol * (64 * 2) + byte);
       }
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < 8; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[i][j] = mfma16g(a_frag[i], g_frag[j], acc[i][j]);
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
     if (rt + 64 < r_end) {
       store_lds();
@@ -357,12 +361,12 @@ __global__ __launch_bounds__(256, 2) void gmm_db_kernel(
   }
 
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < 8; ++i) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int kk = k0 + wm * 64 + i * 16 + (lane >> 4) * 4 + r;
+        const int kk = k0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r;
         const int nn = n0 + wn * 64 + j * 16 + (lane & 15);
         if (kk < K && nn < N) {
           db[((int64_t)e * K + kk) * N + nn] = (bf16_t)acc[i][j][r];
@@ -439,10 +443,10 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
                                   /*descending=*/true).to(torch::kInt32);
   auto expert_order = order_cpu.to(a.device(), /*non_blocking=*/true);
 
-  const dim3 grid((N + 127) / 128, (K + 127) / 128, E);
-  const size_t smem = (2 * 128 * 64) * sizeof(__bf16);
+  const dim3 grid((N + 255) / 256, (K + 255) / 256, E);
+  const size_t smem = (2 * 256 * 64) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(d9d::gmm_db_kernel, grid, dim3(256), smem, stream,
+  hipLaunchKernelGGL(d9d::gmm_db_kernel, grid, dim3(512), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
                      reinterpret_cast<__bf16*>(db.data_ptr()),
