@@ -7,9 +7,9 @@
 // binary-searches its m-tile, so there is no per-expert launch overhead
 // (the eager fallback was E x 3 rocBLAS launches per MoE layer).
 //
-// Geometry: 128x64 tiles, 4 waves (2x2), BK=64, double-buffered LDS with the
-// ((row&15)<<4 style) XOR swizzle for conflict-free ds_read_b128,
-// v_mfma_f32_16x16x32_bf16.
+// Geometry: 256x256 tiles, 8 waves (2 m-halves x 4 n-quarters), BK=64,
+// T14 register-staged LDS with the ((row&7)<<4) XOR swizzle for
+// conflict-free ds_read_b128, v_mfma_f32_16x16x32_bf16.
 #include "common.h"
 
 #include <torch/extension.h>
@@ -25,24 +25,21 @@ D9D_DEVICE f32x4 mfma16g(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-constexpr int kBM = 128;
-constexpr int kBN = 64;
+constexpr int kBM = 256;
+constexpr int kBN = 192;
 constexpr int kBK = 64;
 
 // A LDS tile: [kBM][kBK] row-major, rows kBK*2=128 bytes -> swizzle mask 7.
-// B LDS tile: [kBN][kBK] row-major (B^T layout: we need B[k][n] fragments with
-//   k contiguous... see below), also 128-byte rows.
+// B LDS tile staged TRANSPOSED as bt[n][k] (kBN x kBK): the MFMA B-fragment
+// needs 8 contiguous k at fixed n; b (E,K,N) is n-contiguous, so staging
+// reads 8 consecutive n at fixed k and scatters 8 scalar LDS writes.
 //
-// Fragment needs:
-//   A[m][k]: lane m=l&15, k=(l>>4)*8+j  -> A tile row-major [m][k], contiguous in k. OK.
-//   B[k][n]: lane n=l&15, k=(l>>4)*8+j  -> need k contiguous at fixed n ->
-//   stage B TRANSPOSED: bt[n][k] (64 x 64).
-//
-// b (E,K,N) is row-major in N, so reading b[k][n0..n0+7] is contiguous: we
-// read 8 consecutive n at fixed k and scatter into bt[n][k] (8 scalar LDS
-// writes) — done once per K-tile per block.
+// 256x192 tiles with 8 waves (2 m-halves x 4 n-quarters of 48): this kernel is
+// HBM-bound on A re-reads (once per n-tile), and the big tile quarters that
+// traffic vs the old 128x64 geometry. Register (T14) staging doubles as the
+// second buffer: loads for k-tile t+1 issue early under tile t's MFMAs.
 
-__global__ __launch_bounds__(256, 2) void gmm_kernel(
+__global__ __launch_bounds__(512, 1) void gmm_kernel(
     const bf16_t* __restrict__ a,    // (T, K)
     const bf16_t* __restrict__ b,    // (E, K, N)
     bf16_t* __restrict__ out,        // (T, N)
@@ -50,19 +47,18 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
     const int* __restrict__ mtile_pref,   // (E+1,) prefix of ceil(rows_e/kBM)
     int E, int K, int N) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* a_lds = reinterpret_cast<bf16_t*>(smem);        // 2 x [kBM][kBK]
-  bf16_t* bt_lds = a_lds + 2 * kBM * kBK;                 // 2 x [kBN][kBK]
+  bf16_t* a_lds = reinterpret_cast<bf16_t*>(smem);        // [kBM][kBK]
+  bf16_t* bt_lds = a_lds + kBM * kBK;                     // [kBN][kBK]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = wave >> 1;   // 0..1: wave row (64 rows each)
-  const int wn = wave & 1;    // 0..1: wave col (32 cols each)
+  const int wm = wave >> 2;   // 0..1: wave row (128 rows each)
+  const int wn = wave & 3;    // 0..3: wave col (64 cols each)
 
   // ---- map blockIdx.y -> (expert, m_tile) via binary search ----------------
   // grid.x is the (small) n-tile index: consecutive blocks share the same A
   // rows, so A re-reads hit L2 instead of HBM.
   const int mt_global = blockIdx.y;
-  // largest e with mtile_pref[e] <= mt_global (expert may own several m-tiles)
   int lo = 0, hi = E - 1;
   while (lo < hi) {
     const int mid = (lo + hi + 1) >> 1;
@@ -76,18 +72,23 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
 
   const bf16_t* b_e = b + (int64_t)e * K * N;
 
-  f32x4 acc[4][2];
+  f32x4 acc[8][3];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 8; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < 3; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int n_ktiles = (K + kBK - 1) / kBK;
 
-  // T14 split staging: per k-tile, 4x bf16x8 of A and 2x bf16x8 of B per
+  // T14 split staging: per k-tile, 4x bf16x8 of A and 4x bf16x8 of B per
   // thread are loaded into registers EARLY (issue overlaps the previous
-  // tile's MFMAs) and written to the alternate LDS buffer late.
-  bf16x8 a_reg[4], b_reg[2];
+  // tile's MFMAs) and written to LDS late.
+  union bvec {
+    bf16x8 v;
+    ushort s[8];
+  };
+  bf16x8 a_reg[4];
+  bvec b_reg[2][2];
 
   // Branchless staging: clamp addresses into range, select-zero after the
   // load (a per-element branch around a load makes hipcc serialize every
@@ -96,7 +97,7 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
     const int k0 = kt * kBK;
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
-      const int idx = (threadIdx.x + it * 256) * 8;
+      const int idx = (threadIdx.x + it * 512) * 8;
       const int row = idx / kBK;
       const int col = idx % kBK;
       const int g_row = row0 + row;
@@ -118,108 +119,113 @@ __global__ __launch_bounds__(256, 2) void gmm_kernel(
         a_reg[it] = ev;
       }
     }
+    // B: (kBK/2) x (kBN/8) = 768 k-pair groups; each group loads two
+    // bf16x8 (adjacent k, same 8 n) so the LDS writes pair into u32s.
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
-      const int idx = (threadIdx.x + it * 256) * 8;
-      const int kk = idx / kBN;
-      const int n = idx % kBN;
-      const bool ok = k0 + kk < K && n0 + n + 7 < N;
-      const int64_t sk = min(k0 + kk, K - 1);
-      const int64_t sn = min(n0 + n, max(N - 8, 0));
-      const bf16x8 val = *reinterpret_cast<const bf16x8*>(b_e + sk * N + sn);
-      if (ok) {
-        b_reg[it] = val;
-      } else {
-        // slow edge path (rare): per-element fill
-        bf16x8 ev = {};
-        if (k0 + kk < K) {
+      const int grp = threadIdx.x + it * 512;
+      if (grp >= (kBK / 2) * (kBN / 8)) break;
+      const int kk = (grp / (kBN / 8)) * 2;
+      const int n = (grp % (kBN / 8)) * 8;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            if (n0 + n + j < N) ev[j] = b_e[(int64_t)(k0 + kk) * N + n0 + n + j];
+      for (int h = 0; h < 2; ++h) {
+        const bool ok = k0 + kk + h < K && n0 + n + 7 < N;
+        const int64_t sk = min(k0 + kk + h, K - 1);
+        const int64_t sn = min(n0 + n, max(N - 8, 0));
+        const bf16x8 val = *reinterpret_cast<const bf16x8*>(b_e + sk * N + sn);
+        if (ok) {
+          b_reg[it][h].v = val;
+        } else {
+          // slow edge path (rare): per-element fill
+          bf16x8 ev = {};
+          if (k0 + kk + h < K) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              if (n0 + n + j < N) ev[j] = b_e[(int64_t)(k0 + kk + h) * N + n0 + n + j];
+            }
           }
+          b_reg[it][h].v = ev;
         }
-        b_reg[it] = ev;
       }
     }
   };
 
-  auto store_lds = [&](int buf) {
-    bf16_t* al = a_lds + buf * kBM * kBK;
+  auto store_lds = [&]() {
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
-      const int idx = (threadIdx.x + it * 256) * 8;
+      const int idx = (threadIdx.x + it * 512) * 8;
       const int row = idx / kBK;
       const int col = idx % kBK;
       const int byte = (col * 2) ^ ((row & 7) << 4);
       *reinterpret_cast<bf16x8*>(
-          reinterpret_cast<char*>(al) + row * (kBK * 2) + byte) = a_reg[it];
+          reinterpret_cast<char*>(a_lds) + row * (kBK * 2) + byte) = a_reg[it];
     }
-    bf16_t* bl = bt_lds + buf * kBN * kBK;
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
-      const int idx = (threadIdx.x + it * 256) * 8;
-      const int kk = idx / kBN;
-      const int n = idx % kBN;
+      const int grp = threadIdx.x + it * 512;
+      if (grp >= (kBK / 2) * (kBN / 8)) break;
+      const int kk = (grp / (kBN / 8)) * 2;
+      const int n = (grp % (kBN / 8)) * 8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int nn = n + j;
+        const uint32_t u = (uint32_t)b_reg[it][0].s[j] |
+                           ((uint32_t)b_reg[it][1].s[j] << 16);
         const int byte = (kk * 2) ^ ((nn & 7) << 4);
-        *reinterpret_cast<bf16_t*>(
-            reinterpret_cast<char*>(bl) + nn * (kBK * 2) + byte) = b_reg[it][j];
+        *reinterpret_cast<uint32_t*>(
+            reinterpret_cast<char*>(bt_lds) + nn * (kBK * 2) + byte) = u;
       }
     }
   };
 
   load_regs(0);
-  store_lds(0);
+  store_lds();
   __syncthreads();
 
   for (int kt = 0; kt < n_ktiles; ++kt) {
-    const int buf = kt & 1;
     if (kt + 1 < n_ktiles) load_regs(kt + 1);  // issue early, use late
 
-    const bf16_t* al = a_lds + buf * kBM * kBK;
-    const bf16_t* bl = bt_lds + buf * kBN * kBK;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {  // kBK=64 -> 2 MFMA k-steps
       const int kk = ks * 32 + (lane >> 4) * 8;
-      bf16x8 a_frag[4], b_frag[2];
+      bf16x8 a_frag[8], b_frag[3];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int row = wm * 64 + i * 16 + (lane & 15);
+      for (int i = 0; i < 8; ++i) {
+        const int row = wm * 128 + i * 16 + (lane & 15);
         const int byte = (kk * 2) ^ ((row & 7) << 4);
         a_frag[i] = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(al) + row * (kBK * 2) + byte);
+            reinterpret_cast<const char*>(a_lds) + row * (kBK * 2) + byte);
       }
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
-        const int col = wn * 32 + j * 16 + (lane & 15);
+      for (int j = 0; j < 3; ++j) {
+        const int col = wn * 48 + j * 16 + (lane & 15);
         const int byte = (kk * 2) ^ ((col & 7) << 4);
         b_frag[j] = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(bl) + col * (kBK * 2) + byte);
+            reinterpret_cast<const char*>(bt_lds) + col * (kBK * 2) + byte);
       }
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < 8; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j) acc[i][j] = mfma16g(a_frag[i], b_frag[j], acc[i][j]);
+        for (int j = 0; j < 3; ++j) acc[i][j] = mfma16g(a_frag[i], b_frag[j], acc[i][j]);
     }
     __builtin_amdgcn_s_setprio(0);
-    if (kt + 1 < n_ktiles) {
-      store_lds(buf ^ 1);  // writes into the idle buffer; vmcnt waits land here
-    }
     __syncthreads();
+    if (kt + 1 < n_ktiles) {
+      store_lds();
+      __syncthreads();
+    }
   }
 
   // ---- store ---------------------------------------------------------------
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < 8; ++i) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < 3; ++j) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = row0 + wm * 64 + i * 16 + (lane >> 4) * 4 + r;
-        const int col = n0 + wn * 32 + j * 16 + (lane & 15);
+        const int row = row0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r;
+        const int col = n0 + wn * 48 + j * 16 + (lane & 15);
         if (row < row_end && col < N) {
           out[(int64_t)row * N + col] = (bf16_t)acc[i][j][r];
         }
@@ -418,9 +424,9 @@ torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
   if (total_mtiles == 0) return out;
 
   const dim3 grid((N + d9d::kBN - 1) / d9d::kBN, total_mtiles);
-  const size_t smem = (2 * d9d::kBM * d9d::kBK + 2 * d9d::kBN * d9d::kBK) * sizeof(__bf16);
+  const size_t smem = (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(d9d::gmm_kernel, grid, dim3(256), smem, stream,
+  hipLaunchKernelGGL(d9d::gmm_kernel, grid, dim3(512), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(b.data_ptr()),
                      reinterpret_cast<__bf16*>(out.data_ptr()),
