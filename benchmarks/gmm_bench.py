@@ -23,3 +23,6 @@ def bench(fn, flops, name):
 fl = 2.0 * T * K * N
 bench(lambda: ext.gmm(a, b, sizes), fl, "gmm fwd")
 bench(lambda: ext.gmm_db(a, g, sizes, E), fl, "gmm_db")
+
+w_nt = torch.randn(E, N, K, dtype=torch.bfloat16, device="cuda") * 0.05
+bench(lambda: ext.gmm_nt(a, w_nt, sizes), fl, "gmm_nt")

@@ -27,6 +27,7 @@ torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b);
 
 // gmm.hip
 torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes);
+torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes);
 torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes, int64_t num_experts);
 
 // rope.hip
@@ -44,6 +45,7 @@ torch::Tensor moe_row_dot(torch::Tensor grad_out, torch::Tensor expert_out, torc
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
+  m.def("gmm_nt", &gmm_nt, "CDNA4 grouped GEMM, weight (E,N,K)");
   m.def("gmm_db", &gmm_db, "CDNA4 grouped GEMM weight-grad");
   m.def("rope_qk", &rope_qk, "fused q/k rotary embedding");
   m.def("cce_fwd", &cce_fwd, "fused linear cross-entropy forward (lse + target logit)");

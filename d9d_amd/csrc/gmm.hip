@@ -234,6 +234,176 @@ __global__ __launch_bounds__(512, 1) void gmm_kernel(
   }
 }
 
+// out[rows_e] = a[rows_e] @ w[e]^T for w stored (E, N, K) (nn.Linear layout).
+// Both operands are k-contiguous, so BOTH stage as plain vector copies into
+// row-major swizzled LDS tiles -- no scatter. Same 256x192 / 8-wave geometry
+// as gmm_kernel; this is the fast path for expert FORWARD. The dgrad
+// (dy @ w with w (N,K) read as (red,out)) reuses gmm_kernel unchanged.
+__global__ __launch_bounds__(512, 1) void gmm_nt_kernel(
+    const bf16_t* __restrict__ a,    // (T, K)
+    const bf16_t* __restrict__ w,    // (E, N, K)
+    bf16_t* __restrict__ out,        // (T, N)
+    const int* __restrict__ row_off,
+    const int* __restrict__ mtile_pref,
+    int E, int K, int N) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* a_lds = reinterpret_cast<bf16_t*>(smem);        // [kBM][kBK]
+  bf16_t* bt_lds = a_lds + kBM * kBK;                     // [kBN][kBK]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  const int mt_global = blockIdx.y;
+  int lo = 0, hi = E - 1;
+  while (lo < hi) {
+    const int mid = (lo + hi + 1) >> 1;
+    if (mtile_pref[mid] <= mt_global) lo = mid; else hi = mid - 1;
+  }
+  const int e = lo;
+  const int m_tile = mt_global - mtile_pref[e];
+  const int row0 = row_off[e] + m_tile * kBM;
+  const int row_end = row_off[e + 1];
+  const int n0 = blockIdx.x * kBN;
+
+  const bf16_t* w_e = w + (int64_t)e * N * K;
+
+  f32x4 acc[8][3];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int n_ktiles = (K + kBK - 1) / kBK;
+
+  bf16x8 a_reg[4], b_reg[3];
+
+  auto load_regs = [&](int kt) {
+    const int k0 = kt * kBK;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = (threadIdx.x + it * 512) * 8;
+      const int row = idx / kBK;
+      const int col = idx % kBK;
+      const int g_row = row0 + row;
+      const bool ok = g_row < row_end && k0 + col + 7 < K;
+      const int64_t sr = min(g_row, row_end - 1);
+      const int64_t sc = min(k0 + col, max(K - 8, 0));
+      const bf16x8 val = *reinterpret_cast<const bf16x8*>(a + sr * K + sc);
+      if (ok) {
+        a_reg[it] = val;
+      } else {
+        bf16x8 ev = {};
+        if (g_row < row_end) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            if (k0 + col + j < K) ev[j] = a[(int64_t)g_row * K + k0 + col + j];
+          }
+        }
+        a_reg[it] = ev;
+      }
+    }
+#pragma unroll
+    for (int it = 0; it < 3; ++it) {
+      const int idx = (threadIdx.x + it * 512) * 8;
+      const int n = idx / kBK;
+      const int col = idx % kBK;
+      const bool ok = n0 + n < N && k0 + col + 7 < K;
+      const int64_t sn = min(n0 + n, N - 1);
+      const int64_t sc = min(k0 + col, max(K - 8, 0));
+      const bf16x8 val = *reinterpret_cast<const bf16x8*>(w_e + sn * K + sc);
+      if (ok) {
+        b_reg[it] = val;
+      } else {
+        bf16x8 ev = {};
+        if (n0 + n < N) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            if (k0 + col + j < K) ev[j] = w_e[(int64_t)(n0 + n) * K + k0 + col + j];
+          }
+        }
+        b_reg[it] = ev;
+      }
+    }
+  };
+
+  auto store_lds = [&]() {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = (threadIdx.x + it * 512) * 8;
+      const int row = idx / kBK;
+      const int col = idx % kBK;
+      const int byte = (col * 2) ^ ((row & 7) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(a_lds) + row * (kBK * 2) + byte) = a_reg[it];
+    }
+#pragma unroll
+    for (int it = 0; it < 3; ++it) {
+      const int idx = (threadIdx.x + it * 512) * 8;
+      const int n = idx / kBK;
+      const int col = idx % kBK;
+      const int byte = (col * 2) ^ ((n & 7) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(bt_lds) + n * (kBK * 2) + byte) = b_reg[it];
+    }
+  };
+
+  load_regs(0);
+  store_lds();
+  __syncthreads();
+
+  for (int kt = 0; kt < n_ktiles; ++kt) {
+    if (kt + 1 < n_ktiles) load_regs(kt + 1);
+
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kk = ks * 32 + (lane >> 4) * 8;
+      bf16x8 a_frag[8], b_frag[3];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int row = wm * 128 + i * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((row & 7) << 4);
+        a_frag[i] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(a_lds) + row * (kBK * 2) + byte);
+      }
+#pragma unroll
+      for (int j = 0; j < 3; ++j) {
+        const int col = wn * 48 + j * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((col & 7) << 4);
+        b_frag[j] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(bt_lds) + col * (kBK * 2) + byte);
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 3; ++j) acc[i][j] = mfma16g(a_frag[i], b_frag[j], acc[i][j]);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    if (kt + 1 < n_ktiles) {
+      store_lds();
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r;
+        const int col = n0 + wn * 48 + j * 16 + (lane & 15);
+        if (row < row_end && col < N) {
+          out[(int64_t)row * N + col] = (bf16_t)acc[i][j][r];
+        }
+      }
+    }
+  }
+}
+
 // db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N).
 // Grid (ceil(N/256), ceil(K/256), E): 256x256 tiles per workgroup -- 8 waves
 // as 2 (k-halves of 128) x 4 (n-quarters of 64) -- looping the expert's rows
@@ -429,6 +599,33 @@ torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
   hipLaunchKernelGGL(d9d::gmm_kernel, grid, dim3(512), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(b.data_ptr()),
+                     reinterpret_cast<__bf16*>(out.data_ptr()),
+                     row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
+                     E, K, N);
+  return out;
+}
+
+
+torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 && w.is_contiguous());
+  TORCH_CHECK(batch_sizes.device().is_cpu());
+  const int T = a.size(0), K = a.size(1);
+  const int E = w.size(0), N = w.size(1);
+  TORCH_CHECK(w.size(2) == K, "gmm_nt K mismatch");
+
+  auto out = torch::empty({(int64_t)T, (int64_t)N}, a.options());
+  if (T == 0) return out;
+  auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
+  const int total_mtiles = mtile_pref[E].item<int>();
+  if (total_mtiles == 0) return out;
+
+  const dim3 grid((N + d9d::kBN - 1) / d9d::kBN, total_mtiles);
+  const size_t smem = (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::gmm_nt_kernel, grid, dim3(512), smem, stream,
+                     reinterpret_cast<const __bf16*>(a.data_ptr()),
+                     reinterpret_cast<const __bf16*>(w.data_ptr()),
                      reinterpret_cast<__bf16*>(out.data_ptr()),
                      row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
                      E, K, N);

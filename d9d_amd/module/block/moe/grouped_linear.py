@@ -1,6 +1,8 @@
 """Grouped linear over experts (reference: d9d/module/block/moe/grouped_linear.py:12).
 
-3-D weight (E, in, out); forward is the grouped GEMM. DTensor weights are
+3-D weight (E, out, in) -- nn.Linear layout, matching the HF fused expert
+format and making the forward an NT grouped GEMM (both operands
+k-contiguous on CDNA4, scatter-free staging). DTensor weights are
 unwrapped to the local shard (expert parallelism shards dim 0)."""
 
 import math
@@ -9,7 +11,7 @@ import torch
 from torch import nn
 from torch.distributed.tensor import DTensor
 
-from ....ops import gmm
+from ....ops import gmm_nt
 
 
 class GroupedLinear(nn.Module):
@@ -26,7 +28,7 @@ class GroupedLinear(nn.Module):
         self.in_features = in_features
         self.out_features = out_features
         self.weight = nn.Parameter(
-            torch.empty(num_experts, in_features, out_features, device=device, dtype=dtype)
+            torch.empty(num_experts, out_features, in_features, device=device, dtype=dtype)
         )
 
     def reset_parameters(self) -> None:
@@ -41,4 +43,4 @@ class GroupedLinear(nn.Module):
 
     def forward(self, x: torch.Tensor, batch_sizes: torch.Tensor) -> torch.Tensor:
         """x (sum_T, in); batch_sizes (E_local,) int64 CPU."""
-        return gmm(x, self._local_weight(), batch_sizes)
+        return gmm_nt(x, self._local_weight(), batch_sizes)

@@ -3,7 +3,7 @@
 Two HF expert formats are supported (reference: huggingface.py MODULE_LIST and
 FUSED formats):
   * "module_list": model.layers.N.mlp.experts.M.{gate,up,down}_proj.weight,
-    each (out, in) -> transpose + stack into our (E, in, out).
+    each (out, in) -> stack into our (E, out, in) (same per-expert layout).
   * "fused" (transformers >= 5.x packed experts):
     experts.gate_up_proj (E, 2I, H) and experts.down_proj (E, H, I)
     -> chunk gate/up on dim 1, transpose (1, 2).
@@ -30,7 +30,7 @@ from .params import Qwen3MoEModelParameters
 
 
 class _FusedToStacked(ModelStateMapper):
-    """HF packed experts -> our stacked (E, in, out) weights for one layer."""
+    """HF packed experts -> our stacked (E, out, in) weights for one layer."""
 
     def __init__(self, prefix: str, intermediate: int) -> None:
         self.prefix = prefix
@@ -53,14 +53,15 @@ class _FusedToStacked(ModelStateMapper):
         pre = self.prefix
         if pre + "mlp.experts.gate_up_proj" in tensors:
             gu = tensors[pre + "mlp.experts.gate_up_proj"]  # (E, 2I, H)
-            gate = gu[:, : self.intermediate, :].transpose(1, 2).contiguous()
-            up = gu[:, self.intermediate :, :].transpose(1, 2).contiguous()
+            # our GroupedLinear stores (E, out, in) = (E, I, H): direct split
+            gate = gu[:, : self.intermediate, :].contiguous()
+            up = gu[:, self.intermediate :, :].contiguous()
             return {
                 pre + "mlp.experts.gate_proj.weight": gate,
                 pre + "mlp.experts.up_proj.weight": up,
             }
-        down = tensors[pre + "mlp.experts.down_proj"]  # (E, H, I)
-        return {pre + "mlp.experts.down_proj.weight": down.transpose(1, 2).contiguous()}
+        down = tensors[pre + "mlp.experts.down_proj"]  # (E, H, I) = (E, out, in)
+        return {pre + "mlp.experts.down_proj.weight": down.contiguous()}
 
 
 class _StackedToFused(ModelStateMapper):
@@ -86,11 +87,11 @@ class _StackedToFused(ModelStateMapper):
 
         pre = self.prefix
         if pre + "mlp.experts.gate_proj.weight" in tensors:
-            gate = tensors[pre + "mlp.experts.gate_proj.weight"].transpose(1, 2)
-            up = tensors[pre + "mlp.experts.up_proj.weight"].transpose(1, 2)
+            gate = tensors[pre + "mlp.experts.gate_proj.weight"]
+            up = tensors[pre + "mlp.experts.up_proj.weight"]
             return {pre + "mlp.experts.gate_up_proj": torch.cat([gate, up], dim=1).contiguous()}
         down = tensors[pre + "mlp.experts.down_proj.weight"]
-        return {pre + "mlp.experts.down_proj": down.transpose(1, 2).contiguous()}
+        return {pre + "mlp.experts.down_proj": down.contiguous()}
 
 
 def _vocab_splits(p: Qwen3MoEModelParameters) -> list[tuple[str, int]]:
@@ -137,19 +138,11 @@ def hf_to_d9d_mapper(
             mappers.append(_FusedToStacked(pre, p.intermediate_size))
         else:
             for proj in ("gate_proj", "up_proj", "down_proj"):
-                # HF expert weight (out, in) -> transpose (in, out) -> stack
-                per_expert = [
-                    Transpose(
-                        f"{pre}mlp.experts.{e}.{proj}.weight",
-                        f"{pre}mlp.experts._t{e}.{proj}",
-                    )
-                    for e in range(p.num_experts)
-                ]
+                # HF expert weight (out, in) == our per-expert layout: stack
                 mappers.append(
                     Sequential(
-                        Parallel(*per_expert),
                         StackTensors(
-                            [f"{pre}mlp.experts._t{e}.{proj}" for e in range(p.num_experts)],
+                            [f"{pre}mlp.experts.{e}.{proj}.weight" for e in range(p.num_experts)],
                             f"{pre}mlp.experts.{proj}.weight",
                             dim=0,
                         ),
@@ -198,17 +191,11 @@ def d9d_to_hf_mapper(
             mappers.append(_StackedToFused(pre, p.intermediate_size))
         else:
             for proj in ("gate_proj", "up_proj", "down_proj"):
-                unstack = UnstackTensors(
-                    f"{pre}mlp.experts.{proj}.weight",
-                    [f"{pre}mlp.experts._t{e}.{proj}" for e in range(p.num_experts)],
-                    dim=0,
-                )
-                per_expert = [
-                    Transpose(
-                        f"{pre}mlp.experts._t{e}.{proj}",
-                        f"{pre}mlp.experts.{e}.{proj}.weight",
+                mappers.append(
+                    UnstackTensors(
+                        f"{pre}mlp.experts.{proj}.weight",
+                        [f"{pre}mlp.experts.{e}.{proj}.weight" for e in range(p.num_experts)],
+                        dim=0,
                     )
-                    for e in range(p.num_experts)
-                ]
-                mappers.append(Sequential(unstack, Parallel(*per_expert)))
+                )
     return Parallel(*mappers)

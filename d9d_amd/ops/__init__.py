@@ -3,7 +3,7 @@ from .swiglu import silu_mul
 from .stochastic import copy_fp32_to_bf16_stochastic_, adamw_stochastic_bf16_
 from .attention import flash_attn_func
 from .cce import linear_cross_entropy, LM_IGNORE_INDEX, VocabParallelOptions
-from .gmm import gmm
+from .gmm import gmm, gmm_nt
 from .moe_permute import moe_permute, moe_unpermute
 
 __all__ = [
@@ -16,6 +16,7 @@ __all__ = [
     "LM_IGNORE_INDEX",
     "VocabParallelOptions",
     "gmm",
+    "gmm_nt",
     "moe_permute",
     "moe_unpermute",
 ]

@@ -69,3 +69,52 @@ def gmm(a: torch.Tensor, b: torch.Tensor, batch_sizes: torch.Tensor) -> torch.Te
     """a (sum_T, K); b (E, K, N); batch_sizes (E,) int64 on CPU."""
     assert batch_sizes.device.type == "cpu"
     return _GroupedGemmFunction.apply(a, b, batch_sizes)
+
+
+def _gmm_loop_nt(a: torch.Tensor, w: torch.Tensor, batch_sizes: torch.Tensor) -> torch.Tensor:
+    out = a.new_empty((a.shape[0], w.shape[1]))
+    start = 0
+    for e, n in enumerate(batch_sizes.tolist()):
+        if n:
+            out[start : start + n] = a[start : start + n] @ w[e].t()
+        start += n
+    return out
+
+
+class _GroupedLinearNTFunction(torch.autograd.Function):
+    """Grouped expert linear with weight stored (E, N, K) (nn.Linear layout).
+
+    fwd:   out[rows_e] = a[rows_e] @ w[e]^T          (gmm_nt kernel: both
+           operands k-contiguous, scatter-free LDS staging)
+    dgrad: da[rows_e]  = g[rows_e] @ w[e]            (gmm kernel reads w as
+           (red, out) row-major -- no transpose materialisation)
+    wgrad: dw[e]       = g[rows_e]^T @ a[rows_e]     (gmm_db with swapped args
+           -> (E, N, K) directly)
+    """
+
+    @staticmethod
+    def forward(ctx, a, w, batch_sizes):
+        ctx.save_for_backward(a, w, batch_sizes)
+        if a.is_cuda and has_ext() and hasattr(get_ext(), "gmm_nt"):
+            return get_ext().gmm_nt(a.contiguous(), w.contiguous(), batch_sizes)
+        return _gmm_loop_nt(a, w, batch_sizes)
+
+    @staticmethod
+    def backward(ctx, g):
+        a, w, batch_sizes = ctx.saved_tensors
+        g = g.contiguous()
+        da = dw = None
+        if ctx.needs_input_grad[0] and GLOBAL_GRAD_CONTEXT.computes(GradDirection.INPUTS):
+            da = _gmm_forward_impl(g, w, batch_sizes)
+        if ctx.needs_input_grad[1] and GLOBAL_GRAD_CONTEXT.computes(GradDirection.WEIGHTS):
+            if a.is_cuda and has_ext() and hasattr(get_ext(), "gmm_db"):
+                dw = get_ext().gmm_db(g, a.contiguous(), batch_sizes, w.shape[0])
+            else:
+                dw = _gmm_accum_db(g, a, batch_sizes, w.shape[0])
+        return da, dw, None
+
+
+def gmm_nt(a: torch.Tensor, w: torch.Tensor, batch_sizes: torch.Tensor) -> torch.Tensor:
+    """a (sum_T, K); w (E, N, K) nn.Linear-layout; batch_sizes (E,) int64 CPU."""
+    assert batch_sizes.device.type == "cpu"
+    return _GroupedLinearNTFunction.apply(a, w, batch_sizes)

@@ -77,7 +77,8 @@ class LoRAGroupedLinear(nn.Module):
     def merge_into_base(self) -> GroupedLinear:
         w = self.base.weight
         local = w.to_local() if hasattr(w, "to_local") else w
-        local += torch.bmm(self.lora_A, self.lora_B) * self.scaling
+        # base weight is (E, out, in); A@B is (E, in, out)
+        local += (torch.bmm(self.lora_A, self.lora_B) * self.scaling).transpose(1, 2)
         return self.base
 
 
