@@ -45,7 +45,7 @@ __global__ __launch_bounds__(512, 1) void gmm_kernel(
     bf16_t* __restrict__ out,        // (T, N)
     const int* __restrict__ row_off,      // (E+1,)
     const int* __restrict__ mtile_pref,   // (E+1,) prefix of ceil(rows_e/kBM)
-    int E, int K, int N) {
+    int E, int K, int N, int n_tiles) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* a_lds = reinterpret_cast<bf16_t*>(smem);        // [kBM][kBK]
   bf16_t* bt_lds = a_lds + kBM * kBK;                     // [kBN][kBK]
@@ -55,10 +55,14 @@ __global__ __launch_bounds__(512, 1) void gmm_kernel(
   const int wm = wave >> 2;   // 0..1: wave row (128 rows each)
   const int wn = wave & 3;    // 0..3: wave col (64 cols each)
 
-  // ---- map blockIdx.y -> (expert, m_tile) via binary search ----------------
-  // grid.x is the (small) n-tile index: consecutive blocks share the same A
-  // rows, so A re-reads hit L2 instead of HBM.
-  const int mt_global = blockIdx.y;
+  // ---- XCD-aware expert-major work mapping (see gmm_nt_kernel) -------------
+  const int nwg = gridDim.x;
+  const int xcd = blockIdx.x & 7;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int wid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                  + (blockIdx.x >> 3);
+  // ---- map work id -> (expert, m_tile) via binary search -------------------
+  const int mt_global = wid / n_tiles;
   int lo = 0, hi = E - 1;
   while (lo < hi) {
     const int mid = (lo + hi + 1) >> 1;
@@ -68,7 +72,7 @@ __global__ __launch_bounds__(512, 1) void gmm_kernel(
   const int m_tile = mt_global - mtile_pref[e];
   const int row0 = row_off[e] + m_tile * kBM;
   const int row_end = row_off[e + 1];
-  const int n0 = blockIdx.x * kBN;
+  const int n0 = (wid % n_tiles) * kBN;
 
   const bf16_t* b_e = b + (int64_t)e * K * N;
 
@@ -245,7 +249,7 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_kernel(
     bf16_t* __restrict__ out,        // (T, N)
     const int* __restrict__ row_off,
     const int* __restrict__ mtile_pref,
-    int E, int K, int N) {
+    int E, int K, int N, int n_tiles) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* a_lds = reinterpret_cast<bf16_t*>(smem);        // [kBM][kBK]
   bf16_t* bt_lds = a_lds + kBM * kBK;                     // [kBN][kBK]
@@ -255,7 +259,16 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_kernel(
   const int wm = wave >> 2;
   const int wn = wave & 3;
 
-  const int mt_global = blockIdx.y;
+  // XCD-aware work mapping: hardware deals linear block ids round-robin over
+  // the 8 XCDs (each with a private L2). The bijective remap gives each XCD a
+  // CONTIGUOUS span of expert-major work items, so one expert's weight slice
+  // and A rows are re-read from that XCD's L2 instead of HBM.
+  const int nwg = gridDim.x;
+  const int xcd = blockIdx.x & 7;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int wid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                  + (blockIdx.x >> 3);
+  const int mt_global = wid / n_tiles;
   int lo = 0, hi = E - 1;
   while (lo < hi) {
     const int mid = (lo + hi + 1) >> 1;
@@ -265,7 +278,7 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_kernel(
   const int m_tile = mt_global - mtile_pref[e];
   const int row0 = row_off[e] + m_tile * kBM;
   const int row_end = row_off[e + 1];
-  const int n0 = blockIdx.x * kBN;
+  const int n0 = (wid % n_tiles) * kBN;
 
   const bf16_t* w_e = w + (int64_t)e * N * K;
 
@@ -417,7 +430,7 @@ __global__ __launch_bounds__(512, 1) void gmm_db_kernel(
     bf16_t* __restrict__ db,        // (E, K, N)
     const int* __restrict__ row_off,
     const int* __restrict__ expert_order,  // experts sorted by row count desc
-    int E, int K, int N) {
+    int E, int K, int N, int kt_tiles, int nt_tiles) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* at_lds = reinterpret_cast<bf16_t*>(smem);   // [256 k][64 rows]
   bf16_t* gt_lds = at_lds + 256 * 64;                 // [256 n][64 rows]
@@ -427,10 +440,17 @@ __global__ __launch_bounds__(512, 1) void gmm_db_kernel(
   const int wm = wave >> 2;  // k-half (128 rows of db)
   const int wn = wave & 3;   // n-quarter (64 cols)
 
-  // Heavy experts dispatch first so the tail wave holds the small tiles.
-  const int e = expert_order[blockIdx.z];
-  const int k0 = blockIdx.y * 256;
-  const int n0 = blockIdx.x * 256;
+  // NO XCD grouping here (unlike gmm_nt_kernel): per-tile work scales with
+  // the expert's row count, and giving one XCD a contiguous expert-major
+  // span concentrates the heavy (sorted-first) experts on it -- measured as
+  // an 8-XCD straggler that cost ~15% end-to-end. Round-robin dispatch
+  // balances the ragged loop lengths instead. Heavy experts dispatch first
+  // so the tail wave holds the small tiles.
+  const int wid = blockIdx.x;
+  const int tiles_per_e = kt_tiles * nt_tiles;
+  const int e = expert_order[wid / tiles_per_e];
+  const int k0 = ((wid % tiles_per_e) / nt_tiles) * 256;
+  const int n0 = (wid % nt_tiles) * 256;
   const int r_start = row_off[e];
   const int r_end = row_off[e + 1];
   if (r_start >= r_end) return;  // empty expert: db stays zero (host zeros it)
@@ -593,7 +613,8 @@ torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
       ? mtile_pref[E].item<int>() : 0;
   if (total_mtiles == 0) return out;
 
-  const dim3 grid((N + d9d::kBN - 1) / d9d::kBN, total_mtiles);
+  const int n_tiles = (N + d9d::kBN - 1) / d9d::kBN;
+  const dim3 grid(n_tiles * total_mtiles);
   const size_t smem = (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(d9d::gmm_kernel, grid, dim3(512), smem, stream,
@@ -601,7 +622,7 @@ torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
                      reinterpret_cast<const __bf16*>(b.data_ptr()),
                      reinterpret_cast<__bf16*>(out.data_ptr()),
                      row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
-                     E, K, N);
+                     E, K, N, n_tiles);
   return out;
 }
 
@@ -620,7 +641,8 @@ torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes
   const int total_mtiles = mtile_pref[E].item<int>();
   if (total_mtiles == 0) return out;
 
-  const dim3 grid((N + d9d::kBN - 1) / d9d::kBN, total_mtiles);
+  const int n_tiles = (N + d9d::kBN - 1) / d9d::kBN;
+  const dim3 grid(n_tiles * total_mtiles);
   const size_t smem = (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(d9d::gmm_nt_kernel, grid, dim3(512), smem, stream,
@@ -628,7 +650,7 @@ torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes
                      reinterpret_cast<const __bf16*>(w.data_ptr()),
                      reinterpret_cast<__bf16*>(out.data_ptr()),
                      row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
-                     E, K, N);
+                     E, K, N, n_tiles);
   return out;
 }
 
@@ -646,7 +668,8 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
                                   /*descending=*/true).to(torch::kInt32);
   auto expert_order = order_cpu.to(a.device(), /*non_blocking=*/true);
 
-  const dim3 grid((N + 255) / 256, (K + 255) / 256, E);
+  const int nt_tiles = (N + 255) / 256, kt_tiles = (K + 255) / 256;
+  const dim3 grid(nt_tiles * kt_tiles * E);
   const size_t smem = (2 * 256 * 64) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(d9d::gmm_db_kernel, grid, dim3(512), smem, stream,
@@ -654,6 +677,6 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
                      reinterpret_cast<__bf16*>(db.data_ptr()),
                      row_off.data_ptr<int>(), expert_order.data_ptr<int>(),
-                     E, K, N);
+                     E, K, N, kt_tiles, nt_tiles);
   return db;
 }
