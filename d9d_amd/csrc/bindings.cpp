@@ -7,6 +7,8 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor x, torch::Tensor w, torch:
 
 // silu_mul.hip
 torch::Tensor silu_mul_fwd(torch::Tensor a, torch::Tensor b);
+torch::Tensor silu_mul_packed_fwd(torch::Tensor x);
+torch::Tensor silu_mul_packed_bwd(torch::Tensor x, torch::Tensor g);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor a, torch::Tensor b, torch::Tensor g);
 
 // stochastic.hip
@@ -58,6 +60,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rms_norm_fwd", &rms_norm_fwd, "RMSNorm forward (bf16, CDNA4)");
   m.def("rms_norm_bwd", &rms_norm_bwd, "RMSNorm backward (bf16, CDNA4)");
   m.def("silu_mul_fwd", &silu_mul_fwd, "fused silu(a)*b forward");
+  m.def("silu_mul_packed_fwd", &silu_mul_packed_fwd, "packed SwiGLU fwd");
+  m.def("silu_mul_packed_bwd", &silu_mul_packed_bwd, "packed SwiGLU bwd");
   m.def("silu_mul_bwd", &silu_mul_bwd, "fused silu(a)*b backward");
   m.def("copy_fp32_to_bf16_stochastic_", &copy_fp32_to_bf16_stochastic_,
         "stochastic-rounding fp32->bf16 copy");

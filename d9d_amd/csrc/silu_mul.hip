@@ -78,6 +78,68 @@ __global__ void silu_mul_bwd_kernel(
   }
 }
 
+
+// Packed variant: x (T, 2I) holds [gate | up] from the fused gate_up grouped
+// GEMM; out (T, I). Avoids materialising two separate projections and the
+// dgrad-sum the autograd graph would otherwise insert.
+template <int BLOCK>
+__global__ void silu_mul_packed_fwd_kernel(
+    const ushort* __restrict__ x,
+    ushort* __restrict__ out,
+    int64_t rows, int inter) {
+  const int64_t n_vec = rows * (inter / 8);
+  const int64_t tid = blockIdx.x * static_cast<int64_t>(BLOCK) + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
+  const int iv = inter / 8;
+  for (int64_t v = tid; v < n_vec; v += stride) {
+    const int64_t t = v / iv;
+    const int i = (int)(v % iv) * 8;
+    const ushort* xp = x + t * (int64_t)(2 * inter);
+    Bf16x8 av, bv, ov;
+    av.u = *reinterpret_cast<const ushort8v*>(xp + i);
+    bv.u = *reinterpret_cast<const ushort8v*>(xp + inter + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      ov.s[j] = f32_to_bf16_rne(
+          silu_f(bf16_bits_to_f32(av.s[j])) * bf16_bits_to_f32(bv.s[j]));
+    }
+    *reinterpret_cast<ushort8v*>(out + t * (int64_t)inter + i) = ov.u;
+  }
+}
+
+template <int BLOCK>
+__global__ void silu_mul_packed_bwd_kernel(
+    const ushort* __restrict__ x,
+    const ushort* __restrict__ g,   // (T, I)
+    ushort* __restrict__ dx,        // (T, 2I)
+    int64_t rows, int inter) {
+  const int64_t n_vec = rows * (inter / 8);
+  const int64_t tid = blockIdx.x * static_cast<int64_t>(BLOCK) + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
+  const int iv = inter / 8;
+  for (int64_t v = tid; v < n_vec; v += stride) {
+    const int64_t t = v / iv;
+    const int i = (int)(v % iv) * 8;
+    const ushort* xp = x + t * (int64_t)(2 * inter);
+    ushort* dxp = dx + t * (int64_t)(2 * inter);
+    Bf16x8 av, bv, gv, dav, dbv;
+    av.u = *reinterpret_cast<const ushort8v*>(xp + i);
+    bv.u = *reinterpret_cast<const ushort8v*>(xp + inter + i);
+    gv.u = *reinterpret_cast<const ushort8v*>(g + t * (int64_t)inter + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float af = bf16_bits_to_f32(av.s[j]);
+      const float bf = bf16_bits_to_f32(bv.s[j]);
+      const float gf = bf16_bits_to_f32(gv.s[j]);
+      const float sig = 1.f / (1.f + __expf(-af));
+      dav.s[j] = f32_to_bf16_rne(gf * bf * sig * (1.f + af * (1.f - sig)));
+      dbv.s[j] = f32_to_bf16_rne(gf * af * sig);
+    }
+    *reinterpret_cast<ushort8v*>(dxp + i) = dav.u;
+    *reinterpret_cast<ushort8v*>(dxp + inter + i) = dbv.u;
+  }
+}
+
 }  // namespace d9d
 
 static int silu_grid(int64_t n_vec, int block) {
@@ -120,4 +182,40 @@ std::vector<torch::Tensor> silu_mul_bwd(
       reinterpret_cast<ushort*>(da.data_ptr()),
       reinterpret_cast<ushort*>(db.data_ptr()), n);
   return {da, db};
+}
+
+
+torch::Tensor silu_mul_packed_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+  const int64_t rows = x.size(0);
+  const int inter = (int)x.size(1) / 2;
+  TORCH_CHECK(x.size(1) % 2 == 0 && inter % 8 == 0, "packed silu_mul needs I % 8 == 0");
+  auto out = torch::empty({rows, (int64_t)inter}, x.options());
+  if (out.numel() == 0) return out;
+  constexpr int kBlock = 256;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      (d9d::silu_mul_packed_fwd_kernel<kBlock>),
+      dim3(silu_grid(rows * (inter / 8), kBlock)), dim3(kBlock), 0, stream,
+      reinterpret_cast<const ushort*>(x.data_ptr()),
+      reinterpret_cast<ushort*>(out.data_ptr()), rows, inter);
+  return out;
+}
+
+torch::Tensor silu_mul_packed_bwd(torch::Tensor x, torch::Tensor g) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(g.is_contiguous() && g.scalar_type() == torch::kBFloat16);
+  const int64_t rows = x.size(0);
+  const int inter = (int)x.size(1) / 2;
+  auto dx = torch::empty_like(x);
+  if (x.numel() == 0) return dx;
+  constexpr int kBlock = 256;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      (d9d::silu_mul_packed_bwd_kernel<kBlock>),
+      dim3(silu_grid(rows * (inter / 8), kBlock)), dim3(kBlock), 0, stream,
+      reinterpret_cast<const ushort*>(x.data_ptr()),
+      reinterpret_cast<const ushort*>(g.data_ptr()),
+      reinterpret_cast<ushort*>(dx.data_ptr()), rows, inter);
+  return dx;
 }

@@ -29,71 +29,6 @@ from ....model_state.mapper import (
 from .params import Qwen3MoEModelParameters
 
 
-class _FusedToStacked(ModelStateMapper):
-    """HF packed experts -> our stacked (E, out, in) weights for one layer."""
-
-    def __init__(self, prefix: str, intermediate: int) -> None:
-        self.prefix = prefix
-        self.intermediate = intermediate
-
-    def state_dependency_groups(self):
-        pre = self.prefix
-        return [
-            StateGroup.of(
-                [pre + "mlp.experts.gate_up_proj"],
-                [pre + "mlp.experts.gate_proj.weight", pre + "mlp.experts.up_proj.weight"],
-            ),
-            StateGroup.of(
-                [pre + "mlp.experts.down_proj"],
-                [pre + "mlp.experts.down_proj.weight"],
-            ),
-        ]
-
-    def apply_group(self, group, tensors):
-        pre = self.prefix
-        if pre + "mlp.experts.gate_up_proj" in tensors:
-            gu = tensors[pre + "mlp.experts.gate_up_proj"]  # (E, 2I, H)
-            # our GroupedLinear stores (E, out, in) = (E, I, H): direct split
-            gate = gu[:, : self.intermediate, :].contiguous()
-            up = gu[:, self.intermediate :, :].contiguous()
-            return {
-                pre + "mlp.experts.gate_proj.weight": gate,
-                pre + "mlp.experts.up_proj.weight": up,
-            }
-        down = tensors[pre + "mlp.experts.down_proj"]  # (E, H, I) = (E, out, in)
-        return {pre + "mlp.experts.down_proj.weight": down.contiguous()}
-
-
-class _StackedToFused(ModelStateMapper):
-    def __init__(self, prefix: str, intermediate: int) -> None:
-        self.prefix = prefix
-        self.intermediate = intermediate
-
-    def state_dependency_groups(self):
-        pre = self.prefix
-        return [
-            StateGroup.of(
-                [pre + "mlp.experts.gate_proj.weight", pre + "mlp.experts.up_proj.weight"],
-                [pre + "mlp.experts.gate_up_proj"],
-            ),
-            StateGroup.of(
-                [pre + "mlp.experts.down_proj.weight"],
-                [pre + "mlp.experts.down_proj"],
-            ),
-        ]
-
-    def apply_group(self, group, tensors):
-        import torch
-
-        pre = self.prefix
-        if pre + "mlp.experts.gate_proj.weight" in tensors:
-            gate = tensors[pre + "mlp.experts.gate_proj.weight"]
-            up = tensors[pre + "mlp.experts.up_proj.weight"]
-            return {pre + "mlp.experts.gate_up_proj": torch.cat([gate, up], dim=1).contiguous()}
-        down = tensors[pre + "mlp.experts.down_proj.weight"]
-        return {pre + "mlp.experts.down_proj": down.contiguous()}
-
-
 def _vocab_splits(p: Qwen3MoEModelParameters) -> list[tuple[str, int]]:
     return [(name, p.split_vocab_size[name]) for name in p.split_vocab_order]
 
@@ -135,19 +70,39 @@ def hf_to_d9d_mapper(
             mappers.append(Identity(pre + key))
         mappers.append(Rename(pre + "mlp.gate.weight", pre + "mlp.router.gate.weight"))
         if expert_format == "fused":
-            mappers.append(_FusedToStacked(pre, p.intermediate_size))
+            # HF fused layout (E, 2I, H) / (E, H, I) == ours: pure renames
+            mappers.append(
+                Rename(pre + "mlp.experts.gate_up_proj", pre + "mlp.experts.gate_up_proj.weight")
+            )
+            mappers.append(
+                Rename(pre + "mlp.experts.down_proj", pre + "mlp.experts.down_proj.weight")
+            )
         else:
-            for proj in ("gate_proj", "up_proj", "down_proj"):
-                # HF expert weight (out, in) == our per-expert layout: stack
+            # per-expert (out, in) weights: stack, then concat gate|up on dim 1
+            for proj in ("gate_proj", "up_proj"):
                 mappers.append(
-                    Sequential(
-                        StackTensors(
-                            [f"{pre}mlp.experts.{e}.{proj}.weight" for e in range(p.num_experts)],
-                            f"{pre}mlp.experts.{proj}.weight",
-                            dim=0,
-                        ),
+                    StackTensors(
+                        [f"{pre}mlp.experts.{e}.{proj}.weight" for e in range(p.num_experts)],
+                        f"{pre}mlp.experts._stacked.{proj}",
+                        dim=0,
                     )
                 )
+            mappers.append(
+                Sequential(
+                    ConcatenateTensors(
+                        [pre + "mlp.experts._stacked.gate_proj", pre + "mlp.experts._stacked.up_proj"],
+                        pre + "mlp.experts.gate_up_proj.weight",
+                        dim=1,
+                    ),
+                )
+            )
+            mappers.append(
+                StackTensors(
+                    [f"{pre}mlp.experts.{e}.down_proj.weight" for e in range(p.num_experts)],
+                    pre + "mlp.experts.down_proj.weight",
+                    dim=0,
+                )
+            )
     return Parallel(*mappers)
 
 
@@ -188,14 +143,38 @@ def d9d_to_hf_mapper(
             mappers.append(Identity(pre + key))
         mappers.append(Rename(pre + "mlp.router.gate.weight", pre + "mlp.gate.weight"))
         if expert_format == "fused":
-            mappers.append(_StackedToFused(pre, p.intermediate_size))
+            mappers.append(
+                Rename(pre + "mlp.experts.gate_up_proj.weight", pre + "mlp.experts.gate_up_proj")
+            )
+            mappers.append(
+                Rename(pre + "mlp.experts.down_proj.weight", pre + "mlp.experts.down_proj")
+            )
         else:
-            for proj in ("gate_proj", "up_proj", "down_proj"):
+            mappers.append(
+                Sequential(
+                    SliceRows(
+                        pre + "mlp.experts.gate_up_proj.weight",
+                        [
+                            (pre + "mlp.experts._stacked.gate_proj", p.intermediate_size),
+                            (pre + "mlp.experts._stacked.up_proj", p.intermediate_size),
+                        ],
+                        dim=1,
+                    ),
+                )
+            )
+            for proj in ("gate_proj", "up_proj"):
                 mappers.append(
                     UnstackTensors(
-                        f"{pre}mlp.experts.{proj}.weight",
+                        f"{pre}mlp.experts._stacked.{proj}",
                         [f"{pre}mlp.experts.{e}.{proj}.weight" for e in range(p.num_experts)],
                         dim=0,
                     )
                 )
+            mappers.append(
+                UnstackTensors(
+                    pre + "mlp.experts.down_proj.weight",
+                    [f"{pre}mlp.experts.{e}.down_proj.weight" for e in range(p.num_experts)],
+                    dim=0,
+                )
+            )
     return Parallel(*mappers)

@@ -1,5 +1,5 @@
 from .rms_norm import rms_norm
-from .swiglu import silu_mul
+from .swiglu import silu_mul, silu_mul_packed
 from .stochastic import copy_fp32_to_bf16_stochastic_, adamw_stochastic_bf16_
 from .attention import flash_attn_func
 from .cce import linear_cross_entropy, LM_IGNORE_INDEX, VocabParallelOptions
@@ -9,6 +9,7 @@ from .moe_permute import moe_permute, moe_unpermute
 __all__ = [
     "rms_norm",
     "silu_mul",
+    "silu_mul_packed",
     "copy_fp32_to_bf16_stochastic_",
     "adamw_stochastic_bf16_",
     "flash_attn_func",
