@@ -233,3 +233,42 @@ def test_rope_qk_kernel_parity(rope_dim):
     torch.testing.assert_close(k_out.float(), k_ref.float(), rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(q.grad.float(), q2.grad.float(), rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(k.grad.float(), k2.grad.float(), rtol=2e-2, atol=2e-2)
+
+
+def test_silu_mul_packed_cpu_matches_unpacked():
+    import torch
+
+    from d9d_amd.ops import silu_mul, silu_mul_packed
+
+    torch.manual_seed(0)
+    x = torch.randn(32, 48, requires_grad=True)
+    out = silu_mul_packed(x)
+    ref = silu_mul(x[:, :24], x[:, 24:])
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+    g = torch.randn_like(out)
+    out.backward(g)
+    xg = x.grad.clone()
+    x.grad = None
+    silu_mul(x[:, :24], x[:, 24:]).backward(g)
+    torch.testing.assert_close(xg, x.grad, rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_silu_mul_packed_gpu_parity():
+    import torch
+
+    from d9d_amd.ops import silu_mul_packed
+
+    torch.manual_seed(1)
+    x = torch.randn(500, 2 * 576, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    out = silu_mul_packed(x)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    x32 = x.detach().float().requires_grad_(True)
+    import torch.nn.functional as F
+
+    ref = F.silu(x32[:, :576]) * x32[:, 576:]
+    ref.backward(g.float())
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(x.grad.float(), x32.grad, rtol=3e-2, atol=3e-2)
