@@ -119,6 +119,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     const bf16_t* __restrict__ v,   // (B,Skv,Hkv,D)
     bf16_t* __restrict__ out,       // (B,Sq,Hq,D)
     float* __restrict__ lse,        // (B,Hq,Sq)
+    const float* __restrict__ sinks,  // (Hq,) raw sink logits, or nullptr
     int B, int Sq, int Skv, int Hq, int Hkv,
     float scale, int causal, int window_left, int q_offset) {
   constexpr int kNT = D / 16;   // n-tiles over head dim
@@ -158,13 +159,18 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     }
   }
 
+  // A learnable "sink" is one virtual softmax column per head with logit
+  // sinks[h] and no value row: seed the running max/sum with it and the
+  // online softmax (and the LSE the backward reads) absorbs it for free.
+  const float m_init = sinks ? sinks[h] : -1e30f;
+  const float l_init = sinks ? 1.f : 0.f;
   float m_run[2][4], l_run[2][4];
 #pragma unroll
   for (int m = 0; m < 2; ++m)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      m_run[m][r] = -1e30f;
-      l_run[m][r] = 0.f;
+      m_run[m][r] = m_init;
+      l_run[m][r] = l_init;
     }
   f32x4 o_acc[2][kNT];
 #pragma unroll
@@ -308,13 +314,25 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
         }
         l_add[r] = acc;
       }
+      // Lazy rescale: after the first few KV tiles the running max rarely
+      // moves, so alpha == 1 for every row of the wave most of the time --
+      // skip the 4 exp2s and kNT*4 accumulator multiplies (wave-uniform
+      // branch, ~25% of the softmax VALU work).
+      bool same = true;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const float alpha = __builtin_amdgcn_exp2f((m_run[m][r] - m_new[r]) * kLog2e);
-        l_run[m][r] = l_run[m][r] * alpha + l_add[r];
-        m_run[m][r] = m_new[r];
+      for (int r = 0; r < 4; ++r) same &= (m_new[r] == m_run[m][r]);
+      if (__all(same)) {
 #pragma unroll
-        for (int nt = 0; nt < kNT; ++nt) o_acc[m][nt][r] *= alpha;
+        for (int r = 0; r < 4; ++r) l_run[m][r] += l_add[r];
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float alpha = __builtin_amdgcn_exp2f((m_run[m][r] - m_new[r]) * kLog2e);
+          l_run[m][r] = l_run[m][r] * alpha + l_add[r];
+          m_run[m][r] = m_new[r];
+#pragma unroll
+          for (int nt = 0; nt < kNT; ++nt) o_acc[m][nt][r] *= alpha;
+        }
       }
 
       // P C-layout -> row-major LDS scratch (own-wave slice) -> A fragments
@@ -810,6 +828,7 @@ torch::Tensor maybe_pad_d(torch::Tensor t, int D_pad) {
 
 std::vector<torch::Tensor> flash_attn_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    c10::optional<torch::Tensor> sinks,
     bool causal, double softmax_scale, int64_t window_left, int64_t q_offset) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
@@ -824,6 +843,13 @@ std::vector<torch::Tensor> flash_attn_fwd(
 
   auto out = torch::empty_like(qp);
   auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+  const float* sinks_ptr = nullptr;
+  torch::Tensor sinks_f;
+  if (sinks.has_value()) {
+    sinks_f = sinks->to(torch::kFloat32).contiguous();
+    TORCH_CHECK(sinks_f.numel() == Hq, "sinks must be (Hq,)");
+    sinks_ptr = sinks_f.data_ptr<float>();
+  }
 
   const dim3 grid((Sq + 127) / 128, B * Hq);
   const size_t smem =
@@ -838,7 +864,7 @@ std::vector<torch::Tensor> flash_attn_fwd(
                      reinterpret_cast<const __bf16*>(kp.data_ptr()),          \
                      reinterpret_cast<const __bf16*>(vp.data_ptr()),          \
                      reinterpret_cast<__bf16*>(out.data_ptr()),               \
-                     lse.data_ptr<float>(), B, Sq, Skv, Hq, Hkv,              \
+                     lse.data_ptr<float>(), sinks_ptr, B, Sq, Skv, Hq, Hkv,   \
                      (float)softmax_scale, causal ? 1 : 0, (int)window_left,   \
                      (int)q_offset)
   switch (D_pad) {

@@ -19,6 +19,7 @@ void adamw_stochastic_bf16_(torch::Tensor p, torch::Tensor g, torch::Tensor m, t
 
 // attention.hip
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                           c10::optional<torch::Tensor> sinks,
                                           bool causal, double softmax_scale, int64_t window_left,
                                           int64_t q_offset);
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,

@@ -74,29 +74,38 @@ class _FlashAttnFunction(torch.autograd.Function):
     def forward(ctx, q, k, v, sinks, causal, softmax_scale, window_left, q_offset):
         ext = get_ext()
         out, lse = ext.flash_attn_fwd(
-            q.contiguous(), k.contiguous(), v.contiguous(),
+            q.contiguous(), k.contiguous(), v.contiguous(), sinks,
             causal, softmax_scale, window_left, q_offset,
         )
-        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.save_for_backward(q, k, v, out, lse, *( [sinks] if sinks is not None else [] ))
         ctx.causal = causal
         ctx.softmax_scale = softmax_scale
         ctx.window_left = window_left
         ctx.q_offset = q_offset
         ctx.has_sinks = sinks is not None
-        if ctx.has_sinks:
-            raise NotImplementedError("sinks on the HIP path land with the flash kernel v2")
         return out, lse
 
     @staticmethod
     def backward(ctx, dout, dlse):
-        q, k, v, out, lse = ctx.saved_tensors
+        q, k, v, out, lse = ctx.saved_tensors[:5]
         ext = get_ext()
+        # The sink only enters through the LSE, which the kernel reads back:
+        # dq/dk/dv formulas are unchanged.
         dq, dk, dv = ext.flash_attn_bwd(
             dout.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(),
             out.contiguous(), lse, ctx.causal, ctx.softmax_scale, ctx.window_left,
             ctx.q_offset,
         )
-        return dq, dk, dv, None, None, None, None, None
+        dsinks = None
+        if ctx.has_sinks:
+            sinks = ctx.saved_tensors[5]
+            # p_sink(b,h,t) = exp(sink_h - lse); dsink_h = -sum p_sink * delta
+            # with delta = rowsum(dout * out) (analytic sink grad, reference
+            # d9d/kernel/flash_attn/function.py dsink).
+            delta = (out.float() * dout.float()).sum(-1).permute(0, 2, 1)  # (B,Hq,S)
+            p_sink = torch.exp(sinks.float().view(1, -1, 1) - lse)
+            dsinks = -(p_sink * delta).sum(dim=(0, 2)).to(sinks.dtype)
+        return dq, dk, dv, dsinks, None, None, None, None
 
 
 def flash_attn_func(
@@ -114,9 +123,10 @@ def flash_attn_func(
     """Scaled-dot-product attention. q (B,S,Hq,D); k/v (B,S,Hkv,D)."""
     if softmax_scale is None:
         softmax_scale = 1.0 / math.sqrt(q.shape[-1])
-    if q.is_cuda and sinks is None and window_size == (-1, -1):
+    right_ok = window_size[1] < 0 or causal  # right window only via causality
+    if q.is_cuda and right_ok:
         out, lse = _FlashAttnFunction.apply(
-            q, k, v, None, causal, softmax_scale, -1, q_offset
+            q, k, v, sinks, causal, softmax_scale, window_size[0], q_offset
         )
     else:
         out, lse = _eager_attention(

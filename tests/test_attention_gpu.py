@@ -111,3 +111,55 @@ def test_flash_attn_q_offset_cp_parity():
             part.float(), full[:, r * half : (r + 1) * half].float(),
             rtol=2e-2, atol=2e-2,
         )
+
+
+@pytest.mark.gpu
+def test_flash_sliding_window_matches_eager():
+    torch.manual_seed(11)
+    B, S, Hq, Hkv, D = 2, 512, 8, 2, 64
+    q = torch.randn(B, S, Hq, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    k = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    v = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+
+    out = flash_attn_func(q, k, v, causal=True, window_size=(128, -1))
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    q32 = q.detach().float().requires_grad_(True)
+    k32 = k.detach().float().requires_grad_(True)
+    v32 = v.detach().float().requires_grad_(True)
+    ref, _ = _eager_attention(q32, k32, v32, True, q.shape[-1] ** -0.5, (128, -1), None)
+    ref.backward(g.float())
+
+    torch.testing.assert_close(out.float(), ref.float(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(q.grad.float(), q32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), k32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(v.grad.float(), v32.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.gpu
+def test_flash_attention_sinks_matches_eager():
+    torch.manual_seed(12)
+    B, S, Hq, Hkv, D = 2, 256, 8, 2, 64
+    q = torch.randn(B, S, Hq, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    k = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    v = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    sinks = torch.randn(Hq, dtype=torch.float32, device="cuda", requires_grad=True)
+
+    out, lse = flash_attn_func(q, k, v, causal=True, sinks=sinks, return_lse=True)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    q32 = q.detach().float().requires_grad_(True)
+    k32 = k.detach().float().requires_grad_(True)
+    v32 = v.detach().float().requires_grad_(True)
+    s32 = sinks.detach().clone().requires_grad_(True)
+    ref, ref_lse = _eager_attention(q32, k32, v32, True, q.shape[-1] ** -0.5, (-1, -1), s32)
+    ref.backward(g.float())
+
+    torch.testing.assert_close(out.float(), ref.float(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(lse, ref_lse, rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(q.grad.float(), q32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), k32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(v.grad.float(), v32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(sinks.grad, s32.grad, rtol=5e-2, atol=5e-2)
