@@ -120,6 +120,10 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     bf16_t* __restrict__ out,       // (B,Sq,Hq,D)
     float* __restrict__ lse,        // (B,Hq,Sq)
     const float* __restrict__ sinks,  // (Hq,) raw sink logits, or nullptr
+    const int* __restrict__ cu_q,     // (nseq+1,) packed-seq bounds, or nullptr
+    const int* __restrict__ cu_k,
+    const int* __restrict__ qtile_pref,  // (nseq+1,) prefix of ceil(len_q/128)
+    int nseq,
     int B, int Sq, int Skv, int Hq, int Hkv,
     float scale, int causal, int window_left, int q_offset) {
   constexpr int kNT = D / 16;   // n-tiles over head dim
@@ -135,10 +139,35 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int q_tile = blockIdx.x;
-  const int bh = blockIdx.y;
-  const int b = bh / Hq;
-  const int h = bh % Hq;
+
+  // Varlen (packed sequences): grid.x spans per-sequence q-tiles; binary
+  // search the tile prefix for our sequence, then work in LOCAL coordinates
+  // with the sequence's row ranges. Causal masking aligns the END of q with
+  // the END of kv (flash-attn convention for len_q != len_kv).
+  int b, h, q_tile, q_lo, Sq_loc, kv_lo, Skv_loc, causal_off;
+  if (cu_q != nullptr) {
+    h = blockIdx.y;
+    int lo = 0, hi = nseq - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (qtile_pref[mid] <= (int)blockIdx.x) lo = mid; else hi = mid - 1;
+    }
+    b = 0;
+    q_tile = blockIdx.x - qtile_pref[lo];
+    q_lo = cu_q[lo];
+    Sq_loc = cu_q[lo + 1] - q_lo;
+    kv_lo = cu_k[lo];
+    Skv_loc = cu_k[lo + 1] - kv_lo;
+    causal_off = Skv_loc - Sq_loc + q_offset;
+    if (Sq_loc <= 0 || q_tile * kQB >= Sq_loc) return;
+  } else {
+    const int bh = blockIdx.y;
+    b = bh / Hq;
+    h = bh % Hq;
+    q_tile = blockIdx.x;
+    q_lo = 0; Sq_loc = Sq; kv_lo = 0; Skv_loc = Skv;
+    causal_off = q_offset;
+  }
   const int hkv = h / (Hq / Hkv);
 
   const int64_t q_base = ((int64_t)b * Sq * Hq + h) * D;
@@ -150,8 +179,8 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
   bf16x8 q_frag[2][kKS];
 #pragma unroll
   for (int m = 0; m < 2; ++m) {
-    const int q_row_global = q_tile * kQB + wave * 32 + m * 16 + (lane & 15);
-    const int safe_row = min(q_row_global, Sq - 1);
+    const int q_row_local = q_tile * kQB + wave * 32 + m * 16 + (lane & 15);
+    const int safe_row = q_lo + min(q_row_local, Sq_loc - 1);
     const bf16_t* qp = q + q_base + (int64_t)safe_row * q_row_stride;
 #pragma unroll
     for (int ks = 0; ks < kKS; ++ks) {
@@ -178,10 +207,11 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
 #pragma unroll
     for (int nt = 0; nt < kNT; ++nt) o_acc[m][nt] = {0.f, 0.f, 0.f, 0.f};
 
-  const int q_tile_last_row = min(q_tile * kQB + kQB - 1, Sq - 1);
-  int kv_end = Skv;
-  if (causal) kv_end = min(Skv, q_tile_last_row + q_offset + 1);
+  const int q_tile_last_row = min(q_tile * kQB + kQB - 1, Sq_loc - 1);
+  int kv_end = Skv_loc;
+  if (causal) kv_end = min(Skv_loc, q_tile_last_row + causal_off + 1);
   const int num_kv_tiles = (kv_end + kKvBlk - 1) / kKvBlk;
+  if (num_kv_tiles <= 0) return;
 
   // T14 split staging: next tile's K rows and V (4x2 transpose blocks) are
   // loaded into registers while the current tile's MFMAs run; LDS writes and
@@ -196,7 +226,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       const int idx = (threadIdx.x + it * 256) * 8;
       const int row = idx / D;
       const int col = idx % D;
-      const int g_row = min(kv0 + row, Skv - 1);
+      const int g_row = kv_lo + min(kv0 + row, Skv_loc - 1);
       k_reg[it] = *reinterpret_cast<const bf16x8*>(
           k + kv_base + (int64_t)g_row * kv_row_stride + col);
     }
@@ -208,7 +238,7 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       v_c0[it].u = v_c1[it].u = 0;
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int g_row = min(kv0 + kvb + i, Skv - 1);
+        const int g_row = kv_lo + min(kv0 + kvb + i, Skv_loc - 1);
         const uint32_t pair = *reinterpret_cast<const uint32_t*>(
             v + kv_base + (int64_t)g_row * kv_row_stride + d0);
         v_c0[it].s[i] = (ushort)(pair & 0xffffu);
@@ -280,9 +310,9 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
         const int col = kv0 + nt * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int row = my_q_row + (lane >> 4) * 4 + r + q_offset;
+          const int row = my_q_row + (lane >> 4) * 4 + r + causal_off;
           float s = s_acc[nt][r] * scale;
-          bool masked = col >= Skv;
+          bool masked = col >= kv_end;
           if (causal) masked |= col > row;
           if (window_left >= 0) masked |= col < row - window_left;
           p_val[nt][r] = masked ? -1e30f : s;
@@ -401,8 +431,9 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
         const int row = idx / D;
         const int col = idx % D;
         const int rg = q_tile * kQB + half * 64 + row;
-        if (rg < Sq) {
-          *reinterpret_cast<bf16x8*>(out + q_base + (int64_t)rg * q_row_stride + col) =
+        if (rg < Sq_loc) {
+          *reinterpret_cast<bf16x8*>(
+              out + q_base + (int64_t)(q_lo + rg) * q_row_stride + col) =
               *reinterpret_cast<const bf16x8*>(o_lds + row * D + col);
         }
       }
@@ -413,8 +444,9 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int rg = q_tile * kQB + wave * 32 + m * 16 + (lane >> 4) * 4 + r;
-          if (rg < Sq) {
-            lse[((int64_t)b * Hq + h) * Sq + rg] =
+          if (rg < Sq_loc) {
+            // regular: (B,Hq,Sq); varlen: (Hq,total) with b == 0
+            lse[((int64_t)b * Hq + h) * Sq + q_lo + rg] =
                 m_run[m][r] + __logf(fmaxf(l_run[m][r], 1e-30f));
           }
         }
@@ -464,6 +496,10 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
     float* __restrict__ dq,           // (B,Sq,Hq,D) fp32 accum
     float* __restrict__ dk,           // (B,Skv,Hkv,D) fp32 accum
     float* __restrict__ dv,           // (B,Skv,Hkv,D) fp32 accum
+    const int* __restrict__ cu_q,     // packed-seq bounds, or nullptr
+    const int* __restrict__ cu_k,
+    const int* __restrict__ kvtile_pref,  // prefix of ceil(len_k/kBwdKv)
+    int nseq,
     int B, int Sq, int Skv, int Hq, int Hkv,
     float scale, int causal, int window_left, int q_offset) {
   constexpr int kNT = D / 16;
@@ -479,24 +515,48 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;   // 8 waves x 16 kv rows = 128 kv/block
-  const int kv_tile = blockIdx.x;
-  const int bh = blockIdx.y;
-  const int b = bh / Hq;
-  const int h = bh % Hq;
+
+  // Varlen: grid.x spans per-sequence kv-tiles; local coordinates within
+  // the sequence, causal offset aligns q end to kv end (see fwd kernel).
+  int b, h, kv_tile, q_lo, Sq_loc, kv_lo, Skv_loc, causal_off;
+  if (cu_q != nullptr) {
+    h = blockIdx.y;
+    int lo = 0, hi = nseq - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (kvtile_pref[mid] <= (int)blockIdx.x) lo = mid; else hi = mid - 1;
+    }
+    b = 0;
+    kv_tile = blockIdx.x - kvtile_pref[lo];
+    q_lo = cu_q[lo];
+    Sq_loc = cu_q[lo + 1] - q_lo;
+    kv_lo = cu_k[lo];
+    Skv_loc = cu_k[lo + 1] - kv_lo;
+    causal_off = Skv_loc - Sq_loc + q_offset;
+    if (Skv_loc <= 0 || Sq_loc <= 0 || kv_tile * kBwdKv >= Skv_loc) return;
+  } else {
+    const int bh = blockIdx.y;
+    b = bh / Hq;
+    h = bh % Hq;
+    kv_tile = blockIdx.x;
+    q_lo = 0; Sq_loc = Sq; kv_lo = 0; Skv_loc = Skv;
+    causal_off = q_offset;
+  }
   const int hkv = h / (Hq / Hkv);
 
   const int64_t q_base = ((int64_t)b * Sq * Hq + h) * D;
   const int64_t kv_base = ((int64_t)b * Skv * Hkv + hkv) * D;
   const int64_t q_row_stride = (int64_t)Hq * D;
   const int64_t kv_row_stride = (int64_t)Hkv * D;
-  const float* lse_row = lse + ((int64_t)b * Hq + h) * Sq;
-  const float* delta_row = delta + ((int64_t)b * Hq + h) * Sq;
+  const float* lse_row = lse + ((int64_t)b * Hq + h) * Sq + q_lo;
+  const float* delta_row = delta + ((int64_t)b * Hq + h) * Sq + q_lo;
 
   const int kv0 = kv_tile * kBwdKv;
 
   // This wave's 16 kv rows: K and V A-fragments in registers.
   const int kv_row_local = lane & 15;
-  const int kv_row_global = min(kv0 + wave * 16 + kv_row_local, Skv - 1);
+  const int kv_row_global =
+      kv_lo + min(kv0 + wave * 16 + kv_row_local, Skv_loc - 1);
   bf16x8 k_frag[kKS], v_frag[kKS];
   {
     const bf16_t* kp = k + kv_base + (int64_t)kv_row_global * kv_row_stride;
@@ -511,7 +571,8 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 
   // Stage K^T once (for dQ's B fragments).
   stage_transposed_tile<D, kBwdKv, kBwdThreads>(
-      kt_lds, k + kv_base, kv_row_stride, kv0, Skv - 1, threadIdx.x);
+      kt_lds, k + kv_base + (int64_t)kv_lo * kv_row_stride, kv_row_stride, kv0,
+      Skv_loc - 1, threadIdx.x);
 
   f32x4 dk_acc[kNT], dv_acc[kNT];
 #pragma unroll
@@ -522,10 +583,10 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 
   int q_start = 0;
   if (causal) {
-    // first q tile whose last GLOBAL position reaches kv0
-    q_start = (max(kv0 - q_offset, 0) / kQBlk) * kQBlk;
+    // first q tile whose last aligned position reaches kv0
+    q_start = (max(kv0 - causal_off, 0) / kQBlk) * kQBlk;
   }
-  if (window_left >= 0) q_start = max(q_start, 0);
+  if (q_start >= Sq_loc) return;
 
   // T14: each q-tile's Q/dO rows are prefetched into registers while the
   // previous tile's MFMAs run; the transposed images are derived from the
@@ -541,7 +602,7 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
       if (idx >= kQBlk * D) break;
       const int row = idx / D;
       const int col = idx % D;
-      const int g_row = min(qt_next + row, Sq - 1);
+      const int g_row = q_lo + min(qt_next + row, Sq_loc - 1);
       q_reg[it] = *reinterpret_cast<const bf16x8*>(
           q + q_base + (int64_t)g_row * q_row_stride + col);
       do_reg[it] = *reinterpret_cast<const bf16x8*>(
@@ -567,8 +628,8 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
   store_qdo_lds();
   __syncthreads();
 
-  for (int qt = q_start; qt < Sq; qt += kQBlk) {
-    if (qt + kQBlk < Sq) load_qdo_regs(qt + kQBlk);  // issue early
+  for (int qt = q_start; qt < Sq_loc; qt += kQBlk) {
+    if (qt + kQBlk < Sq_loc) load_qdo_regs(qt + kQBlk);  // issue early
     // dO^T for dV's B fragments, from the row-major LDS image
     transpose_lds_tile<D, kBwdThreads>(t_lds, do_lds, threadIdx.x);
     __syncthreads();
@@ -597,14 +658,14 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
       const int q_glob = qt + nt * 16 + (lane & 15);
-      const float l = (q_glob < Sq) ? lse_row[min(q_glob, Sq - 1)] : 1e30f;
+      const float l = (q_glob < Sq_loc) ? lse_row[min(q_glob, Sq_loc - 1)] : 1e30f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int kv_glob = kv0 + wave * 16 + (lane >> 4) * 4 + r;
         float s = st_acc[nt][r] * scale;
-        bool masked = (kv_glob >= Skv) || (q_glob >= Sq);
-        if (causal) masked |= kv_glob > q_glob + q_offset;
-        if (window_left >= 0) masked |= kv_glob < q_glob + q_offset - window_left;
+        bool masked = (kv_glob >= Skv_loc) || (q_glob >= Sq_loc);
+        if (causal) masked |= kv_glob > q_glob + causal_off;
+        if (window_left >= 0) masked |= kv_glob < q_glob + causal_off - window_left;
         pt_val[nt][r] =
             masked ? 0.f : __builtin_amdgcn_exp2f((s - l) * kLog2e);
       }
@@ -667,7 +728,7 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
       const int q_glob = qt + nt * 16 + (lane & 15);
-      const float dlt = (q_glob < Sq) ? delta_row[min(q_glob, Sq - 1)] : 0.f;
+      const float dlt = (q_glob < Sq_loc) ? delta_row[min(q_glob, Sq_loc - 1)] : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         dst_val[nt][r] = pt_val[nt][r] * (dpt_acc[nt][r] - dlt) * scale;
@@ -747,10 +808,11 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int q_glob = qt + wave * 16 + (lane >> 4) * 4 + r;
-          if (q_glob < Sq) {
+          if (q_glob < Sq_loc) {
 #pragma unroll
             for (int nt = 0; nt < kNT; ++nt) {
-              atomicAdd(dq + q_base + (int64_t)q_glob * q_row_stride + nt * 16 + (lane & 15),
+              atomicAdd(dq + q_base +
+                            (int64_t)(q_lo + q_glob) * q_row_stride + nt * 16 + (lane & 15),
                         dq_acc[nt][r]);
             }
           }
@@ -758,7 +820,7 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
       }
     }
     __syncthreads();
-    if (qt + kQBlk < Sq) {
+    if (qt + kQBlk < Sq_loc) {
       store_qdo_lds();  // prefetched next Q/dO land after all phases read LDS
     }
     __syncthreads();
@@ -768,12 +830,14 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int kv_glob = kv0 + wave * 16 + (lane >> 4) * 4 + r;
-    if (kv_glob < Skv) {
+    if (kv_glob < Skv_loc) {
 #pragma unroll
       for (int nt = 0; nt < kNT; ++nt) {
         const int d = nt * 16 + (lane & 15);
-        atomicAdd(dk + kv_base + (int64_t)kv_glob * kv_row_stride + d, dk_acc[nt][r]);
-        atomicAdd(dv + kv_base + (int64_t)kv_glob * kv_row_stride + d, dv_acc[nt][r]);
+        atomicAdd(dk + kv_base + (int64_t)(kv_lo + kv_glob) * kv_row_stride + d,
+                  dk_acc[nt][r]);
+        atomicAdd(dv + kv_base + (int64_t)(kv_lo + kv_glob) * kv_row_stride + d,
+                  dv_acc[nt][r]);
       }
     }
   }
@@ -824,13 +888,37 @@ torch::Tensor maybe_pad_d(torch::Tensor t, int D_pad) {
   return padded;
 }
 
+// Per-sequence tile-count prefix for varlen grids (CPU int32 cu_seqlens).
+std::pair<torch::Tensor, int> tile_prefix(torch::Tensor cu_cpu, int tile) {
+  const int n = cu_cpu.numel() - 1;
+  auto pref = torch::empty({n + 1}, torch::dtype(torch::kInt32));
+  const int* cu = cu_cpu.data_ptr<int>();
+  int* p = pref.data_ptr<int>();
+  int total = 0;
+  for (int i = 0; i < n; ++i) {
+    p[i] = total;
+    total += (cu[i + 1] - cu[i] + tile - 1) / tile;
+  }
+  p[n] = total;
+  return {pref, total};
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> flash_attn_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     c10::optional<torch::Tensor> sinks,
+    c10::optional<torch::Tensor> cu_seqlens_q,
+    c10::optional<torch::Tensor> cu_seqlens_k,
     bool causal, double softmax_scale, int64_t window_left, int64_t q_offset) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  const bool varlen = cu_seqlens_q.has_value();
+  if (varlen) {
+    TORCH_CHECK(q.dim() == 3, "varlen expects (total, H, D)");
+    q = q.unsqueeze(0);
+    k = k.unsqueeze(0);
+    v = v.unsqueeze(0);
+  }
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
   const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Skv = k.size(1), Hkv = k.size(2);
@@ -841,8 +929,28 @@ std::vector<torch::Tensor> flash_attn_fwd(
   auto kp = maybe_pad_d(k, D_pad);
   auto vp = maybe_pad_d(v, D_pad);
 
+  torch::Tensor cu_q_gpu, cu_k_gpu, qpref_gpu;
+  const int* cu_q_ptr = nullptr;
+  const int* cu_k_ptr = nullptr;
+  const int* qpref_ptr = nullptr;
+  int nseq = 0, n_qtiles = 0;
+  if (varlen) {
+    auto cu_q_cpu = cu_seqlens_q->to(torch::kInt32).cpu().contiguous();
+    auto cu_k_cpu = cu_seqlens_k->to(torch::kInt32).cpu().contiguous();
+    nseq = cu_q_cpu.numel() - 1;
+    auto [pref, total] = tile_prefix(cu_q_cpu, 128);
+    n_qtiles = total;
+    cu_q_gpu = cu_q_cpu.to(q.device(), true);
+    cu_k_gpu = cu_k_cpu.to(q.device(), true);
+    qpref_gpu = pref.to(q.device(), true);
+    cu_q_ptr = cu_q_gpu.data_ptr<int>();
+    cu_k_ptr = cu_k_gpu.data_ptr<int>();
+    qpref_ptr = qpref_gpu.data_ptr<int>();
+  }
+
   auto out = torch::empty_like(qp);
-  auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+  auto lse = torch::empty({varlen ? 1 : B, Hq, Sq},
+                          q.options().dtype(torch::kFloat32));
   const float* sinks_ptr = nullptr;
   torch::Tensor sinks_f;
   if (sinks.has_value()) {
@@ -851,7 +959,8 @@ std::vector<torch::Tensor> flash_attn_fwd(
     sinks_ptr = sinks_f.data_ptr<float>();
   }
 
-  const dim3 grid((Sq + 127) / 128, B * Hq);
+  const dim3 grid(varlen ? n_qtiles : (Sq + 127) / 128,
+                  varlen ? Hq : B * Hq);
   const size_t smem =
       (size_t)(d9d::kKvBlk * D_pad + D_pad * d9d::kKvBlk + 128 * (d9d::kKvBlk + 8)) *
       sizeof(__bf16);
@@ -864,7 +973,8 @@ std::vector<torch::Tensor> flash_attn_fwd(
                      reinterpret_cast<const __bf16*>(kp.data_ptr()),          \
                      reinterpret_cast<const __bf16*>(vp.data_ptr()),          \
                      reinterpret_cast<__bf16*>(out.data_ptr()),               \
-                     lse.data_ptr<float>(), sinks_ptr, B, Sq, Skv, Hq, Hkv,   \
+                     lse.data_ptr<float>(), sinks_ptr, cu_q_ptr, cu_k_ptr,    \
+                     qpref_ptr, nseq, B, Sq, Skv, Hq, Hkv,                    \
                      (float)softmax_scale, causal ? 1 : 0, (int)window_left,   \
                      (int)q_offset)
   switch (D_pad) {
@@ -876,16 +986,50 @@ std::vector<torch::Tensor> flash_attn_fwd(
 #undef LAUNCH_FWD
 
   if (D_pad != D) out = out.narrow(-1, 0, D).contiguous();
+  if (varlen) {
+    out = out.squeeze(0);
+    lse = lse.squeeze(0);  // (Hq, total)
+  }
   return {out, lse};
 }
 
 std::vector<torch::Tensor> flash_attn_bwd(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor out, torch::Tensor lse,
+    c10::optional<torch::Tensor> cu_seqlens_q,
+    c10::optional<torch::Tensor> cu_seqlens_k,
     bool causal, double softmax_scale, int64_t window_left, int64_t q_offset) {
+  const bool varlen = cu_seqlens_q.has_value();
+  if (varlen) {
+    dout = dout.unsqueeze(0);
+    q = q.unsqueeze(0);
+    k = k.unsqueeze(0);
+    v = v.unsqueeze(0);
+    out = out.unsqueeze(0);
+    lse = lse.unsqueeze(0);
+  }
   const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Skv = k.size(1), Hkv = k.size(2);
   const int D_pad = padded_head_dim(D);
+
+  torch::Tensor cu_q_gpu, cu_k_gpu, kvpref_gpu;
+  const int* cu_q_ptr = nullptr;
+  const int* cu_k_ptr = nullptr;
+  const int* kvpref_ptr = nullptr;
+  int nseq = 0, n_kvtiles = 0;
+  if (varlen) {
+    auto cu_q_cpu = cu_seqlens_q->to(torch::kInt32).cpu().contiguous();
+    auto cu_k_cpu = cu_seqlens_k->to(torch::kInt32).cpu().contiguous();
+    nseq = cu_q_cpu.numel() - 1;
+    auto [pref, total] = tile_prefix(cu_k_cpu, d9d::kBwdKv);
+    n_kvtiles = total;
+    cu_q_gpu = cu_q_cpu.to(q.device(), true);
+    cu_k_gpu = cu_k_cpu.to(q.device(), true);
+    kvpref_gpu = pref.to(q.device(), true);
+    cu_q_ptr = cu_q_gpu.data_ptr<int>();
+    cu_k_ptr = cu_k_gpu.data_ptr<int>();
+    kvpref_ptr = kvpref_gpu.data_ptr<int>();
+  }
 
   auto qp = maybe_pad_d(q, D_pad);
   auto kp = maybe_pad_d(k, D_pad);
@@ -908,7 +1052,8 @@ std::vector<torch::Tensor> flash_attn_bwd(
                        delta.data_ptr<float>(), B, Sq, Hq, D_pad);
   }
 
-  const dim3 grid((Skv + d9d::kBwdKv - 1) / d9d::kBwdKv, B * Hq);
+  const dim3 grid(varlen ? n_kvtiles : (Skv + d9d::kBwdKv - 1) / d9d::kBwdKv,
+                  varlen ? Hq : B * Hq);
   const size_t smem =
       (size_t)(3 * d9d::kQBlk * D_pad + D_pad * d9d::kBwdKv +
                d9d::kBwdKv * (64 + 8)) *
@@ -923,7 +1068,8 @@ std::vector<torch::Tensor> flash_attn_bwd(
                      reinterpret_cast<const __bf16*>(dop.data_ptr()),         \
                      lse.data_ptr<float>(), delta.data_ptr<float>(),          \
                      dq32.data_ptr<float>(), dk32.data_ptr<float>(),          \
-                     dv32.data_ptr<float>(), B, Sq, Skv, Hq, Hkv,             \
+                     dv32.data_ptr<float>(), cu_q_ptr, cu_k_ptr, kvpref_ptr,  \
+                     nseq, B, Sq, Skv, Hq, Hkv,                               \
                      (float)softmax_scale, causal ? 1 : 0, (int)window_left,   \
                      (int)q_offset)
   switch (D_pad) {
@@ -934,10 +1080,15 @@ std::vector<torch::Tensor> flash_attn_bwd(
   }
 #undef LAUNCH_BWD
 
-  auto dq = dq32.narrow(-1, 0, D).to(torch::kBFloat16);
-  auto dk = dk32.narrow(-1, 0, D).to(torch::kBFloat16);
-  auto dv = dv32.narrow(-1, 0, D).to(torch::kBFloat16);
-  return {dq.contiguous(), dk.contiguous(), dv.contiguous()};
+  auto dq = dq32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
+  auto dk = dk32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
+  auto dv = dv32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
+  if (varlen) {
+    dq = dq.squeeze(0);
+    dk = dk.squeeze(0);
+    dv = dv.squeeze(0);
+  }
+  return {dq, dk, dv};
 }
 
 torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b) {

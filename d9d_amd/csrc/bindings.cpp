@@ -20,10 +20,14 @@ void adamw_stochastic_bf16_(torch::Tensor p, torch::Tensor g, torch::Tensor m, t
 // attention.hip
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                            c10::optional<torch::Tensor> sinks,
+                                           c10::optional<torch::Tensor> cu_seqlens_q,
+                                           c10::optional<torch::Tensor> cu_seqlens_k,
                                           bool causal, double softmax_scale, int64_t window_left,
                                           int64_t q_offset);
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, torch::Tensor out, torch::Tensor lse,
+                                          c10::optional<torch::Tensor> cu_seqlens_q,
+                                          c10::optional<torch::Tensor> cu_seqlens_k,
                                           bool causal, double softmax_scale, int64_t window_left,
                                           int64_t q_offset);
 torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b);

@@ -74,7 +74,7 @@ class _FlashAttnFunction(torch.autograd.Function):
     def forward(ctx, q, k, v, sinks, causal, softmax_scale, window_left, q_offset):
         ext = get_ext()
         out, lse = ext.flash_attn_fwd(
-            q.contiguous(), k.contiguous(), v.contiguous(), sinks,
+            q.contiguous(), k.contiguous(), v.contiguous(), sinks, None, None,
             causal, softmax_scale, window_left, q_offset,
         )
         ctx.save_for_backward(q, k, v, out, lse, *( [sinks] if sinks is not None else [] ))
@@ -93,8 +93,8 @@ class _FlashAttnFunction(torch.autograd.Function):
         # dq/dk/dv formulas are unchanged.
         dq, dk, dv = ext.flash_attn_bwd(
             dout.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(),
-            out.contiguous(), lse, ctx.causal, ctx.softmax_scale, ctx.window_left,
-            ctx.q_offset,
+            out.contiguous(), lse, None, None,
+            ctx.causal, ctx.softmax_scale, ctx.window_left, ctx.q_offset,
         )
         dsinks = None
         if ctx.has_sinks:
@@ -131,6 +131,96 @@ def flash_attn_func(
     else:
         out, lse = _eager_attention(
             q, k, v, causal, softmax_scale, window_size, sinks, q_offset
+        )
+    if return_lse:
+        return out, lse
+    return out
+
+
+def _eager_varlen(q, k, v, cu_q, cu_k, causal, softmax_scale, window_size, sinks):
+    """Per-sequence eager oracle for packed (total, H, D) inputs."""
+    outs, lses = [], []
+    for s in range(cu_q.numel() - 1):
+        q_s = q[cu_q[s] : cu_q[s + 1]].unsqueeze(0)
+        k_s = k[cu_k[s] : cu_k[s + 1]].unsqueeze(0)
+        v_s = v[cu_k[s] : cu_k[s + 1]].unsqueeze(0)
+        # causal aligns the END of q with the END of kv
+        off = (cu_k[s + 1] - cu_k[s]) - (cu_q[s + 1] - cu_q[s])
+        o, l = _eager_attention(
+            q_s, k_s, v_s, causal, softmax_scale, window_size, sinks, int(off)
+        )
+        outs.append(o.squeeze(0))
+        lses.append(l.squeeze(0))
+    return torch.cat(outs, dim=0), torch.cat(lses, dim=1)
+
+
+class _FlashAttnVarlenFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, cu_q, cu_k, sinks, causal, softmax_scale, window_left):
+        ext = get_ext()
+        out, lse = ext.flash_attn_fwd(
+            q.contiguous(), k.contiguous(), v.contiguous(), sinks, cu_q, cu_k,
+            causal, softmax_scale, window_left, 0,
+        )
+        to_save = [q, k, v, out, lse, cu_q, cu_k]
+        if sinks is not None:
+            to_save.append(sinks)
+        ctx.save_for_backward(*to_save)
+        ctx.causal = causal
+        ctx.softmax_scale = softmax_scale
+        ctx.window_left = window_left
+        ctx.has_sinks = sinks is not None
+        return out, lse
+
+    @staticmethod
+    def backward(ctx, dout, dlse):
+        q, k, v, out, lse, cu_q, cu_k = ctx.saved_tensors[:7]
+        ext = get_ext()
+        dq, dk, dv = ext.flash_attn_bwd(
+            dout.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(),
+            out.contiguous(), lse, cu_q, cu_k,
+            ctx.causal, ctx.softmax_scale, ctx.window_left, 0,
+        )
+        dsinks = None
+        if ctx.has_sinks:
+            sinks = ctx.saved_tensors[7]
+            delta = (out.float() * dout.float()).sum(-1).permute(1, 0)  # (Hq,total)
+            p_sink = torch.exp(sinks.float().view(-1, 1) - lse)
+            dsinks = -(p_sink * delta).sum(dim=1).to(sinks.dtype)
+        return dq, dk, dv, None, None, dsinks, None, None, None
+
+
+def flash_attn_varlen_func(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cu_seqlens_q: torch.Tensor,
+    cu_seqlens_k: torch.Tensor | None = None,
+    *,
+    causal: bool = True,
+    softmax_scale: float | None = None,
+    window_size: tuple[int, int] = (-1, -1),
+    sinks: torch.Tensor | None = None,
+    return_lse: bool = False,
+):
+    """Packed-sequence attention. q (total_q, Hq, D); k/v (total_k, Hkv, D);
+    cu_seqlens (nseq+1,) int32. lse is (Hq, total_q). Causal masking aligns
+    the end of each sequence's q with the end of its kv (flash-attn
+    convention). Reference: d9d/kernel/flash_attn/function.py varlen path."""
+    if cu_seqlens_k is None:
+        cu_seqlens_k = cu_seqlens_q
+    if softmax_scale is None:
+        softmax_scale = 1.0 / math.sqrt(q.shape[-1])
+    right_ok = window_size[1] < 0 or causal
+    if q.is_cuda and right_ok:
+        out, lse = _FlashAttnVarlenFunction.apply(
+            q, k, v, cu_seqlens_q, cu_seqlens_k, sinks, causal, softmax_scale,
+            window_size[0],
+        )
+    else:
+        out, lse = _eager_varlen(
+            q, k, v, cu_seqlens_q.cpu(), cu_seqlens_k.cpu(), causal,
+            softmax_scale, window_size, sinks,
         )
     if return_lse:
         return out, lse

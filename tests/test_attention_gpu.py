@@ -163,3 +163,59 @@ def test_flash_attention_sinks_matches_eager():
     torch.testing.assert_close(k.grad.float(), k32.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(v.grad.float(), v32.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(sinks.grad, s32.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.gpu
+def test_flash_varlen_matches_eager():
+    from d9d_amd.ops.attention import _eager_varlen, flash_attn_varlen_func
+
+    torch.manual_seed(13)
+    lens_q = [100, 260, 37, 512]
+    lens_k = [100, 260, 37, 512]
+    cu_q = torch.tensor([0] + list(torch.tensor(lens_q).cumsum(0)), dtype=torch.int32)
+    cu_k = torch.tensor([0] + list(torch.tensor(lens_k).cumsum(0)), dtype=torch.int32)
+    Hq, Hkv, D = 8, 2, 64
+    tq, tk = int(cu_q[-1]), int(cu_k[-1])
+    q = torch.randn(tq, Hq, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    k = torch.randn(tk, Hkv, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    v = torch.randn(tk, Hkv, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+
+    out, lse = flash_attn_varlen_func(
+        q, k, v, cu_q.cuda(), cu_k.cuda(), causal=True, return_lse=True
+    )
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    q32 = q.detach().float().requires_grad_(True)
+    k32 = k.detach().float().requires_grad_(True)
+    v32 = v.detach().float().requires_grad_(True)
+    ref, ref_lse = _eager_varlen(q32, k32, v32, cu_q, cu_k, True, D ** -0.5, (-1, -1), None)
+    ref.backward(g.float())
+
+    torch.testing.assert_close(out.float(), ref.float(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(lse, ref_lse, rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(q.grad.float(), q32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), k32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(v.grad.float(), v32.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.gpu
+def test_flash_varlen_cross_attention_lengths():
+    """len_q != len_k: causal aligns sequence ends."""
+    from d9d_amd.ops.attention import _eager_varlen, flash_attn_varlen_func
+
+    torch.manual_seed(14)
+    lens_q = [64, 130]
+    lens_k = [200, 300]
+    cu_q = torch.tensor([0] + list(torch.tensor(lens_q).cumsum(0)), dtype=torch.int32)
+    cu_k = torch.tensor([0] + list(torch.tensor(lens_k).cumsum(0)), dtype=torch.int32)
+    Hq, Hkv, D = 4, 4, 128
+    q = torch.randn(int(cu_q[-1]), Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(int(cu_k[-1]), Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(int(cu_k[-1]), Hkv, D, dtype=torch.bfloat16, device="cuda")
+
+    out = flash_attn_varlen_func(q, k, v, cu_q.cuda(), cu_k.cuda(), causal=True)
+    ref, _ = _eager_varlen(
+        q.float(), k.float(), v.float(), cu_q, cu_k, True, D ** -0.5, (-1, -1), None
+    )
+    torch.testing.assert_close(out.float(), ref.float(), rtol=3e-2, atol=3e-2)
