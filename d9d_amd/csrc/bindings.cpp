@@ -32,6 +32,12 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q, t
                                           int64_t q_offset);
 torch::Tensor mfma_selfcheck(torch::Tensor a, torch::Tensor b);
 
+// moe_router.hip
+std::vector<torch::Tensor> router_topk_fwd(torch::Tensor logits, c10::optional<torch::Tensor> bias,
+                                           int64_t K, bool renormalize);
+torch::Tensor router_topk_bwd(torch::Tensor logits, torch::Tensor top_idx, torch::Tensor dtop,
+                              bool renormalize);
+
 // gmm.hip
 torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes);
 torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes);
@@ -51,6 +57,8 @@ torch::Tensor moe_csr_combine(torch::Tensor expert_out, c10::optional<torch::Ten
 torch::Tensor moe_row_dot(torch::Tensor grad_out, torch::Tensor expert_out, torch::Tensor row_to_token);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("router_topk_fwd", &router_topk_fwd, "fused MoE router fwd");
+  m.def("router_topk_bwd", &router_topk_bwd, "fused MoE router bwd");
   m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
   m.def("gmm_nt", &gmm_nt, "CDNA4 grouped GEMM, weight (E,N,K)");
   m.def("gmm_db", &gmm_db, "CDNA4 grouped GEMM weight-grad");

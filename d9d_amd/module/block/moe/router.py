@@ -8,6 +8,8 @@ the selected k.
 import torch
 from torch import nn
 
+from ....ops import router_topk
+
 
 class TopKRouter(nn.Module):
     def __init__(
@@ -42,13 +44,4 @@ class TopKRouter(nn.Module):
     def forward(self, x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
         """x (T, H) -> (probs (T, k) fp32, indices (T, k) int64)."""
         logits = self.gate(x).float()
-        probs = torch.softmax(logits, dim=-1)
-        if self.expert_bias is not None:
-            select_scores = probs + self.expert_bias
-        else:
-            select_scores = probs
-        _, indices = torch.topk(select_scores, self.top_k, dim=-1)
-        top_probs = probs.gather(-1, indices)
-        if self.renormalize:
-            top_probs = top_probs / top_probs.sum(dim=-1, keepdim=True).clamp_min(1e-20)
-        return top_probs, indices
+        return router_topk(logits, self.expert_bias, self.top_k, self.renormalize)

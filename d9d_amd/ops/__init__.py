@@ -5,9 +5,11 @@ from .attention import flash_attn_func, flash_attn_varlen_func
 from .cce import linear_cross_entropy, LM_IGNORE_INDEX, VocabParallelOptions
 from .gmm import gmm, gmm_nt
 from .moe_permute import moe_permute, moe_unpermute
+from .router import router_topk
 
 __all__ = [
     "rms_norm",
+    "router_topk",
     "silu_mul",
     "silu_mul_packed",
     "copy_fp32_to_bf16_stochastic_",

@@ -26,6 +26,7 @@ _SOURCES = [
     "stochastic.hip",
     "rope.hip",
     "moe_permute.hip",
+    "moe_router.hip",
     "gmm.hip",
     "attention.hip",
     "cce.hip",

@@ -272,3 +272,32 @@ def test_silu_mul_packed_gpu_parity():
     ref.backward(g.float())
     torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(x.grad.float(), x32.grad, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.gpu
+def test_router_topk_gpu_matches_torch():
+    import torch
+
+    from d9d_amd.ops.router import _torch_router, router_topk
+
+    torch.manual_seed(3)
+    for (T, E, K, use_bias, renorm) in [
+        (1000, 128, 8, False, True),
+        (1000, 128, 8, True, True),
+        (257, 100, 4, False, False),
+        (64, 256, 16, True, True),
+    ]:
+        logits = torch.randn(T, E, device="cuda", requires_grad=True)
+        bias = torch.randn(E, device="cuda") * 0.01 if use_bias else None
+        tp, idx = router_topk(logits, bias, K, renorm)
+        g = torch.randn_like(tp)
+        tp.backward(g)
+        got_grad = logits.grad.clone()
+
+        logits2 = logits.detach().clone().requires_grad_(True)
+        rp, ridx = _torch_router(logits2, bias, K, renorm)
+        rp.backward(g)
+
+        assert torch.equal(idx, ridx), (T, E, K)
+        torch.testing.assert_close(tp, rp, rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(got_grad, logits2.grad, rtol=1e-4, atol=1e-6)
