@@ -108,10 +108,17 @@ __global__ void router_topk_fwd_kernel(
     ssum += tp[k];
   }
 
+  // constant-index writes: tp/ti are identical on every lane (shfl
+  // broadcast), so lane 0 writes them all -- a tp[lane] dynamic index would
+  // push the whole array to scratch.
   const float norm = renorm ? 1.f / fmaxf(ssum, 1e-20f) : 1.f;
-  if (lane < K) {
-    top_probs[row * K + lane] = tp[lane] * norm;
-    top_idx[row * K + lane] = ti[lane];
+  if (lane == 0) {
+#pragma unroll
+    for (int k = 0; k < kMaxK; ++k) {
+      if (k >= K) break;
+      top_probs[row * K + k] = tp[k] * norm;
+      top_idx[row * K + k] = ti[k];
+    }
   }
 }
 
@@ -159,27 +166,21 @@ __global__ void router_topk_bwd_kernel(
   // Selected-entry grads. S = sum p_sel; t_k = p_k / S;
   // renorm:  dp_k = (dtop_k - A) / S with A = sum_m dtop_m * t_m
   // no-norm: dp_k = dtop_k
+  // Uniform loop over the k selected entries: ds_bpermute under divergence
+  // returns undefined data from inactive source lanes, so every shfl here
+  // runs with the full wave active. All lanes accumulate identical S/A.
   float S = 0.f, A = 0.f;
-  if (lane < K) {
-    // lane k holds entry k's (idx, p, dtop)
-    const int idx = (int)top_idx[row * K + lane];
-    float p_k = 0.f;
+  for (int k = 0; k < K; ++k) {
+    const int idx = (int)top_idx[row * K + k];
+    float cand = 0.f;
 #pragma unroll
     for (int i = 0; i < kMaxCols; ++i) {
       if (i >= nv) break;
-      const float o = __shfl(v[i], idx & 63, 64);
-      if (i == (idx >> 6)) p_k = o;
+      if (i == (idx >> 6)) cand = v[i];
     }
-    S = p_k;
-    A = dtop[row * K + lane] * p_k;
-  }
-  // (the shfl above reads pre-divergence register values: well-defined on
-  // CDNA where ds_bpermute addresses the register file directly)
-  // reduce S and A over the wave
-#pragma unroll
-  for (int off = 1; off < 64; off <<= 1) {
-    S += __shfl_xor(S, off, 64);
-    A += __shfl_xor(A, off, 64);
+    const float p_k = __shfl(cand, idx & 63, 64);
+    S += p_k;
+    A += dtop[row * K + k] * p_k;
   }
   A /= fmaxf(S, 1e-20f);  // now A = sum dtop_m t_m
 
