@@ -486,7 +486,7 @@ constexpr int kBwdKv = 128;      // kv rows per block (8 waves x 16)
 constexpr int kBwdThreads = 512;
 
 template <int D>
-__global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
+__global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
     const bf16_t* __restrict__ q,
     const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v,
@@ -507,11 +507,20 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
   constexpr int kRowBytes = D * 2;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // Separate buffers for every staged operand: the three MFMA phases
+  // (dV, dK, dQ) then run back-to-back with NO barrier between them, and the
+  // whole q-tile iteration needs 4 __syncthreads instead of 8 (PMC showed
+  // ~50% of bwd wave cycles parked on barriers with the shared-buffer
+  // schedule). 149 KB LDS at D=128 -- one workgroup per CU, which this
+  // kernel is at anyway.
   bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);   // [64][D] swizzled rows
   bf16_t* do_lds = q_lds + kQBlk * D;                // [64][D]
-  bf16_t* t_lds = do_lds + kQBlk * D;                // [D][64]: dO^T then Q^T
-  bf16_t* kt_lds = t_lds + D * kQBlk;                // [D][kBwdKv]
-  bf16_t* x_lds = kt_lds + D * kBwdKv;               // [kBwdKv][64+8] scratch
+  bf16_t* tdo_lds = do_lds + kQBlk * D;              // [D][64]: dO^T
+  bf16_t* tq_lds = tdo_lds + D * kQBlk;              // [D][64]: Q^T
+  bf16_t* kt_lds = tq_lds + D * kQBlk;               // [D][kBwdKv]
+  bf16_t* x1_lds = kt_lds + D * kBwdKv;              // [kBwdKv][64+8]: P^T
+  bf16_t* x2_lds = x1_lds + kBwdKv * (kQBlk + 8);    // [kBwdKv][64+8]: dS^T
+  bf16_t* x3_lds = x2_lds + kBwdKv * (kQBlk + 8);    // [64][kBwdKv+8]: dS
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;   // 8 waves x 16 kv rows = 128 kv/block
@@ -630,8 +639,9 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
 
   for (int qt = q_start; qt < Sq_loc; qt += kQBlk) {
     if (qt + kQBlk < Sq_loc) load_qdo_regs(qt + kQBlk);  // issue early
-    // dO^T for dV's B fragments, from the row-major LDS image
-    transpose_lds_tile<D, kBwdThreads>(t_lds, do_lds, threadIdx.x);
+    // dO^T and Q^T images for the dV/dK B fragments (cheap LDS-to-LDS)
+    transpose_lds_tile<D, kBwdThreads>(tdo_lds, do_lds, threadIdx.x);
+    transpose_lds_tile<D, kBwdThreads>(tq_lds, q_lds, threadIdx.x);
     __syncthreads();
 
     // ---- S^T = K @ Q^T (per wave: 16 kv x 64 q) ----------------------------
@@ -690,39 +700,6 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- dV += P^T @ dO  (A = P^T via x_lds, B = dO^T) --------------------
-    {
-      // write this wave's P^T rows (kv-local) into x_lds [kv 64][q 64+8]
-      bf16_t* xw = x_lds;
-#pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
-          const int q_l = nt * 16 + (lane & 15);
-          xw[kv_l * (kQBlk + 8) + q_l] = (bf16_t)pt_val[nt][r];
-        }
-      }
-      __syncthreads();
-      const bf16_t* xr = x_lds + (wave * 16) * (kQBlk + 8);
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int ks2 = 0; ks2 < 2; ++ks2) {
-        const int q_off = ks2 * 32 + (lane >> 4) * 8;
-        const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-            xr + (lane & 15) * (kQBlk + 8) + q_off);
-#pragma unroll
-        for (int nt = 0; nt < kNT; ++nt) {
-          const int d = nt * 16 + (lane & 15);
-          const int byte = (q_off * 2) ^ ((d & 7) << 4);
-          const bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + byte);
-          dv_acc[nt] = mfma16(pa, dob, dv_acc[nt]);
-        }
-      }
-      __builtin_amdgcn_s_setprio(0);
-    }
-
     // ---- dS^T = P^T * (dP^T - delta[q]) * scale ---------------------------
     float dst_val[4][4];
 #pragma unroll
@@ -735,60 +712,57 @@ __global__ __launch_bounds__(512, 2) void flash_bwd_kernel(
       }
     }
 
-    // ---- dK += dS^T @ Q  (A = dS^T via x_lds, B = Q^T re-staged) ----------
+    // ---- stage P^T, dS^T (kv-major) and dS (q-major), ONE barrier ----------
     {
-      __syncthreads();  // x_lds + t_lds reuse
-      transpose_lds_tile<D, kBwdThreads>(t_lds, q_lds, threadIdx.x);
-      bf16_t* xw = x_lds;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
           const int q_l = nt * 16 + (lane & 15);
-          xw[kv_l * (kQBlk + 8) + q_l] = (bf16_t)dst_val[nt][r];
+          x1_lds[kv_l * (kQBlk + 8) + q_l] = (bf16_t)pt_val[nt][r];
+          x2_lds[kv_l * (kQBlk + 8) + q_l] = (bf16_t)dst_val[nt][r];
+          x3_lds[q_l * (kBwdKv + 8) + kv_l] = (bf16_t)dst_val[nt][r];
         }
       }
       __syncthreads();
-      const bf16_t* xr = x_lds + (wave * 16) * (kQBlk + 8);
+    }
+
+    // ---- dV += P^T @ dO; dK += dS^T @ Q: back-to-back, no barrier ----------
+    {
+      const bf16_t* x1r = x1_lds + (wave * 16) * (kQBlk + 8);
+      const bf16_t* x2r = x2_lds + (wave * 16) * (kQBlk + 8);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         const int q_off = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+            x1r + (lane & 15) * (kQBlk + 8) + q_off);
         const bf16x8 da = *reinterpret_cast<const bf16x8*>(
-            xr + (lane & 15) * (kQBlk + 8) + q_off);
+            x2r + (lane & 15) * (kQBlk + 8) + q_off);
 #pragma unroll
         for (int nt = 0; nt < kNT; ++nt) {
           const int d = nt * 16 + (lane & 15);
           const int byte = (q_off * 2) ^ ((d & 7) << 4);
+          const bf16x8 dob = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(tdo_lds) + d * (kQBlk * 2) + byte);
           const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(t_lds) + d * (kQBlk * 2) + byte);
+              reinterpret_cast<char*>(tq_lds) + d * (kQBlk * 2) + byte);
+          dv_acc[nt] = mfma16(pa, dob, dv_acc[nt]);
           dk_acc[nt] = mfma16(da, qb, dk_acc[nt]);
         }
       }
       __builtin_amdgcn_s_setprio(0);
     }
 
-    // ---- dQ partial = dS @ K; atomicAdd (A = dS q-major via x_lds) --------
+    // ---- dQ partial = dS @ K; atomicAdd (A = dS q-major via x3) -----------
     {
-      __syncthreads();
-      bf16_t* xw = x_lds;  // now [q 64][kv kBwdKv+8]
-#pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
-          const int q_l = nt * 16 + (lane & 15);
-          xw[q_l * (kBwdKv + 8) + kv_l] = (bf16_t)dst_val[nt][r];
-        }
-      }
-      __syncthreads();
       // waves 0-3: one 16-row q m-tile each, K = kBwdKv kv, N = D
       if (wave < 4) {
         f32x4 dq_acc[kNT];
 #pragma unroll
         for (int nt = 0; nt < kNT; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
-        const bf16_t* xr = x_lds + (wave * 16) * (kBwdKv + 8);
+        const bf16_t* xr = x3_lds + (wave * 16) * (kBwdKv + 8);
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks2 = 0; ks2 < kBwdKv / 32; ++ks2) {
@@ -1055,8 +1029,9 @@ std::vector<torch::Tensor> flash_attn_bwd(
   const dim3 grid(varlen ? n_kvtiles : (Skv + d9d::kBwdKv - 1) / d9d::kBwdKv,
                   varlen ? Hq : B * Hq);
   const size_t smem =
-      (size_t)(3 * d9d::kQBlk * D_pad + D_pad * d9d::kBwdKv +
-               d9d::kBwdKv * (64 + 8)) *
+      (size_t)(4 * d9d::kQBlk * D_pad + D_pad * d9d::kBwdKv +
+               2 * d9d::kBwdKv * (d9d::kQBlk + 8) +
+               d9d::kQBlk * (d9d::kBwdKv + 8)) *
       sizeof(__bf16);
 
 #define LAUNCH_BWD(DP)                                                        \
