@@ -5,8 +5,10 @@ beta = sigmoid, GQA head expansion, chunked gated delta rule recurrence,
 per-head RMSNorm and silu-mul output gate.
 
 The chunked delta-rule recurrence (replacing fla-core's CUDA/Triton
-`chunk_gated_delta_rule`) is implemented as a torch chunked scan — matmuls
-land on MFMA via rocBLAS; a fused HIP kernel is the planned follow-up.
+`chunk_gated_delta_rule`) uses the chunked-parallel WY formulation: a
+unit-lower triangular solve plus a handful of batched matmuls per 64-token
+chunk, all landing on MFMA via rocBLAS. A per-timestep reference recurrence
+(`step_gated_delta_rule`) is kept as the numerics oracle.
 """
 
 import math
@@ -42,6 +44,35 @@ class CausalShortDepthwiseConv1d(nn.Module):
         return F.silu(out.transpose(1, 2))
 
 
+def step_gated_delta_rule(
+    q: torch.Tensor,  # (B, H, S, Dk)
+    k: torch.Tensor,  # (B, H, S, Dk)
+    v: torch.Tensor,  # (B, H, S, Dv)
+    beta: torch.Tensor,  # (B, H, S) write strength in [0, 1]
+    decay_log: torch.Tensor,  # (B, H, S) log decay (<= 0)
+) -> torch.Tensor:
+    """Per-timestep reference recurrence (correctness oracle):
+    S_t = gamma_t S_{t-1} + beta_t k_t (v_t - gamma_t k_t^T S_{t-1})^T;
+    o_t = S_t^T q_t."""
+    B, H, S, Dk = q.shape
+    Dv = v.shape[-1]
+    out = torch.zeros(B, H, S, Dv, dtype=torch.float32, device=q.device)
+    state = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+    q32, k32, v32 = q.float(), k.float(), v.float()
+    beta32 = beta.float()
+    g32 = decay_log.float()
+    for t in range(S):
+        kt = k32[:, :, t]  # (B,H,Dk)
+        vt = v32[:, :, t]  # (B,H,Dv)
+        bt = beta32[:, :, t].unsqueeze(-1)  # (B,H,1)
+        gt = g32[:, :, t].exp().unsqueeze(-1).unsqueeze(-1)
+        state = state * gt
+        pred = torch.einsum("bhk,bhkv->bhv", kt, state)
+        state = state + torch.einsum("bhk,bhv->bhkv", kt, bt * (vt - pred))
+        out[:, :, t] = torch.einsum("bhk,bhkv->bhv", q32[:, :, t], state)
+    return out.to(v.dtype)
+
+
 def chunk_gated_delta_rule(
     q: torch.Tensor,  # (B, H, S, Dk)
     k: torch.Tensor,  # (B, H, S, Dk)
@@ -50,34 +81,64 @@ def chunk_gated_delta_rule(
     decay_log: torch.Tensor,  # (B, H, S) log decay (<= 0)
     chunk_size: int = 64,
 ) -> torch.Tensor:
-    """Gated delta rule: S_t = S_{t-1} * exp(g_t) * (I - beta_t k_t k_t^T) +
-    beta_t k_t v_t^T; o_t = q_t @ S_t.
+    """Chunked-parallel gated delta rule (WY form, replacing fla-core's
+    `chunk_gated_delta_rule` CUDA path with MFMA matmuls via rocBLAS).
 
-    Sequential over chunks, exact recurrence within each chunk (the
-    correctness oracle the HIP kernel will be tested against).
-    """
+    Writing r_t = beta_t (v_t - gamma_t k_t^T S_{t-1}) and unrolling inside a
+    chunk with cumulative decays L_t = exp(cumsum g):
+
+        (I + M) R = diag(beta) (V - (L*K) S_0),
+            M[t,j] = beta_t (k_t.k_j) exp(gc_t - gc_j)   (strictly lower)
+        O = (L*Q) S_0 + N R,
+            N[t,j] = (q_t.k_j) exp(gc_t - gc_j)          (incl. diagonal)
+        S_C = exp(gc_C) S_0 + (K * exp(gc_C - gc))^T R
+
+    All decay ratios have t >= j so exp(gc_t - gc_j) <= 1: numerically safe.
+    One unit-lower triangular solve + ~4 matmuls per chunk; chunks run
+    sequentially over S. Matches step_gated_delta_rule to fp32 tolerance."""
     B, H, S, Dk = q.shape
     Dv = v.shape[-1]
-    out = torch.zeros(B, H, S, Dv, dtype=torch.float32, device=q.device)
-    state = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+    C = min(chunk_size, S)
     q32, k32, v32 = q.float(), k.float(), v.float()
     beta32 = beta.float()
     g32 = decay_log.float()
-    for s0 in range(0, S, chunk_size):
-        s1 = min(s0 + chunk_size, S)
-        for t in range(s0, s1):
-            kt = k32[:, :, t]  # (B,H,Dk)
-            vt = v32[:, :, t]  # (B,H,Dv)
-            bt = beta32[:, :, t].unsqueeze(-1)  # (B,H,1)
-            gt = g32[:, :, t].exp().unsqueeze(-1).unsqueeze(-1)
-            state = state * gt
-            # delta update: remove old association along k_t, write new one
-            pred = torch.einsum("bhk,bhkv->bhv", kt, state)
-            state = state + torch.einsum(
-                "bhk,bhv->bhkv", kt, bt * (vt - pred)
-            )
-            out[:, :, t] = torch.einsum("bhk,bhkv->bhv", q32[:, :, t], state)
-    return out.to(v.dtype)
+
+    outs = []
+    state = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+    eye = None
+    for s0 in range(0, S, C):
+        s1 = min(s0 + C, S)
+        c = s1 - s0
+        Qc = q32[:, :, s0:s1]
+        Kc = k32[:, :, s0:s1]
+        Vc = v32[:, :, s0:s1]
+        bc = beta32[:, :, s0:s1]  # (B,H,c)
+        gc = g32[:, :, s0:s1].cumsum(dim=-1)  # (B,H,c) cumulative log decay
+
+        # decay-ratio factor exp(gc_t - gc_j) as an outer difference
+        ratio = torch.exp(gc.unsqueeze(-1) - gc.unsqueeze(-2))  # (B,H,c,c)
+        kk = torch.matmul(Kc, Kc.transpose(-1, -2))  # (B,H,c,c)
+        M = (bc.unsqueeze(-1) * kk * ratio).tril(-1)
+        if eye is None or eye.shape[-1] != c:
+            eye = torch.eye(c, dtype=torch.float32, device=q.device)
+        rhs = bc.unsqueeze(-1) * (
+            Vc - torch.matmul(gc.exp().unsqueeze(-1) * Kc, state)
+        )
+        R = torch.linalg.solve_triangular(
+            M + eye, rhs, upper=False, unitriangular=False
+        )  # (B,H,c,Dv)
+
+        qk = torch.matmul(Qc, Kc.transpose(-1, -2))
+        N = (qk * ratio).tril(0)
+        O = torch.matmul(gc.exp().unsqueeze(-1) * Qc, state) + torch.matmul(N, R)
+        outs.append(O)
+
+        g_tot = gc[:, :, -1].unsqueeze(-1)  # (B,H,1)
+        k_scaled = Kc * torch.exp(g_tot - gc).unsqueeze(-1)
+        state = torch.exp(g_tot).unsqueeze(-1) * state + torch.matmul(
+            k_scaled.transpose(-1, -2), R
+        )
+    return torch.cat(outs, dim=2).to(v.dtype)
 
 
 class LogSigmoidDecayGate(nn.Module):

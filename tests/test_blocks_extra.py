@@ -96,3 +96,40 @@ def test_hidden_states_aggregator():
     assert a.shape == (2, 2, 8)
     torch.testing.assert_close(a[0], h.mean(1))
     assert HiddenStatesAggregator("none").append(None, h) is None
+
+
+def test_chunked_delta_rule_matches_step_oracle():
+    import torch.nn.functional as F
+
+    from d9d_amd.module.block.attention.linear.gated_deltanet import (
+        step_gated_delta_rule,
+    )
+
+    torch.manual_seed(1)
+    B, H, S, Dk, Dv = 2, 3, 130, 16, 24
+    q = F.normalize(torch.randn(B, H, S, Dk), dim=-1)
+    k = F.normalize(torch.randn(B, H, S, Dk), dim=-1)
+    v = torch.randn(B, H, S, Dv)
+    beta = torch.rand(B, H, S)
+    g = -torch.rand(B, H, S) * 0.2
+
+    ref = step_gated_delta_rule(q, k, v, beta, g)
+    for cs in (32, 64, 130):
+        got = chunk_gated_delta_rule(q, k, v, beta, g, chunk_size=cs)
+        torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_chunked_delta_rule_backward_flows():
+    import torch.nn.functional as F
+
+    torch.manual_seed(2)
+    B, H, S, Dk, Dv = 1, 2, 64, 8, 8
+    q = F.normalize(torch.randn(B, H, S, Dk), dim=-1).requires_grad_(True)
+    k = F.normalize(torch.randn(B, H, S, Dk), dim=-1).requires_grad_(True)
+    v = torch.randn(B, H, S, Dv, requires_grad=True)
+    beta = torch.rand(B, H, S, requires_grad=True)
+    g = (-torch.rand(B, H, S) * 0.2).requires_grad_(True)
+    out = chunk_gated_delta_rule(q, k, v, beta, g, chunk_size=16)
+    out.sum().backward()
+    for t in (q, k, v, beta, g):
+        assert t.grad is not None and torch.isfinite(t.grad).all()
