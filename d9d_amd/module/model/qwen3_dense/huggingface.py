@@ -14,19 +14,23 @@ def _vocab_splits(p: Qwen3DenseModelParameters) -> list[tuple[str, int]]:
     return [(name, p.split_vocab_size[name]) for name in p.split_vocab_order]
 
 
-_LAYER_KEYS = (
-    "self_attn.q_proj.weight",
-    "self_attn.k_proj.weight",
-    "self_attn.v_proj.weight",
-    "self_attn.o_proj.weight",
-    "self_attn.q_norm.weight",
-    "self_attn.k_norm.weight",
-    "input_layernorm.weight",
-    "post_attention_layernorm.weight",
-    "mlp.gate_proj.weight",
-    "mlp.up_proj.weight",
-    "mlp.down_proj.weight",
-)
+def _layer_keys(p: Qwen3DenseModelParameters) -> tuple[str, ...]:
+    keys = [
+        "self_attn.q_proj.weight",
+        "self_attn.k_proj.weight",
+        "self_attn.v_proj.weight",
+        "self_attn.o_proj.weight",
+    ]
+    if p.use_qk_norm:
+        keys += ["self_attn.q_norm.weight", "self_attn.k_norm.weight"]
+    keys += [
+        "input_layernorm.weight",
+        "post_attention_layernorm.weight",
+        "mlp.gate_proj.weight",
+        "mlp.up_proj.weight",
+        "mlp.down_proj.weight",
+    ]
+    return tuple(keys)
 
 
 def hf_to_d9d_mapper(p: Qwen3DenseModelParameters) -> ModelStateMapper:
@@ -45,7 +49,7 @@ def hf_to_d9d_mapper(p: Qwen3DenseModelParameters) -> ModelStateMapper:
         Identity("model.norm.weight"),
     ]
     for i in range(p.num_hidden_layers):
-        for key in _LAYER_KEYS:
+        for key in _layer_keys(p):
             mappers.append(Identity(f"model.layers.{i}.{key}"))
     return Parallel(*mappers)
 
@@ -66,6 +70,6 @@ def d9d_to_hf_mapper(p: Qwen3DenseModelParameters) -> ModelStateMapper:
         Identity("model.norm.weight"),
     ]
     for i in range(p.num_hidden_layers):
-        for key in _LAYER_KEYS:
+        for key in _layer_keys(p):
             mappers.append(Identity(f"model.layers.{i}.{key}"))
     return Parallel(*mappers)

@@ -28,6 +28,7 @@ class Qwen3DenseDecoderLayer(nn.Module):
             num_key_value_heads=p.num_key_value_heads,
             head_dim=p.head_dim,
             rms_norm_eps=p.rms_norm_eps,
+            use_qk_norm=p.use_qk_norm,
             **kw,
         )
         self.post_attention_layernorm = RMSNorm(p.hidden_size, eps=p.rms_norm_eps, **kw)

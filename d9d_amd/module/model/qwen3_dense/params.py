@@ -12,6 +12,7 @@ class Qwen3DenseModelParameters:
     head_dim: int = 128
     num_hidden_layers: int = 28
     rms_norm_eps: float = 1e-6
+    use_qk_norm: bool = True
     rope_base: float = 1_000_000.0
     max_position_ids: int = 40_960
     split_vocab_size: dict = field(

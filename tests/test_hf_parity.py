@@ -136,3 +136,54 @@ def test_hf_parity_qwen3_moe():
     back = d9d_to_hf_mapper(p).apply(mapped)
     for k, v in hf.state_dict().items():
         torch.testing.assert_close(back[k], v, rtol=1e-6, atol=1e-7)
+
+
+@pytest.mark.filterwarnings("ignore")
+def test_hf_parity_llama3():
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    from d9d_amd.module.model.llama3 import (
+        Llama3ForCausalLM,
+        Llama3ModelParameters,
+        hf_to_d9d_mapper,
+    )
+
+    p = Llama3ModelParameters(
+        hidden_size=64,
+        intermediate_size=128,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=16,
+        num_hidden_layers=2,
+        rope_base=10000.0,
+        split_vocab_size={"regular": 480, "special": 32},
+    )
+    hf_cfg = LlamaConfig(
+        vocab_size=512,
+        hidden_size=64,
+        intermediate_size=128,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=16,
+        rope_theta=10000.0,
+        rms_norm_eps=1e-6,
+        attention_bias=False,
+        mlp_bias=False,
+        tie_word_embeddings=False,
+    )
+    torch.manual_seed(0)
+    hf = LlamaForCausalLM(hf_cfg).eval()
+
+    ours = Llama3ForCausalLM(p).eval()
+    mapped = hf_to_d9d_mapper(p).apply(dict(hf.state_dict()))
+    missing = set(ours.state_dict()) - set(mapped)
+    assert not missing, f"unmapped keys: {sorted(missing)[:8]}"
+    ours.load_state_dict(mapped)
+
+    ids = torch.randint(0, 512, (2, 24))
+    with torch.no_grad():
+        hf_logits = hf(ids).logits
+        h = ours.model(input_ids=ids)["hidden_states"]
+        our_logits = ours.lm_head.logits(h)
+    assert _kl(hf_logits, our_logits) < 1e-4
