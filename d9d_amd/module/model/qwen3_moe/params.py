@@ -48,3 +48,18 @@ class Qwen3MoEModelParameters:
     def example_pretrain() -> "Qwen3MoEModelParameters":
         """The reference example config (example/qwen3_moe/pretrain.json)."""
         return Qwen3MoEModelParameters()
+
+    @staticmethod
+    def qwen3_30b_a3b() -> "Qwen3MoEModelParameters":
+        """Qwen3-30B-A3B (BASELINE.json config: MoE pretrain, EP=8 over xGMI)."""
+        return Qwen3MoEModelParameters(
+            hidden_size=2048,
+            intermediate_size=768,
+            num_experts=128,
+            top_k=8,
+            num_attention_heads=32,
+            num_key_value_heads=4,
+            head_dim=128,
+            num_hidden_layers=48,
+            split_vocab_size={"regular": 151_643, "special": 26},
+        )
