@@ -56,7 +56,7 @@ class Qwen3MoEModelParameters:
             hidden_size=2048,
             intermediate_size=768,
             num_experts=128,
-            top_k=8,
+            experts_top_k=8,
             num_attention_heads=32,
             num_key_value_heads=4,
             head_dim=128,
