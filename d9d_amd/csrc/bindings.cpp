@@ -38,6 +38,9 @@ std::vector<torch::Tensor> router_topk_fwd(torch::Tensor logits, c10::optional<t
 torch::Tensor router_topk_bwd(torch::Tensor logits, torch::Tensor top_idx, torch::Tensor dtop,
                               bool renormalize);
 
+torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse, torch::Tensor targets,
+                           torch::Tensor dl, int64_t vocab_start, int64_t ignore_index);
+
 // gmm.hip
 torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes);
 torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes);
@@ -59,6 +62,7 @@ torch::Tensor moe_row_dot(torch::Tensor grad_out, torch::Tensor expert_out, torc
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("router_topk_fwd", &router_topk_fwd, "fused MoE router fwd");
   m.def("router_topk_bwd", &router_topk_bwd, "fused MoE router bwd");
+  m.def("cce_dlogits_", &cce_dlogits_, "fused CCE dlogits (in-place)");
   m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
   m.def("gmm_nt", &gmm_nt, "CDNA4 grouped GEMM, weight (E,N,K)");
   m.def("gmm_db", &gmm_db, "CDNA4 grouped GEMM weight-grad");

@@ -253,3 +253,70 @@ std::vector<torch::Tensor> cce_fwd(
       lse.data_ptr<float>(), tgt.data_ptr<float>(), T, V, K);
   return {lse, tgt};
 }
+
+namespace d9d {
+
+// In-place CCE backward dlogits: p = exp(logit - lse) (fp32 math), minus the
+// one-hot target, scaled by the per-row upstream grad -- replaces a 4-pass
+// torch chain (sub, exp_, scatter_add_, mul_) over the (Tc, V) chunk with
+// one read+write.
+__global__ void cce_dlogits_kernel(
+    ushort* __restrict__ logits,       // (R, V) bf16, in/out
+    const float* __restrict__ lse,     // (R,)
+    const int64_t* __restrict__ targets,  // (R,) GLOBAL vocab ids (or ignore)
+    const float* __restrict__ dl,      // (R,) upstream grad (0 for ignored)
+    int64_t R, int64_t V, int64_t vocab_start, int64_t ignore_index) {
+  const int64_t row = blockIdx.y;
+  if (row >= R) return;
+  const float l = lse[row];
+  const float g = dl[row];
+  const int64_t tgt_global = targets[row];
+  const int64_t tgt =
+      (tgt_global == ignore_index) ? -1 : tgt_global - vocab_start;
+  ushort* rowp = logits + row * V;
+
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t base = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+       base < V; base += stride) {
+    if (base + 8 <= V) {
+      Bf16x8 x;
+      x.u = *reinterpret_cast<const ushort8v*>(rowp + base);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float p = __builtin_amdgcn_exp2f(
+            (bf16_bits_to_f32(x.s[j]) - l) * 1.44269504089f);
+        p -= (base + j == tgt) ? 1.f : 0.f;
+        x.s[j] = f32_to_bf16_rne(p * g);
+      }
+      *reinterpret_cast<ushort8v*>(rowp + base) = x.u;
+    } else {
+      for (int64_t i = base; i < V; ++i) {
+        float p = __builtin_amdgcn_exp2f(
+            (bf16_bits_to_f32(rowp[i]) - l) * 1.44269504089f);
+        p -= (i == tgt) ? 1.f : 0.f;
+        rowp[i] = f32_to_bf16_rne(p * g);
+      }
+    }
+  }
+}
+
+}  // namespace d9d
+
+torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse,
+                           torch::Tensor targets, torch::Tensor dl,
+                           int64_t vocab_start, int64_t ignore_index) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kBFloat16 &&
+              logits.is_contiguous());
+  const int64_t R = logits.size(0), V = logits.size(1);
+  if (R == 0) return logits;
+  const int gx = (int)std::min<int64_t>((V + 256 * 8 - 1) / (256 * 8), 128);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::cce_dlogits_kernel, dim3(gx, (unsigned)R), dim3(256),
+                     0, stream,
+                     reinterpret_cast<ushort*>(logits.data_ptr()),
+                     lse.contiguous().data_ptr<float>(),
+                     targets.contiguous().data_ptr<int64_t>(),
+                     dl.contiguous().data_ptr<float>(), R, V, vocab_start,
+                     ignore_index);
+  return logits;
+}

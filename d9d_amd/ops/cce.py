@@ -12,6 +12,8 @@ logsumexp-merge and the target logit summed (exactly one rank owns a target).
 
 from dataclasses import dataclass
 
+from ._ext import get_ext, has_ext
+
 import torch
 import torch.distributed as dist
 
@@ -42,7 +44,6 @@ def _chunk_fwd(e32, c, targets, vocab_start):
 
 def _kernel_forward(e, c, targets, vocab_start):
     """Fused CDNA4 forward: one kernel computes per-token lse + target logit."""
-    from ._ext import get_ext
 
     local_targets = (targets - vocab_start).clamp(min=-1)  # out-of-shard -> -1
     lse, tgt = get_ext().cce_fwd(e.contiguous(), c.contiguous(), local_targets)
@@ -54,7 +55,6 @@ def _kernel_forward(e, c, targets, vocab_start):
 
 
 def _can_use_kernel(e, c):
-    from ._ext import has_ext
 
     return (
         e.is_cuda
@@ -121,19 +121,12 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
             in_shard = (local_targets >= 0) & (local_targets < V) & (~ignored[sl])
             safe = local_targets.clamp(0, V - 1)
             if bf16_fast:
-                # softmax in bf16 (exp in (0, 1], ~3 digits: grads are bf16
-                # downstream anyway) — avoids two fp32 copies of (Tc, V).
-                logits -= lse[sl].unsqueeze(1).to(logits.dtype)
-                p = logits.exp_()
-                p.scatter_add_(
-                    1, safe.unsqueeze(1),
-                    torch.where(
-                        in_shard, -torch.ones_like(safe, dtype=p.dtype),
-                        torch.zeros_like(safe, dtype=p.dtype),
-                    ).unsqueeze(1),
+                # fused in-place dlogits kernel: p = exp(logit - lse) - onehot,
+                # scaled by the row grad, fp32 math, one pass over (Tc, V)
+                pb = get_ext().cce_dlogits_(
+                    logits, lse[sl].float(), targets[sl], dl[sl],
+                    vocab_start, LM_IGNORE_INDEX,
                 )
-                p *= dl[sl].unsqueeze(1).to(p.dtype)
-                pb = p
             else:
                 p = torch.exp(logits.float() - lse[sl].unsqueeze(1))
                 p.scatter_add_(
