@@ -266,8 +266,8 @@ __global__ void cce_dlogits_kernel(
     const int64_t* __restrict__ targets,  // (R,) GLOBAL vocab ids (or ignore)
     const float* __restrict__ dl,      // (R,) upstream grad (0 for ignored)
     int64_t R, int64_t V, int64_t vocab_start, int64_t ignore_index) {
-  const int64_t row = blockIdx.y;
-  if (row >= R) return;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t row = blockIdx.y; row < R; row += gridDim.y) {
   const float l = lse[row];
   const float g = dl[row];
   const int64_t tgt_global = targets[row];
@@ -275,7 +275,6 @@ __global__ void cce_dlogits_kernel(
       (tgt_global == ignore_index) ? -1 : tgt_global - vocab_start;
   ushort* rowp = logits + row * V;
 
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (int64_t base = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
        base < V; base += stride) {
     if (base + 8 <= V) {
@@ -298,6 +297,7 @@ __global__ void cce_dlogits_kernel(
       }
     }
   }
+  }
 }
 
 }  // namespace d9d
@@ -310,8 +310,9 @@ torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse,
   const int64_t R = logits.size(0), V = logits.size(1);
   if (R == 0) return logits;
   const int gx = (int)std::min<int64_t>((V + 256 * 8 - 1) / (256 * 8), 128);
+  const unsigned gy = (unsigned)std::min<int64_t>(R, 4096 / gx + 1);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(d9d::cce_dlogits_kernel, dim3(gx, (unsigned)R), dim3(256),
+  hipLaunchKernelGGL(d9d::cce_dlogits_kernel, dim3(gx, gy), dim3(256),
                      0, stream,
                      reinterpret_cast<ushort*>(logits.data_ptr()),
                      lse.contiguous().data_ptr<float>(),
