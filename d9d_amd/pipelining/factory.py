@@ -19,6 +19,7 @@ from .programs import (
     build_gpipe,
     build_looped_bfs,
     build_zb1p,
+    build_dualpipev,
     build_zbv,
     local_stages,
     loop_stage_to_rank,
@@ -55,8 +56,11 @@ class PipelineScheduleZBVConfig(BaseModel):
 
 
 class PipelineScheduleDualPipeVConfig(BaseModel):
-    """Accepted for config compatibility; currently executes the ZBV ordering
-    (the F+B compose overlap of true DualPipeV is not yet implemented)."""
+    """DualPipeV (DeepSeek bidirectional V schedule): interleaved F/B steady
+    state with a zero-bubble weight-grad ramp in the drain. Requires 2 stages
+    per rank and num_microbatches >= 2*pp. The reference runtime executes its
+    F/B compose pairs sequentially too; communication overlap comes from the
+    batched async P2P layer."""
 
     schedule: Literal["dual_pipe_v"] = "dual_pipe_v"
 
@@ -134,8 +138,10 @@ def build_schedule(
                           zero_bubble=config.zero_bubble)
     elif name == "zb1p":
         prog = build_zb1p(pp_rank, pp_size, num_stages, num_microbatches)
-    elif name in _V_SCHEDULES:
+    elif name == "zero_bubble_v":
         prog = build_zbv(pp_rank, pp_size, num_stages, num_microbatches)
+    elif name == "dual_pipe_v":
+        prog = build_dualpipev(pp_rank, pp_size, num_stages, num_microbatches)
     else:
         raise ValueError(f"unknown schedule {name!r}")
 

@@ -149,3 +149,85 @@ def build_zbv(rank: int, pp: int, num_stages: int, num_microbatches: int) -> Pro
     for wli, wmb in w_queue:
         prog.append(Action(ActionKind.BACKWARD_WEIGHT, wli, wmb))
     return prog
+
+
+def build_dualpipev(rank: int, pp: int, num_stages: int, num_microbatches: int) -> Program:
+    """DualPipeV: bidirectional V schedule with interleaved forward/backward
+    in the steady state and a zero-bubble weight-grad ramp in the drain
+    (DeepSeek DualPipe, V variant; reference: dualpipev.py -- which also
+    executes its F/B "compose" pairs sequentially at runtime, so a flattened
+    F-then-B emission is behaviorally identical while the batched async P2P
+    layer provides the communication overlap).
+
+    Local stage 0 = down leg (global stage `rank`), 1 = up leg (global
+    `2*pp-1-rank`). Requires num_microbatches >= 2*pp.
+    """
+    assert num_stages == 2 * pp, "DualPipeV requires exactly 2 stages per rank"
+    assert num_microbatches >= num_stages, (
+        f"DualPipeV requires num_microbatches ({num_microbatches}) >= "
+        f"num_stages ({num_stages})"
+    )
+    prog: Program = []
+    f = [0, 0]
+    b = [0, 0]
+    w_queue: list[tuple[int, int]] = []
+
+    def add_f(s: int) -> None:
+        prog.append(Action(ActionKind.FORWARD_COMPUTE, s, f[s]))
+        f[s] += 1
+
+    def add_b_full(s: int) -> None:
+        prog.append(Action(ActionKind.BACKWARD_COMPUTE, s, b[s]))
+        b[s] += 1
+
+    def add_b_input(s: int) -> None:
+        prog.append(Action(ActionKind.BACKWARD_INPUT, s, b[s]))
+        w_queue.append((s, b[s]))
+        b[s] += 1
+
+    def pop_w() -> None:
+        if w_queue:
+            ws, wmb = w_queue.pop(0)
+            prog.append(Action(ActionKind.BACKWARD_WEIGHT, ws, wmb))
+
+    # 1: startup on the down leg
+    for _ in range((pp - rank - 1) * 2):
+        add_f(0)
+    # 2: fill both legs
+    for _ in range(rank + 1):
+        add_f(0)
+        add_f(1)
+    # 3: up-leg input-grad / deferred-weight / forward mix
+    for _ in range(pp - rank - 1):
+        add_b_input(1)
+        pop_w()
+        add_f(1)
+    # 4: steady state -- interleave F(down)/B(up) and F(up)/B(down)
+    for _ in range(num_microbatches - 2 * pp + rank + 1):
+        add_f(0)
+        add_b_full(1)
+        add_f(1)
+        add_b_full(0)
+    # 5: cooldown with remaining up-leg forwards
+    for _ in range(pp - rank - 1):
+        add_b_full(1)
+        add_f(1)
+        add_b_full(0)
+    # 6: drain both legs, ramping into input-only backwards (zero bubble)
+    enable_zb = False
+    n6 = rank + 1
+    for i in range(n6):
+        if i == n6 // 2 and rank % 2 == 1:
+            enable_zb = True
+        (add_b_input if enable_zb else add_b_full)(1)
+        if i == n6 // 2 and rank % 2 == 0:
+            enable_zb = True
+        (add_b_input if enable_zb else add_b_full)(0)
+    # 7: deferred weights interleaved with down-leg input backwards
+    for _ in range(pp - rank - 1):
+        pop_w()
+        add_b_input(0)
+    # 8: flush remaining weight grads
+    for _ in range(rank + 1):
+        pop_w()
+    return prog

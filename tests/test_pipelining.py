@@ -197,7 +197,8 @@ def _pp2_schedule_case(rank, world_size, schedule_name, num_stages_per_rank, num
         ("zb1p", 1, 6),
         ("looped_bfs", 2, 8),
         ("zbv", 2, 4),
-        ("dualpipev", 2, 3),
+        ("dualpipev", 2, 4),
+        ("dualpipev", 2, 6),
     ],
 )
 def test_pp2_gradient_exact(schedule_name, stages_per_rank, num_mb):
