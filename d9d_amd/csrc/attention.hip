@@ -305,18 +305,34 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       float m_new[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) m_new[r] = m_run[m][r];
+      // Tile-level mask skip: most KV tiles sit fully below the causal
+      // diagonal and inside the window for every row of this m-tile -- the
+      // per-element mask chain (3 compares + select x16) only runs on the
+      // diagonal/edge tiles (wave-uniform branch).
+      const int row_lo = my_q_row + causal_off;
+      const bool any_mask =
+          (kv0 + kKvBlk > kv_end) ||
+          (causal && kv0 + kKvBlk - 1 > row_lo) ||
+          (window_left >= 0 && kv0 < row_lo + 15 - window_left);
+      if (any_mask) {
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-        const int col = kv0 + nt * 16 + (lane & 15);
+        for (int nt = 0; nt < 4; ++nt) {
+          const int col = kv0 + nt * 16 + (lane & 15);
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = my_q_row + (lane >> 4) * 4 + r + causal_off;
-          float s = s_acc[nt][r] * scale;
-          bool masked = col >= kv_end;
-          if (causal) masked |= col > row;
-          if (window_left >= 0) masked |= col < row - window_left;
-          p_val[nt][r] = masked ? -1e30f : s;
+          for (int r = 0; r < 4; ++r) {
+            const int row = my_q_row + (lane >> 4) * 4 + r + causal_off;
+            float s = s_acc[nt][r] * scale;
+            bool masked = col >= kv_end;
+            if (causal) masked |= col > row;
+            if (window_left >= 0) masked |= col < row - window_left;
+            p_val[nt][r] = masked ? -1e30f : s;
+          }
         }
+      } else {
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) p_val[nt][r] = s_acc[nt][r] * scale;
       }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -664,20 +680,41 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     // ---- P^T = exp(S^T * scale - lse[q]); masked entries 0 -----------------
+    // Tile-level mask skip (see fwd): interior tiles take the select-free
+    // path, only diagonal/edge tiles run the per-element mask chain.
+    const bool any_mask =
+        (kv0 + kBwdKv > Skv_loc) || (qt + kQBlk > Sq_loc) ||
+        (causal && kv0 + kBwdKv - 1 > qt + causal_off) ||
+        (window_left >= 0 &&
+         kv0 < qt + kQBlk - 1 + causal_off - window_left);
     float pt_val[4][4];  // [nt][r]
+    if (any_mask) {
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
-      const int q_glob = qt + nt * 16 + (lane & 15);
-      const float l = (q_glob < Sq_loc) ? lse_row[min(q_glob, Sq_loc - 1)] : 1e30f;
+      for (int nt = 0; nt < 4; ++nt) {
+        const int q_glob = qt + nt * 16 + (lane & 15);
+        const float l = (q_glob < Sq_loc) ? lse_row[min(q_glob, Sq_loc - 1)] : 1e30f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int kv_glob = kv0 + wave * 16 + (lane >> 4) * 4 + r;
-        float s = st_acc[nt][r] * scale;
-        bool masked = (kv_glob >= Skv_loc) || (q_glob >= Sq_loc);
-        if (causal) masked |= kv_glob > q_glob + causal_off;
-        if (window_left >= 0) masked |= kv_glob < q_glob + causal_off - window_left;
-        pt_val[nt][r] =
-            masked ? 0.f : __builtin_amdgcn_exp2f((s - l) * kLog2e);
+        for (int r = 0; r < 4; ++r) {
+          const int kv_glob = kv0 + wave * 16 + (lane >> 4) * 4 + r;
+          float s = st_acc[nt][r] * scale;
+          bool masked = (kv_glob >= Skv_loc) || (q_glob >= Sq_loc);
+          if (causal) masked |= kv_glob > q_glob + causal_off;
+          if (window_left >= 0) masked |= kv_glob < q_glob + causal_off - window_left;
+          pt_val[nt][r] =
+              masked ? 0.f : __builtin_amdgcn_exp2f((s - l) * kLog2e);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        // clamp even on the fast path: the compiler may speculate this load
+        // across the branch, and edge tiles would read past the lse buffer
+        const float l = lse_row[min(qt + nt * 16 + (lane & 15), Sq_loc - 1)];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          pt_val[nt][r] =
+              __builtin_amdgcn_exp2f((st_acc[nt][r] * scale - l) * kLog2e);
+        }
       }
     }
 
