@@ -1,3 +1,8 @@
+from .visualizer import (
+    render_lr_ascii,
+    simulate_lr_history,
+    visualize_lr_scheduler,
+)
 from .piecewise import (
     Curve,
     ConstantCurve,
@@ -20,4 +25,7 @@ __all__ = [
     "Phase",
     "PiecewiseLRScheduler",
     "piecewise_schedule",
+    "simulate_lr_history",
+    "render_lr_ascii",
+    "visualize_lr_scheduler",
 ]

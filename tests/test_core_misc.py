@@ -74,3 +74,27 @@ def test_main_process_stateful():
     o.v = 5
     wrapped.load_state_dict(sd)
     assert o.v == 1
+
+
+def test_lr_visualizer_simulation(tmp_path):
+    from d9d_amd.lr_scheduler import (
+        LinearCurve,
+        Phase,
+        PiecewiseLRScheduler,
+        render_lr_ascii,
+        simulate_lr_history,
+        visualize_lr_scheduler,
+    )
+
+    def factory(opt):
+        return PiecewiseLRScheduler(
+            opt, [Phase(10, LinearCurve(0.0, 1.0)), Phase(10, LinearCurve(1.0, 0.1))]
+        )
+
+    lrs = simulate_lr_history(factory, 20, init_lr=1.0)
+    assert len(lrs) == 20
+    assert max(lrs) <= 1.0 + 1e-6
+    out = visualize_lr_scheduler(factory, 20, csv_path=str(tmp_path / "lr.csv"))
+    assert "lr [" in out
+    assert (tmp_path / "lr.csv").read_text().count("\n") == 21
+    assert render_lr_ascii([]) == ""
