@@ -265,7 +265,8 @@ __global__ void cce_dlogits_kernel(
     const float* __restrict__ lse,     // (R,)
     const int64_t* __restrict__ targets,  // (R,) GLOBAL vocab ids (or ignore)
     const float* __restrict__ dl,      // (R,) upstream grad (0 for ignored)
-    int64_t R, int64_t V, int64_t vocab_start, int64_t ignore_index) {
+    int64_t R, int64_t V, int64_t vocab_start, int64_t ignore_index,
+    float filter_eps) {  // zero non-target dlogits with p < eps (<=0: off)
   const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (int64_t row = blockIdx.y; row < R; row += gridDim.y) {
   const float l = lse[row];
@@ -284,7 +285,9 @@ __global__ void cce_dlogits_kernel(
       for (int j = 0; j < 8; ++j) {
         float p = __builtin_amdgcn_exp2f(
             (bf16_bits_to_f32(x.s[j]) - l) * 1.44269504089f);
-        p -= (base + j == tgt) ? 1.f : 0.f;
+        const bool is_tgt = base + j == tgt;
+        if (p < filter_eps && !is_tgt) p = 0.f;
+        p -= is_tgt ? 1.f : 0.f;
         x.s[j] = f32_to_bf16_rne(p * g);
       }
       *reinterpret_cast<ushort8v*>(rowp + base) = x.u;
@@ -292,7 +295,9 @@ __global__ void cce_dlogits_kernel(
       for (int64_t i = base; i < V; ++i) {
         float p = __builtin_amdgcn_exp2f(
             (bf16_bits_to_f32(rowp[i]) - l) * 1.44269504089f);
-        p -= (i == tgt) ? 1.f : 0.f;
+        const bool is_tgt = i == tgt;
+        if (p < filter_eps && !is_tgt) p = 0.f;
+        p -= is_tgt ? 1.f : 0.f;
         rowp[i] = f32_to_bf16_rne(p * g);
       }
     }
@@ -304,7 +309,8 @@ __global__ void cce_dlogits_kernel(
 
 torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse,
                            torch::Tensor targets, torch::Tensor dl,
-                           int64_t vocab_start, int64_t ignore_index) {
+                           int64_t vocab_start, int64_t ignore_index,
+                           double filter_eps) {
   TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kBFloat16 &&
               logits.is_contiguous());
   const int64_t R = logits.size(0), V = logits.size(1);
@@ -318,6 +324,6 @@ torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse,
                      lse.contiguous().data_ptr<float>(),
                      targets.contiguous().data_ptr<int64_t>(),
                      dl.contiguous().data_ptr<float>(), R, V, vocab_start,
-                     ignore_index);
+                     ignore_index, (float)filter_eps);
   return logits;
 }

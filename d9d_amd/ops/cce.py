@@ -73,7 +73,7 @@ def _can_use_kernel(e, c):
 
 class _LinearCrossEntropyFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, e, c, targets, vp_group, vocab_start, vocab_end):
+    def forward(ctx, e, c, targets, vp_group, vocab_start, vocab_end, filter_eps):
         T = e.shape[0]
         if _can_use_kernel(e, c):
             lse, tgt_logit = _kernel_forward(e, c, targets, vocab_start)
@@ -104,6 +104,7 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
 
         ctx.save_for_backward(e, c, targets, lse)
         ctx.vp = (vp_group, vocab_start, vocab_end)
+        ctx.filter_eps = filter_eps
         return loss
 
     @staticmethod
@@ -134,7 +135,7 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
                 # scaled by the row grad, fp32 math, one pass over (Tc, V)
                 pb = get_ext().cce_dlogits_(
                     logits, lse[sl].float(), targets[sl], dl[sl],
-                    vocab_start, LM_IGNORE_INDEX,
+                    vocab_start, LM_IGNORE_INDEX, ctx.filter_eps,
                 )
             else:
                 p = torch.exp(logits.float() - lse[sl].unsqueeze(1))
@@ -149,10 +150,10 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
                 pb = p.to(c.dtype)
             de[sl] = torch.matmul(pb, c).to(e.dtype)
             if single:
-                return de, torch.matmul(pb.t(), e_chunk.to(c.dtype)), None, None, None, None
+                return de, torch.matmul(pb.t(), e_chunk.to(c.dtype)), None, None, None, None, None
             dc += torch.matmul(pb.t(), e_chunk.to(c.dtype)).float()
 
-        return de, dc.to(c.dtype), None, None, None, None
+        return de, dc.to(c.dtype), None, None, None, None, None
 
 
 def linear_cross_entropy(
@@ -160,11 +161,75 @@ def linear_cross_entropy(
     classifier: torch.Tensor,  # (V_local, H)
     targets: torch.Tensor,  # (T,) int64; LM_IGNORE_INDEX skipped
     vocab_parallel: VocabParallelOptions | None = None,
-) -> torch.Tensor:
-    """Per-token negative log-likelihood, zeros at ignored positions."""
+    *,
+    shift: bool | int = 0,
+    reduction: str = "none",
+    return_lse: bool = False,
+    filter_eps: float | str | None = "auto",
+):
+    """Per-token negative log-likelihood, zeros at ignored positions.
+
+    API mirrors the reference wrapper (d9d/kernel/cce/main.py): `shift` rolls
+    targets left by 1 (or `shift` positions) for causal LMs; `reduction` in
+    {"none", "mean", "sum"} ("mean" averages over non-ignored tokens);
+    `filter_eps` zeroes negligible non-target probabilities in the backward
+    ("auto" = bf16-epsilon-scaled threshold, None/0 disables). `return_lse`
+    additionally returns the per-token logsumexp.
+    """
+    if shift:
+        n = 1 if shift is True else int(shift)
+        embeddings = embeddings[..., :-n, :] if embeddings.dim() == 3 else embeddings[:-n]
+        targets = targets[..., n:] if targets.dim() > 1 else targets[n:]
+    if filter_eps == "auto":
+        eps = 2.0 ** -12  # ~bf16 relative epsilon: drops sub-rounding grads
+    elif filter_eps is None:
+        eps = 0.0
+    else:
+        eps = float(filter_eps)
     if vocab_parallel is None:
-        return _LinearCrossEntropyFunction.apply(embeddings, classifier, targets, None, 0, classifier.shape[0])
-    return _LinearCrossEntropyFunction.apply(
-        embeddings, classifier, targets,
-        vocab_parallel.group, vocab_parallel.vocab_start, vocab_parallel.vocab_end,
-    )
+        loss = _LinearCrossEntropyFunction.apply(
+            embeddings, classifier, targets, None, 0, classifier.shape[0], eps
+        )
+    else:
+        loss = _LinearCrossEntropyFunction.apply(
+            embeddings, classifier, targets,
+            vocab_parallel.group, vocab_parallel.vocab_start,
+            vocab_parallel.vocab_end, eps,
+        )
+    if reduction == "mean":
+        valid = (targets != LM_IGNORE_INDEX).sum().clamp_min(1)
+        loss = loss.sum() / valid
+    elif reduction == "sum":
+        loss = loss.sum()
+    if return_lse:
+        # lse is returned detached (diagnostic use, matching how the
+        # reference's lse output is consumed); the loss path carries grads.
+        with torch.no_grad():
+            if vocab_parallel is None:
+                lse = _lse_only(embeddings, classifier, targets)
+            else:
+                lse = _lse_only(
+                    embeddings, classifier, targets,
+                    vocab_parallel.group, vocab_parallel.vocab_start,
+                )
+        return loss, lse
+    return loss
+
+
+@torch.no_grad()
+def _lse_only(e, c, targets, vp_group=None, vocab_start=0):
+    if _can_use_kernel(e, c):
+        lse, _ = _kernel_forward(e, c, targets, vocab_start)
+    else:
+        parts = []
+        for s in range(0, e.shape[0], _ROW_CHUNK):
+            sl = slice(s, min(s + _ROW_CHUNK, e.shape[0]))
+            lse_c, _ = _chunk_fwd(e[sl], c, targets[sl], vocab_start)
+            parts.append(lse_c)
+        lse = torch.cat(parts)
+    if vp_group is not None:
+        world = dist.get_world_size(vp_group)
+        all_lse = torch.empty(world * lse.numel(), dtype=lse.dtype, device=lse.device)
+        dist.all_gather_into_tensor(all_lse, lse.contiguous(), group=vp_group)
+        lse = torch.logsumexp(all_lse.view(world, -1), dim=0)
+    return lse

@@ -307,3 +307,28 @@ def test_router_topk_gpu_matches_torch():
         torch.testing.assert_close(
             got_grad[same], logits2.grad[same], rtol=1e-4, atol=1e-5
         )
+
+
+def test_cce_api_shift_reduction_lse():
+    import torch
+
+    from d9d_amd.ops.cce import linear_cross_entropy
+
+    torch.manual_seed(4)
+    T, H, V = 10, 16, 32
+    e = torch.randn(T, H)
+    c = torch.randn(V, H) * 0.1
+    tg = torch.randint(0, V, (T,))
+
+    base = linear_cross_entropy(e, c, tg)
+    assert base.shape == (T,)
+    mean = linear_cross_entropy(e, c, tg, reduction="mean")
+    torch.testing.assert_close(mean, base.mean())
+
+    shifted = linear_cross_entropy(e, c, tg, shift=True)
+    ref = linear_cross_entropy(e[:-1], c, tg[1:])
+    torch.testing.assert_close(shifted, ref)
+
+    loss, lse = linear_cross_entropy(e, c, tg, return_lse=True)
+    logits = e @ c.t()
+    torch.testing.assert_close(lse, torch.logsumexp(logits, -1), rtol=1e-4, atol=1e-5)
