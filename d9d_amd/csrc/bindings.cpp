@@ -42,6 +42,12 @@ torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse, torch::Tenso
                            torch::Tensor dl, int64_t vocab_start, int64_t ignore_index,
                            double filter_eps);
 
+void adamw_stochastic_bf16_multi_(
+    std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
+    std::vector<torch::Tensor> exp_avgs, std::vector<torch::Tensor> exp_avg_sqs,
+    double lr, double beta1, double beta2, double eps, double weight_decay,
+    std::vector<int64_t> steps, std::vector<int64_t> seeds);
+
 // gmm.hip
 torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes);
 torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes);
@@ -64,6 +70,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("router_topk_fwd", &router_topk_fwd, "fused MoE router fwd");
   m.def("router_topk_bwd", &router_topk_bwd, "fused MoE router bwd");
   m.def("cce_dlogits_", &cce_dlogits_, "fused CCE dlogits (in-place)");
+  m.def("adamw_stochastic_bf16_multi_", &adamw_stochastic_bf16_multi_, "multi-tensor fused SR-AdamW");
   m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
   m.def("gmm_nt", &gmm_nt, "CDNA4 grouped GEMM, weight (E,N,K)");
   m.def("gmm_db", &gmm_db, "CDNA4 grouped GEMM weight-grad");

@@ -89,6 +89,83 @@ __global__ void adamw_sr_bf16_kernel(
   }
 }
 
+
+// Multi-tensor fused AdamW: one launch for the whole parameter list (the
+// per-tensor variant costs ~360 launches per optimizer step). Slots of 4
+// elements; each slot binary-searches its tensor in the prefix table.
+// Per-slot RNG matches the single-tensor kernel (seed[t] ^ local_slot).
+template <int BLOCK>
+__global__ void adamw_sr_bf16_multi_kernel(
+    const int64_t* __restrict__ meta,  // prefix(N+1) | count(N) | p|g|m|v ptrs (N each) | seed(N)
+    const float* __restrict__ bc,      // (N, 2): bias_corr1, bias_corr2
+    int n_tensors, int64_t total_slots,
+    float lr, float beta1, float beta2, float eps, float weight_decay) {
+  const int64_t* prefix = meta;
+  const int64_t* count = meta + n_tensors + 1;
+  const int64_t* p_ptrs = count + n_tensors;
+  const int64_t* g_ptrs = p_ptrs + n_tensors;
+  const int64_t* m_ptrs = g_ptrs + n_tensors;
+  const int64_t* v_ptrs = m_ptrs + n_tensors;
+  const int64_t* seeds = v_ptrs + n_tensors;
+  const float decay_mul = 1.f - lr * weight_decay;
+
+  const int64_t tid = blockIdx.x * static_cast<int64_t>(BLOCK) + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
+  for (int64_t slot = tid; slot < total_slots; slot += stride) {
+    int lo = 0, hi = n_tensors - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (prefix[mid] <= slot) lo = mid; else hi = mid - 1;
+    }
+    const int t = lo;
+    const int64_t vi = slot - prefix[t];
+    const int64_t n = count[t];
+    const int64_t base = vi * 4;
+    ushort* p = reinterpret_cast<ushort*>(p_ptrs[t]);
+    const ushort* g = reinterpret_cast<const ushort*>(g_ptrs[t]);
+    float* m = reinterpret_cast<float*>(m_ptrs[t]);
+    float* v = reinterpret_cast<float*>(v_ptrs[t]);
+    const float inv_bc1 = 1.f / bc[t * 2];
+    const float inv_sqrt_bc2 = rsqrtf(bc[t * 2 + 1]);
+    const uint64_t r = splitmix64((uint64_t)seeds[t] ^ (uint64_t)vi);
+
+    if (base + 4 <= n) {
+      ushort4v pv = *reinterpret_cast<const ushort4v*>(p + base);
+      const ushort4v gv = *reinterpret_cast<const ushort4v*>(g + base);
+      float4v mv = *reinterpret_cast<const float4v*>(m + base);
+      float4v vv = *reinterpret_cast<const float4v*>(v + base);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float pf = bf16_bits_to_f32(pv[j]) * decay_mul;
+        const float gf = bf16_bits_to_f32(gv[j]);
+        float mf = beta1 * mv[j] + (1.f - beta1) * gf;
+        float vf = beta2 * vv[j] + (1.f - beta2) * gf * gf;
+        pf -= lr * (mf * inv_bc1) / (sqrtf(vf) * inv_sqrt_bc2 + eps);
+        mv[j] = mf;
+        vv[j] = vf;
+        pv[j] = f32_to_bf16_stochastic(pf, static_cast<uint32_t>(r >> (16 * j)));
+      }
+      *reinterpret_cast<ushort4v*>(p + base) = pv;
+      *reinterpret_cast<float4v*>(m + base) = mv;
+      *reinterpret_cast<float4v*>(v + base) = vv;
+    } else {
+      // tail slot: per-element RNG exactly as the single-tensor kernel
+      for (int64_t i = base; i < n; ++i) {
+        float pf = bf16_bits_to_f32(p[i]) * decay_mul;
+        const float gf = bf16_bits_to_f32(g[i]);
+        float mf = beta1 * m[i] + (1.f - beta1) * gf;
+        float vf = beta2 * v[i] + (1.f - beta2) * gf * gf;
+        pf -= lr * (mf * inv_bc1) / (sqrtf(vf) * inv_sqrt_bc2 + eps);
+        m[i] = mf;
+        v[i] = vf;
+        const uint64_t rt = splitmix64(
+            (uint64_t)seeds[t] ^ (0x8000000000000000ull | (uint64_t)i));
+        p[i] = f32_to_bf16_stochastic(pf, static_cast<uint32_t>(rt));
+      }
+    }
+  }
+}
+
 }  // namespace d9d
 
 void copy_fp32_to_bf16_stochastic_(
@@ -136,4 +213,44 @@ void adamw_stochastic_bf16_(
       static_cast<float>(lr), static_cast<float>(beta1),
       static_cast<float>(beta2), static_cast<float>(eps),
       static_cast<float>(weight_decay), bc1, bc2, static_cast<uint64_t>(seed));
+}
+
+
+void adamw_stochastic_bf16_multi_(
+    std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
+    std::vector<torch::Tensor> exp_avgs, std::vector<torch::Tensor> exp_avg_sqs,
+    double lr, double beta1, double beta2, double eps, double weight_decay,
+    std::vector<int64_t> steps, std::vector<int64_t> seeds) {
+  const int n = (int)params.size();
+  if (n == 0) return;
+  auto meta_cpu = torch::empty({(int64_t)n * 7 + 1}, torch::dtype(torch::kInt64));
+  auto bc_cpu = torch::empty({(int64_t)n * 2}, torch::dtype(torch::kFloat32));
+  int64_t* mp = meta_cpu.data_ptr<int64_t>();
+  float* bp = bc_cpu.data_ptr<float>();
+  int64_t slots = 0;
+  for (int i = 0; i < n; ++i) {
+    TORCH_CHECK(params[i].is_cuda() && params[i].scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(grads[i].is_contiguous() && params[i].is_contiguous());
+    const int64_t cnt = params[i].numel();
+    mp[i] = slots;                          // prefix
+    mp[n + 1 + i] = cnt;                    // count
+    mp[2 * n + 1 + i] = (int64_t)params[i].data_ptr();
+    mp[3 * n + 1 + i] = (int64_t)grads[i].data_ptr();
+    mp[4 * n + 1 + i] = (int64_t)exp_avgs[i].data_ptr();
+    mp[5 * n + 1 + i] = (int64_t)exp_avg_sqs[i].data_ptr();
+    mp[6 * n + 1 + i] = seeds[i];
+    bp[i * 2] = 1.f - powf((float)beta1, (float)steps[i]);
+    bp[i * 2 + 1] = 1.f - powf((float)beta2, (float)steps[i]);
+    slots += (cnt + 3) / 4;
+  }
+  mp[n] = slots;
+  auto meta = meta_cpu.to(params[0].device(), /*non_blocking=*/true);
+  auto bc = bc_cpu.to(params[0].device(), /*non_blocking=*/true);
+  constexpr int kBlock = 256;
+  const int grid = (int)std::min<int64_t>((slots + kBlock - 1) / kBlock, 4096);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL((d9d::adamw_sr_bf16_multi_kernel<kBlock>), dim3(grid),
+                     dim3(kBlock), 0, stream, meta.data_ptr<int64_t>(),
+                     bc.data_ptr<float>(), n, slots, (float)lr, (float)beta1,
+                     (float)beta2, (float)eps, (float)weight_decay);
 }

@@ -43,6 +43,10 @@ class StochasticAdamW(torch.optim.Optimizer):
         idx = 0
         for group in self.param_groups:
             beta1, beta2 = group["betas"]
+            # GPU params with identical hyperparams batch into ONE multi-tensor
+            # launch (the per-tensor kernel costs ~360 launches/step on the
+            # bench model); CPU / odd params fall back to the single kernel.
+            batch: list[tuple] = []
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -65,18 +69,41 @@ class StochasticAdamW(torch.optim.Optimizer):
                     + self._step_count * 0x85EBCA6B
                     + idx * 0xC2B2AE35
                 ) & 0x7FFFFFFFFFFFFFFF
-                adamw_stochastic_bf16_(
-                    local_p.view(-1),
-                    local_g.reshape(-1).to(torch.bfloat16),
-                    state["exp_avg"].view(-1),
-                    state["exp_avg_sq"].view(-1),
-                    lr=group["lr"],
-                    beta1=beta1,
-                    beta2=beta2,
-                    eps=group["eps"],
-                    weight_decay=group["weight_decay"],
-                    step=state["step"],
-                    seed=seed,
+                if local_p.is_cuda:
+                    batch.append((
+                        local_p.view(-1),
+                        local_g.reshape(-1).to(torch.bfloat16).contiguous(),
+                        state["exp_avg"].view(-1),
+                        state["exp_avg_sq"].view(-1),
+                        state["step"],
+                        seed,
+                    ))
+                else:
+                    adamw_stochastic_bf16_(
+                        local_p.view(-1),
+                        local_g.reshape(-1).to(torch.bfloat16),
+                        state["exp_avg"].view(-1),
+                        state["exp_avg_sq"].view(-1),
+                        lr=group["lr"],
+                        beta1=beta1,
+                        beta2=beta2,
+                        eps=group["eps"],
+                        weight_decay=group["weight_decay"],
+                        step=state["step"],
+                        seed=seed,
+                    )
+            if batch:
+                from ..ops._ext import get_ext
+
+                get_ext().adamw_stochastic_bf16_multi_(
+                    [b[0] for b in batch],
+                    [b[1] for b in batch],
+                    [b[2] for b in batch],
+                    [b[3] for b in batch],
+                    group["lr"], beta1, beta2, group["eps"],
+                    group["weight_decay"],
+                    [b[4] for b in batch],
+                    [b[5] for b in batch],
                 )
         return loss
 

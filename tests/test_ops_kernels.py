@@ -332,3 +332,34 @@ def test_cce_api_shift_reduction_lse():
     loss, lse = linear_cross_entropy(e, c, tg, return_lse=True)
     logits = e @ c.t()
     torch.testing.assert_close(lse, torch.logsumexp(logits, -1), rtol=1e-4, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_adamw_multi_matches_single():
+    import torch
+
+    from d9d_amd.ops._ext import get_ext
+
+    torch.manual_seed(7)
+    ext = get_ext()
+    shapes = [(1000,), (257,), (64, 64), (3,)]
+    ps = [torch.randn(s, dtype=torch.bfloat16, device="cuda") for s in shapes]
+    gs = [torch.randn(s, dtype=torch.bfloat16, device="cuda") for s in shapes]
+    ms = [torch.randn(s, dtype=torch.float32, device="cuda").abs() for s in shapes]
+    vs = [torch.randn(s, dtype=torch.float32, device="cuda").abs() for s in shapes]
+    ps2 = [p.clone() for p in ps]
+    ms2 = [m.clone() for m in ms]
+    vs2 = [v.clone() for v in vs]
+    seeds = [11, 22, 33, 44]
+    for i in range(4):
+        ext.adamw_stochastic_bf16_(
+            ps[i].view(-1), gs[i].contiguous().view(-1), ms[i].view(-1),
+            vs[i].view(-1), 1e-3, 0.9, 0.95, 1e-8, 0.01, 3, seeds[i])
+    ext.adamw_stochastic_bf16_multi_(
+        [p.view(-1) for p in ps2], [g.contiguous().view(-1) for g in gs],
+        [m.view(-1) for m in ms2], [v.view(-1) for v in vs2],
+        1e-3, 0.9, 0.95, 1e-8, 0.01, [3, 3, 3, 3], seeds)
+    for i in range(4):
+        assert torch.equal(ps[i], ps2[i]), i  # bitwise: same per-slot RNG
+        torch.testing.assert_close(ms[i], ms2[i])
+        torch.testing.assert_close(vs[i], vs2[i])
