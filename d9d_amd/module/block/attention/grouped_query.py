@@ -10,6 +10,8 @@ import math
 import torch
 from torch import nn
 
+from ..linear import KernelLinear
+
 from ....ops import flash_attn_func
 from ..normalization import RMSNorm
 from ..positional import RopeLayout, apply_rotary_emb
@@ -44,10 +46,10 @@ class GroupedQueryAttention(nn.Module):
 
         kw = {"device": device, "dtype": dtype, "bias": False}
         q_out = num_attention_heads * head_dim
-        self.q_proj = nn.Linear(hidden_size, q_out * (2 if use_output_gate else 1), **kw)
-        self.k_proj = nn.Linear(hidden_size, num_key_value_heads * head_dim, **kw)
-        self.v_proj = nn.Linear(hidden_size, num_key_value_heads * head_dim, **kw)
-        self.o_proj = nn.Linear(q_out, hidden_size, **kw)
+        self.q_proj = KernelLinear(hidden_size, q_out * (2 if use_output_gate else 1), **kw)
+        self.k_proj = KernelLinear(hidden_size, num_key_value_heads * head_dim, **kw)
+        self.v_proj = KernelLinear(hidden_size, num_key_value_heads * head_dim, **kw)
+        self.o_proj = KernelLinear(q_out, hidden_size, **kw)
         self.use_output_gate = use_output_gate
 
         if use_qk_norm:

@@ -5,6 +5,8 @@ import math
 import torch
 from torch import nn
 
+from ..linear import KernelLinear
+
 from ....ops import silu_mul
 
 
@@ -20,9 +22,9 @@ class SwiGLU(nn.Module):
         self.hidden_size = hidden_size
         self.intermediate_size = intermediate_size
         kw = {"device": device, "dtype": dtype, "bias": False}
-        self.gate_proj = nn.Linear(hidden_size, intermediate_size, **kw)
-        self.up_proj = nn.Linear(hidden_size, intermediate_size, **kw)
-        self.down_proj = nn.Linear(intermediate_size, hidden_size, **kw)
+        self.gate_proj = KernelLinear(hidden_size, intermediate_size, **kw)
+        self.up_proj = KernelLinear(hidden_size, intermediate_size, **kw)
+        self.down_proj = KernelLinear(intermediate_size, hidden_size, **kw)
 
     def reset_parameters(self) -> None:
         with torch.no_grad():
