@@ -576,7 +576,7 @@ __global__ __launch_bounds__(512, 1) void gmm_db_kernel(
 
 // ---------------------------------------------------------------------------
 
-static std::pair<torch::Tensor, torch::Tensor> build_offsets(
+static std::tuple<torch::Tensor, torch::Tensor, int> build_offsets(
     torch::Tensor batch_sizes, torch::Device device, int tile_m) {
   const int E = batch_sizes.numel();
   auto row_off = torch::empty({E + 1}, torch::dtype(torch::kInt32));
@@ -594,8 +594,11 @@ static std::pair<torch::Tensor, torch::Tensor> build_offsets(
   }
   ro[E] = rows;
   mp[E] = tiles;
+  // total tile count returned from the CPU copy: reading it from the device
+  // tensor (`mtile_pref[E].item()`) is a D2H sync on EVERY call, which
+  // serializes the launch pipeline (measured ~0.5 ms/call end-to-end).
   return {row_off.to(device, /*non_blocking=*/true),
-          mtile_pref.to(device, /*non_blocking=*/true)};
+          mtile_pref.to(device, /*non_blocking=*/true), tiles};
 }
 
 torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
@@ -608,9 +611,8 @@ torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
 
   auto out = torch::empty({(int64_t)T, (int64_t)N}, a.options());
   if (T == 0) return out;
-  auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
-  const int total_mtiles = ((int64_t)T + d9d::kBM * E) > 0
-      ? mtile_pref[E].item<int>() : 0;
+  auto [row_off, mtile_pref, total_mtiles] =
+      build_offsets(batch_sizes, a.device(), d9d::kBM);
   if (total_mtiles == 0) return out;
 
   const int n_tiles = (N + d9d::kBN - 1) / d9d::kBN;
@@ -637,8 +639,8 @@ torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes
 
   auto out = torch::empty({(int64_t)T, (int64_t)N}, a.options());
   if (T == 0) return out;
-  auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
-  const int total_mtiles = mtile_pref[E].item<int>();
+  auto [row_off, mtile_pref, total_mtiles] =
+      build_offsets(batch_sizes, a.device(), d9d::kBM);
   if (total_mtiles == 0) return out;
 
   const int n_tiles = (N + d9d::kBN - 1) / d9d::kBN;
@@ -663,7 +665,9 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
 
   auto db = torch::zeros({(int64_t)E, (int64_t)K, (int64_t)N}, a.options());
   if (a.size(0) == 0) return db;
-  auto [row_off, mtile_pref] = build_offsets(batch_sizes, a.device(), d9d::kBM);
+  auto [row_off, mtile_pref, total_mtiles_unused] =
+      build_offsets(batch_sizes, a.device(), d9d::kBM);
+  (void)total_mtiles_unused;
   auto order_cpu = torch::argsort(batch_sizes.to(torch::kInt64), /*dim=*/0,
                                   /*descending=*/true).to(torch::kInt32);
   auto expert_order = order_cpu.to(a.device(), /*non_blocking=*/true);

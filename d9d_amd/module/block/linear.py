@@ -9,11 +9,18 @@ other case (bias, CPU, exotic dtype, missing extension) falls back to the
 stock matmul path.
 """
 
+import os
+
 import torch
 from torch import nn
 
 from ...ops import gmm_nt
 from ...ops._ext import has_ext
+
+# Opt-in: the NT kernel beats rocBLAS ~2x at the projection shapes in
+# isolation, but per-call host overhead (group-offset build + H2D) cancels
+# the win end-to-end at bench scale; measured A/B kept rocBLAS as default.
+_ENABLED = bool(int(os.environ.get("D9D_KERNEL_LINEAR", "0")))
 
 
 class KernelLinear(nn.Linear):
@@ -22,7 +29,8 @@ class KernelLinear(nn.Linear):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         w = self.weight
         if (
-            x.is_cuda
+            _ENABLED
+            and x.is_cuda
             and self.bias is None
             and x.dtype == torch.bfloat16
             and w.dtype == torch.bfloat16

@@ -104,10 +104,20 @@ class _GroupedLinearNTFunction(torch.autograd.Function):
         a, w, batch_sizes = ctx.saved_tensors
         g = g.contiguous()
         da = dw = None
+        single = w.shape[0] == 1
         if ctx.needs_input_grad[0] and GLOBAL_GRAD_CONTEXT.computes(GradDirection.INPUTS):
-            da = _gmm_forward_impl(g, w, batch_sizes)
+            if single and a.is_cuda:
+                # dense (E=1) dgrad: one big NN GEMM -- rocBLAS is fine here
+                da = torch.matmul(g, w[0])
+            else:
+                da = _gmm_forward_impl(g, w, batch_sizes)
         if ctx.needs_input_grad[1] and GLOBAL_GRAD_CONTEXT.computes(GradDirection.WEIGHTS):
-            if a.is_cuda and has_ext() and hasattr(get_ext(), "gmm_db"):
+            if single and a.is_cuda:
+                # gmm_db's expert-tile grid collapses to ~24 workgroups at
+                # E=1 (it parallelizes over experts); the TN GEMM via rocBLAS
+                # keeps the chip full instead
+                dw = torch.matmul(g.t(), a).unsqueeze(0)
+            elif a.is_cuda and has_ext() and hasattr(get_ext(), "gmm_db"):
                 dw = get_ext().gmm_db(g, a.contiguous(), batch_sizes, w.shape[0])
             else:
                 dw = _gmm_accum_db(g, a, batch_sizes, w.shape[0])
