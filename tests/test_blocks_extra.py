@@ -133,3 +133,35 @@ def test_chunked_delta_rule_backward_flows():
     out.sum().backward()
     for t in (q, k, v, beta, g):
         assert t.grad is not None and torch.isfinite(t.grad).all()
+
+
+import pytest
+
+
+@pytest.mark.gpu
+def test_gated_deltanet_module_gpu():
+    torch.manual_seed(5)
+    m = GatedDeltaNet(64, num_heads=4, head_k_dim=16, head_v_dim=16).cuda().bfloat16()
+    m.reset_parameters()
+    x = torch.randn(2, 128, 64, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = m(x)
+    assert out.shape == (2, 128, 64)
+    out.float().sum().backward()
+    assert x.grad is not None and torch.isfinite(x.grad.float()).all()
+
+
+@pytest.mark.gpu
+def test_mla_module_gpu():
+    torch.manual_seed(6)
+    mla = MultiHeadLatentAttention(
+        64, 4, qk_nope_head_dim=16, qk_rope_head_dim=8, v_head_dim=16,
+        kv_lora_rank=32, q_lora_rank=24,
+    ).cuda().bfloat16()
+    mla.reset_parameters()
+    prov = RotaryEmbeddingProvider(rope_dim=8)
+    pos = torch.arange(32).unsqueeze(0).expand(2, 32).cuda()
+    x = torch.randn(2, 32, 64, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    cos_sin = prov(pos)
+    out = mla(x, cos_sin)
+    out.float().sum().backward()
+    assert x.grad is not None
