@@ -125,9 +125,30 @@ def flash_attn_func(
         softmax_scale = 1.0 / math.sqrt(q.shape[-1])
     right_ok = window_size[1] < 0 or causal  # right window only via causality
     if q.is_cuda and right_ok:
-        out, lse = _FlashAttnFunction.apply(
-            q, k, v, sinks, causal, softmax_scale, window_size[0], q_offset
-        )
+        Dq, Dv = q.shape[-1], v.shape[-1]
+        if Dv != Dq:
+            # asymmetric head dims (e.g. MLA: qk 24, v 16): zero-pad the
+            # smaller side -- zero V columns add nothing to PV, zero q/k
+            # columns add nothing to the dots -- and slice the output.
+            # F.pad keeps the whole thing differentiable.
+            import torch.nn.functional as F
+
+            if Dv < Dq:
+                v_in = F.pad(v, (0, Dq - Dv))
+                out, lse = _FlashAttnFunction.apply(
+                    q, k, v_in, sinks, causal, softmax_scale, window_size[0], q_offset
+                )
+                out = out[..., :Dv]
+            else:
+                q_in = F.pad(q, (0, Dv - Dq))
+                k_in = F.pad(k, (0, Dv - Dq))
+                out, lse = _FlashAttnFunction.apply(
+                    q_in, k_in, v, sinks, causal, softmax_scale, window_size[0], q_offset
+                )
+        else:
+            out, lse = _FlashAttnFunction.apply(
+                q, k, v, sinks, causal, softmax_scale, window_size[0], q_offset
+            )
     else:
         out, lse = _eager_attention(
             q, k, v, causal, softmax_scale, window_size, sinks, q_offset

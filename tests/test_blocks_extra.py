@@ -158,7 +158,7 @@ def test_mla_module_gpu():
         kv_lora_rank=32, q_lora_rank=24,
     ).cuda().bfloat16()
     mla.reset_parameters()
-    prov = RotaryEmbeddingProvider(rope_dim=8)
+    prov = RotaryEmbeddingProvider(rope_dim=8).cuda()
     pos = torch.arange(32).unsqueeze(0).expand(2, 32).cuda()
     x = torch.randn(2, 32, 64, device="cuda", dtype=torch.bfloat16, requires_grad=True)
     cos_sin = prov(pos)
