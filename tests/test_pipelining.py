@@ -209,3 +209,79 @@ def test_pp2_gradient_exact(schedule_name, stages_per_rank, num_mb):
     covered = set(results[0]) | set(results[1])
     num_stages = 2 * stages_per_rank
     assert len(covered) == num_stages * 2  # weight+bias per 1-layer stage
+
+
+def _pp2_dp2_case(rank, world_size):
+    """Composed PP(2) x DP(2) on 4 gloo ranks: 1F1B over the pp dim, replicated
+    stage weights + bucketed grad all-reduce over the dp dim; gradients must
+    match the sequential single-process run over ALL ranks' data."""
+    import torch.distributed as dist
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import DTensor
+
+    from d9d_amd.internals.grad_sync import GradientSynchronizer
+    from d9d_amd.parallel import parallelize_replicate
+    from d9d_amd.pipelining.factory import (
+        PipelineSchedule1F1BConfig,
+        build_schedule,
+    )
+
+    mesh = init_device_mesh("cpu", (2, 2), mesh_dim_names=("pp", "dp"))
+    pp_rank = mesh.get_coordinate()[0]
+    dp_rank = mesh.get_coordinate()[1]
+    pp_group = mesh["pp"].get_group()
+    dp_mesh = mesh["dp"]
+
+    seed, num_mb, num_stages = 31, 4, 2
+    torch.manual_seed(500 + dp_rank)  # different data per dp rank
+    x = torch.randn(4 * num_mb, 16)
+    target = torch.randn(4 * num_mb, 16)
+
+    info = build_schedule(
+        PipelineSchedule1F1BConfig(),
+        _make_provider(seed, num_stages, layers_per_stage=1),
+        num_microbatches=num_mb,
+        device=torch.device("cpu"),
+        pp_rank=pp_rank,
+        pp_size=2,
+        pp_group=pp_group,
+    )
+    for module in info.modules:
+        parallelize_replicate(module, dp_mesh)
+    sync = GradientSynchronizer(
+        [(n, p) for m in info.modules for n, p in m.named_parameters()],
+        accumulation_steps=num_mb,
+        bucket_bytes=1 << 20,
+    )
+    info.schedule.configure_buffers({"x": x, "target": target})
+    info.schedule.step({"x": x, "target": target}, loss_fn=_loss_fn_factory())
+    sync.wait()
+
+    # reference: sequential model over BOTH dp ranks' datasets, summed grads
+    ref_grads: dict = {}
+    for r in range(2):
+        torch.manual_seed(500 + r)
+        xr = torch.randn(4 * num_mb, 16)
+        tr = torch.randn(4 * num_mb, 16)
+        for s0 in range(num_mb):
+            sl = slice(s0 * 4, s0 * 4 + 4)
+            _, g = _sequential_reference(seed, num_stages, xr[sl], tr[sl], layers_per_stage=1)
+            for k, v in g.items():
+                ref_grads[k] = ref_grads.get(k, 0) + v
+
+    for local_idx, module in enumerate(info.modules):
+        gidx = info.stages[local_idx].stage_index
+        for n, p in module.named_parameters():
+            grad = p.grad.to_local() if isinstance(p.grad, DTensor) else p.grad
+            torch.testing.assert_close(
+                grad, ref_grads[f"s{gidx}.{n}"], rtol=1e-4, atol=1e-5,
+                msg=lambda m: f"stage {gidx} {n}: {m}",
+            )
+    sync.remove()
+    return True
+
+
+@pytest.mark.distributed
+def test_pp2_dp2_composed_gradient_exact():
+    results = run_distributed(_pp2_dp2_case, world_size=4)
+    assert all(results)
