@@ -161,10 +161,19 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     causal_off = Skv_loc - Sq_loc + q_offset;
     if (Sq_loc <= 0 || q_tile * kQB >= Sq_loc) return;
   } else {
-    const int bh = blockIdx.y;
+    // XCD-aware mapping: group one (b, h)'s q-tiles onto one XCD so the
+    // KV stream they share is served from that XCD's L2 instead of every
+    // XCD re-streaming it from HBM (see gmm_nt_kernel for the bijection).
+    const int nwg = gridDim.x * gridDim.y;
+    const int lin = blockIdx.y * gridDim.x + blockIdx.x;
+    const int xcd = lin & 7;
+    const int qd = nwg >> 3, rd = nwg & 7;
+    const int wid = (xcd < rd ? xcd * (qd + 1) : rd * (qd + 1) + (xcd - rd) * qd)
+                    + (lin >> 3);
+    const int bh = wid / gridDim.x;
     b = bh / Hq;
     h = bh % Hq;
-    q_tile = blockIdx.x;
+    q_tile = wid % gridDim.x;
     q_lo = 0; Sq_loc = Sq; kv_lo = 0; Skv_loc = Skv;
     causal_off = q_offset;
   }
@@ -560,10 +569,18 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
     causal_off = Skv_loc - Sq_loc + q_offset;
     if (Skv_loc <= 0 || Sq_loc <= 0 || kv_tile * kBwdKv >= Skv_loc) return;
   } else {
-    const int bh = blockIdx.y;
+    // XCD-aware mapping (see fwd): one (b, h)'s kv-tiles share its Q/dO/lse
+    // streams through a single XCD's L2.
+    const int nwg = gridDim.x * gridDim.y;
+    const int lin = blockIdx.y * gridDim.x + blockIdx.x;
+    const int xcd = lin & 7;
+    const int qd = nwg >> 3, rd = nwg & 7;
+    const int wid = (xcd < rd ? xcd * (qd + 1) : rd * (qd + 1) + (xcd - rd) * qd)
+                    + (lin >> 3);
+    const int bh = wid / gridDim.x;
     b = bh / Hq;
     h = bh % Hq;
-    kv_tile = blockIdx.x;
+    kv_tile = wid % gridDim.x;
     q_lo = 0; Sq_loc = Sq; kv_lo = 0; Skv_loc = Skv;
     causal_off = q_offset;
   }
