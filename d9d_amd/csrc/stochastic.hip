@@ -96,10 +96,18 @@ __global__ void adamw_sr_bf16_kernel(
 // Per-slot RNG matches the single-tensor kernel (seed[t] ^ local_slot).
 template <int BLOCK>
 __global__ void adamw_sr_bf16_multi_kernel(
-    const int64_t* __restrict__ meta,  // prefix(N+1) | count(N) | p|g|m|v ptrs (N each) | seed(N)
-    const float* __restrict__ bc,      // (N, 2): bias_corr1, bias_corr2
+    const int64_t* __restrict__ meta_g,  // prefix(N+1) | count(N) | p|g|m|v ptrs (N each) | seed(N)
+    const float* __restrict__ bc_g,      // (N, 2): bias_corr1, bias_corr2
     int n_tensors, int64_t total_slots,
     float lr, float beta1, float beta2, float eps, float weight_decay) {
+  // Stage the lookup tables in LDS: the per-slot binary search is a chain of
+  // dependent loads, and from global memory it dominates the whole kernel.
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  int64_t* meta = reinterpret_cast<int64_t*>(smem_raw);
+  float* bc = reinterpret_cast<float*>(meta + 7 * n_tensors + 1);
+  for (int i = threadIdx.x; i < 7 * n_tensors + 1; i += BLOCK) meta[i] = meta_g[i];
+  for (int i = threadIdx.x; i < 2 * n_tensors; i += BLOCK) bc[i] = bc_g[i];
+  __syncthreads();
   const int64_t* prefix = meta;
   const int64_t* count = meta + n_tensors + 1;
   const int64_t* p_ptrs = count + n_tensors;
@@ -248,9 +256,11 @@ void adamw_stochastic_bf16_multi_(
   auto bc = bc_cpu.to(params[0].device(), /*non_blocking=*/true);
   constexpr int kBlock = 256;
   const int grid = (int)std::min<int64_t>((slots + kBlock - 1) / kBlock, 4096);
+  const size_t smem = (7 * (size_t)n + 1) * 8 + 2 * (size_t)n * 4;
+  TORCH_CHECK(smem <= 64 * 1024, "adamw multi: too many tensors for LDS");
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL((d9d::adamw_sr_bf16_multi_kernel<kBlock>), dim3(grid),
-                     dim3(kBlock), 0, stream, meta.data_ptr<int64_t>(),
+                     dim3(kBlock), smem, stream, meta.data_ptr<int64_t>(),
                      bc.data_ptr<float>(), n, slots, (float)lr, (float)beta1,
                      (float)beta2, (float)eps, (float)weight_decay);
 }
