@@ -98,3 +98,22 @@ def test_lr_visualizer_simulation(tmp_path):
     assert "lr [" in out
     assert (tmp_path / "lr.csv").read_text().count("\n") == 21
     assert render_lr_ascii([]) == ""
+
+
+def test_auroc_matches_sklearn():
+    import numpy as np
+    import torch
+    from sklearn.metrics import roc_auc_score
+
+    from d9d_amd.metric import BinaryAUROC
+
+    rng = np.random.default_rng(0)
+    scores = rng.random(500).astype("float32")
+    labels = (rng.random(500) < scores).astype("int64")  # correlated labels
+
+    m = BinaryAUROC(num_bins=4096)
+    m.update(torch.from_numpy(scores), torch.from_numpy(labels))
+    m.sync()
+    got = float(m.compute())
+    ref = roc_auc_score(labels, scores)
+    assert abs(got - ref) < 5e-3, (got, ref)
