@@ -32,6 +32,11 @@ D9D_DEVICE f32x4 mfma16c(bf16x8 a, bf16x8 b, f32x4 c) {
 
 constexpr float kLog2eC = 1.44269504088896340736f;
 
+// KT = k-tile width (128 when K % 128 == 0): wider k-tiles double the MFMA
+// work between barriers (16 vs 8 per wave); the classifier tile is
+// single-buffered with T14 register staging (the registers are the second
+// buffer), like the grouped-GEMM kernels.
+template <int KT>
 __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
     const bf16_t* __restrict__ e,   // (T, K)
     const bf16_t* __restrict__ c,   // (V, K)
@@ -41,7 +46,8 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
     int T, int V, int K) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* e_lds = reinterpret_cast<bf16_t*>(smem);       // [64][K] swizzled
-  bf16_t* c_lds = e_lds + 64 * K;                        // 2 x [128][64]
+  bf16_t* c_lds = e_lds + 64 * K;                        // [128][KT] swizzled
+  constexpr int kSwz = (KT == 64) ? 7 : 15;  // XOR stays within the KT*2 row
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;   // 8 waves: (vocab half) x (4 m-tiles)
@@ -49,7 +55,39 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
   const int vhalf = wave >> 2;
   const int t0 = blockIdx.x * 64;
 
-  // ---- stage e block (row-major [64][K], ((row&15)<<4) swizzle) ------------
+  const int n_vtiles = (V + 127) / 128;  // 128-wide vocab tiles (2 x 64 halves)
+  const int n_ktiles = K / KT;
+  const int total_iters = n_vtiles * n_ktiles;
+
+  constexpr int kCRegs = KT / 32;  // (128 * KT) / (512 threads * 8)
+  bf16x8 c_reg[kCRegs];
+  auto load_c = [&](int it_lin) {
+    const int vt = it_lin / n_ktiles;
+    const int kt = it_lin % n_ktiles;
+#pragma unroll
+    for (int i = 0; i < kCRegs; ++i) {
+      const int idx = (threadIdx.x + i * 512) * 8;
+      const int vrow = idx / KT;
+      const int col = idx % KT;
+      const int g_v = min(vt * 128 + vrow, V - 1);
+      c_reg[i] = *reinterpret_cast<const bf16x8*>(
+          c + (int64_t)g_v * K + kt * KT + col);
+    }
+  };
+  auto store_c = [&]() {
+#pragma unroll
+    for (int i = 0; i < kCRegs; ++i) {
+      const int idx = (threadIdx.x + i * 512) * 8;
+      const int vrow = idx / KT;
+      const int col = idx % KT;
+      const int byte = (col * 2) ^ ((vrow & kSwz) << 4);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(c_lds) + vrow * (KT * 2) + byte) = c_reg[i];
+    }
+  };
+
+  // ---- stage e block (row-major [64][K], ((row&15)<<4) swizzle) and the
+  // first classifier tile; ONE barrier covers both ---------------------------
   for (int idx = threadIdx.x * 8; idx < 64 * K; idx += 512 * 8) {
     const int row = idx / K;
     const int col = idx % K;
@@ -59,6 +97,8 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
     *reinterpret_cast<bf16x8*>(
         reinterpret_cast<char*>(e_lds) + row * (K * 2) + byte) = val;
   }
+  load_c(0);
+  store_c();
   __syncthreads();
 
   const int my_row_local = m_tile * 16 + (lane & 15);
@@ -71,42 +111,6 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
     tgt[r] = -1e30f;
   }
 
-  const int n_vtiles = (V + 127) / 128;  // 128-wide vocab tiles (2 x 64 halves)
-  const int n_ktiles = K / 64;  // K % 64 == 0 enforced by the host
-
-  // T14 c staging: per (vtile, ktile) iteration each thread owns 2 bf16x8.
-  bf16x8 c_reg[2];
-  auto load_c = [&](int it_lin) {
-    const int vt = it_lin / n_ktiles;
-    const int kt = it_lin % n_ktiles;
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const int idx = (threadIdx.x + i * 512) * 8;
-      const int vrow = idx / 64;  // vocab row within the 128-wide tile
-      const int col = idx % 64;
-      const int g_v = min(vt * 128 + vrow, V - 1);
-      c_reg[i] = *reinterpret_cast<const bf16x8*>(
-          c + (int64_t)g_v * K + kt * 64 + col);
-    }
-  };
-  auto store_c = [&](int buf) {
-    bf16_t* cl = c_lds + buf * 128 * 64;
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const int idx = (threadIdx.x + i * 512) * 8;
-      const int vrow = idx / 64;
-      const int col = idx % 64;
-      const int byte = (col * 2) ^ ((vrow & 7) << 4);
-      *reinterpret_cast<bf16x8*>(
-          reinterpret_cast<char*>(cl) + vrow * 128 + byte) = c_reg[i];
-    }
-  };
-
-  const int total_iters = n_vtiles * n_ktiles;
-  load_c(0);
-  store_c(0);
-  __syncthreads();
-
   f32x4 acc[4];
 #pragma unroll
   for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
@@ -114,25 +118,26 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
   for (int it = 0; it < total_iters; ++it) {
     const int vt = it / n_ktiles;
     const int kt = it % n_ktiles;
-    const int buf = it & 1;
-    if (it + 1 < total_iters) load_c(it + 1);
+    if (it + 1 < total_iters) load_c(it + 1);  // issue early, store late
 
-    const bf16_t* cl = c_lds + buf * 128 * 64;
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      const int kk = kt * 64 + ks * 32 + (lane >> 4) * 8;
+    for (int ks = 0; ks < KT / 32; ++ks) {
+      const int kk = kt * KT + ks * 32 + (lane >> 4) * 8;
       const int ebyte = (kk * 2) ^ ((my_row_local & 15) << 4);
       const bf16x8 ea = *reinterpret_cast<const bf16x8*>(
           reinterpret_cast<char*>(e_lds) + my_row_local * (K * 2) + ebyte);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         const int vrow = vhalf * 64 + nt * 16 + (lane & 15);
-        const int cbyte = ((ks * 32 + (lane >> 4) * 8) * 2) ^ ((vrow & 7) << 4);
+        const int cbyte =
+            ((ks * 32 + (lane >> 4) * 8) * 2) ^ ((vrow & kSwz) << 4);
         const bf16x8 cb = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(cl) + vrow * 128 + cbyte);
+            reinterpret_cast<const char*>(c_lds) + vrow * (KT * 2) + cbyte);
         acc[nt] = mfma16c(ea, cb, acc[nt]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     if (kt == n_ktiles - 1) {
       // logits tile complete: online lse + target capture, reset acc.
@@ -192,9 +197,9 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
 
     __syncthreads();
     if (it + 1 < total_iters) {
-      store_c(buf ^ 1);
+      store_c();
+      __syncthreads();
     }
-    __syncthreads();
   }
 
   // ---- merge the two vocab halves, write lse + target logit ---------------
@@ -238,19 +243,26 @@ std::vector<torch::Tensor> cce_fwd(
   TORCH_CHECK(targets.scalar_type() == torch::kInt64);
   const int T = e.size(0), K = e.size(1), V = c.size(0);
   TORCH_CHECK(K % 64 == 0, "cce_fwd requires hidden % 64 == 0");
-  const size_t smem = (size_t)(64 * K + 2 * 128 * 64) * sizeof(__bf16);
+  const bool wide = (K % 128 == 0) &&
+      (size_t)(64 * K + 128 * 128) * sizeof(__bf16) <= 160 * 1024;
+  const size_t smem =
+      (size_t)(64 * K + 128 * (wide ? 128 : 64)) * sizeof(__bf16);
   TORCH_CHECK(smem <= 160 * 1024, "cce_fwd: hidden too large for LDS: ", K);
 
   auto lse = torch::empty({T}, e.options().dtype(torch::kFloat32));
   auto tgt = torch::empty({T}, e.options().dtype(torch::kFloat32));
   if (T == 0) return {lse, tgt};
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(
-      d9d::cce_fwd_kernel, dim3((T + 63) / 64), dim3(512), smem, stream,
-      reinterpret_cast<const __bf16*>(e.data_ptr()),
-      reinterpret_cast<const __bf16*>(c.data_ptr()),
-      targets.contiguous().data_ptr<int64_t>(),
-      lse.data_ptr<float>(), tgt.data_ptr<float>(), T, V, K);
+  auto tgts = targets.contiguous();
+#define LAUNCH_CCE(KT)                                                        \
+  hipLaunchKernelGGL(d9d::cce_fwd_kernel<KT>, dim3((T + 63) / 64), dim3(512), \
+                     smem, stream,                                            \
+                     reinterpret_cast<const __bf16*>(e.data_ptr()),           \
+                     reinterpret_cast<const __bf16*>(c.data_ptr()),           \
+                     tgts.data_ptr<int64_t>(), lse.data_ptr<float>(),         \
+                     tgt.data_ptr<float>(), T, V, K)
+  if (wide) LAUNCH_CCE(128); else LAUNCH_CCE(64);
+#undef LAUNCH_CCE
   return {lse, tgt};
 }
 
