@@ -45,6 +45,7 @@ class PipelineScheduleLoopedBFSConfig(BaseModel):
 class PipelineSchedule1F1BConfig(BaseModel):
     schedule: Literal["1f1b"] = "1f1b"
     zero_bubble: bool = False
+    num_stages_per_rank: int = 1  # > 1: interleaved (Megatron virtual pipeline)
 
 
 class PipelineScheduleZB1PConfig(BaseModel):

@@ -63,7 +63,8 @@ def build_1f1b(rank: int, pp: int, num_stages: int, num_microbatches: int,
     """Non-interleaved 1F1B (num_stages == pp). Warmup = pp-1-rank forwards,
     then steady 1F1B, then drain. zero_bubble: split B into B_in at B's slot
     and B_w deferred to the tail (ZB1P, reference: interleaved.py)."""
-    assert num_stages == pp, "1F1B requires one stage per rank (use looped_bfs otherwise)"
+    if num_stages != pp:
+        return build_interleaved_1f1b(rank, pp, num_stages, num_microbatches, zero_bubble)
     warmup = min(pp - 1 - rank, num_microbatches)
     prog: Program = []
     b_kind = ActionKind.BACKWARD_INPUT if zero_bubble else ActionKind.BACKWARD_COMPUTE
@@ -230,4 +231,68 @@ def build_dualpipev(rank: int, pp: int, num_stages: int, num_microbatches: int) 
     # 8: flush remaining weight grads
     for _ in range(rank + 1):
         pop_w()
+    return prog
+
+
+def build_interleaved_1f1b(rank: int, pp: int, num_stages: int,
+                           num_microbatches: int, zero_bubble: bool = False) -> Program:
+    """Interleaved 1F1B (Megatron-style virtual pipeline; reference:
+    interleaved.py Interleaved1F1BPipelineProgramBuilder). v = num_stages/pp
+    chunks per rank on a LOOP topology (rank r owns global stages r, r+pp,
+    ...; local chunk index == position in that list). Forward walks chunks in
+    pp-sized rounds, backward walks them reversed; warmup depth
+    (pp-1-rank)*2 + (v-1)*pp keeps every rank one chunk ahead.
+
+    zero_bubble splits each backward into input-grad at its slot + deferred
+    weight-grad at the tail."""
+    assert num_stages % pp == 0
+    v = num_stages // pp
+    total = num_microbatches * v
+    # the round-robin chunk walk needs whole pp-rounds; microbatch counts that
+    # are not multiples of pp would interleave incorrectly
+    assert num_microbatches % pp == 0 or v == 1, (
+        "interleaved 1F1B needs num_microbatches % pp == 0"
+    )
+
+    def fwd_unit(i: int) -> tuple[int, int]:
+        chunk = (i % (pp * v)) // pp
+        round_base = (i // (pp * v)) * pp
+        mb = round_base + i % pp
+        return chunk, mb
+
+    def bwd_unit(j: int) -> tuple[int, int]:
+        chunk = v - 1 - (j % (pp * v)) // pp
+        round_base = (j // (pp * v)) * pp
+        mb = round_base + j % pp
+        return chunk, mb
+
+    b_kind = ActionKind.BACKWARD_INPUT if zero_bubble else ActionKind.BACKWARD_COMPUTE
+    prog: Program = []
+    w_queue: list[tuple[int, int]] = []
+
+    def emit_f(i: int) -> None:
+        chunk, mb = fwd_unit(i)
+        prog.append(Action(ActionKind.FORWARD_COMPUTE, chunk, mb))
+
+    def emit_b(j: int) -> None:
+        chunk, mb = bwd_unit(j)
+        prog.append(Action(b_kind, chunk, mb))
+        if zero_bubble:
+            w_queue.append((chunk, mb))
+
+    warmup = min((pp - rank - 1) * 2 + (v - 1) * pp, total)
+    f = b = 0
+    for _ in range(warmup):
+        emit_f(f)
+        f += 1
+    while f < total:
+        emit_f(f)
+        f += 1
+        emit_b(b)
+        b += 1
+    while b < total:
+        emit_b(b)
+        b += 1
+    for chunk, mb in w_queue:
+        prog.append(Action(ActionKind.BACKWARD_WEIGHT, chunk, mb))
     return prog

@@ -145,6 +145,12 @@ def _pp2_schedule_case(rank, world_size, schedule_name, num_stages_per_rank, num
         "looped_bfs": PipelineScheduleLoopedBFSConfig(num_stages_per_rank=num_stages_per_rank),
         "1f1b": PipelineSchedule1F1BConfig(),
         "1f1b_zb": PipelineSchedule1F1BConfig(zero_bubble=True),
+        "1f1b_interleaved": PipelineSchedule1F1BConfig(
+            num_stages_per_rank=num_stages_per_rank
+        ),
+        "1f1b_interleaved_zb": PipelineSchedule1F1BConfig(
+            num_stages_per_rank=num_stages_per_rank, zero_bubble=True
+        ),
         "zb1p": PipelineScheduleZB1PConfig(),
         "zbv": PipelineScheduleZBVConfig(),
         "dualpipev": PipelineScheduleDualPipeVConfig(),
@@ -199,6 +205,8 @@ def _pp2_schedule_case(rank, world_size, schedule_name, num_stages_per_rank, num
         ("zbv", 2, 4),
         ("dualpipev", 2, 4),
         ("dualpipev", 2, 6),
+        ("1f1b_interleaved", 2, 4),
+        ("1f1b_interleaved_zb", 2, 6),
     ],
 )
 def test_pp2_gradient_exact(schedule_name, stages_per_rank, num_mb):
