@@ -155,3 +155,56 @@ def _dp2_trainer(rank, world_size):
 def test_trainer_dp2_replicate_consistent():
     results = run_distributed(_dp2_trainer, world_size=2, timeout=300)
     assert results[0] == results[1]
+
+
+def test_inference_configurator_runs():
+    import torch
+
+    from d9d_amd.core.dist_context import DeviceMeshParameters
+    from d9d_amd.loop.config import BatchingConfig, TrainerConfig
+    from d9d_amd.loop.control import DatasetProvider, InferenceTask, ModelProvider
+    from d9d_amd.loop.inference import InferenceConfigurator
+    from d9d_amd.module.model.qwen3_dense import (
+        Qwen3DenseForCausalLM,
+        Qwen3DenseModelParameters,
+    )
+
+    p = Qwen3DenseModelParameters.tiny()
+
+    class Provider(ModelProvider):
+        def initialize_model_stage(self, stage_info):
+            return Qwen3DenseForCausalLM(p, stage_info)
+
+        def parallelize_model_stage(self, module, ctx):
+            return module
+
+    class DS(torch.utils.data.Dataset):
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            g = torch.Generator().manual_seed(i)
+            ids = torch.randint(0, p.vocab_size, (12,), generator=g)
+            return {"input_ids": ids[:-1], "labels": ids[1:]}
+
+    class DP(DatasetProvider):
+        def build_dataset(self, ctx):
+            return DS()
+
+    class Task(InferenceTask):
+        def build_forward_inputs(self, batch):
+            return {"input_ids": batch["input_ids"], "labels": batch["labels"]}
+
+        def process_outputs(self, outputs, mb_inputs):
+            return outputs["logps"].detach()
+
+    cfg = TrainerConfig(
+        batching=BatchingConfig(global_batch_size=4, microbatch_size=2),
+        total_steps=1,
+    )
+    runner = InferenceConfigurator(
+        cfg, DeviceMeshParameters(), Provider(), DP(), Task()
+    ).configure()
+    results = runner.run()
+    assert len(results) >= 2  # 8 rows / batch 4 -> 2 batches, >= 1 mb each
+    assert all(torch.isfinite(r).all() for r in results)
