@@ -91,6 +91,7 @@ class GroupedQueryAttention(nn.Module):
         self,
         hidden_states: torch.Tensor,  # (B, S, H)
         rotary_cos_sin: tuple[torch.Tensor, torch.Tensor],
+        cu_seqlens: torch.Tensor | None = None,  # packed documents (B == 1)
     ) -> torch.Tensor:
         B, S, _ = hidden_states.shape
         q = self.q_proj(hidden_states)
@@ -116,6 +117,20 @@ class GroupedQueryAttention(nn.Module):
             k = apply_rotary_emb(k, cos, sin, self.rope_layout)
 
         window = (-1, -1) if self.sliding_window is None else (self.sliding_window, -1)
+        if cu_seqlens is not None:
+            # packed documents: attention is causal WITHIN each document
+            # (varlen path; rotary positions must already restart per doc)
+            assert B == 1, "packed cu_seqlens input expects batch dim 1"
+            from ....ops import flash_attn_varlen_func
+
+            attn = flash_attn_varlen_func(
+                q.squeeze(0), k.squeeze(0), v.squeeze(0), cu_seqlens,
+                causal=True, window_size=window, sinks=self.sinks,
+            ).unsqueeze(0)
+            attn = attn.reshape(B, S, self.num_heads * self.head_dim)
+            if self.use_output_gate:
+                attn = attn * torch.sigmoid(gate)
+            return self.o_proj(attn)
         q_offset = 0
         if self._cp_group is not None and self._cp_size > 1:
             from ....parallel.tensor import _AllGatherSeq

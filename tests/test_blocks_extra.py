@@ -165,3 +165,28 @@ def test_mla_module_gpu():
     out = mla(x, cos_sin)
     out.float().sum().backward()
     assert x.grad is not None
+
+
+def test_gqa_packed_documents_matches_per_doc():
+    """cu_seqlens path == running each document separately."""
+    from d9d_amd.module.block.attention import GroupedQueryAttention
+
+    torch.manual_seed(9)
+    attn = GroupedQueryAttention(32, 4, 2, 8, use_qk_norm=False)
+    attn.reset_parameters()
+    prov = RotaryEmbeddingProvider(rope_dim=8)
+    lens = [5, 9, 3]
+    cu = torch.tensor([0, 5, 14, 17], dtype=torch.int32)
+    xs = [torch.randn(1, n, 32) for n in lens]
+    packed = torch.cat(xs, dim=1)
+
+    # rotary positions restart per document
+    pos = torch.cat([torch.arange(n) for n in lens]).unsqueeze(0)
+    out_packed = attn(packed, prov(pos), cu_seqlens=cu)
+
+    outs = []
+    for x, n in zip(xs, lens):
+        p = torch.arange(n).unsqueeze(0)
+        outs.append(attn(x, prov(p)))
+    ref = torch.cat(outs, dim=1)
+    torch.testing.assert_close(out_packed, ref, rtol=1e-4, atol=1e-5)

@@ -117,3 +117,24 @@ def test_auroc_matches_sklearn():
     got = float(m.compute())
     ref = roc_auc_score(labels, scores)
     assert abs(got - ref) < 5e-3, (got, ref)
+
+
+def test_pack_documents():
+    import torch
+
+    from d9d_amd.dataset import PackedDocumentDataset, pack_documents
+
+    docs = [torch.arange(n) for n in (5, 9, 3, 12, 2, 40)]
+    packs = list(pack_documents(docs, tokens_per_pack=16))
+    total = sum(int(p["cu_seqlens"][-1]) for p in packs)
+    assert total == 5 + 9 + 3 + 12 + 2  # 40-token doc dropped
+    for p in packs:
+        assert int(p["cu_seqlens"][-1]) <= 16
+        assert len(p["input_ids"]) == int(p["cu_seqlens"][-1])
+        # positions restart at each boundary
+        for s, e in zip(p["cu_seqlens"][:-1], p["cu_seqlens"][1:]):
+            seg = p["position_ids"][s:e]
+            assert seg[0] == 0 and (seg == torch.arange(e - s)).all()
+
+    ds = PackedDocumentDataset([{"input_ids": d} for d in docs], 16)
+    assert sum(1 for _ in ds) == len(packs)
