@@ -118,7 +118,9 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
         ignored = targets == LM_IGNORE_INDEX
         dl = torch.where(ignored, torch.zeros_like(dloss), dloss).float()
 
-        bf16_fast = e.dtype == torch.bfloat16 and c.dtype == torch.bfloat16
+        bf16_fast = (
+            e.is_cuda and e.dtype == torch.bfloat16 and c.dtype == torch.bfloat16
+        )
         chunk = _ROW_CHUNK
         single = chunk >= T
         if not single:

@@ -52,7 +52,8 @@ class Qwen3MoEProvider(ModelProvider):
         self.params = params
 
     def initialize_model_stage(self, stage_info):
-        return Qwen3MoEForCausalLM(self.params, stage_info)
+        # StochasticAdamW requires bf16 parameters (fp32 moments + SR writes)
+        return Qwen3MoEForCausalLM(self.params, stage_info).to(torch.bfloat16)
 
     def parallelize_model_stage(self, module, ctx):
         if not ctx.is_distributed:
@@ -90,7 +91,12 @@ def main():
     import os
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
-    params = Qwen3MoEModelParameters.example_pretrain()
+    # full pretrain config on GPUs; tiny config for CPU smoke runs
+    params = (
+        Qwen3MoEModelParameters.example_pretrain()
+        if torch.cuda.is_available()
+        else Qwen3MoEModelParameters.tiny()
+    )
     mesh = DeviceMeshParameters(
         data_parallel_replicate=world,
         expert_parallel=world if world > 1 else 1,
