@@ -14,8 +14,12 @@ from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
 
 def _apply_sum_reduction(module: nn.Module) -> None:
     # FSDP2 defaults to averaging over the shard group; d9d semantics are SUM
-    # (the trainer divides by the accumulated loss weight itself).
-    state = getattr(module, "_get_fsdp_state", None)
+    # (the trainer divides by the accumulated loss weight itself). Forcing
+    # plain-SUM comms makes the divide factor a host-side scale instead of an
+    # NCCL premul-sum reduce op, which gloo (the CPU test backend) lacks.
+    set_force = getattr(module, "set_force_sum_reduction_for_comms", None)
+    if callable(set_force):
+        set_force(True)
     set_factor = getattr(module, "set_gradient_divide_factor", None)
     if callable(set_factor):
         set_factor(1.0)

@@ -225,3 +225,78 @@ def _vocab_parallel_cce(rank, world_size):
 @pytest.mark.distributed
 def test_vocab_parallel_cce_ws2():
     assert all(run_distributed(_vocab_parallel_cce, world_size=2))
+
+
+def _fsdp_grad_matches_replicated(rank, world_size):
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import DTensor
+
+    from d9d_amd.parallel import parallelize_fsdp
+
+    torch.manual_seed(21)
+    model = nn.Sequential(nn.Linear(16, 32), nn.Tanh(), nn.Linear(32, 8))
+    ref = nn.Sequential(nn.Linear(16, 32), nn.Tanh(), nn.Linear(32, 8))
+    ref.load_state_dict(model.state_dict())
+
+    mesh = init_device_mesh("cpu", (world_size,), mesh_dim_names=("dp_shard",))
+    parallelize_fsdp(model, mesh, shard_units=[model[0], model[2]])
+    # params are sharded DTensors now
+    assert isinstance(next(model.parameters()), DTensor)
+
+    torch.manual_seed(300 + rank)
+    x = torch.randn(6, 16)
+    model(x).pow(2).sum().backward()
+
+    # reference: sum of both ranks' grads (sum reduction, divide factor 1)
+    for r in range(world_size):
+        torch.manual_seed(300 + r)
+        xr = torch.randn(6, 16)
+        ref(xr).pow(2).sum().backward()
+
+    for (n, p), (_, pr) in zip(model.named_parameters(), ref.named_parameters()):
+        assert p.grad is not None, n
+        full = p.grad.full_tensor() if isinstance(p.grad, DTensor) else p.grad
+        torch.testing.assert_close(full, pr.grad, rtol=1e-4, atol=1e-5)
+    return True
+
+
+@pytest.mark.distributed
+def test_fsdp_grad_sum_ws2():
+    assert all(run_distributed(_fsdp_grad_matches_replicated, world_size=2))
+
+
+def _hsdp_grad_matches(rank, world_size):
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import DTensor
+
+    from d9d_amd.parallel import parallelize_hsdp
+
+    torch.manual_seed(22)
+    model = nn.Sequential(nn.Linear(16, 32), nn.Tanh(), nn.Linear(32, 8))
+    ref = nn.Sequential(nn.Linear(16, 32), nn.Tanh(), nn.Linear(32, 8))
+    ref.load_state_dict(model.state_dict())
+
+    mesh = init_device_mesh(
+        "cpu", (2, 2), mesh_dim_names=("dp_replicate", "dp_shard")
+    )
+    parallelize_hsdp(model, mesh, shard_units=[model[0], model[2]])
+    assert isinstance(next(model.parameters()), DTensor)
+
+    torch.manual_seed(400 + rank)
+    x = torch.randn(6, 16)
+    model(x).pow(2).sum().backward()
+
+    for r in range(world_size):
+        torch.manual_seed(400 + r)
+        xr = torch.randn(6, 16)
+        ref(xr).pow(2).sum().backward()
+
+    for (n, p), (_, pr) in zip(model.named_parameters(), ref.named_parameters()):
+        full = p.grad.full_tensor() if isinstance(p.grad, DTensor) else p.grad
+        torch.testing.assert_close(full, pr.grad, rtol=1e-4, atol=1e-5)
+    return True
+
+
+@pytest.mark.distributed
+def test_hsdp_grad_sum_ws4():
+    assert all(run_distributed(_hsdp_grad_matches, world_size=4))
