@@ -417,6 +417,153 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_kernel(
   }
 }
 
+// glds variant of gmm_nt: async global->LDS staging
+// (global_load_lds_dwordx4) into a double buffer -- no staging registers, no
+// ds_write pass, loads are fire-and-forget behind a counted s_waitcnt
+// vmcnt(N). Raw s_barrier (NOT __syncthreads) keeps hipcc from draining the
+// in-flight glds at each barrier. The swizzled LDS image is produced by
+// PRE-SWIZZLING the per-lane global source address (glds writes linearly:
+// wave-uniform base + lane*16). Requires K % 64 == 0 (no k-tail masking is
+// possible with direct-to-LDS loads); the register-staged kernel remains the
+// fallback.
+__global__ __launch_bounds__(512, 1) void gmm_nt_glds_kernel(
+    const bf16_t* __restrict__ a,    // (T, K)
+    const bf16_t* __restrict__ w,    // (E, N, K)
+    bf16_t* __restrict__ out,        // (T, N)
+    const int* __restrict__ row_off,
+    const int* __restrict__ mtile_pref,
+    int E, int K, int N, int n_tiles) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // dbuf: A [256][64] (32 KB) + B [192][64] (24 KB) per buffer
+  char* a_buf[2] = {smem, smem + 2 * kBM * kBK};
+  char* b_buf[2] = {smem + kBM * kBK, smem + 3 * kBM * kBK};
+  // (interleave A0,B0,A1,B1 as: smem | A0 32K | B0 24K-pad... use explicit)
+  a_buf[0] = smem;
+  b_buf[0] = smem + kBM * kBK * 2;
+  a_buf[1] = b_buf[0] + kBN * kBK * 2;
+  b_buf[1] = a_buf[1] + kBM * kBK * 2;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  const int nwg = gridDim.x;
+  const int xcd = blockIdx.x & 7;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int wid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                  + (blockIdx.x >> 3);
+  const int mt_global = wid / n_tiles;
+  int lo = 0, hi = E - 1;
+  while (lo < hi) {
+    const int mid = (lo + hi + 1) >> 1;
+    if (mtile_pref[mid] <= mt_global) lo = mid; else hi = mid - 1;
+  }
+  const int e = lo;
+  const int m_tile = mt_global - mtile_pref[e];
+  const int row0 = row_off[e] + m_tile * kBM;
+  const int row_end = row_off[e + 1];
+  const int n0 = (wid % n_tiles) * kBN;
+
+  const bf16_t* w_e = w + (int64_t)e * N * K;
+
+  f32x4 acc[8][3];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int n_ktiles = K / kBK;
+
+  // per k-tile: A = 32 wave-chunks of 1 KB, B = 24; wave 'wave' stages
+  // A chunks {wave + 8i} and B chunks {wave + 8i}.
+  auto stage = [&](int kt, int buf) {
+    const int k0 = kt * kBK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int chunk = wave + i * 8;
+      const int l = chunk * 1024 + lane * 16;   // linear LDS byte
+      const int arow = l >> 7;                  // 128-byte rows
+      const int colb = (l & 127) ^ ((arow & 7) << 4);  // source pre-swizzle
+      const int64_t sr = min(row0 + arow, row_end - 1);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const uint32_t*>(
+              reinterpret_cast<const char*>(a) + (sr * (int64_t)K + k0) * 2 + colb),
+          reinterpret_cast<uint32_t*>(a_buf[buf] + chunk * 1024), 16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < 3; ++i) {
+      const int chunk = wave + i * 8;
+      const int l = chunk * 1024 + lane * 16;
+      const int nrow = l >> 7;
+      const int colb = (l & 127) ^ ((nrow & 7) << 4);
+      const int64_t sn = min(n0 + nrow, N - 1);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const uint32_t*>(
+              reinterpret_cast<const char*>(w_e) + (sn * (int64_t)K + k0) * 2 + colb),
+          reinterpret_cast<uint32_t*>(b_buf[buf] + chunk * 1024), 16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  if (n_ktiles > 1) stage(1, 1);
+
+  for (int kt = 0; kt < n_ktiles; ++kt) {
+    const int buf = kt & 1;
+    // drain THIS tile's 7 glds: the next tile's 7 may stay in flight
+    if (kt + 1 < n_ktiles) {
+      asm volatile("s_waitcnt vmcnt(7)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    const char* al = a_buf[buf];
+    const char* bl = b_buf[buf];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kk = ks * 32 + (lane >> 4) * 8;
+      bf16x8 a_frag[8], b_frag[3];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int row = wm * 128 + i * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((row & 7) << 4);
+        a_frag[i] = *reinterpret_cast<const bf16x8*>(al + row * (kBK * 2) + byte);
+      }
+#pragma unroll
+      for (int j = 0; j < 3; ++j) {
+        const int col = wn * 48 + j * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((col & 7) << 4);
+        b_frag[j] = *reinterpret_cast<const bf16x8*>(bl + col * (kBK * 2) + byte);
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 3; ++j) acc[i][j] = mfma16g(a_frag[i], b_frag[j], acc[i][j]);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 2 < n_ktiles) stage(kt + 2, buf);
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+#pragma unroll
+      for (int r2 = 0; r2 < 4; ++r2) {
+        const int row = row0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r2;
+        const int col = n0 + wn * 48 + j * 16 + (lane & 15);
+        if (row < row_end && col < N) {
+          out[(int64_t)row * N + col] = (bf16_t)acc[i][j][r2];
+        }
+      }
+    }
+  }
+}
+
 // db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N).
 // Grid (ceil(N/256), ceil(K/256), E): 256x256 tiles per workgroup -- 8 waves
 // as 2 (k-halves of 128) x 4 (n-quarters of 64) -- looping the expert's rows
@@ -645,8 +792,28 @@ torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes
 
   const int n_tiles = (N + d9d::kBN - 1) / d9d::kBN;
   const dim3 grid(n_tiles * total_mtiles);
-  const size_t smem = (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
+  // glds staging measured 526 vs 609 TF/s for the register-T14 pipeline: at
+  // 2-deep tile prefetch one iteration of MFMAs (~0.3 us) cannot cover the
+  // HBM latency, so the counted vmcnt stalls every tile. The win the guide
+  // documents needs the finer 8-phase quadrant interleave (round-2 work);
+  // the kernel stays available for experiments via D9D_GMM_GLDS=1.
+  static const bool use_glds = []() {
+    const char* v = getenv("D9D_GMM_GLDS");
+    return v != nullptr && v[0] == '1';
+  }();
+  if (use_glds && K % 64 == 0) {
+    const size_t smem =
+        2 * (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
+    hipLaunchKernelGGL(d9d::gmm_nt_glds_kernel, grid, dim3(512), smem, stream,
+                       reinterpret_cast<const __bf16*>(a.data_ptr()),
+                       reinterpret_cast<const __bf16*>(w.data_ptr()),
+                       reinterpret_cast<__bf16*>(out.data_ptr()),
+                       row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
+                       E, K, N, n_tiles);
+    return out;
+  }
+  const size_t smem = (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
   hipLaunchKernelGGL(d9d::gmm_nt_kernel, grid, dim3(512), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(w.data_ptr()),
