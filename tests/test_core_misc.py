@@ -138,3 +138,30 @@ def test_pack_documents():
 
     ds = PackedDocumentDataset([{"input_ids": d} for d in docs], 16)
     assert sum(1 for _ in ds) == len(packs)
+
+
+def test_pooling_masks():
+    import torch
+
+    from d9d_amd.dataset import last_token_pooling_mask, mean_pooling_mask
+
+    lengths = torch.tensor([3, 1, 5])
+    m = mean_pooling_mask(lengths, 5)
+    assert m.shape == (3, 5)
+    assert m.sum(dim=1).tolist() == [3.0, 1.0, 5.0]
+    lt = last_token_pooling_mask(lengths, 5)
+    assert lt.sum().item() == 3.0
+    assert lt[0, 2] == 1.0 and lt[1, 0] == 1.0 and lt[2, 4] == 1.0
+
+
+def test_sdpa_env_backend_selection(monkeypatch):
+    from d9d_amd.module.block.attention.sdpa import (
+        SDPA_ENV_VAR,
+        build_sdpa_backend,
+    )
+
+    monkeypatch.setenv(SDPA_ENV_VAR, "eager")
+    backend = build_sdpa_backend(None)
+    assert getattr(backend, "__name__", "") == "_eager"
+    monkeypatch.delenv(SDPA_ENV_VAR)
+    assert getattr(build_sdpa_backend(None), "__name__", "") == "_cdna4_flash"
