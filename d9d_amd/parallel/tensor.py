@@ -155,6 +155,24 @@ def parallelize_tp_attention(
     attn.num_heads //= tp_size
     attn.num_kv_heads //= tp_size
 
+    # Replicated params INSIDE the tp region see only the local heads'
+    # activations, so their gradients are partial: sum them over tp.
+    # (q_norm/k_norm weights are shared across heads; learnable sinks are
+    # per-head and follow the q-head shard instead.)
+    def _allreduce_grad(p):
+        if p.grad is not None:
+            dist.all_reduce(p.grad, group=group)
+
+    for norm in (attn.q_norm, attn.k_norm):
+        if norm is not None:
+            norm.weight.register_post_accumulate_grad_hook(_allreduce_grad)
+    if attn.sinks is not None:
+        rank = dist.get_rank(group)
+        shard = attn.sinks.shape[0] // tp_size
+        with torch.no_grad():
+            local_sinks = attn.sinks[rank * shard : (rank + 1) * shard].clone()
+        attn.sinks = nn.Parameter(local_sinks)
+
     orig_forward = attn.forward
 
     def forward(hidden_states, rotary_cos_sin):
