@@ -106,3 +106,84 @@ def _tp2_whole_model(rank, world_size):
 @pytest.mark.distributed
 def test_tp2_whole_model_golden():
     assert all(run_distributed(_tp2_whole_model, world_size=2))
+
+
+def _ep_dp_whole_model(rank, world_size):
+    """bench.py --parallelism ep composition: experts sharded over ep, the
+    rest replicated over dp, bucketed grad sync; gradients must equal the
+    global model run over BOTH ranks' batches (summed)."""
+    from torch.distributed.tensor import DTensor
+
+    from d9d_amd.core.dist_context import DeviceMeshParameters
+    from d9d_amd.internals.grad_sync import GradientSynchronizer
+    from d9d_amd.module.model.qwen3_moe import (
+        Qwen3MoEForCausalLM,
+        Qwen3MoEModelParameters,
+    )
+    from d9d_amd.parallel import (
+        parallelize_expert_parallel,
+        parallelize_replicate,
+    )
+
+    p = Qwen3MoEModelParameters.tiny()
+    torch.manual_seed(88)
+    global_model = Qwen3MoEForCausalLM(p)
+    global_model.init_weights()
+
+    torch.manual_seed(88)
+    model = Qwen3MoEForCausalLM(p)
+    model.init_weights()
+
+    ctx = DeviceMeshParameters(
+        data_parallel_replicate=world_size, expert_parallel=world_size
+    ).build(device_type="cpu")
+    parallelize_expert_parallel(model, ctx.mesh_for("expert"))
+    parallelize_replicate(model, ctx.mesh_for("dense"))
+
+    # copy the global weights into the local shards (experts: dim-0 slice)
+    gsd = {n: t for n, t in global_model.state_dict().items()}
+    with torch.no_grad():
+        for name, param in model.named_parameters():
+            src = gsd[name]
+            local = param.to_local() if isinstance(param, DTensor) else param
+            if local.shape == src.shape:
+                local.copy_(src)
+            else:  # expert shard on dim 0
+                e_local = local.shape[0]
+                local.copy_(src[rank * e_local : (rank + 1) * e_local])
+
+    sync = GradientSynchronizer(
+        list(model.named_parameters()), accumulation_steps=1, bucket_bytes=1 << 20
+    )
+
+    torch.manual_seed(600 + rank)
+    ids = torch.randint(0, p.vocab_size, (2, 16))
+    model(input_ids=ids, labels=ids)["loss"].mean().backward()
+    sync.wait()
+
+    # global reference over both ranks' batches
+    for r in range(world_size):
+        torch.manual_seed(600 + r)
+        idr = torch.randint(0, p.vocab_size, (2, 16))
+        global_model(input_ids=idr, labels=idr)["loss"].mean().backward()
+
+    for name, param in model.named_parameters():
+        if param.grad is None:
+            continue
+        g = param.grad.to_local() if isinstance(param.grad, DTensor) else param.grad
+        ref = gsd[name]  # shapes
+        rg = dict(global_model.named_parameters())[name].grad
+        if g.shape != rg.shape:  # expert shard
+            e_local = g.shape[0]
+            rg = rg[rank * e_local : (rank + 1) * e_local]
+        ang = grad_angle(g, rg)
+        nrm = grad_norm_ratio(g, rg)
+        assert ang <= 1e-2, f"{name}: angle {ang:.2e}"
+        assert nrm <= 1e-2, f"{name}: norm ratio {nrm:.2e}"
+    sync.remove()
+    return True
+
+
+@pytest.mark.distributed
+def test_ep_dp_whole_model_golden():
+    assert all(run_distributed(_ep_dp_whole_model, world_size=2))
