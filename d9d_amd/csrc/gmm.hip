@@ -175,7 +175,11 @@ __global__ __launch_bounds__(512, 1) void gmm_kernel(
         const int nn = n + j;
         const uint32_t u = (uint32_t)b_reg[it][0].s[j] |
                            ((uint32_t)b_reg[it][1].s[j] << 16);
-        const int byte = (kk * 2) ^ ((nn & 7) << 4);
+        // swizzle on (nn ^ nn>>3): the paired writes walk nn in +8 steps
+        // inside one instruction (nn&7 alone is constant there -> 24-way
+        // bank conflict, PMC-measured 21% of wave cycles); the reads walk
+        // nn in +1 steps, and the combined mask spreads both 8 ways.
+        const int byte = (kk * 2) ^ ((((nn >> 3) ^ nn) & 7) << 4);
         *reinterpret_cast<uint32_t*>(
             reinterpret_cast<char*>(bt_lds) + nn * (kBK * 2) + byte) = u;
       }
@@ -204,7 +208,7 @@ __global__ __launch_bounds__(512, 1) void gmm_kernel(
 #pragma unroll
       for (int j = 0; j < 3; ++j) {
         const int col = wn * 48 + j * 16 + (lane & 15);
-        const int byte = (kk * 2) ^ ((col & 7) << 4);
+        const int byte = (kk * 2) ^ ((((col >> 3) ^ col) & 7) << 4);
         b_frag[j] = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const char*>(bt_lds) + col * (kBK * 2) + byte);
       }
