@@ -438,10 +438,9 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_glds_kernel(
     const int* __restrict__ mtile_pref,
     int E, int K, int N, int n_tiles) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // dbuf: A [256][64] (32 KB) + B [192][64] (24 KB) per buffer
-  char* a_buf[2] = {smem, smem + 2 * kBM * kBK};
-  char* b_buf[2] = {smem + kBM * kBK, smem + 3 * kBM * kBK};
-  // (interleave A0,B0,A1,B1 as: smem | A0 32K | B0 24K-pad... use explicit)
+  // layout: A0 (32 KB) | B0 (24 KB) | A1 | B1
+  char* a_buf[2];
+  char* b_buf[2];
   a_buf[0] = smem;
   b_buf[0] = smem + kBM * kBK * 2;
   a_buf[1] = b_buf[0] + kBN * kBK * 2;

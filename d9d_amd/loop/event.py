@@ -1,7 +1,7 @@
 """Typed pub/sub event bus (reference: d9d/loop/event/core.py:10-71 + catalogue)."""
 
 from contextlib import contextmanager
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Callable, Generic, Iterator, TypeVar
 
 TContext = TypeVar("TContext")

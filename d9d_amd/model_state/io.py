@@ -10,7 +10,6 @@ and merges indices on rank 0 (reference: io/writer.py:61-309).
 """
 
 import json
-import os
 from pathlib import Path
 from typing import Iterator
 

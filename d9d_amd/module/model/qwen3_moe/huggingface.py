@@ -14,7 +14,6 @@ Common differences from HF:
 """
 
 from ....model_state.mapper import (
-    StateGroup,
     ConcatenateTensors,
     Identity,
     ModelStateMapper,
@@ -23,7 +22,6 @@ from ....model_state.mapper import (
     Sequential,
     SliceRows,
     StackTensors,
-    Transpose,
     UnstackTensors,
 )
 from .params import Qwen3MoEModelParameters

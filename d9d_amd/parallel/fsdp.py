@@ -9,7 +9,7 @@ reduction with divide factor 1 so the trainer scales gradients itself
 
 from torch import nn
 from torch.distributed.device_mesh import DeviceMesh
-from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
+from torch.distributed.fsdp import fully_shard
 
 
 def _apply_sum_reduction(module: nn.Module) -> None:

@@ -16,7 +16,6 @@ from .api import PipelineStageInfo
 from .executor import OfflinePipelineExecutor, PipelineScheduleExecutor
 from .programs import (
     build_1f1b,
-    build_gpipe,
     build_looped_bfs,
     build_zb1p,
     build_dualpipev,

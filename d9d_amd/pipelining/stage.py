@@ -5,7 +5,7 @@ from meta-device shape inference; forward caches (inputs, outputs) per
 microbatch; backward supports full or input/weight-split (splitgrad).
 """
 
-from typing import Any, Callable
+from typing import Any
 
 import torch
 from torch import nn
