@@ -187,3 +187,35 @@ def test_hf_parity_llama3():
         h = ours.model(input_ids=ids)["hidden_states"]
         our_logits = ours.lm_head.logits(h)
     assert _kl(hf_logits, our_logits) < 1e-4
+
+
+@pytest.mark.filterwarnings("ignore")
+def test_llama3_export_roundtrip():
+    """d9d -> HF -> d9d mapping is the identity on Llama-3 state."""
+    from d9d_amd.module.model.llama3 import (
+        Llama3ForCausalLM,
+        Llama3ModelParameters,
+        d9d_to_hf_mapper,
+        hf_to_d9d_mapper,
+    )
+
+    p = Llama3ModelParameters(
+        hidden_size=32,
+        intermediate_size=64,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=16,
+        num_hidden_layers=2,
+        split_vocab_size={"regular": 96, "special": 8},
+    )
+    torch.manual_seed(1)
+    m = Llama3ForCausalLM(p)
+    m.init_weights()
+    sd = {k: v.clone() for k, v in m.state_dict().items()}
+    hf_sd = d9d_to_hf_mapper(p).apply(dict(sd))
+    assert "model.layers.0.self_attn.q_proj.weight" in hf_sd
+    assert not any("q_norm" in k for k in hf_sd)
+    back = hf_to_d9d_mapper(p).apply(hf_sd)
+    assert set(back) == set(sd)
+    for k in sd:
+        torch.testing.assert_close(back[k], sd[k], msg=lambda m_: f"{k}: {m_}")
