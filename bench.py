@@ -64,7 +64,8 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world > 1
 
-    torch.manual_seed(1234 + rank)
+    # identical weights on every DP replica; per-rank data seeds come later
+    torch.manual_seed(1234)
     device = torch.device("cuda", local_rank)
     torch.cuda.set_device(device)
 
@@ -121,7 +122,9 @@ def main():
     tokens_per_step_per_rank = B * S * n_micro
     vocab = params.vocab_size
 
-    # Pre-generate synthetic microbatches (same shapes as tokenized pretrain data).
+    # Pre-generate synthetic microbatches (same shapes as tokenized pretrain
+    # data); per-rank seed so DP ranks train on different tokens.
+    torch.manual_seed(1234 + rank)
     batches = [
         torch.randint(0, vocab, (B, S + 1), device=device) for _ in range(2)
     ]

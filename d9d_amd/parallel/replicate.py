@@ -6,12 +6,39 @@ NOT done here — the bucketed `GradientSynchronizer`
 dedicated HIP stream, overlapped with backward.
 """
 
+import torch
+import torch.distributed as dist
 from torch import nn
 from torch.distributed.device_mesh import DeviceMesh
+from torch.distributed.tensor import DTensor
 
 from .style import distribute_module_params
 
 
-def parallelize_replicate(module: nn.Module, mesh: DeviceMesh) -> nn.Module:
-    """Replicate every parameter of `module` over all dims of `mesh`."""
-    return distribute_module_params(module, mesh, placement_fn=None, recurse=True)
+def parallelize_replicate(
+    module: nn.Module, mesh: DeviceMesh, broadcast_params: bool = True
+) -> nn.Module:
+    """Replicate every parameter of `module` over all dims of `mesh`.
+
+    `broadcast_params` copies group-rank-0's values over each mesh dim so
+    replicas agree even when the caller seeded per-rank (DDP semantics);
+    pass False when weights are known-identical (e.g. loaded checkpoint).
+    """
+    distribute_module_params(module, mesh, placement_fn=None, recurse=True)
+    if broadcast_params:
+        with torch.no_grad():
+            for dim in range(mesh.ndim):
+                group = mesh.get_group(dim)
+                if dist.get_world_size(group) == 1:
+                    continue
+                for p in module.parameters():
+                    local = p.to_local() if isinstance(p, DTensor) else p
+                    dist.broadcast(
+                        local, src=dist.get_global_rank(group, 0), group=group
+                    )
+                for b in module.buffers():
+                    local = b.to_local() if isinstance(b, DTensor) else b
+                    dist.broadcast(
+                        local, src=dist.get_global_rank(group, 0), group=group
+                    )
+    return module
