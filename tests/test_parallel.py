@@ -300,3 +300,28 @@ def _hsdp_grad_matches(rank, world_size):
 @pytest.mark.distributed
 def test_hsdp_grad_sum_ws4():
     assert all(run_distributed(_hsdp_grad_matches, world_size=4))
+
+
+def _replicate_broadcast(rank, world_size):
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import DTensor
+
+    from d9d_amd.parallel import parallelize_replicate
+
+    torch.manual_seed(900 + rank)  # deliberately DIFFERENT init per rank
+    model = nn.Linear(4, 4)
+    mesh = init_device_mesh("cpu", (2,), mesh_dim_names=("dp",))
+    parallelize_replicate(model, mesh)
+    import torch.distributed as dist
+
+    w = model._parameters["weight"]
+    local = w.to_local() if isinstance(w, DTensor) else w
+    gathered = [torch.empty_like(local) for _ in range(world_size)]
+    dist.all_gather(gathered, local)
+    assert torch.equal(gathered[0], gathered[1]), "replicas must agree after broadcast"
+    return True
+
+
+@pytest.mark.distributed
+def test_replicate_broadcasts_initial_weights_ws2():
+    assert all(run_distributed(_replicate_broadcast, world_size=2))
