@@ -26,19 +26,23 @@ def parallelize_replicate(
     """
     distribute_module_params(module, mesh, placement_fn=None, recurse=True)
     if broadcast_params:
+        # Only tensors replicated over THIS mesh: params parallelized earlier
+        # by another strategy (e.g. EP-sharded experts on the expert mesh)
+        # hold different data per rank by design and must not be overwritten.
+        def _broadcastable(t):
+            if not isinstance(t, DTensor):
+                return True
+            return t.device_mesh is mesh
+
         with torch.no_grad():
             for dim in range(mesh.ndim):
                 group = mesh.get_group(dim)
                 if dist.get_world_size(group) == 1:
                     continue
-                for p in module.parameters():
-                    local = p.to_local() if isinstance(p, DTensor) else p
-                    dist.broadcast(
-                        local, src=dist.get_global_rank(group, 0), group=group
-                    )
-                for b in module.buffers():
-                    local = b.to_local() if isinstance(b, DTensor) else b
-                    dist.broadcast(
-                        local, src=dist.get_global_rank(group, 0), group=group
-                    )
+                src_rank = dist.get_global_rank(group, 0)
+                for t in list(module.parameters()) + list(module.buffers()):
+                    if not _broadcastable(t):
+                        continue
+                    local = t.to_local() if isinstance(t, DTensor) else t
+                    dist.broadcast(local, src=src_rank, group=group)
     return module
