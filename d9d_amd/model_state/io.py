@@ -98,10 +98,17 @@ def write_model_state(
     shard_size_gb: float = 4.0,
     file_prefix: str = "model",
     write_index: bool = True,
+    persist: bool = True,
 ) -> dict[str, str]:
-    """Run `source` through `mapper`, writing outputs as safetensors shards."""
+    """Run `source` through `mapper`, writing outputs as safetensors shards.
+
+    `persist=False` runs the full traversal WITHOUT buffering or writing:
+    mapper leaves may issue collectives (DTensor `full_tensor`), so in a
+    distributed export every rank must traverse every group in the same
+    order even when only some ranks write."""
     path = Path(path)
-    path.mkdir(parents=True, exist_ok=True)
+    if persist:
+        path.mkdir(parents=True, exist_ok=True)
     applier = _StreamingApplier(mapper)
     items = source.items() if isinstance(source, dict) else source
 
@@ -121,7 +128,9 @@ def write_model_state(
         for outs in applier.offer(key, tensor):
             for ok, ov in outs.items():
                 if isinstance(ov, DTensor):
-                    ov = ov.full_tensor()
+                    ov = ov.full_tensor()  # collective: runs on EVERY rank
+                if not persist:
+                    continue
                 ov = ov.detach().cpu().contiguous()
                 buffer[ok] = ov
                 buffered += ov.numel() * ov.element_size()
@@ -134,6 +143,8 @@ def write_model_state(
         )[:10]
         raise ValueError(f"model_state write: missing inputs for groups, e.g. {missing}")
 
+    if not persist:
+        return {}
     total = len(shards)
     weight_map: dict[str, str] = {}
     for i, shard in enumerate(shards):
@@ -159,14 +170,15 @@ def write_model_state_distributed(
     rank 0 merges the index (reference: io/writer.py:252-309)."""
     path = Path(path)
     rank = dist.get_rank(group) if dist.is_initialized() else 0
-    maps: dict[str, str] = {}
-    if is_writer:
-        maps = write_model_state(
-            mapper, source, path,
-            shard_size_gb=shard_size_gb,
-            file_prefix=f"model-rank{rank}",
-            write_index=False,
-        )
+    # EVERY rank traverses the mapper (DTensor gathers inside are
+    # collectives); only writer ranks persist shards.
+    maps: dict[str, str] = write_model_state(
+        mapper, source, path,
+        shard_size_gb=shard_size_gb,
+        file_prefix=f"model-rank{rank}",
+        write_index=False,
+        persist=is_writer,
+    )
     if dist.is_initialized():
         from ..core.dist_ops import all_gather_object
 

@@ -1,7 +1,10 @@
 """model_state mapper DAG + streaming safetensors IO tests."""
 
+import pytest
 import torch
 from torch import nn
+
+from tests.helpers import run_distributed
 
 from d9d_amd.model_state import (
     ChunkTensors,
@@ -119,3 +122,45 @@ def test_identity_mapper_from_module():
     mapper = identity_mapper_from_module(m)
     groups = mapper.state_dependency_groups()
     assert len(groups) == 2
+
+
+def _distributed_export_ep(rank, world_size, tmp_dir):
+    import torch
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import DTensor, Shard, distribute_tensor
+
+    from d9d_amd.model_state.io import write_model_state_distributed
+    from d9d_amd.model_state.mapper import GatherFullTensor, Parallel
+
+    mesh = init_device_mesh("cpu", (world_size,), mesh_dim_names=("ep",))
+    torch.manual_seed(1)
+    full = torch.randn(4, 6)
+    sharded = distribute_tensor(full, mesh, (Shard(0),))
+    state = {"experts.weight": sharded, "norm.weight": torch.ones(6)}
+    mapper = Parallel(*[GatherFullTensor(k) for k in state])
+
+    # only rank 0 persists, but the gather inside is collective: every rank
+    # must traverse (regression: this deadlocked / mismatched before)
+    write_model_state_distributed(
+        mapper, state, tmp_dir, is_writer=(rank == 0)
+    )
+    import torch.distributed as dist
+
+    dist.barrier()
+    if rank == 0:
+        from safetensors.torch import load_file
+        import json, pathlib
+
+        index = json.loads((pathlib.Path(tmp_dir) / "model.safetensors.index.json").read_text())
+        fname = index["weight_map"]["experts.weight"]
+        data = load_file(str(pathlib.Path(tmp_dir) / fname))
+        torch.testing.assert_close(data["experts.weight"], full)
+    return True
+
+
+@pytest.mark.distributed
+def test_distributed_export_with_sharded_dtensor_ws2(tmp_path):
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as d:
+        assert all(run_distributed(_distributed_export_ep, world_size=2, args=(d,)))
