@@ -136,6 +136,15 @@ class Qwen3DenseModel(nn.Module):
             names.add("input_ids")
         return names
 
+    def _stage_dtype(self) -> torch.dtype:
+        # P2P buffers must match the dtype ACTUALLY sent: the model's
+        # parameter dtype (a hardcoded bf16 here sized CPU/fp32 pipeline
+        # recv buffers at half the sender's bytes -- found via the ws=2
+        # PP trainer test).
+        for p in self.parameters():
+            return p.dtype
+        return torch.bfloat16
+
     def infer_stage_inputs_from_pipeline_inputs(
         self, pipeline_inputs: dict[str, Any], num_microbatches: int
     ) -> dict[str, torch.Tensor]:
@@ -145,7 +154,8 @@ class Qwen3DenseModel(nn.Module):
         B, S = ids.shape[0] // num_microbatches, ids.shape[1]
         return {
             "hidden_states": torch.empty(
-                B, S, self.params.hidden_size, dtype=torch.bfloat16, device="meta"
+                B, S, self.params.hidden_size, dtype=self._stage_dtype(),
+                device="meta",
             )
         }
 
@@ -158,7 +168,8 @@ class Qwen3DenseModel(nn.Module):
         B, S = ids.shape[0] // num_microbatches, ids.shape[1]
         return {
             "hidden_states": torch.empty(
-                B, S, self.params.hidden_size, dtype=torch.bfloat16, device="meta"
+                B, S, self.params.hidden_size, dtype=self._stage_dtype(),
+                device="meta",
             )
         }
 

@@ -39,8 +39,11 @@ def run_distributed(fn, world_size: int = 2, args: tuple = (), timeout: float = 
     `fn` must be a module-level function (pickled into the workers).
     Returns the list of per-rank results ordered by rank.
     """
+    import queue as queue_mod
+    import time
+
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
+    q = ctx.Queue()
     port = _free_port()
     fn_bytes = pickle.dumps(fn)
     procs = [
@@ -51,17 +54,30 @@ def run_distributed(fn, world_size: int = 2, args: tuple = (), timeout: float = 
         p.start()
     results: dict[int, object] = {}
     errors: list[str] = []
-    for _ in range(world_size):
-        rank, status, payload = q.get()
+    # bounded wait: a deadlocked rank must FAIL the test, not hang the suite
+    deadline = time.time() + timeout
+    received = 0
+    while received < world_size and time.time() < deadline:
+        try:
+            rank, status, payload = q.get(timeout=1.0)
+        except queue_mod.Empty:
+            if all(not p.is_alive() for p in procs):
+                break  # every process died without reporting
+            continue
+        received += 1
         if status == "error":
             errors.append(f"rank {rank}:\n{payload}")
         else:
             results[rank] = payload
+    if received < world_size:
+        errors.append(
+            f"only {received}/{world_size} ranks reported within {timeout}s "
+            "(deadlock or crash)"
+        )
     for p in procs:
-        p.join(timeout=timeout)
+        p.join(timeout=5.0)
         if p.is_alive():
             p.terminate()
-            errors.append("process did not exit in time")
     if errors:
         raise RuntimeError("distributed test failed:\n" + "\n".join(errors))
     return [results[r] for r in range(world_size)]

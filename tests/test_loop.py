@@ -208,3 +208,34 @@ def test_inference_configurator_runs():
     results = runner.run()
     assert len(results) >= 2  # 8 rows / batch 4 -> 2 batches, >= 1 mb each
     assert all(torch.isfinite(r).all() for r in results)
+
+
+def _pp2_trainer(rank, world_size):
+    """Trainer with pipeline_parallel=2 (1F1B): per-stage modules via the
+    provider's stage_info, pipelined optimizer, last-stage loss."""
+    from d9d_amd.loop.config import PipeliningConfig
+    from d9d_amd.pipelining.factory import PipelineSchedule1F1BConfig
+
+    params = Qwen3DenseModelParameters.tiny()
+    cfg = _make_config(total_steps=2)
+    cfg = cfg.model_copy(
+        update={"pipelining": PipeliningConfig(schedule=PipelineSchedule1F1BConfig())}
+    )
+    mesh = DeviceMeshParameters(pipeline_parallel=2)
+    trainer = TrainingConfigurator(
+        cfg,
+        mesh,
+        _LMModelProvider(params),
+        _LMDatasetProvider(params),
+        AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+        AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+        _LMTask(),
+    ).configure(device_type="cpu")
+    trainer.train()
+    # both ranks completed 2 steps; last stage saw losses
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_pp2_runs():
+    assert all(run_distributed(_pp2_trainer, world_size=2, timeout=90))
