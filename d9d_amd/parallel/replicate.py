@@ -23,7 +23,14 @@ def parallelize_replicate(
     `broadcast_params` copies group-rank-0's values over each mesh dim so
     replicas agree even when the caller seeded per-rank (DDP semantics);
     pass False when weights are known-identical (e.g. loaded checkpoint).
+
+    A "pp" dim in the mesh is dropped automatically: pipeline stages hold
+    DIFFERENT modules, so parameters are never replicated across pp (the
+    dense/batch domains carry pp for other consumers).
     """
+    if mesh.mesh_dim_names and "pp" in mesh.mesh_dim_names and mesh.ndim > 1:
+        keep = tuple(n for n in mesh.mesh_dim_names if n != "pp")
+        mesh = mesh[keep]
     distribute_module_params(module, mesh, placement_fn=None, recurse=True)
     if broadcast_params:
         # Only tensors replicated over THIS mesh: params parallelized earlier
