@@ -25,6 +25,11 @@ def parallelize_expert_parallel(
     Must run BEFORE materialization: GroupedLinear weights are re-created with
     the local expert count so each rank only allocates its shard.
     """
+    # Pipeline stages hold different modules: drop the pp dim (placements
+    # over pp would make grad sync all-reduce across stages -- a deadlock).
+    if mesh.mesh_dim_names and "pp" in mesh.mesh_dim_names and mesh.ndim > 1:
+        keep = tuple(n for n in mesh.mesh_dim_names if n != "pp")
+        mesh = mesh[keep]
     shard_dim = mesh.mesh_dim_names.index(shard_dim_name)
     ep_size = mesh.shape[shard_dim]
     ep_group = mesh.get_group(shard_dim)

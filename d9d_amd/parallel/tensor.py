@@ -223,6 +223,9 @@ def parallelize_tensor_parallel(
     sequence_parallel: bool = False,
 ) -> nn.Module:
     """Apply TP (optionally SP) to every GQA attention and SwiGLU FFN block."""
+    if mesh.mesh_dim_names and "pp" in mesh.mesh_dim_names and mesh.ndim > 1:
+        # stages hold different modules; placements must not span pp
+        mesh = mesh[tuple(n for n in mesh.mesh_dim_names if n != "pp")]
     for sub in module.modules():
         if isinstance(sub, GroupedQueryAttention):
             parallelize_tp_attention(sub, mesh, tp_dim_name, sequence_parallel)

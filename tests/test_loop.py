@@ -288,3 +288,92 @@ def test_trainer_pp2_dp2():
 
     counts = Counter(results)
     assert all(v == 2 for v in counts.values()), results
+
+
+def _pp2_ep2_trainer(rank, world_size):
+    """ws=4 trainer: pipeline_parallel=2 x (dp_replicate=2 with ep=2): the
+    full 3D composition of the MoE flagship (schedule P2P + expert all-to-all
+    + replicated-grad sync in one step)."""
+    from d9d_amd.loop.config import PipeliningConfig
+    from d9d_amd.module.model.qwen3_moe import (
+        Qwen3MoEForCausalLM,
+        Qwen3MoEModelParameters,
+    )
+    from d9d_amd.parallel import (
+        parallelize_expert_parallel,
+        parallelize_replicate,
+    )
+    from d9d_amd.pipelining.factory import PipelineSchedule1F1BConfig
+
+    params = Qwen3MoEModelParameters.tiny()
+
+    class Provider(ModelProvider):
+        def initialize_model_stage(self, stage_info):
+            return Qwen3MoEForCausalLM(params, stage_info)
+
+        def parallelize_model_stage(self, module, ctx):
+            parallelize_expert_parallel(module, ctx.mesh_for("expert"))
+            parallelize_replicate(module, ctx.mesh_for("dense"))
+            return module
+
+    class DS(DatasetProvider):
+        def build_dataset(self, ctx):
+            return _SyntheticLM(params.vocab_size, 16)
+
+    cfg = TrainerConfig(
+        batching=BatchingConfig(global_batch_size=8, microbatch_size=2),
+        total_steps=2,
+        pipelining=PipeliningConfig(schedule=PipelineSchedule1F1BConfig()),
+    )
+    mesh = DeviceMeshParameters(
+        pipeline_parallel=2, data_parallel_replicate=2, expert_parallel=2
+    )
+    trainer = TrainingConfigurator(
+        cfg,
+        mesh,
+        Provider(),
+        DS(),
+        AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+        AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+        _LMTask(),
+    ).configure(device_type="cpu")
+    trainer.train()
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_pp2_ep2_dp2():
+    assert all(run_distributed(_pp2_ep2_trainer, world_size=4, timeout=240))
+
+
+def _pp2_tp2_trainer(rank, world_size):
+    """ws=4 trainer: pipeline_parallel=2 x tensor_parallel=2."""
+    from d9d_amd.loop.config import PipeliningConfig
+    from d9d_amd.parallel import parallelize_tensor_parallel
+    from d9d_amd.pipelining.factory import PipelineSchedule1F1BConfig
+
+    def parallelize(module, ctx):
+        return parallelize_tensor_parallel(module, ctx.mesh_for("regular"))
+
+    params = Qwen3DenseModelParameters.tiny()
+    cfg = _make_config(total_steps=2)
+    cfg = cfg.model_copy(
+        update={"pipelining": PipeliningConfig(schedule=PipelineSchedule1F1BConfig())}
+    )
+    mesh = DeviceMeshParameters(pipeline_parallel=2, tensor_parallel=2)
+    trainer = TrainingConfigurator(
+        cfg,
+        mesh,
+        _LMModelProvider(params, parallelize),
+        _LMDatasetProvider(params),
+        AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+        AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+        _LMTask(),
+    ).configure(device_type="cpu")
+    trainer.train()
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_pp2_tp2():
+    assert all(run_distributed(_pp2_tp2_trainer, world_size=4, timeout=240))
