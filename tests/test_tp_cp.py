@@ -114,3 +114,45 @@ def _cp2_attention(rank, world_size):
 @pytest.mark.distributed
 def test_cp2_attention_exact():
     assert all(run_distributed(_cp2_attention, world_size=2))
+
+
+def _tp2_sp_layer(rank, world_size):
+    """TP + sequence parallel: sequence-sharded activations between blocks;
+    outputs and input grads must match the replicated reference."""
+    from torch.distributed.device_mesh import init_device_mesh
+
+    from d9d_amd.parallel.tensor import parallelize_tensor_parallel
+
+    p, ref_layer = _make_layer(seed=6)
+    _, layer = _make_layer(seed=6)
+
+    mesh = init_device_mesh("cpu", (2,), mesh_dim_names=("tp",))
+    parallelize_tensor_parallel(layer, mesh, sequence_parallel=True)
+
+    torch.manual_seed(78)
+    S = 8
+    x = torch.randn(2, S, p.hidden_size)
+    cos_sin = _rotary(p, 2, S)
+
+    # local input = this rank's sequence shard
+    x_local = x.chunk(2, dim=1)[rank].clone().requires_grad_(True)
+    out_local = layer(x_local, cos_sin)
+    assert out_local.shape[1] == S // 2
+    out_local.sum().backward()
+
+    x_ref = x.detach().clone().requires_grad_(True)
+    ref_out = ref_layer(x_ref, cos_sin)
+    ref_out.sum().backward()
+
+    torch.testing.assert_close(
+        out_local, ref_out.chunk(2, dim=1)[rank], rtol=1e-4, atol=1e-5
+    )
+    torch.testing.assert_close(
+        x_local.grad, x_ref.grad.chunk(2, dim=1)[rank], rtol=1e-4, atol=1e-5
+    )
+    return True
+
+
+@pytest.mark.distributed
+def test_tp2_sequence_parallel_layer_exact():
+    assert all(run_distributed(_tp2_sp_layer, world_size=2))
