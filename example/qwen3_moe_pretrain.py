@@ -110,7 +110,8 @@ def main():
         config,
         mesh,
         Qwen3MoEProvider(params),
-        LMDatasetProvider(params),
+        # short sequences for the CPU demo path; 2048+ on GPUs
+        LMDatasetProvider(params, seq_len=2048 if torch.cuda.is_available() else 256),
         AutoOptimizerProvider(OptimizerConfig(optimizer="stochastic_adamw", lr=3e-4)),
         AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=5, decay_steps=100)),
         PretrainTask(),
