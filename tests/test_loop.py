@@ -377,3 +377,56 @@ def _pp2_tp2_trainer(rank, world_size):
 @pytest.mark.distributed
 def test_trainer_pp2_tp2():
     assert all(run_distributed(_pp2_tp2_trainer, world_size=4, timeout=240))
+
+
+def _pp2_inference(rank, world_size):
+    from d9d_amd.loop.control import InferenceTask
+    from d9d_amd.loop.inference import InferenceConfigurator
+
+    params = Qwen3DenseModelParameters.tiny()
+
+    class Task(InferenceTask):
+        def build_forward_inputs(self, batch):
+            return {"input_ids": batch[:, :-1], "labels": batch[:, 1:]}
+
+        def process_outputs(self, outputs, mb_inputs):
+            return outputs["logps"].detach().sum()
+
+    cfg = TrainerConfig(
+        batching=BatchingConfig(global_batch_size=4, microbatch_size=2),
+        total_steps=1,
+    )
+    mesh = DeviceMeshParameters(pipeline_parallel=2)
+    runner = InferenceConfigurator(
+        cfg, mesh, _LMModelProvider(params), _LMDatasetProvider(params), Task()
+    ).configure(device_type="cpu")
+    results = runner.run()
+    # only the LAST pipeline stage produces results
+    return len(results)
+
+
+@pytest.mark.distributed
+def test_inference_pp2_forward_only():
+    results = run_distributed(_pp2_inference, world_size=2, timeout=180)
+    assert results[0] == 0 and results[1] > 0
+
+
+def _sleep_wake_dp2(rank, world_size):
+    from d9d_amd.parallel import parallelize_replicate
+
+    def par(module, ctx):
+        return parallelize_replicate(module, ctx.mesh_for("dense"))
+
+    mesh = DeviceMeshParameters(data_parallel_replicate=2)
+    tr = _build_trainer(total_steps=1, parallelize=par, mesh=mesh)
+    tr.train()
+    tr.sleep()
+    assert tr.is_sleeping
+    tr.wake()
+    tr.train()
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_sleep_wake_dp2():
+    assert all(run_distributed(_sleep_wake_dp2, world_size=2, timeout=240))
