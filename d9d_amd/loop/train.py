@@ -283,14 +283,21 @@ class Trainer:
     # -- checkpoint state schema (reference: loop/state.py:29-150) ------------
 
     def _job_state(self) -> _StatefulDict:
+        # optimizer / lr-scheduler state is per-PIPELINE-STAGE (different
+        # parameter shapes on each pp rank): the keys must be pp-qualified or
+        # DCP's global planner sees one fqn with conflicting tensors
+        # (AssertionError "item.index.fqn not in md").
+        pp = self.ctx.pp_rank
         return _StatefulDict(
-            stepper=self.stepper,
-            tracked_modules=_ModuleStates(self.modules_by_key),
-            data_loader=self.data_loader,
-            optimizer=self.optimizer,
-            lr_scheduler=self.lr_scheduler,
-            metrics=self.collector,
-            tracker=self.tracker,
+            **{
+                "stepper": self.stepper,
+                "tracked_modules": _ModuleStates(self.modules_by_key),
+                "data_loader": self.data_loader,
+                f"pp_{pp}_optimizer": self.optimizer,
+                f"pp_{pp}_lr_scheduler": self.lr_scheduler,
+                "metrics": self.collector,
+                "tracker": self.tracker,
+            }
         )
 
     # -- the loop --------------------------------------------------------------

@@ -430,3 +430,86 @@ def _sleep_wake_dp2(rank, world_size):
 @pytest.mark.distributed
 def test_trainer_sleep_wake_dp2():
     assert all(run_distributed(_sleep_wake_dp2, world_size=2, timeout=240))
+
+
+def _pp2_schedule_trainer(rank, world_size, schedule_name):
+    """Trainer e2e with V-topology schedules (2 stages/rank)."""
+    from d9d_amd.loop.config import PipeliningConfig
+    from d9d_amd.pipelining.factory import (
+        PipelineScheduleDualPipeVConfig,
+        PipelineScheduleZBVConfig,
+    )
+
+    sched = {
+        "zbv": PipelineScheduleZBVConfig(),
+        "dualpipev": PipelineScheduleDualPipeVConfig(),
+    }[schedule_name]
+    params = Qwen3DenseModelParameters.tiny()
+    cfg = _make_config(total_steps=2)
+    cfg = cfg.model_copy(
+        update={
+            "pipelining": PipeliningConfig(schedule=sched),
+            "batching": BatchingConfig(global_batch_size=8, microbatch_size=1),
+        }
+    )
+    mesh = DeviceMeshParameters(pipeline_parallel=2)
+    trainer = TrainingConfigurator(
+        cfg,
+        mesh,
+        _LMModelProvider(params),
+        _LMDatasetProvider(params),
+        AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+        AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+        _LMTask(),
+    ).configure(device_type="cpu")
+    trainer.train()
+    return True
+
+
+@pytest.mark.distributed
+@pytest.mark.parametrize("schedule_name", ["zbv", "dualpipev"])
+def test_trainer_pp2_v_schedules(schedule_name):
+    assert all(
+        run_distributed(
+            _pp2_schedule_trainer, world_size=2, args=(schedule_name,), timeout=240
+        )
+    )
+
+
+def _pp2_ckpt_resume(rank, world_size, save_dir):
+    """DCP save/resume under PP: per-stage optimizer state must round-trip
+    (regression: un-qualified optimizer fqns broke DCP's global planner)."""
+    from d9d_amd.loop.config import PipeliningConfig
+    from d9d_amd.pipelining.factory import PipelineSchedule1F1BConfig
+
+    params = Qwen3DenseModelParameters.tiny()
+
+    def build(steps):
+        cfg = _make_config(total_steps=steps, save_dir=save_dir)
+        cfg = cfg.model_copy(
+            update={"pipelining": PipeliningConfig(schedule=PipelineSchedule1F1BConfig())}
+        )
+        mesh = DeviceMeshParameters(pipeline_parallel=2)
+        return TrainingConfigurator(
+            cfg,
+            mesh,
+            _LMModelProvider(params),
+            _LMDatasetProvider(params),
+            AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+            AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+            _LMTask(),
+        ).configure(device_type="cpu")
+
+    build(2).train()      # saves at step 2
+    build(4).train()      # resumes from save-2, continues to 4
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_pp2_checkpoint_resume(tmp_path):
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as d:
+        assert all(
+            run_distributed(_pp2_ckpt_resume, world_size=2, args=(d,), timeout=300)
+        )
