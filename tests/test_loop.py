@@ -513,3 +513,31 @@ def test_trainer_pp2_checkpoint_resume(tmp_path):
         assert all(
             run_distributed(_pp2_ckpt_resume, world_size=2, args=(d,), timeout=300)
         )
+
+
+def _fsdp2_trainer_resume(rank, world_size, save_dir):
+    """FSDP2-sharded trainer with DCP save + resume (sharded DTensor state)."""
+    from d9d_amd.parallel import parallelize_fsdp
+
+    def par(module, ctx):
+        units = None
+        if hasattr(module.model, "layers"):
+            units = [l for l in module.model.layers.values()]
+        return parallelize_fsdp(
+            module, ctx.mesh_for("dense")["dp_cp_shard"], shard_units=units
+        )
+
+    mesh = DeviceMeshParameters(data_parallel_shard=2)
+    _build_trainer(total_steps=2, parallelize=par, mesh=mesh, save_dir=save_dir).train()
+    _build_trainer(total_steps=4, parallelize=par, mesh=mesh, save_dir=save_dir).train()
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_fsdp2_checkpoint_resume():
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as d:
+        assert all(
+            run_distributed(_fsdp2_trainer_resume, world_size=2, args=(d,), timeout=300)
+        )
