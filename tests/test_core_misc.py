@@ -165,3 +165,44 @@ def test_sdpa_env_backend_selection(monkeypatch):
     assert getattr(backend, "__name__", "") == "_eager"
     monkeypatch.delenv(SDPA_ENV_VAR)
     assert getattr(build_sdpa_backend(None), "__name__", "") == "_cdna4_flash"
+
+
+def test_jsonl_tracker_roundtrip(tmp_path):
+    import json
+
+    from d9d_amd.tracker import JsonlTracker
+
+    tracker = JsonlTracker(str(tmp_path))
+    run = tracker.new_run("exp1", "desc")
+    run.set_step(1)
+    run.set_context(phase="train")
+    run.scalar("loss", 1.5)
+    run.set_step(2)
+    run.scalar("loss", 1.2)
+    run.bins("hist", [1.0, 2.0])
+    run.hparams({"lr": 0.1})
+    run.close()
+
+    files = list(tmp_path.glob("*.jsonl"))
+    assert files, "tracker wrote no file"
+    rows = [json.loads(l) for l in files[0].read_text().splitlines()]
+    losses = [r for r in rows if r.get("name") == "loss"]
+    assert len(losses) == 2 and losses[1]["step"] == 2
+
+    sd = tracker.state_dict()
+    tracker2 = JsonlTracker(str(tmp_path))
+    tracker2.load_state_dict(sd)
+
+
+def test_profiling_wrapper_smoke(tmp_path):
+    import torch
+
+    from d9d_amd.internals.profiling import Profiler
+
+    prof = Profiler(tmp_path, rank_tag="r0", wait=0, warmup=1, active=1)
+    prof.open()
+    for _ in range(3):
+        torch.randn(8, 8) @ torch.randn(8, 8)
+        prof.step()
+    prof.close()
+    assert any(tmp_path.glob("trace_r0_*.json.tar.gz")), "no trace written"
