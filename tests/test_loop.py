@@ -541,3 +541,35 @@ def test_trainer_fsdp2_checkpoint_resume():
         assert all(
             run_distributed(_fsdp2_trainer_resume, world_size=2, args=(d,), timeout=300)
         )
+
+
+def _hsdp_trainer(rank, world_size):
+    """ws=4 trainer: HSDP (dp_replicate=2 x dp_shard=2)."""
+    from d9d_amd.parallel import parallelize_hsdp
+
+    def par(module, ctx):
+        mesh = ctx.mesh_for("dense")[("dp_replicate", "dp_cp_shard")]
+        units = [l for l in module.model.layers.values()]
+        return parallelize_hsdp(module, mesh, shard_units=units)
+
+    params = Qwen3DenseModelParameters.tiny()
+    cfg = TrainerConfig(
+        batching=BatchingConfig(global_batch_size=16, microbatch_size=2),
+        total_steps=2,
+    )
+    mesh = DeviceMeshParameters(data_parallel_replicate=2, data_parallel_shard=2)
+    TrainingConfigurator(
+        cfg,
+        mesh,
+        _LMModelProvider(params, par),
+        _LMDatasetProvider(params),
+        AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+        AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+        _LMTask(),
+    ).configure(device_type="cpu").train()
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_hsdp_ws4():
+    assert all(run_distributed(_hsdp_trainer, world_size=4, timeout=240))
