@@ -79,13 +79,26 @@ class GradientManager:
     def add_loss_weight(self, weight: float) -> None:
         self._loss_weight_total += weight
 
+    @property
+    def loss_weight_total(self) -> float:
+        return self._loss_weight_total
+
     def sync_and_scale(self, group=None) -> None:
         """Wait grad comms; scale grads by 1/total weight."""
         if self._sync is not None:
             self._sync.wait()
         total = self._loss_weight_total
         if dist.is_initialized() and group is not None:
-            t = torch.tensor([total])
+            # The dp group is RCCL on GPU runs: the reduce tensor must live on
+            # the same device as the gradients or the collective rejects it.
+            device = None
+            for _, p in self._named_params:
+                if p.grad is not None:
+                    device = p.grad.device
+                    break
+                if device is None:
+                    device = p.device
+            t = torch.tensor([total], dtype=torch.float64, device=device)
             dist.all_reduce(t, group=group)
             total = t.item()
         if total > 0:

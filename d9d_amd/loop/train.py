@@ -279,6 +279,7 @@ class Trainer:
         self.model_provider: ModelProvider = kw["model_provider"]
         self._run = None
         self.last_losses: list[float] = []
+        self.last_loss_weight: float = 0.0
 
     # -- checkpoint state schema (reference: loop/state.py:29-150) ------------
 
@@ -320,6 +321,9 @@ class Trainer:
         with self.bus.bounded(ev.TRAIN_FORWARD_BACKWARD_PRE, ev.TRAIN_FORWARD_BACKWARD_POST):
             losses = schedule.step(inputs, loss_fn=loss_fn)
         self.last_losses = [float(l) for l in losses]
+        # this rank's Σweight for the step — captured before sync_and_scale
+        # resets it, so logging can report a per-unit-weight loss
+        self.last_loss_weight = self.grad_manager.loss_weight_total
 
     def train(self) -> None:
         resumed = self.checkpointer.load_last(self._job_state())
@@ -372,7 +376,10 @@ class Trainer:
         if self._run is not None:
             self._run.set_step(step)
             if self.last_losses:
-                self._run.scalar("loss", sum(self.last_losses) / len(self.last_losses))
+                # losses are weight-scaled (loss*weight per microbatch):
+                # divide by the step's Σweight for a per-unit (per-token) loss
+                denom = self.last_loss_weight if self.last_loss_weight > 0 else len(self.last_losses)
+                self._run.scalar("loss", sum(self.last_losses) / denom)
             if grad_norm is not None:
                 self._run.scalar("l2_grad_norm_total", grad_norm)
             self._run.scalar("lr", self.lr_scheduler.get_last_lr()[0])

@@ -95,7 +95,10 @@ class PiecewiseLRScheduler:
         self._apply()
 
     def _apply(self) -> None:
-        m = self._multiplier(self._step)
+        # The trainer steps the scheduler AFTER optimizer.step(), so optimizer
+        # step k (1-indexed) runs at multiplier(k): with the auto warmup phase
+        # Linear(0, 1) the first step trains at 1/warmup_steps instead of 0.
+        m = self._multiplier(self._step + 1)
         for group, base in zip(self.optimizer.param_groups, self._base_lrs):
             group["lr"] = base * m
 

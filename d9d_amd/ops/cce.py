@@ -152,9 +152,17 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
                 pb = p.to(c.dtype)
             de[sl] = torch.matmul(pb, c).to(e.dtype)
             if single:
-                return de, torch.matmul(pb.t(), e_chunk.to(c.dtype)), None, None, None, None, None
+                dc = torch.matmul(pb.t(), e_chunk.to(c.dtype))
+                if vp_group is not None:
+                    # de from the local vocab shard is a partial sum over
+                    # vocab: full de = sum_r pb_r @ c_r (reference
+                    # reduce_e_grad, d9d/kernel/cce/cce.py:190-198).
+                    dist.all_reduce(de, group=vp_group)
+                return de, dc, None, None, None, None, None
             dc += torch.matmul(pb.t(), e_chunk.to(c.dtype)).float()
 
+        if vp_group is not None:
+            dist.all_reduce(de, group=vp_group)
         return de, dc.to(c.dtype), None, None, None, None, None
 
 

@@ -43,6 +43,51 @@ class _CopyToTP(torch.autograd.Function):
         return grad, None
 
 
+class _SumGradFromTP(torch.autograd.Function):
+    """fwd identity; bwd all-reduce — for params replicated inside the tp
+    region (e.g. q/k norm weights) whose per-microbatch grads are partial."""
+
+    @staticmethod
+    def forward(ctx, w, group):
+        ctx.group = group
+        return w.view_as(w)
+
+    @staticmethod
+    def backward(ctx, grad):
+        grad = grad.contiguous()
+        dist.all_reduce(grad, group=ctx.group)
+        return grad, None
+
+
+_SUMGRAD_CLASS_CACHE: dict[tuple, type] = {}
+
+
+def _install_tp_sumgrad(module: nn.Module, param_names: tuple[str, ...], group) -> None:
+    """Swap `module`'s class so each named param reads through
+    `_SumGradFromTP` during grad-enabled forward (identity otherwise)."""
+    module._d9d_tp_group = group
+
+    def _make_getter(name: str):
+        def getter(self):
+            p = self._parameters.get(name)
+            if p is None:
+                raise AttributeError(name)
+            if torch.is_grad_enabled() and p.requires_grad:
+                return _SumGradFromTP.apply(p, self._d9d_tp_group)
+            return p
+
+        return property(getter)
+
+    key = (type(module), param_names)
+    if key not in _SUMGRAD_CLASS_CACHE:
+        ns = {name: _make_getter(name) for name in param_names}
+        ns["_d9d_tp_sumgrad_params"] = param_names
+        _SUMGRAD_CLASS_CACHE[key] = type(
+            f"TPSumGrad{type(module).__name__}", (type(module),), ns
+        )
+    module.__class__ = _SUMGRAD_CLASS_CACHE[key]
+
+
 class _ReduceFromTP(torch.autograd.Function):
     """fwd all-reduce (partial rowwise outputs); bwd identity."""
 
@@ -159,13 +204,15 @@ def parallelize_tp_attention(
     # activations, so their gradients are partial: sum them over tp.
     # (q_norm/k_norm weights are shared across heads; learnable sinks are
     # per-head and follow the q-head shard instead.)
-    def _allreduce_grad(p):
-        if p.grad is not None:
-            dist.all_reduce(p.grad, group=group)
-
+    #
+    # The reduction lives IN the autograd graph (`_SumGradFromTP` wraps the
+    # weight each forward), so each microbatch's partial grad is reduced
+    # exactly once before it accumulates into .grad — a post-accumulate hook
+    # would re-reduce earlier microbatches' already-summed contributions
+    # under gradient accumulation.
     for norm in (attn.q_norm, attn.k_norm):
         if norm is not None:
-            norm.weight.register_post_accumulate_grad_hook(_allreduce_grad)
+            _install_tp_sumgrad(norm, ("weight",), group)
     if attn.sinks is not None:
         rank = dist.get_rank(group)
         shard = attn.sinks.shape[0] // tp_size
