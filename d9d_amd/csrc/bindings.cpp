@@ -39,7 +39,8 @@ torch::Tensor router_topk_bwd(torch::Tensor logits, torch::Tensor top_idx, torch
                               bool renormalize);
 
 torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse, torch::Tensor targets,
-                           torch::Tensor dl, int64_t vocab_start, int64_t ignore_index,
+                           torch::Tensor dl, c10::optional<torch::Tensor> dlse,
+                           int64_t vocab_start, int64_t ignore_index,
                            double filter_eps);
 
 void adamw_stochastic_bf16_multi_(

@@ -277,12 +277,16 @@ __global__ void cce_dlogits_kernel(
     const float* __restrict__ lse,     // (R,)
     const int64_t* __restrict__ targets,  // (R,) GLOBAL vocab ids (or ignore)
     const float* __restrict__ dl,      // (R,) upstream grad (0 for ignored)
+    const float* __restrict__ dlse,    // (R,) upstream lse grad, or nullptr
     int64_t R, int64_t V, int64_t vocab_start, int64_t ignore_index,
     float filter_eps) {  // zero non-target dlogits with p < eps (<=0: off)
   const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (int64_t row = blockIdx.y; row < R; row += gridDim.y) {
   const float l = lse[row];
   const float g = dl[row];
+  // d(lse)/d(logit) = p: a differentiable lse output simply adds its grad
+  // to the softmax term's scale (loss term keeps -onehot * g).
+  const float gp = g + (dlse ? dlse[row] : 0.f);
   const int64_t tgt_global = targets[row];
   const int64_t tgt =
       (tgt_global == ignore_index) ? -1 : tgt_global - vocab_start;
@@ -299,8 +303,8 @@ __global__ void cce_dlogits_kernel(
             (bf16_bits_to_f32(x.s[j]) - l) * 1.44269504089f);
         const bool is_tgt = base + j == tgt;
         if (p < filter_eps && !is_tgt) p = 0.f;
-        p -= is_tgt ? 1.f : 0.f;
-        x.s[j] = f32_to_bf16_rne(p * g);
+        float out = p * gp - (is_tgt ? g : 0.f);
+        x.s[j] = f32_to_bf16_rne(out);
       }
       *reinterpret_cast<ushort8v*>(rowp + base) = x.u;
     } else {
@@ -309,8 +313,8 @@ __global__ void cce_dlogits_kernel(
             (bf16_bits_to_f32(rowp[i]) - l) * 1.44269504089f);
         const bool is_tgt = i == tgt;
         if (p < filter_eps && !is_tgt) p = 0.f;
-        p -= is_tgt ? 1.f : 0.f;
-        rowp[i] = f32_to_bf16_rne(p * g);
+        float out = p * gp - (is_tgt ? g : 0.f);
+        rowp[i] = f32_to_bf16_rne(out);
       }
     }
   }
@@ -321,6 +325,7 @@ __global__ void cce_dlogits_kernel(
 
 torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse,
                            torch::Tensor targets, torch::Tensor dl,
+                           c10::optional<torch::Tensor> dlse,
                            int64_t vocab_start, int64_t ignore_index,
                            double filter_eps) {
   TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kBFloat16 &&
@@ -330,12 +335,19 @@ torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse,
   const int gx = (int)std::min<int64_t>((V + 256 * 8 - 1) / (256 * 8), 128);
   const unsigned gy = (unsigned)std::min<int64_t>(R, 4096 / gx + 1);
   auto stream = at::hip::getCurrentHIPStream();
+  torch::Tensor dlse_c;
+  const float* dlse_ptr = nullptr;
+  if (dlse.has_value()) {
+    dlse_c = dlse->contiguous();
+    dlse_ptr = dlse_c.data_ptr<float>();
+  }
   hipLaunchKernelGGL(d9d::cce_dlogits_kernel, dim3(gx, gy), dim3(256),
                      0, stream,
                      reinterpret_cast<ushort*>(logits.data_ptr()),
                      lse.contiguous().data_ptr<float>(),
                      targets.contiguous().data_ptr<int64_t>(),
-                     dl.contiguous().data_ptr<float>(), R, V, vocab_start,
+                     dl.contiguous().data_ptr<float>(), dlse_ptr,
+                     R, V, vocab_start,
                      ignore_index, (float)filter_eps);
   return logits;
 }

@@ -3,11 +3,18 @@
 Replaces the reference's vendored cut-cross-entropy
 (d9d/kernel/cce/cce.py:47-216): per-token loss = lse(e @ c^T) - logit[target],
 with the logits recomputed in vocab/row chunks in backward. Matmuls run on
-MFMA via rocBLAS (library GEMM); a fully fused HIP kernel is the follow-up.
+MFMA via rocBLAS (library GEMM); the hot fwd/dlogits passes are fused HIP
+kernels (csrc/cce.hip).
 
 Supports vocab-parallel reduction (VocabParallelOptions analog): pass a
 process group and each rank's local vocab shard; lse is all-reduced via
-logsumexp-merge and the target logit summed (exactly one rank owns a target).
+logsumexp-merge, the target logit summed (exactly one rank owns a target),
+and the embedding grad all-reduced in backward (reference reduce_e_grad).
+
+The `lse` output is a REAL autograd output (d(lse)/d(logit) = softmax adds
+its upstream grad to the dlogits scale), so distillation-style losses can
+differentiate through it. `softcap` applies z = cap * tanh(logit / cap)
+before the softmax (reference kernel/cce/main.py:59).
 """
 
 from dataclasses import dataclass
@@ -33,16 +40,18 @@ class VocabParallelOptions:
     vocab_end: int
 
 
-def _chunk_fwd(e32, c, targets, vocab_start):
+def _chunk_fwd(e32, c, targets, vocab_start, softcap=None):
     """Return (lse, target_logit) for a row chunk; target_logit=0 for ignored/out-of-shard."""
-    logits = torch.matmul(e32, c.t().to(e32.dtype))  # (Tc, V_local)
-    lse = torch.logsumexp(logits.float(), dim=-1)
+    logits = torch.matmul(e32, c.t().to(e32.dtype)).float()  # (Tc, V_local)
+    if softcap is not None:
+        logits = torch.tanh(logits / softcap) * softcap
+    lse = torch.logsumexp(logits, dim=-1)
     local_targets = targets - vocab_start
     in_shard = (local_targets >= 0) & (local_targets < logits.shape[-1]) & (
         targets != LM_IGNORE_INDEX
     )
     safe = local_targets.clamp(0, logits.shape[-1] - 1)
-    tgt_logit = logits.float().gather(1, safe.unsqueeze(1)).squeeze(1)
+    tgt_logit = logits.gather(1, safe.unsqueeze(1)).squeeze(1)
     tgt_logit = torch.where(in_shard, tgt_logit, torch.zeros_like(tgt_logit))
     return lse, tgt_logit
 
@@ -73,20 +82,20 @@ def _can_use_kernel(e, c):
 
 class _LinearCrossEntropyFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, e, c, targets, vp_group, vocab_start, vocab_end, filter_eps):
+    def forward(ctx, e, c, targets, vp_group, vocab_start, vocab_end, filter_eps, softcap):
         T = e.shape[0]
-        if _can_use_kernel(e, c):
+        if softcap is None and _can_use_kernel(e, c):
             lse, tgt_logit = _kernel_forward(e, c, targets, vocab_start)
         else:
             lse_parts = []
             tgt_parts = []
             for s in range(0, T, _ROW_CHUNK):
                 sl = slice(s, min(s + _ROW_CHUNK, T))
-                lse_c, tgt_c = _chunk_fwd(e[sl], c, targets[sl], vocab_start)
+                lse_c, tgt_c = _chunk_fwd(e[sl], c, targets[sl], vocab_start, softcap)
                 lse_parts.append(lse_c)
                 tgt_parts.append(tgt_c)
-            lse = torch.cat(lse_parts)
-            tgt_logit = torch.cat(tgt_parts)
+            lse = torch.cat(lse_parts) if lse_parts else e.new_zeros(0, dtype=torch.float32)
+            tgt_logit = torch.cat(tgt_parts) if tgt_parts else e.new_zeros(0, dtype=torch.float32)
 
         if vp_group is not None:
             # Merge lse across vocab shards: lse_full = log sum_r exp(lse_r).
@@ -105,21 +114,29 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
         ctx.save_for_backward(e, c, targets, lse)
         ctx.vp = (vp_group, vocab_start, vocab_end)
         ctx.filter_eps = filter_eps
-        return loss
+        ctx.softcap = softcap
+        ctx.set_materialize_grads(False)
+        return loss, lse
 
     @staticmethod
-    def backward(ctx, dloss):
+    def backward(ctx, dloss, dlse):
         e, c, targets, lse = ctx.saved_tensors
         vp_group, vocab_start, _ = ctx.vp
+        softcap = ctx.softcap
         T, H = e.shape
         V = c.shape[0]
         de = torch.empty_like(e)
         dc = None
         ignored = targets == LM_IGNORE_INDEX
-        dl = torch.where(ignored, torch.zeros_like(dloss), dloss).float()
+        if dloss is None:
+            dl = torch.zeros(T, dtype=torch.float32, device=e.device)
+        else:
+            dl = torch.where(ignored, torch.zeros_like(dloss), dloss).float()
+        dlse_f = dlse.float() if dlse is not None else None
 
         bf16_fast = (
             e.is_cuda and e.dtype == torch.bfloat16 and c.dtype == torch.bfloat16
+            and softcap is None and has_ext()
         )
         chunk = _ROW_CHUNK
         single = chunk >= T
@@ -133,23 +150,32 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
             in_shard = (local_targets >= 0) & (local_targets < V) & (~ignored[sl])
             safe = local_targets.clamp(0, V - 1)
             if bf16_fast:
-                # fused in-place dlogits kernel: p = exp(logit - lse) - onehot,
-                # scaled by the row grad, fp32 math, one pass over (Tc, V)
+                # fused in-place dlogits kernel: p = exp(logit - lse), scaled
+                # by (dl + dlse) with the one-hot target column getting -dl;
+                # fp32 math, one pass over (Tc, V)
                 pb = get_ext().cce_dlogits_(
                     logits, lse[sl].float(), targets[sl], dl[sl],
+                    dlse_f[sl] if dlse_f is not None else None,
                     vocab_start, LM_IGNORE_INDEX, ctx.filter_eps,
                 )
             else:
-                p = torch.exp(logits.float() - lse[sl].unsqueeze(1))
-                p.scatter_add_(
+                z = logits.float()
+                if softcap is not None:
+                    z = torch.tanh(z / softcap) * softcap
+                p = torch.exp(z - lse[sl].unsqueeze(1))
+                scale = dl[sl]
+                if dlse_f is not None:
+                    scale = scale + dlse_f[sl]
+                grad_z = p * scale.unsqueeze(1)
+                grad_z.scatter_add_(
                     1, safe.unsqueeze(1),
                     torch.where(
-                        in_shard, -torch.ones_like(safe, dtype=p.dtype),
-                        torch.zeros_like(safe, dtype=p.dtype),
+                        in_shard, -dl[sl], torch.zeros_like(dl[sl])
                     ).unsqueeze(1),
                 )
-                p *= dl[sl].unsqueeze(1)
-                pb = p.to(c.dtype)
+                if softcap is not None:
+                    grad_z = grad_z * (1.0 - (z / softcap) ** 2)
+                pb = grad_z.to(c.dtype)
             de[sl] = torch.matmul(pb, c).to(e.dtype)
             if single:
                 dc = torch.matmul(pb.t(), e_chunk.to(c.dtype))
@@ -158,12 +184,14 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
                     # vocab: full de = sum_r pb_r @ c_r (reference
                     # reduce_e_grad, d9d/kernel/cce/cce.py:190-198).
                     dist.all_reduce(de, group=vp_group)
-                return de, dc, None, None, None, None, None
+                return de, dc, None, None, None, None, None, None
             dc += torch.matmul(pb.t(), e_chunk.to(c.dtype)).float()
 
         if vp_group is not None:
             dist.all_reduce(de, group=vp_group)
-        return de, dc.to(c.dtype), None, None, None, None, None
+        if dc is None:  # T == 0
+            dc = torch.zeros_like(c)
+        return de, dc.to(c.dtype), None, None, None, None, None, None
 
 
 def linear_cross_entropy(
@@ -176,6 +204,7 @@ def linear_cross_entropy(
     reduction: str = "none",
     return_lse: bool = False,
     filter_eps: float | str | None = "auto",
+    softcap: float | None = None,
 ):
     """Per-token negative log-likelihood, zeros at ignored positions.
 
@@ -184,7 +213,8 @@ def linear_cross_entropy(
     {"none", "mean", "sum"} ("mean" averages over non-ignored tokens);
     `filter_eps` zeroes negligible non-target probabilities in the backward
     ("auto" = bf16-epsilon-scaled threshold, None/0 disables). `return_lse`
-    additionally returns the per-token logsumexp.
+    additionally returns the per-token logsumexp as a DIFFERENTIABLE output.
+    `softcap` applies cap * tanh(logit / cap) before the softmax.
     """
     if shift:
         n = 1 if shift is True else int(shift)
@@ -197,14 +227,14 @@ def linear_cross_entropy(
     else:
         eps = float(filter_eps)
     if vocab_parallel is None:
-        loss = _LinearCrossEntropyFunction.apply(
-            embeddings, classifier, targets, None, 0, classifier.shape[0], eps
+        loss, lse = _LinearCrossEntropyFunction.apply(
+            embeddings, classifier, targets, None, 0, classifier.shape[0], eps, softcap
         )
     else:
-        loss = _LinearCrossEntropyFunction.apply(
+        loss, lse = _LinearCrossEntropyFunction.apply(
             embeddings, classifier, targets,
             vocab_parallel.group, vocab_parallel.vocab_start,
-            vocab_parallel.vocab_end, eps,
+            vocab_parallel.vocab_end, eps, softcap,
         )
     if reduction == "mean":
         valid = (targets != LM_IGNORE_INDEX).sum().clamp_min(1)
@@ -212,34 +242,5 @@ def linear_cross_entropy(
     elif reduction == "sum":
         loss = loss.sum()
     if return_lse:
-        # lse is returned detached (diagnostic use, matching how the
-        # reference's lse output is consumed); the loss path carries grads.
-        with torch.no_grad():
-            if vocab_parallel is None:
-                lse = _lse_only(embeddings, classifier, targets)
-            else:
-                lse = _lse_only(
-                    embeddings, classifier, targets,
-                    vocab_parallel.group, vocab_parallel.vocab_start,
-                )
         return loss, lse
     return loss
-
-
-@torch.no_grad()
-def _lse_only(e, c, targets, vp_group=None, vocab_start=0):
-    if _can_use_kernel(e, c):
-        lse, _ = _kernel_forward(e, c, targets, vocab_start)
-    else:
-        parts = []
-        for s in range(0, e.shape[0], _ROW_CHUNK):
-            sl = slice(s, min(s + _ROW_CHUNK, e.shape[0]))
-            lse_c, _ = _chunk_fwd(e[sl], c, targets[sl], vocab_start)
-            parts.append(lse_c)
-        lse = torch.cat(parts)
-    if vp_group is not None:
-        world = dist.get_world_size(vp_group)
-        all_lse = torch.empty(world * lse.numel(), dtype=lse.dtype, device=lse.device)
-        dist.all_gather_into_tensor(all_lse, lse.contiguous(), group=vp_group)
-        lse = torch.logsumexp(all_lse.view(world, -1), dim=0)
-    return lse

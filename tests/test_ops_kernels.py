@@ -334,6 +334,63 @@ def test_cce_api_shift_reduction_lse():
     torch.testing.assert_close(lse, torch.logsumexp(logits, -1), rtol=1e-4, atol=1e-5)
 
 
+def test_cce_lse_differentiable():
+    """The lse output carries gradients (distillation-style losses)."""
+    import torch
+
+    from d9d_amd.ops.cce import linear_cross_entropy
+
+    torch.manual_seed(5)
+    T, H, V = 8, 16, 24
+    e = torch.randn(T, H, requires_grad=True)
+    c = torch.randn(V, H, requires_grad=True) * 0.1
+    c.retain_grad()
+    tg = torch.randint(0, V, (T,))
+
+    loss, lse = linear_cross_entropy(e, c, tg, return_lse=True)
+    mix = loss.sum() + 0.7 * (lse ** 2).sum()
+    mix.backward()
+
+    e_ref = e.detach().clone().requires_grad_(True)
+    c_ref = c.detach().clone().requires_grad_(True)
+    logits = e_ref @ c_ref.t()
+    lse_ref = torch.logsumexp(logits, -1)
+    loss_ref = lse_ref - logits.gather(1, tg.unsqueeze(1)).squeeze(1)
+    (loss_ref.sum() + 0.7 * (lse_ref ** 2).sum()).backward()
+
+    torch.testing.assert_close(e.grad, e_ref.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(c.grad, c_ref.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_cce_softcap():
+    """softcap applies cap*tanh(logit/cap) before softmax, fwd + bwd."""
+    import torch
+
+    from d9d_amd.ops.cce import linear_cross_entropy
+
+    torch.manual_seed(6)
+    T, H, V, cap = 8, 16, 24, 5.0
+    e = torch.randn(T, H, requires_grad=True) * 2
+    e.retain_grad()
+    c = torch.randn(V, H, requires_grad=True)
+    c.retain_grad()
+    tg = torch.randint(0, V, (T,))
+    tg[0] = -100
+
+    loss = linear_cross_entropy(e, c, tg, softcap=cap, filter_eps=None)
+    loss.sum().backward()
+
+    e_ref = e.detach().clone().requires_grad_(True)
+    c_ref = c.detach().clone().requires_grad_(True)
+    z = torch.tanh((e_ref @ c_ref.t()) / cap) * cap
+    ref = torch.nn.functional.cross_entropy(z, tg, ignore_index=-100, reduction="none")
+    assert loss.shape == ref.shape
+    torch.testing.assert_close(loss, ref, rtol=1e-4, atol=1e-5)
+    ref.sum().backward()
+    torch.testing.assert_close(e.grad, e_ref.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(c.grad, c_ref.grad, rtol=1e-4, atol=1e-5)
+
+
 @pytest.mark.gpu
 def test_adamw_multi_matches_single():
     import torch
