@@ -1,0 +1,18 @@
+import sys, pathlib
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
+import torch, time, statistics
+from d9d_amd.ops import _ext
+ext = _ext.get_ext()
+torch.manual_seed(0)
+def bench_nt(K, N, T=262144, E=128):
+    sizes = torch.full((E,), T // E, dtype=torch.int64)
+    a = torch.randn(T, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(E, N, K, dtype=torch.bfloat16, device="cuda") * 0.05
+    ext.gmm_nt(a, w, sizes); torch.cuda.synchronize()
+    ts = []
+    for _ in range(10):
+        torch.cuda.synchronize(); t0 = time.perf_counter(); ext.gmm_nt(a, w, sizes); torch.cuda.synchronize(); ts.append(time.perf_counter()-t0)
+    t = statistics.median(ts)
+    print(f"gmm_nt K={K} N={N}: {t*1e3:.2f} ms {2.0*T*K*N/t/1e12:.0f} TF/s", flush=True)
+for K, N in ((768,576),(768,768),(768,1536),(768,2048),(288,768),(2048,1536)):
+    bench_nt(K, N)

@@ -567,6 +567,243 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_glds_kernel(
   }
 }
 
+// 8-phase counted-vmcnt gmm_nt (the guide's verified 256-wide template
+// adapted to grouped ragged rows), templated on the n-tile width BN8
+// (256 or 192 so the production shapes N=576/768/1536/2048 tile exactly).
+// 256(M) x BN8(N) tile, BK=64, 8 waves as 2(M) x 4(N), per-wave
+// 128 x BN8/4 output. LDS holds exactly TWO k-tiles
+// (2 x (A 32KB + B BN8*128B)); staging is global_load_lds_dwordx4:
+// A in 128-row halves (2 glds/thread), B in 64-row chunks (1 glds/thread),
+// issued inside the compute phases so 2-3 staging units stay in flight
+// behind ONE counted s_waitcnt vmcnt(NJ) per k-tile (NJ = BN8/64).
+// Each k-tile runs as 4 phases: phase p ds_reads the A fragments of
+// C-quadrant p (i in {2p, 2p+1}; B fragments for the whole k-tile are read
+// once at phase 0 and live in registers), stages its share of upcoming
+// tiles, raw-barriers, and issues 4*NJ MFMAs.
+// Death/arrival schedule (slot parity = kt & 1):
+//   A(kt+1) goes to the OTHER slot whose tenant A(kt-1) is fully dead
+//     -> staged at p0, p1 of kt (one half each)
+//   B(kt) region (this slot) dies after phase 0 (all B frags read there)
+//     -> B chunks of kt+2 staged at p1..p3
+// Per-thread issue order makes the boundary wait vmcnt(NJ): the NJ
+// outstanding loads are B chunks of kt+1; everything older has landed.
+template <int BN8>
+__global__ __launch_bounds__(512, 1) void gmm_nt_8phase_kernel(
+    const bf16_t* __restrict__ a,    // (T, K)
+    const bf16_t* __restrict__ w,    // (E, N, K)
+    bf16_t* __restrict__ out,        // (T, N)
+    const int* __restrict__ row_off,
+    const int* __restrict__ mtile_pref,
+    int E, int K, int N, int n_tiles) {
+  constexpr int NJ = BN8 / 64;           // B frags per wave = B chunks
+  constexpr int SLOT = (256 + BN8) * 128;  // bytes per k-tile slot
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // slot s (s = kt & 1): A at smem + s*SLOT (32 KB), B right after
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  const int nwg = gridDim.x;
+  const int xcd = blockIdx.x & 7;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int wid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                  + (blockIdx.x >> 3);
+  const int mt_global = wid / n_tiles;
+  int lo = 0, hi = E - 1;
+  while (lo < hi) {
+    const int mid = (lo + hi + 1) >> 1;
+    if (mtile_pref[mid] <= mt_global) lo = mid; else hi = mid - 1;
+  }
+  const int e = lo;
+  const int m_tile = mt_global - mtile_pref[e];
+  const int row0 = row_off[e] + m_tile * kBM;
+  const int row_end = row_off[e + 1];
+  const int n0 = (wid % n_tiles) * BN8;
+
+  const bf16_t* w_e = w + (int64_t)e * N * K;
+
+  f32x4 acc[8][NJ];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // K % 64 == 32 tail: the last k-tile stages the OVERLAPPING window
+  // [K-64, K) (all loads in bounds, no masking -- glds cannot mask) and the
+  // MFMA phases run only its upper half (ks = 1), so columns [K-64, K-32)
+  // are not double-counted. Host guarantees K % 32 == 0.
+  const int n_ktiles = (K + kBK - 1) / kBK;
+  const bool k_tail = (K % kBK) != 0;
+  const int last_kt = n_ktiles - 1;
+
+  auto ktile_k0 = [&](int kt) {
+    return (k_tail && kt == last_kt) ? K - kBK : kt * kBK;
+  };
+
+  // A half (128 rows = 16 KB): 2 glds per thread.
+  auto stage_a = [&](int kt, int half) {
+    const int k0 = ktile_k0(kt);
+    char* base = smem + (size_t)(kt & 1) * SLOT + half * (16 * 1024);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int chunk = wave + i * 8;           // 16 chunks of 1 KB
+      const int l = chunk * 1024 + lane * 16;   // linear LDS byte in half
+      const int arow = half * 128 + (l >> 7);   // tile row (128-B rows)
+      const int colb = (l & 127) ^ ((arow & 7) << 4);  // source pre-swizzle
+      const int64_t sr = min(row0 + arow, row_end - 1);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const uint32_t*>(
+              reinterpret_cast<const char*>(a) + (sr * (int64_t)K + k0) * 2 + colb),
+          reinterpret_cast<uint32_t*>(base + l), 16, 0, 0);
+    }
+  };
+  // B chunk (64 rows = 8 KB): 1 glds per thread.
+  auto stage_b = [&](int kt, int chunk) {
+    const int k0 = ktile_k0(kt);
+    char* base = smem + (size_t)(kt & 1) * SLOT + (32 * 1024)
+                 + chunk * (8 * 1024);
+    const int l = wave * 1024 + lane * 16;
+    const int nrow = chunk * 64 + (l >> 7);
+    const int colb = (l & 127) ^ ((nrow & 7) << 4);
+    const int64_t sn = min(n0 + nrow, N - 1);
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const uint32_t*>(
+            reinterpret_cast<const char*>(w_e) + (sn * (int64_t)K + k0) * 2 + colb),
+        reinterpret_cast<uint32_t*>(base + l), 16, 0, 0);
+  };
+
+  // Prologue: issue order matches the steady-state boundary count.
+#pragma unroll
+  for (int c = 0; c < NJ; ++c) stage_b(0, c);
+  stage_a(0, 0); stage_a(0, 1);
+  if (n_ktiles > 1) {
+#pragma unroll
+    for (int c = 0; c < NJ; ++c) stage_b(1, c);
+  }
+
+  const int n_full = k_tail ? n_ktiles - 1 : n_ktiles;
+  for (int kt = 0; kt < n_full; ++kt) {
+    const char* a_base = smem + (size_t)(kt & 1) * SLOT;
+    const char* b_base = a_base + 32 * 1024;
+
+    // k-tile boundary: this tile's staging landed; <= NJ newer in flight.
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NJ) : "memory");
+    __builtin_amdgcn_s_barrier();
+
+    bf16x8 b_frag[NJ][2];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      // -- ds_read this phase's fragments ---------------------------------
+      bf16x8 a_frag[2][2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kk = ks * 32 + (lane >> 4) * 8;
+#pragma unroll
+        for (int il = 0; il < 2; ++il) {
+          const int row = wm * 128 + (2 * p + il) * 16 + (lane & 15);
+          const int byte = (kk * 2) ^ ((row & 7) << 4);
+          a_frag[il][ks] = *reinterpret_cast<const bf16x8*>(
+              a_base + row * (kBK * 2) + byte);
+        }
+        if (p == 0) {
+#pragma unroll
+          for (int j = 0; j < NJ; ++j) {
+            const int col = wn * (BN8 / 4) + j * 16 + (lane & 15);
+            const int byte = (kk * 2) ^ ((col & 7) << 4);
+            b_frag[j][ks] = *reinterpret_cast<const bf16x8*>(
+                b_base + col * (kBK * 2) + byte);
+          }
+        }
+      }
+      // -- stage upcoming tiles into dead rows ----------------------------
+      if (p == 0) {
+        if (kt + 1 < n_ktiles) stage_a(kt + 1, 0);
+      } else if (p == 1) {
+        if (kt + 1 < n_ktiles) stage_a(kt + 1, 1);
+        if (kt + 2 < n_ktiles) {
+          stage_b(kt + 2, 0);
+          if (NJ == 4) stage_b(kt + 2, 1);
+        }
+      } else if (p == 2) {
+        if (kt + 2 < n_ktiles) {
+          if (NJ == 4) { stage_b(kt + 2, 2); stage_b(kt + 2, 3); }
+          else stage_b(kt + 2, 1);
+        }
+      } else {
+        if (NJ == 3 && kt + 2 < n_ktiles) stage_b(kt + 2, 2);
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int il = 0; il < 2; ++il)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[2 * p + il][j] =
+                mfma16g(a_frag[il][ks], b_frag[j][ks], acc[2 * p + il][j]);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // Peeled tail k-tile (K % 64 == 32): only its upper half (ks = 1)
+  // contributes -- the staged window [K-64, K) overlaps the previous tile.
+  if (k_tail) {
+    const int kt = last_kt;
+    const char* a_base = smem + (size_t)(kt & 1) * SLOT;
+    const char* b_base = a_base + 32 * 1024;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    bf16x8 b_frag_t[NJ];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      bf16x8 a_frag_t[2];
+      const int kk = 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int il = 0; il < 2; ++il) {
+        const int row = wm * 128 + (2 * p + il) * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((row & 7) << 4);
+        a_frag_t[il] = *reinterpret_cast<const bf16x8*>(
+            a_base + row * (kBK * 2) + byte);
+      }
+      if (p == 0) {
+#pragma unroll
+        for (int j = 0; j < NJ; ++j) {
+          const int col = wn * (BN8 / 4) + j * 16 + (lane & 15);
+          const int byte = (kk * 2) ^ ((col & 7) << 4);
+          b_frag_t[j] = *reinterpret_cast<const bf16x8*>(
+              b_base + col * (kBK * 2) + byte);
+        }
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int il = 0; il < 2; ++il)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j)
+          acc[2 * p + il][j] =
+              mfma16g(a_frag_t[il], b_frag_t[j], acc[2 * p + il][j]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) {
+#pragma unroll
+      for (int r2 = 0; r2 < 4; ++r2) {
+        const int row = row0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r2;
+        const int col = n0 + wn * (BN8 / 4) + j * 16 + (lane & 15);
+        if (row < row_end && col < N) {
+          out[(int64_t)row * N + col] = (bf16_t)acc[i][j][r2];
+        }
+      }
+    }
+  }
+}
+
 // db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N).
 // Grid (ceil(N/256), ceil(K/256), E): 256x256 tiles per workgroup -- 8 waves
 // as 2 (k-halves of 128) x 4 (n-quarters of 64) -- looping the expert's rows
@@ -793,14 +1030,68 @@ torch::Tensor gmm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor batch_sizes
       build_offsets(batch_sizes, a.device(), d9d::kBM);
   if (total_mtiles == 0) return out;
 
+  auto stream = at::hip::getCurrentHIPStream();
+  // 8-phase counted-vmcnt schedule (glds half-tile staging interleaved with
+  // the MFMA quadrants, one vmcnt(4) per k-tile): the phase-level interleave
+  // the plain 2-buffer glds kernel below was missing. Opt out with
+  // D9D_GMM_8PHASE=0.
+  static const bool use_8phase = []() {
+    const char* v = getenv("D9D_GMM_8PHASE");
+    return v == nullptr || v[0] != '0';
+  }();
+  if (use_8phase && K % 32 == 0 && K >= 64) {
+    // pick the n-tile with the least padded waste (ties -> measured faster).
+    // Production shapes: 576 -> 192 exact (654 vs 552 TF/s), 768 -> both
+    // exact (256 faster), 1536/2048 -> 256. D9D_GMM_NT_BN forces a tile.
+    static const int force_bn = []() {
+      const char* v = getenv("D9D_GMM_NT_BN");
+      return v ? atoi(v) : 0;
+    }();
+    const int pad256 = ((N + 255) / 256) * 256 - N;
+    const int pad192 = ((N + 191) / 192) * 192 - N;
+    bool use256 = pad256 <= pad192;  // ties -> 256 (measured: 729 vs 655
+                                     // TF/s at N=768, 807 vs 742 at 1536)
+    if (force_bn == 256) use256 = true;
+    if (force_bn == 192) use256 = false;
+    const int bn = use256 ? 256 : 192;
+    const int n_tiles8 = (N + bn - 1) / bn;
+    const dim3 grid8(n_tiles8 * total_mtiles);
+    const size_t smem8 = (size_t)2 * (256 + bn) * 128;
+    static bool attr_set = false;
+    if (!attr_set) {
+      hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&d9d::gmm_nt_8phase_kernel<256>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 128 * 1024);
+      hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&d9d::gmm_nt_8phase_kernel<192>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 112 * 1024);
+      attr_set = true;
+    }
+    if (use256) {
+      hipLaunchKernelGGL(d9d::gmm_nt_8phase_kernel<256>, grid8, dim3(512),
+                         smem8, stream,
+                         reinterpret_cast<const __bf16*>(a.data_ptr()),
+                         reinterpret_cast<const __bf16*>(w.data_ptr()),
+                         reinterpret_cast<__bf16*>(out.data_ptr()),
+                         row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
+                         E, K, N, n_tiles8);
+    } else {
+      hipLaunchKernelGGL(d9d::gmm_nt_8phase_kernel<192>, grid8, dim3(512),
+                         smem8, stream,
+                         reinterpret_cast<const __bf16*>(a.data_ptr()),
+                         reinterpret_cast<const __bf16*>(w.data_ptr()),
+                         reinterpret_cast<__bf16*>(out.data_ptr()),
+                         row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
+                         E, K, N, n_tiles8);
+    }
+    return out;
+  }
   const int n_tiles = (N + d9d::kBN - 1) / d9d::kBN;
   const dim3 grid(n_tiles * total_mtiles);
-  auto stream = at::hip::getCurrentHIPStream();
-  // glds staging measured 526 vs 609 TF/s for the register-T14 pipeline: at
-  // 2-deep tile prefetch one iteration of MFMAs (~0.3 us) cannot cover the
-  // HBM latency, so the counted vmcnt stalls every tile. The win the guide
-  // documents needs the finer 8-phase quadrant interleave (round-2 work);
-  // the kernel stays available for experiments via D9D_GMM_GLDS=1.
+  // plain glds staging measured 526 vs 609 TF/s for the register-T14
+  // pipeline: at 2-deep tile prefetch one iteration of MFMAs (~0.3 us)
+  // cannot cover the HBM latency, so the counted vmcnt stalls every tile.
+  // Kept for experiments via D9D_GMM_GLDS=1.
   static const bool use_glds = []() {
     const char* v = getenv("D9D_GMM_GLDS");
     return v != nullptr && v[0] == '1';
