@@ -811,12 +811,17 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
 
     // ---- dQ partial = dS @ K; atomicAdd (A = dS q-major via x3) -----------
     {
-      // waves 0-3: one 16-row q m-tile each, K = kBwdKv kv, N = D
-      if (wave < 4) {
-        f32x4 dq_acc[kNT];
+      // ALL 8 waves: wave & 3 picks the 16-row q m-tile, wave >> 2 the
+      // d-half -- the old waves-0-3-only split parked half the block
+      // through an entire MFMA phase.
+      constexpr int kNTH = kNT > 1 ? kNT / 2 : 1;  // nt per d-half
+      const int q_mt = wave & 3;
+      const int nt0 = kNT > 1 ? (wave >> 2) * kNTH : 0;
+      if (kNT > 1 || wave < 4) {
+        f32x4 dq_acc[kNTH];
 #pragma unroll
-        for (int nt = 0; nt < kNT; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
-        const bf16_t* xr = x3_lds + (wave * 16) * (kBwdKv + 8);
+        for (int nt = 0; nt < kNTH; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+        const bf16_t* xr = x3_lds + (q_mt * 16) * (kBwdKv + 8);
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks2 = 0; ks2 < kBwdKv / 32; ++ks2) {
@@ -824,8 +829,8 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
           const bf16x8 da = *reinterpret_cast<const bf16x8*>(
               xr + (lane & 15) * (kBwdKv + 8) + kv_off);
 #pragma unroll
-          for (int nt = 0; nt < kNT; ++nt) {
-            const int d = nt * 16 + (lane & 15);
+          for (int nt = 0; nt < kNTH; ++nt) {
+            const int d = (nt0 + nt) * 16 + (lane & 15);
             const int byte = (kv_off * 2) ^ ((d & 7) << 4);
             const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
                 reinterpret_cast<char*>(kt_lds) + d * (kBwdKv * 2) + byte);
@@ -835,12 +840,13 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
         __builtin_amdgcn_s_setprio(0);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int q_glob = qt + wave * 16 + (lane >> 4) * 4 + r;
+          const int q_glob = qt + q_mt * 16 + (lane >> 4) * 4 + r;
           if (q_glob < Sq_loc) {
 #pragma unroll
-            for (int nt = 0; nt < kNT; ++nt) {
+            for (int nt = 0; nt < kNTH; ++nt) {
               atomicAdd(dq + q_base +
-                            (int64_t)(q_lo + q_glob) * q_row_stride + nt * 16 + (lane & 15),
+                            (int64_t)(q_lo + q_glob) * q_row_stride +
+                            (nt0 + nt) * 16 + (lane & 15),
                         dq_acc[nt][r]);
             }
           }
