@@ -132,8 +132,17 @@ class Qwen3MoEModel(nn.Module):
 
         B, S, _ = hidden_states.shape
         if position_ids is None:
+            # Under sequence parallelism the inter-block hidden states carry
+            # only S/tp rows, but RoPE is applied inside the attention region
+            # AFTER the all-gather: positions must cover the FULL sequence.
+            S_full = (
+                input_ids.shape[1]
+                if input_ids is not None
+                else S * getattr(self, "_d9d_sp_factor", 1)
+            )
             position_ids = (
-                torch.arange(S, device=hidden_states.device).unsqueeze(0).expand(B, S)
+                torch.arange(S_full, device=hidden_states.device)
+                .unsqueeze(0).expand(B, S_full)
             )
         rotary_cos_sin = self.rotary(position_ids)
 
@@ -174,6 +183,8 @@ class Qwen3MoEModel(nn.Module):
             return {}
         ids = pipeline_inputs["input_ids"]
         B, S = ids.shape[0] // num_microbatches, ids.shape[1]
+        # sequence parallelism: inter-stage hidden states carry S/tp rows
+        S //= getattr(self, "_d9d_sp_factor", 1)
         return {
             "hidden_states": torch.empty(
                 B, S, self.params.hidden_size, dtype=self._stage_dtype(),
@@ -188,6 +199,8 @@ class Qwen3MoEModel(nn.Module):
             return {}
         ids = pipeline_inputs["input_ids"]
         B, S = ids.shape[0] // num_microbatches, ids.shape[1]
+        # sequence parallelism: inter-stage hidden states carry S/tp rows
+        S //= getattr(self, "_d9d_sp_factor", 1)
         return {
             "hidden_states": torch.empty(
                 B, S, self.params.hidden_size, dtype=self._stage_dtype(),

@@ -573,3 +573,51 @@ def _hsdp_trainer(rank, world_size):
 @pytest.mark.distributed
 def test_trainer_hsdp_ws4():
     assert all(run_distributed(_hsdp_trainer, world_size=4, timeout=240))
+
+
+def _tp4_pp2_llama_trainer(rank, world_size):
+    """ws=8 trainer on a Llama-3-shaped model: tensor_parallel=4 (+SP) x
+    pipeline_parallel=2 — the BASELINE config #4 topology (Llama-3-70B
+    TP4+SP+PP2) at tiny scale on gloo."""
+    from d9d_amd.loop.config import PipeliningConfig
+    from d9d_amd.module.model.llama3 import Llama3ModelParameters
+    from d9d_amd.parallel import parallelize_tensor_parallel
+    from d9d_amd.pipelining.factory import PipelineSchedule1F1BConfig
+
+    params = Llama3ModelParameters(
+        hidden_size=64,
+        intermediate_size=128,
+        num_attention_heads=8,
+        num_key_value_heads=4,
+        head_dim=8,
+        num_hidden_layers=2,
+        split_vocab_size={"regular": 128, "special": 8},
+    )
+
+    def parallelize(module, ctx):
+        return parallelize_tensor_parallel(
+            module, ctx.mesh_for("regular"), sequence_parallel=True
+        )
+
+    cfg = _make_config(total_steps=2)
+    cfg = cfg.model_copy(
+        update={"pipelining": PipeliningConfig(schedule=PipelineSchedule1F1BConfig())}
+    )
+    mesh = DeviceMeshParameters(pipeline_parallel=2, tensor_parallel=4)
+    trainer = TrainingConfigurator(
+        cfg,
+        mesh,
+        _LMModelProvider(params, parallelize),
+        _LMDatasetProvider(params),
+        AutoOptimizerProvider(OptimizerConfig(optimizer="adamw", lr=1e-3)),
+        AutoLRSchedulerProvider(LRSchedulerConfig(warmup_steps=1, decay_steps=10)),
+        _LMTask(),
+    ).configure(device_type="cpu")
+    trainer.train()
+    assert trainer.stepper.step == 2
+    return True
+
+
+@pytest.mark.distributed
+def test_trainer_tp4_sp_pp2_llama():
+    assert all(run_distributed(_tp4_pp2_llama_trainer, world_size=8, timeout=300))

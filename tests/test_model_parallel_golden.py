@@ -76,6 +76,11 @@ def _tp2_whole_model(rank, world_size):
                     local.copy_(src.narrow(dim, rank * size, size))
                 else:
                     local.copy_(src)
+            elif param.shape != src.shape:
+                # plain local shard (vocab-parallel LM head): rows
+                # [rank*sh, (rank+1)*sh) of the segment
+                sh = param.shape[0]
+                param.copy_(src.narrow(0, rank * sh, sh))
             else:
                 param.copy_(src)
 
@@ -91,15 +96,23 @@ def _tp2_whole_model(rank, world_size):
 
     ref = {n: q.grad for n, q in global_model.named_parameters() if q.grad is not None}
     got = {}
+    ref_cmp = {}
     for name, param in par_model.named_parameters():
         if param.grad is None:
             continue
         g = param.grad
         if isinstance(g, DTensor):
             g = g.full_tensor()
+            ref_cmp[name] = ref[name]
+        elif g.shape != ref[name].shape:
+            # plain local shard (vocab-parallel LM head): compare the slice
+            sh = g.shape[0]
+            ref_cmp[name] = ref[name].narrow(0, rank * sh, sh)
+        else:
+            ref_cmp[name] = ref[name]
         got[name] = g
     # TP-sharded grads gathered to full; compare with angle + norm metrics
-    assert_grads_close(got, {n: ref[n] for n in got})
+    assert_grads_close(got, ref_cmp)
     return True
 
 

@@ -156,3 +156,161 @@ def _tp2_sp_layer(rank, world_size):
 @pytest.mark.distributed
 def test_tp2_sequence_parallel_layer_exact():
     assert all(run_distributed(_tp2_sp_layer, world_size=2))
+
+
+# -- whole-model TP: embeddings + blocks (incl. MoE) + vocab-parallel head ----
+
+
+def _tp2_whole_moe_model(rank, world_size):
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import DTensor
+
+    from d9d_amd.module.model.qwen3_moe import (
+        Qwen3MoEForCausalLM,
+        Qwen3MoEModelParameters,
+    )
+    from d9d_amd.parallel.tensor import parallelize_tensor_parallel
+
+    p = Qwen3MoEModelParameters.tiny()
+    torch.manual_seed(11)
+    ref = Qwen3MoEForCausalLM(p)
+    ref.init_weights()
+    torch.manual_seed(11)
+    model = Qwen3MoEForCausalLM(p)
+    model.init_weights()
+
+    mesh = init_device_mesh("cpu", (2,), mesh_dim_names=("tp",))
+    parallelize_tensor_parallel(model, mesh)
+
+    torch.manual_seed(55)
+    ids = torch.randint(0, p.vocab_size, (2, 10))
+    labels = torch.randint(0, p.vocab_size, (2, 10))
+    labels[0, 0] = -100
+
+    out = model(input_ids=ids, labels=labels)
+    out["loss"].sum().backward()
+    ref_out = ref(input_ids=ids, labels=labels)
+    ref_out["loss"].sum().backward()
+
+    torch.testing.assert_close(out["loss"], ref_out["loss"], rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(out["logps"], ref_out["logps"], rtol=1e-4, atol=1e-4)
+
+    # replicated-inside-tp param (router gate): full grad on every rank
+    moe = model.model.layers["0"].mlp
+    ref_moe = ref.model.layers["0"].mlp
+    torch.testing.assert_close(
+        moe.router.gate._parameters["weight"].grad,
+        ref_moe.router.gate.weight.grad, rtol=1e-4, atol=1e-5,
+    )
+    # expert shards: gate slice of the packed gate_up grad
+    inter = p.intermediate_size
+    sh = inter // 2
+    gu = moe.experts.gate_up_proj.weight
+    ref_gu_grad = ref_moe.experts.gate_up_proj.weight.grad
+    expected = torch.cat(
+        [ref_gu_grad[:, rank * sh : (rank + 1) * sh],
+         ref_gu_grad[:, inter + rank * sh : inter + (rank + 1) * sh]], dim=1)
+    torch.testing.assert_close(gu.grad, expected, rtol=1e-4, atol=1e-5)
+    # vocab-parallel embedding shard grad
+    seg = model.model.embed_tokens.embeddings["regular"]
+    ref_seg = ref.model.embed_tokens.embeddings["regular"]
+    local = seg._parameters["weight"].grad
+    assert isinstance(local, DTensor)
+    n = ref_seg.weight.shape[0] // 2
+    torch.testing.assert_close(
+        local.to_local(), ref_seg.weight.grad[rank * n : (rank + 1) * n],
+        rtol=1e-4, atol=1e-5,
+    )
+    # vocab-parallel head shard grad
+    hw = model.lm_head.weights["regular"]
+    ref_hw = ref.lm_head.weights["regular"]
+    torch.testing.assert_close(
+        hw.grad, ref_hw.grad[rank * n : (rank + 1) * n], rtol=1e-4, atol=1e-5
+    )
+    return True
+
+
+@pytest.mark.distributed
+def test_tp2_whole_moe_model_exact():
+    assert all(run_distributed(_tp2_whole_moe_model, world_size=2))
+
+
+def _tp2_grad_accumulation_exact(rank, world_size):
+    """Regression for the round-1 double-reduce bug: replicated-inside-tp
+    params (q/k norm, router gate) must accumulate EXACTLY over microbatches."""
+    from torch.distributed.device_mesh import init_device_mesh
+
+    from d9d_amd.parallel.tensor import parallelize_tensor_parallel
+
+    p, ref_layer = _make_layer(seed=9)
+    _, layer = _make_layer(seed=9)
+
+    mesh = init_device_mesh("cpu", (2,), mesh_dim_names=("tp",))
+    parallelize_tensor_parallel(layer, mesh)
+
+    cos_sin = _rotary(p, 2, 8)
+    torch.manual_seed(31)
+    for mb in range(3):  # 3 microbatches, grads accumulate
+        x = torch.randn(2, 8, p.hidden_size)
+        layer(x, cos_sin).sum().backward()
+        ref_layer(x, cos_sin).sum().backward()
+
+    qn = layer.self_attn.q_norm
+    ref_qn = ref_layer.self_attn.q_norm
+    if qn is not None:
+        torch.testing.assert_close(
+            qn._parameters["weight"].grad, ref_qn.weight.grad,
+            rtol=1e-4, atol=1e-5,
+        )
+    return True
+
+
+@pytest.mark.distributed
+def test_tp2_grad_accumulation_exact():
+    assert all(run_distributed(_tp2_grad_accumulation_exact, world_size=2))
+
+
+def _tp2_sp_whole_moe_model(rank, world_size):
+    """Sequence-parallel whole model: embedding exit reduce-scatters the
+    sequence, every block all-gathers/reduce-scatters, head gathers."""
+    from torch.distributed.device_mesh import init_device_mesh
+
+    from d9d_amd.module.model.qwen3_moe import (
+        Qwen3MoEForCausalLM,
+        Qwen3MoEModelParameters,
+    )
+    from d9d_amd.parallel.tensor import parallelize_tensor_parallel
+
+    p = Qwen3MoEModelParameters.tiny()
+    torch.manual_seed(13)
+    ref = Qwen3MoEForCausalLM(p)
+    ref.init_weights()
+    torch.manual_seed(13)
+    model = Qwen3MoEForCausalLM(p)
+    model.init_weights()
+
+    mesh = init_device_mesh("cpu", (2,), mesh_dim_names=("tp",))
+    parallelize_tensor_parallel(model, mesh, sequence_parallel=True)
+
+    torch.manual_seed(57)
+    ids = torch.randint(0, p.vocab_size, (2, 8))
+    labels = torch.randint(0, p.vocab_size, (2, 8))
+
+    out = model(input_ids=ids, labels=labels)
+    out["loss"].sum().backward()
+    ref_out = ref(input_ids=ids, labels=labels)
+    ref_out["loss"].sum().backward()
+
+    torch.testing.assert_close(out["loss"], ref_out["loss"], rtol=1e-4, atol=1e-5)
+    moe = model.model.layers["0"].mlp
+    ref_moe = ref.model.layers["0"].mlp
+    torch.testing.assert_close(
+        moe.router.gate._parameters["weight"].grad,
+        ref_moe.router.gate.weight.grad, rtol=1e-4, atol=1e-5,
+    )
+    return True
+
+
+@pytest.mark.distributed
+def test_tp2_sp_whole_moe_model_exact():
+    assert all(run_distributed(_tp2_sp_whole_moe_model, world_size=2))
