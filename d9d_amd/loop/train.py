@@ -153,11 +153,28 @@ class TrainingConfigurator:
         )
 
         def provider_fn(stage_info):
-            module = self.model_provider.initialize_model_stage(stage_info)
-            module = self.model_provider.parallelize_model_stage(module, ctx)
-            module = module.to(device=device)
+            # Reference flow (d9d/loop/component/model_stage_factory.py:218-249):
+            # meta init -> parallelize on meta -> to_empty(device) ->
+            # reset_parameters -> streamed load. A 70B-class stage never
+            # materializes outside its own shard. Providers can opt out with
+            # `meta_device_init = False` (e.g. when initialize copies real
+            # pretrained weights directly).
+            use_meta = getattr(self.model_provider, "meta_device_init", True)
+            if use_meta:
+                with torch.device("meta"):
+                    module = self.model_provider.initialize_model_stage(stage_info)
+                module = self.model_provider.parallelize_model_stage(module, ctx)
+                module = module.to_empty(device=device)
+            else:
+                module = self.model_provider.initialize_model_stage(stage_info)
+                module = self.model_provider.parallelize_model_stage(module, ctx)
+                module = module.to(device=device)
             if hasattr(module, "reset_parameters"):
                 module.reset_parameters()
+            # parallelize steps that need real data (replicate's broadcast-
+            # init) defer themselves when they ran on meta tensors
+            for cb in getattr(module, "_d9d_post_materialize", []):
+                cb(module)
             source = self.model_provider.source_checkpoint()
             if source:
                 from ..model_state import load_model_state

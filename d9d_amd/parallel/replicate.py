@@ -41,15 +41,28 @@ def parallelize_replicate(
                 return True
             return t.device_mesh is mesh
 
-        with torch.no_grad():
-            for dim in range(mesh.ndim):
-                group = mesh.get_group(dim)
-                if dist.get_world_size(group) == 1:
-                    continue
-                src_rank = dist.get_global_rank(group, 0)
-                for t in list(module.parameters()) + list(module.buffers()):
-                    if not _broadcastable(t):
+        def _do_broadcast(mod):
+            with torch.no_grad():
+                for dim in range(mesh.ndim):
+                    group = mesh.get_group(dim)
+                    if dist.get_world_size(group) == 1:
                         continue
-                    local = t.to_local() if isinstance(t, DTensor) else t
-                    dist.broadcast(local, src=src_rank, group=group)
+                    src_rank = dist.get_global_rank(group, 0)
+                    for t in list(mod.parameters()) + list(mod.buffers()):
+                        if not _broadcastable(t):
+                            continue
+                        local = t.to_local() if isinstance(t, DTensor) else t
+                        dist.broadcast(local, src=src_rank, group=group)
+
+        params = list(module.parameters())
+        if any(p.is_meta for p in params):
+            # meta-device build flow: real values exist only after
+            # to_empty + reset_parameters — defer the broadcast there
+            cbs = getattr(module, "_d9d_post_materialize", None)
+            if cbs is None:
+                cbs = []
+                module._d9d_post_materialize = cbs
+            cbs.append(_do_broadcast)
+        else:
+            _do_broadcast(module)
     return module
