@@ -83,24 +83,33 @@ class GradientManager:
     def loss_weight_total(self) -> float:
         return self._loss_weight_total
 
-    def sync_and_scale(self, group=None) -> None:
-        """Wait grad comms; scale grads by 1/total weight."""
+    def sync_and_scale(self, group=None, cp_group=None) -> None:
+        """Wait grad comms; scale grads by 1/total weight (summed over dp
+        and, under context parallelism, over cp — orthogonal groups, so two
+        all-reduces compose to the product)."""
         if self._sync is not None:
             self._sync.wait()
         total = self._loss_weight_total
-        if dist.is_initialized() and group is not None:
-            # The dp group is RCCL on GPU runs: the reduce tensor must live on
-            # the same device as the gradients or the collective rejects it.
-            device = None
-            for _, p in self._named_params:
-                if p.grad is not None:
-                    device = p.grad.device
-                    break
-                if device is None:
-                    device = p.device
-            t = torch.tensor([total], dtype=torch.float64, device=device)
-            dist.all_reduce(t, group=group)
-            total = t.item()
+        for g in (group, cp_group):
+            if dist.is_initialized() and g is not None:
+                total = self._reduce_total(total, g)
+        self._finish_scale(total)
+
+    def _reduce_total(self, total, group):
+        # The group is RCCL on GPU runs: the reduce tensor must live on
+        # the same device as the gradients or the collective rejects it.
+        device = None
+        for _, p in self._named_params:
+            if p.grad is not None:
+                device = p.grad.device
+                break
+            if device is None:
+                device = p.device
+        t = torch.tensor([total], dtype=torch.float64, device=device)
+        dist.all_reduce(t, group=group)
+        return t.item()
+
+    def _finish_scale(self, total: float) -> None:
         if total > 0:
             scale = 1.0 / total
             grads = []

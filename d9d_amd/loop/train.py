@@ -210,6 +210,12 @@ class TrainingConfigurator:
         dp_group = None
         if ctx.is_distributed and dp > 1:
             dp_group = ctx.mesh_for("batch").get_group("dp")
+        cp_group = None
+        cp_rank, cp_size = 0, self.mesh.context_parallel_shard
+        if ctx.is_distributed and cp_size > 1:
+            batch_mesh = ctx.mesh_for("batch")
+            cp_group = batch_mesh.get_group("cp")
+            cp_rank = batch_mesh.get_local_rank("cp")
         clipper = GradientClipper(
             [p for _, p in named_params],
             cfg.gradient_clipping.max_norm,
@@ -266,6 +272,9 @@ class TrainingConfigurator:
             sleeper=sleeper,
             modules_by_key=modules_by_key,
             dp_group=dp_group,
+            cp_group=cp_group,
+            cp_rank=cp_rank,
+            cp_size=cp_size,
             model_provider=self.model_provider,
         )
 
@@ -293,6 +302,9 @@ class Trainer:
         self.sleeper: TrainSleeper = kw["sleeper"]
         self.modules_by_key = kw["modules_by_key"]
         self.dp_group = kw["dp_group"]
+        self.cp_group = kw["cp_group"]
+        self.cp_rank = kw["cp_rank"]
+        self.cp_size = kw["cp_size"]
         self.model_provider: ModelProvider = kw["model_provider"]
         self._run = None
         self.last_losses: list[float] = []
@@ -326,6 +338,20 @@ class Trainer:
             k: (v.to(self.ctx.device) if isinstance(v, torch.Tensor) else v)
             for k, v in inputs.items()
         }
+        if self.cp_size > 1:
+            # context parallelism: each cp rank trains on its contiguous
+            # sequence chunk (dim 1 of every 2D+ tensor input); the model
+            # offsets positions and ring attention spans the full context
+            from ..parallel.context import shard_sequence
+
+            inputs = {
+                k: (
+                    shard_sequence(v, self.cp_rank, self.cp_size, dim=1)
+                    if isinstance(v, torch.Tensor) and v.ndim >= 2
+                    else v
+                )
+                for k, v in inputs.items()
+            }
         schedule = self.schedule_info.schedule
         schedule.configure_buffers(inputs)
 
@@ -382,7 +408,7 @@ class Trainer:
     def _train_step(self, batch) -> None:
         self._forward_backward(batch)
         self.collector.trigger_sync()
-        self.grad_manager.sync_and_scale(self.dp_group)
+        self.grad_manager.sync_and_scale(self.dp_group, self.cp_group)
         grad_norm = self.clipper.clip_and_log()
         with self.bus.bounded(ev.TRAIN_OPTIMIZER_STEP_PRE, ev.TRAIN_OPTIMIZER_STEP_POST):
             self.optimizer.step()

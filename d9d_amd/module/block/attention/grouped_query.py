@@ -76,6 +76,7 @@ class GroupedQueryAttention(nn.Module):
         self._cp_group = None
         self._cp_rank = 0
         self._cp_size = 1
+        self._cp_mode = "ring"
 
     def reset_parameters(self) -> None:
         with torch.no_grad():
@@ -133,6 +134,19 @@ class GroupedQueryAttention(nn.Module):
             return self.o_proj(attn)
         q_offset = 0
         if self._cp_group is not None and self._cp_size > 1:
+            if getattr(self, "_cp_mode", "ring") == "ring":
+                # ring attention: KV chunks travel the ring, partials merge
+                # by LSE — O(S/cp) resident memory
+                assert self.sinks is None and self.sliding_window is None, (
+                    "ring CP supports plain causal attention (no sinks/window)"
+                )
+                from ....parallel.context import ring_attention
+
+                attn = ring_attention(q, k, v, group=self._cp_group, causal=True)
+                attn = attn.reshape(B, S, self.num_heads * self.head_dim)
+                if self.use_output_gate:
+                    attn = attn * torch.sigmoid(gate)
+                return self.o_proj(attn)
             from ....parallel.tensor import _AllGatherSeq
 
             k = _AllGatherSeq.apply(k, self._cp_group, 1)

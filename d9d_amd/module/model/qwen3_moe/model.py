@@ -140,8 +140,12 @@ class Qwen3MoEModel(nn.Module):
                 if input_ids is not None
                 else S * getattr(self, "_d9d_sp_factor", 1)
             )
+            # context parallelism: this rank holds the cp_rank-th contiguous
+            # sequence chunk — positions offset to the global range
+            cp_rank, _cp = getattr(self, "_d9d_cp", (0, 1))
             position_ids = (
-                torch.arange(S_full, device=hidden_states.device)
+                (torch.arange(S_full, device=hidden_states.device)
+                 + cp_rank * S_full)
                 .unsqueeze(0).expand(B, S_full)
             )
         rotary_cos_sin = self.rotary(position_ids)

@@ -54,8 +54,16 @@ def parallelize_replicate(
                         local = t.to_local() if isinstance(t, DTensor) else t
                         dist.broadcast(local, src=src_rank, group=group)
 
-        params = list(module.parameters())
-        if any(p.is_meta for p in params):
+        def _is_meta(t):
+            # a DTensor over a cpu/cuda mesh whose LOCAL tensor is meta
+            # reports wrapper-level is_meta False — check the local
+            lt = getattr(t, "_local_tensor", None)
+            if lt is not None:
+                return lt.is_meta
+            return t.is_meta
+
+        tensors = list(module.parameters()) + list(module.buffers())
+        if any(_is_meta(t) for t in tensors):
             # meta-device build flow: real values exist only after
             # to_empty + reset_parameters — defer the broadcast there
             cbs = getattr(module, "_d9d_post_materialize", None)
