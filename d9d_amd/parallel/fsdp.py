@@ -10,6 +10,19 @@ reduction with divide factor 1 so the trainer scales gradients itself
 from torch import nn
 from torch.distributed.device_mesh import DeviceMesh
 from torch.distributed.fsdp import fully_shard
+from torch.distributed.tensor import DTensor
+
+
+def _foreign_dtensor_params(module: nn.Module, mesh: DeviceMesh) -> set[nn.Parameter]:
+    """Params already DTensor-sharded on ANOTHER mesh (e.g. EP experts on the
+    expert domain, TP shards on the regular domain): fully_shard cannot
+    compose overlapping meshes, so these are ignored — their cross-replica
+    grad sync rides the GradientSynchronizer's Replicate-dim reduction."""
+    out: set[nn.Parameter] = set()
+    for p in module.parameters():
+        if isinstance(p, DTensor) and p.device_mesh is not mesh:
+            out.add(p)
+    return out
 
 
 def _apply_sum_reduction(module: nn.Module) -> None:
@@ -38,10 +51,13 @@ def parallelize_fsdp(
     always wrapped last.
     """
     assert mesh.ndim == 1, "parallelize_fsdp expects a 1-D mesh"
+    ignored = _foreign_dtensor_params(module, mesh)
     for unit in shard_units or []:
-        fully_shard(unit, mesh=mesh, reshard_after_forward=reshard_after_forward)
+        fully_shard(unit, mesh=mesh, reshard_after_forward=reshard_after_forward,
+                    ignored_params=ignored)
         _apply_sum_reduction(unit)
-    fully_shard(module, mesh=mesh, reshard_after_forward=reshard_after_forward)
+    fully_shard(module, mesh=mesh, reshard_after_forward=reshard_after_forward,
+                ignored_params=ignored)
     _apply_sum_reduction(module)
     return module
 
@@ -54,9 +70,12 @@ def parallelize_hsdp(
 ) -> nn.Module:
     """Hybrid sharding over a 2-D (replicate, shard) mesh."""
     assert mesh.ndim == 2, "parallelize_hsdp expects a 2-D (replicate, shard) mesh"
+    ignored = _foreign_dtensor_params(module, mesh)
     for unit in shard_units or []:
-        fully_shard(unit, mesh=mesh, reshard_after_forward=reshard_after_forward)
+        fully_shard(unit, mesh=mesh, reshard_after_forward=reshard_after_forward,
+                    ignored_params=ignored)
         _apply_sum_reduction(unit)
-    fully_shard(module, mesh=mesh, reshard_after_forward=reshard_after_forward)
+    fully_shard(module, mesh=mesh, reshard_after_forward=reshard_after_forward,
+                ignored_params=ignored)
     _apply_sum_reduction(module)
     return module
