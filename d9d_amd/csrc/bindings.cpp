@@ -38,6 +38,14 @@ std::vector<torch::Tensor> router_topk_fwd(torch::Tensor logits, c10::optional<t
 torch::Tensor router_topk_bwd(torch::Tensor logits, torch::Tensor top_idx, torch::Tensor dtop,
                               bool renormalize);
 
+torch::Tensor causal_conv_silu_fwd(torch::Tensor x, torch::Tensor w);
+std::vector<torch::Tensor> causal_conv_silu_bwd(torch::Tensor x, torch::Tensor w,
+                                                torch::Tensor dy);
+std::vector<torch::Tensor> gdn_chunk_fwd(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v, torch::Tensor beta,
+                                         torch::Tensor decay_log,
+                                         bool return_state);
+
 torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse, torch::Tensor targets,
                            torch::Tensor dl, c10::optional<torch::Tensor> dlse,
                            int64_t vocab_start, int64_t ignore_index,
@@ -71,6 +79,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("router_topk_fwd", &router_topk_fwd, "fused MoE router fwd");
   m.def("router_topk_bwd", &router_topk_bwd, "fused MoE router bwd");
   m.def("cce_dlogits_", &cce_dlogits_, "fused CCE dlogits (in-place)");
+  m.def("causal_conv_silu_fwd", &causal_conv_silu_fwd, "fused causal depthwise conv + SiLU fwd");
+  m.def("causal_conv_silu_bwd", &causal_conv_silu_bwd, "fused causal depthwise conv + SiLU bwd");
+  m.def("gdn_chunk_fwd", &gdn_chunk_fwd, "chunked gated delta rule forward (GDN)");
   m.def("adamw_stochastic_bf16_multi_", &adamw_stochastic_bf16_multi_, "multi-tensor fused SR-AdamW");
   m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
   m.def("gmm_nt", &gmm_nt, "CDNA4 grouped GEMM, weight (E,N,K)");

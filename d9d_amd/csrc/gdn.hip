@@ -1,0 +1,537 @@
+// CDNA4 kernels for GatedDeltaNet (the fla-core replacement:
+// reference d9d/module/block/attention/linear/gated_deltanet.py:6-8,360-370).
+//
+// 1. causal_conv_silu fwd/bwd: per-channel causal conv (kernel<=4) fused
+//    with SiLU over (B, S, C) — one HBM read + write instead of the
+//    pad/conv1d/silu chain.
+// 2. gdn_chunk_fwd: the chunked-parallel WY gated delta rule, one
+//    workgroup per (b, h), chunks of 64 sequential with the fp32 state
+//    resident in LDS. MFMA bf16 for the chunk GEMMs (KK^T, QK^T, K@S,
+//    Q@S, N@R, K^T@R), fp32 blocked forward-substitution for the
+//    unit-lower triangular solve. Dk = Dv = 64.
+//    (The python wrapper pairs this forward with a recompute-through-the
+//    torch-WY-graph backward; a native bwd kernel is the follow-up.)
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace d9d {
+
+typedef __bf16 bf16_t;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+D9D_DEVICE f32x4 mfma16gdn(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// ---------------------------------------------------------------------------
+// causal depthwise conv (kernel <= 4) + SiLU, bf16
+// ---------------------------------------------------------------------------
+
+__global__ void causal_conv_silu_fwd_kernel(
+    const bf16_t* __restrict__ x,   // (B, S, C)
+    const bf16_t* __restrict__ w,   // (C, K)
+    bf16_t* __restrict__ out,       // (B, S, C)
+    int64_t B, int64_t S, int64_t C, int K) {
+  const int64_t total = B * S * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t c = i % C;
+    const int64_t s = (i / C) % S;
+    const int64_t b = i / (C * S);
+    float acc = 0.f;
+#pragma unroll 4
+    for (int t = 0; t < K; ++t) {
+      const int64_t sp = s - (K - 1 - t);
+      if (sp >= 0) {
+        acc += (float)x[(b * S + sp) * C + c] * (float)w[c * K + t];
+      }
+    }
+    const float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-acc * 1.44269504f));
+    out[i] = (bf16_t)(acc * sig);
+  }
+}
+
+// dx and dw for conv+silu. dpre = dy * silu'(pre); dx[s] = sum_t dpre[s+K-1-t]*w[t]
+// (within bounds); dw[c,t] = sum_{b,s} dpre[b,s,c] * x[b, s-(K-1-t), c].
+__global__ void causal_conv_silu_bwd_kernel(
+    const bf16_t* __restrict__ x,
+    const bf16_t* __restrict__ w,
+    const bf16_t* __restrict__ dy,
+    bf16_t* __restrict__ dx,        // (B, S, C)
+    float* __restrict__ dw,         // (C, K) fp32 (atomic)
+    int64_t B, int64_t S, int64_t C, int K) {
+  const int64_t total = B * S * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t c = i % C;
+    const int64_t s = (i / C) % S;
+    const int64_t b = i / (C * S);
+
+    // recompute pre-activation at s (cheap: K<=4 reads)
+    auto pre_at = [&](int64_t sq) -> float {
+      float acc = 0.f;
+#pragma unroll 4
+      for (int t = 0; t < K; ++t) {
+        const int64_t sp = sq - (K - 1 - t);
+        if (sp >= 0) acc += (float)x[(b * S + sp) * C + c] * (float)w[c * K + t];
+      }
+      return acc;
+    };
+    auto dsilu = [&](float a) -> float {
+      const float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-a * 1.44269504f));
+      return sig * (1.f + a * (1.f - sig));
+    };
+
+    // dx[s]: gather from future positions using dpre there
+    float acc_dx = 0.f;
+#pragma unroll 4
+    for (int t = 0; t < K; ++t) {
+      const int64_t sq = s + (K - 1 - t);
+      if (sq < S) {
+        const float dpre = (float)dy[(b * S + sq) * C + c] * dsilu(pre_at(sq));
+        acc_dx += dpre * (float)w[c * K + t];
+      }
+    }
+    dx[i] = (bf16_t)acc_dx;
+
+    // dw contributions from THIS position's dpre
+    const float dpre_s = (float)dy[i] * dsilu(pre_at(s));
+#pragma unroll 4
+    for (int t = 0; t < K; ++t) {
+      const int64_t sp = s - (K - 1 - t);
+      if (sp >= 0) {
+        atomicAdd(&dw[c * K + t], dpre_s * (float)x[(b * S + sp) * C + c]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// chunked gated delta rule forward, Dk = Dv = 64, chunk C = 64
+// ---------------------------------------------------------------------------
+//
+// One workgroup (4 waves, 256 threads) per (b, h). Chunks run sequentially
+// with the fp32 state resident in LDS. Per chunk (t, j in [0, 64); rows
+// past the sequence end are ZERO-padded so every product is exact):
+//   gc      = cumsum(g)                                   (serial scan)
+//   M[t,j]  = beta_t (K K^T)[t,j] e^{gc_t-gc_j}  (j < t)  (MFMA)
+//   rhs     = diag(beta) (V - (e^{gc} K) @ S)             (MFMA, S as bf16)
+//   R: (I + M) R = rhs    forward substitution, fp32, 4-way j-split
+//   N[t,j]  = (Q K^T)[t,j] e^{gc_t-gc_j}  (j <= t)        (MFMA)
+//   O       = (e^{gc} Q) @ S + N @ R                      (MFMA, R as bf16)
+//   S       = e^{gc_last} S + (K e^{gc_last-gc})^T @ R    (fp32 VALU)
+//
+// MFMA fragment map (verified by attention.hip's mfma_selfcheck):
+//   mfma16gdn(a, b, acc): A[m][k] lane m = l&15, k-slice (l>>4)*8;
+//   B[k][n] lane n = l&15, same k-slice; C[m][n] lane: m = (l>>4)*4 + r,
+//   n = l&15. All tiles here are row-major [row][64] so A fragments are
+//   contiguous row reads; B fragments read the row of the TRANSPOSED
+//   operand copy (kc rows for K^T, sbT rows for S, rbT rows for R).
+
+constexpr int kGdnC = 64;   // chunk rows
+constexpr int kGdnD = 64;   // Dk = Dv
+
+struct GdnLds {
+  float state[kGdnD][kGdnD];      // S (k, v) fp32
+  float M[kGdnC][kGdnC + 1];      // solve matrix / N scratch (+1 pad)
+  float R[kGdnC][kGdnD];          // solve result fp32
+  float rhs[kGdnC][kGdnD];        // rhs / partials scratch
+  bf16_t kc[kGdnC][kGdnD];        // K chunk
+  bf16_t qc[kGdnC][kGdnD];        // Q chunk
+  bf16_t vc[kGdnC][kGdnD];        // V chunk
+  bf16_t ks[kGdnC][kGdnD];        // scaled K / scaled Q / scaled-K2 scratch
+  bf16_t sbT[kGdnD][kGdnD];       // state^T cast to bf16: sbT[v][k]
+  bf16_t nb[kGdnC][kGdnC];        // N cast to bf16
+  bf16_t rbT[kGdnD][kGdnC];       // R^T cast to bf16: rbT[v][t]
+  float gc[kGdnC];
+  float beta[kGdnC];
+};
+
+__global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
+    const bf16_t* __restrict__ q,   // (B, H, S, D)
+    const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v,
+    const float* __restrict__ beta,      // (B, H, S)
+    const float* __restrict__ decay_log, // (B, H, S)
+    bf16_t* __restrict__ out,            // (B, H, S, D)
+    float* __restrict__ final_state,     // (B, H, D, D) or nullptr
+    int64_t BH, int64_t S) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  GdnLds& L = *reinterpret_cast<GdnLds*>(smem_raw);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;   // 4 waves; wave w owns m-rows 16w..16w+15
+  const int64_t bh = blockIdx.x;
+  if (bh >= BH) return;
+
+  const bf16_t* qp = q + bh * S * kGdnD;
+  const bf16_t* kp = k + bh * S * kGdnD;
+  const bf16_t* vp = v + bh * S * kGdnD;
+  const float* bp = beta + bh * S;
+  const float* gp = decay_log + bh * S;
+  bf16_t* op = out + bh * S * kGdnD;
+
+  for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
+    (&L.state[0][0])[i] = 0.f;
+  }
+  __syncthreads();
+
+  const float kLog2eG = 1.44269504f;
+  const int n_chunks = (int)((S + kGdnC - 1) / kGdnC);
+  for (int ch = 0; ch < n_chunks; ++ch) {
+    const int s0 = ch * kGdnC;
+    const int c_rows = min((int)(S - (int64_t)s0), kGdnC);
+
+    // ---- stage chunk tiles (zero-pad past c_rows) + state^T bf16 -------
+    for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+      const int row = (i * 8) / kGdnD;
+      const int col = (i * 8) % kGdnD;
+      if (row < c_rows) {
+        const int64_t off = (int64_t)(s0 + row) * kGdnD + col;
+        *reinterpret_cast<bf16x8*>(&L.kc[row][col]) =
+            *reinterpret_cast<const bf16x8*>(kp + off);
+        *reinterpret_cast<bf16x8*>(&L.qc[row][col]) =
+            *reinterpret_cast<const bf16x8*>(qp + off);
+        *reinterpret_cast<bf16x8*>(&L.vc[row][col]) =
+            *reinterpret_cast<const bf16x8*>(vp + off);
+      } else {
+        bf16x8 z = {};
+        *reinterpret_cast<bf16x8*>(&L.kc[row][col]) = z;
+        *reinterpret_cast<bf16x8*>(&L.qc[row][col]) = z;
+        *reinterpret_cast<bf16x8*>(&L.vc[row][col]) = z;
+      }
+    }
+    for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
+      const int vcol = i / kGdnD;
+      const int dk = i % kGdnD;
+      L.sbT[vcol][dk] = (bf16_t)L.state[dk][vcol];
+    }
+    if (threadIdx.x == 0) {
+      float run = 0.f;
+      for (int t = 0; t < kGdnC; ++t) {
+        if (t < c_rows) run += gp[s0 + t];
+        L.gc[t] = run;                       // clamped-constant past end
+        L.beta[t] = (t < c_rows) ? bp[s0 + t] : 0.f;
+      }
+    }
+    __syncthreads();
+
+    // ---- M = tril(beta * KK^T * ratio, -1) -----------------------------
+    {
+      f32x4 acc[4];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      const int arow = wave * 16 + (lane & 15);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int d0 = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.kc[arow][d0]);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int brow = nt * 16 + (lane & 15);   // B = K^T: col j = K row j
+          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.kc[brow][d0]);
+          acc[nt] = mfma16gdn(a, b, acc[nt]);
+        }
+      }
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int t = wave * 16 + (lane >> 4) * 4 + r;
+          const int j = nt * 16 + (lane & 15);
+          float m = 0.f;
+          if (j < t) {
+            m = L.beta[t] * acc[nt][r] *
+                __builtin_amdgcn_exp2f((L.gc[t] - L.gc[j]) * kLog2eG);
+          }
+          L.M[t][j] = m;
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- rhs = diag(beta) (V - (e^{gc} K) @ S) -------------------------
+    for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+      const int row = (i * 8) / kGdnD;
+      const int col = (i * 8) % kGdnD;
+      const float sc = __builtin_amdgcn_exp2f(L.gc[row] * kLog2eG);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        L.ks[row][col + j] = (bf16_t)((float)L.kc[row][col + j] * sc);
+      }
+    }
+    __syncthreads();
+    {
+      f32x4 acc[4];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      const int arow = wave * 16 + (lane & 15);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int d0 = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.ks[arow][d0]);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int vcol = nt * 16 + (lane & 15);
+          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.sbT[vcol][d0]);
+          acc[nt] = mfma16gdn(a, b, acc[nt]);
+        }
+      }
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int t = wave * 16 + (lane >> 4) * 4 + r;
+          const int vcol = nt * 16 + (lane & 15);
+          L.rhs[t][vcol] =
+              L.beta[t] * ((float)L.vc[t][vcol] - acc[nt][r]);
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- forward substitution: (I + M) R = rhs -------------------------
+    // wave w owns Dv columns [16w, 16w+16); its 64 lanes split as
+    // (j-quarter, col): lane = qj*16 + c. Row t: each lane partial-sums its
+    // j-quarter of M[t,:]*R[:,col], a 2-step xor-shuffle combines the 4
+    // partials, and the qj==0 lane writes R[t][col] (+ bf16 R^T copy).
+    {
+      const int c = lane & 15;
+      const int qj = lane >> 4;
+      const int col = wave * 16 + c;
+      for (int t = 0; t < kGdnC; ++t) {
+        float part = 0.f;
+        const int jend = min(t, (qj + 1) * 16);
+        for (int j = qj * 16; j < jend; ++j) {
+          part += L.M[t][j] * L.R[j][col];
+        }
+        // all 64 lanes active and convergent: xor-shuffles are safe
+        part += __shfl_xor(part, 16, 64);
+        part += __shfl_xor(part, 32, 64);
+        const float r_t = L.rhs[t][col] - part;
+        if (qj == 0) {
+          L.R[t][col] = r_t;
+          L.rbT[col][t] = (bf16_t)r_t;
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+      }
+    }
+    __syncthreads();
+
+    // ---- N (bf16) and O = (e^{gc} Q) @ S + N @ R -----------------------
+    for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+      const int row = (i * 8) / kGdnD;
+      const int col = (i * 8) % kGdnD;
+      const float sc = __builtin_amdgcn_exp2f(L.gc[row] * kLog2eG);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        L.ks[row][col + j] = (bf16_t)((float)L.qc[row][col + j] * sc);
+      }
+    }
+    __syncthreads();
+    {
+      // N = tril(QK^T * ratio, 0) -> nb (bf16)
+      f32x4 acc[4];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      const int arow = wave * 16 + (lane & 15);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int d0 = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.qc[arow][d0]);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int brow = nt * 16 + (lane & 15);
+          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.kc[brow][d0]);
+          acc[nt] = mfma16gdn(a, b, acc[nt]);
+        }
+      }
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int t = wave * 16 + (lane >> 4) * 4 + r;
+          const int j = nt * 16 + (lane & 15);
+          float n = 0.f;
+          if (j <= t) {
+            n = acc[nt][r] *
+                __builtin_amdgcn_exp2f((L.gc[t] - L.gc[j]) * kLog2eG);
+          }
+          L.nb[t][j] = (bf16_t)n;
+        }
+      }
+    }
+    __syncthreads();
+    {
+      f32x4 acc[4];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      const int arow = wave * 16 + (lane & 15);
+      // (e^{gc} Q) @ S
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int d0 = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.ks[arow][d0]);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int vcol = nt * 16 + (lane & 15);
+          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.sbT[vcol][d0]);
+          acc[nt] = mfma16gdn(a, b, acc[nt]);
+        }
+      }
+      // + N @ R  (k dim = chunk rows j, 64 wide)
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int j0 = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.nb[arow][j0]);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int vcol = nt * 16 + (lane & 15);
+          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.rbT[vcol][j0]);
+          acc[nt] = mfma16gdn(a, b, acc[nt]);
+        }
+      }
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int t = wave * 16 + (lane >> 4) * 4 + r;
+          const int vcol = nt * 16 + (lane & 15);
+          if (t < c_rows) {
+            op[(int64_t)(s0 + t) * kGdnD + vcol] = (bf16_t)acc[nt][r];
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- state = e^{gc_last} S + (K e^{gc_last - gc})^T @ R ------------
+    {
+      const float g_tot = L.gc[c_rows - 1];
+      for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+        const int row = (i * 8) / kGdnD;
+        const int col = (i * 8) % kGdnD;
+        const float sc =
+            (row < c_rows)
+                ? __builtin_amdgcn_exp2f((g_tot - L.gc[row]) * kLog2eG)
+                : 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          L.ks[row][col + j] = (bf16_t)((float)L.kc[row][col + j] * sc);
+        }
+      }
+      __syncthreads();
+      const float e_tot = __builtin_amdgcn_exp2f(g_tot * kLog2eG);
+      for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
+        const int dk = i / kGdnD;
+        const int dv = i % kGdnD;
+        float acc = L.state[dk][dv] * e_tot;
+        for (int t = 0; t < c_rows; ++t) {
+          acc += (float)L.ks[t][dk] * L.R[t][dv];
+        }
+        L.state[dk][dv] = acc;
+      }
+    }
+    __syncthreads();
+  }
+
+  if (final_state != nullptr) {
+    float* fs = final_state + bh * kGdnD * kGdnD;
+    for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
+      fs[i] = (&L.state[0][0])[i];
+    }
+  }
+}
+
+}  // namespace d9d
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+torch::Tensor causal_conv_silu_fwd(torch::Tensor x, torch::Tensor w) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(w.size(1) <= 4, "conv kernel must be <= 4");
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  auto out = torch::empty_like(xc);
+  const int64_t B = xc.size(0), S = xc.size(1), C = xc.size(2);
+  const int64_t total = B * S * C;
+  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::causal_conv_silu_fwd_kernel, dim3(blocks), dim3(256),
+                     0, stream,
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                     reinterpret_cast<__bf16*>(out.data_ptr()),
+                     B, S, C, (int)w.size(1));
+  return out;
+}
+
+std::vector<torch::Tensor> causal_conv_silu_bwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy) {
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  auto dyc = dy.contiguous();
+  auto dx = torch::empty_like(xc);
+  auto dw = torch::zeros(
+      {w.size(0), w.size(1)},
+      torch::dtype(torch::kFloat32).device(w.device()));
+  const int64_t B = xc.size(0), S = xc.size(1), C = xc.size(2);
+  const int64_t total = B * S * C;
+  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::causal_conv_silu_bwd_kernel, dim3(blocks), dim3(256),
+                     0, stream,
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(dyc.data_ptr()),
+                     reinterpret_cast<__bf16*>(dx.data_ptr()),
+                     dw.data_ptr<float>(), B, S, C, (int)w.size(1));
+  return {dx, dw.to(w.scalar_type())};
+}
+
+std::vector<torch::Tensor> gdn_chunk_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor beta, torch::Tensor decay_log, bool return_state) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.size(-1) == 64 && v.size(-1) == 64,
+              "gdn_chunk_fwd supports Dk = Dv = 64");
+  auto qc = q.contiguous();
+  auto kc = k.contiguous();
+  auto vc = v.contiguous();
+  auto bc = beta.to(torch::kFloat32).contiguous();
+  auto gc = decay_log.to(torch::kFloat32).contiguous();
+  const int64_t B = qc.size(0), H = qc.size(1), S = qc.size(2);
+  auto out = torch::empty_like(vc);
+  torch::Tensor fs;
+  float* fs_ptr = nullptr;
+  if (return_state) {
+    fs = torch::empty({B, H, 64, 64},
+                      torch::dtype(torch::kFloat32).device(q.device()));
+    fs_ptr = fs.data_ptr<float>();
+  }
+  const size_t smem = sizeof(d9d::GdnLds);
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&d9d::gdn_chunk_fwd_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);
+    attr_set = true;
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(d9d::gdn_chunk_fwd_kernel, dim3((unsigned)(B * H)),
+                     dim3(256), smem, stream,
+                     reinterpret_cast<const __bf16*>(qc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(kc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(vc.data_ptr()),
+                     bc.data_ptr<float>(), gc.data_ptr<float>(),
+                     reinterpret_cast<__bf16*>(out.data_ptr()),
+                     fs_ptr, B * H, S);
+  if (return_state) return {out, fs};
+  return {out};
+}

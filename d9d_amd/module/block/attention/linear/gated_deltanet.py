@@ -18,7 +18,12 @@ import torch.nn.functional as F
 from torch import nn
 
 from .....ops import silu_mul
+from .....ops._ext import get_ext, has_ext
 from ...normalization import RMSNorm
+
+
+def _use_gdn_kernels(t: torch.Tensor) -> bool:
+    return t.is_cuda and has_ext()
 
 
 class CausalShortDepthwiseConv1d(nn.Module):
@@ -37,11 +42,30 @@ class CausalShortDepthwiseConv1d(nn.Module):
             nn.init.normal_(self.weight, std=1.0 / math.sqrt(self.kernel_size))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:  # (B, S, C)
+        if _use_gdn_kernels(x) and x.dtype == torch.bfloat16:
+            return _CausalConvSiluFunction.apply(x, self.weight)
         B, S, C = x.shape
         xt = x.transpose(1, 2)  # (B, C, S)
         xt = F.pad(xt, (self.kernel_size - 1, 0))
         out = F.conv1d(xt, self.weight.unsqueeze(1), groups=C)
         return F.silu(out.transpose(1, 2))
+
+
+class _CausalConvSiluFunction(torch.autograd.Function):
+    """Fused causal depthwise conv + SiLU (csrc/gdn.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        return get_ext().causal_conv_silu_fwd(x.contiguous(), w.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dx, dw = get_ext().causal_conv_silu_bwd(
+            x.contiguous(), w.contiguous(), dy.contiguous()
+        )
+        return dx, dw
 
 
 def step_gated_delta_rule(
@@ -79,6 +103,57 @@ def chunk_gated_delta_rule(
     v: torch.Tensor,  # (B, H, S, Dv)
     beta: torch.Tensor,  # (B, H, S) write strength in [0, 1]
     decay_log: torch.Tensor,  # (B, H, S) log decay (<= 0)
+    chunk_size: int = 64,
+) -> torch.Tensor:
+    """GPU (Dk = Dv = 64): fused CDNA4 chunk kernel (csrc/gdn.hip) — the
+    fp32 state stays LDS-resident across the chunk loop and every chunk
+    GEMM lands on MFMA; backward recomputes through the torch WY graph
+    (a native bwd kernel is the follow-up). Other shapes / CPU: the
+    chunked-parallel WY torch path below."""
+    if (
+        _use_gdn_kernels(q)
+        and q.shape[-1] == 64
+        and v.shape[-1] == 64
+        and chunk_size == 64
+    ):
+        return _GdnChunkFunction.apply(q, k, v, beta, decay_log)
+    return _chunk_gated_delta_rule_torch(q, k, v, beta, decay_log, chunk_size)
+
+
+class _GdnChunkFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, beta, decay_log):
+        out = get_ext().gdn_chunk_fwd(
+            q.to(torch.bfloat16).contiguous(),
+            k.to(torch.bfloat16).contiguous(),
+            v.to(torch.bfloat16).contiguous(),
+            beta.contiguous(),
+            decay_log.contiguous(),
+            False,
+        )[0]
+        ctx.save_for_backward(q, k, v, beta, decay_log)
+        return out.to(v.dtype)
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, beta, decay_log = ctx.saved_tensors
+        with torch.enable_grad():
+            qs = q.detach().requires_grad_(True)
+            ks = k.detach().requires_grad_(True)
+            vs = v.detach().requires_grad_(True)
+            bs = beta.detach().requires_grad_(True)
+            gs = decay_log.detach().requires_grad_(True)
+            ref = _chunk_gated_delta_rule_torch(qs, ks, vs, bs, gs)
+            grads = torch.autograd.grad(ref, (qs, ks, vs, bs, gs), dout)
+        return grads
+
+
+def _chunk_gated_delta_rule_torch(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    beta: torch.Tensor,
+    decay_log: torch.Tensor,
     chunk_size: int = 64,
 ) -> torch.Tensor:
     """Chunked-parallel gated delta rule (WY form, replacing fla-core's

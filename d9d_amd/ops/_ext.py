@@ -30,6 +30,7 @@ _SOURCES = [
     "gmm.hip",
     "attention.hip",
     "cce.hip",
+    "gdn.hip",
 ]
 
 _ext_module = None
