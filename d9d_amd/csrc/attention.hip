@@ -672,10 +672,12 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
 
   for (int qt = q_start; qt < Sq_loc; qt += kQBlk) {
     if (qt + kQBlk < Sq_loc) load_qdo_regs(qt + kQBlk);  // issue early
-    // dO^T and Q^T images for the dV/dK B fragments (cheap LDS-to-LDS)
+    // dO^T and Q^T images for the dV/dK B fragments (cheap LDS-to-LDS).
+    // NO barrier here: tdo/tq are first read after the x-stage barrier
+    // below, which also publishes these writes (2 barriers per q-tile
+    // instead of 4 — the PMC wave-parked share was the top bwd cost).
     transpose_lds_tile<D, kBwdThreads>(tdo_lds, do_lds, threadIdx.x);
     transpose_lds_tile<D, kBwdThreads>(tq_lds, q_lds, threadIdx.x);
-    __syncthreads();
 
     // ---- S^T = K @ Q^T (per wave: 16 kv x 64 q) ----------------------------
     f32x4 st_acc[4];
@@ -853,9 +855,12 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
         }
       }
     }
-    __syncthreads();
+    // The Q/dO store overlaps the dV/dK/dQ MFMA phases above: nothing in
+    // them reads q_lds/do_lds (their consumers — S^T, dP^T and the
+    // transposes — all ran before the x-stage barrier), so the write only
+    // needs the end-of-iteration barrier to publish for the next tile.
     if (qt + kQBlk < Sq_loc) {
-      store_qdo_lds();  // prefetched next Q/dO land after all phases read LDS
+      store_qdo_lds();
     }
     __syncthreads();
   }
