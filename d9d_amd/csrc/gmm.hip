@@ -15,6 +15,10 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <map>
+#include <mutex>
+#include <tuple>
+
 namespace d9d {
 
 typedef __bf16 bf16_t;
@@ -963,9 +967,31 @@ __global__ __launch_bounds__(512, 1) void gmm_db_kernel(
 
 // ---------------------------------------------------------------------------
 
+// E == 1 (dense KernelLinear) offsets depend only on the row count: cache
+// the device tensors so the hot path pays no per-call CPU alloc + H2D
+// (the round-1 measured overhead that cancelled the kernel's 600-vs-260
+// TF/s win at bench scale). MoE sizes change per routing step: no cache.
+static std::map<std::tuple<int64_t, int, int>,
+                std::tuple<torch::Tensor, torch::Tensor, int>> g_e1_offsets;
+static std::mutex g_e1_offsets_mu;
+
 static std::tuple<torch::Tensor, torch::Tensor, int> build_offsets(
     torch::Tensor batch_sizes, torch::Device device, int tile_m) {
   const int E = batch_sizes.numel();
+  if (E == 1) {
+    const int64_t rows = batch_sizes.to(torch::kInt64).item<int64_t>();
+    const auto key = std::make_tuple(rows, tile_m, (int)device.index());
+    std::lock_guard<std::mutex> lock(g_e1_offsets_mu);
+    auto it = g_e1_offsets.find(key);
+    if (it != g_e1_offsets.end()) return it->second;
+    auto ro = torch::tensor({(int)0, (int)rows},
+                            torch::dtype(torch::kInt32)).to(device);
+    const int tiles = (int)((rows + tile_m - 1) / tile_m);
+    auto mp = torch::tensor({0, tiles}, torch::dtype(torch::kInt32)).to(device);
+    auto val = std::make_tuple(ro, mp, tiles);
+    g_e1_offsets.emplace(key, val);
+    return val;
+  }
   auto row_off = torch::empty({E + 1}, torch::dtype(torch::kInt32));
   auto mtile_pref = torch::empty({E + 1}, torch::dtype(torch::kInt32));
   auto bs = batch_sizes.to(torch::kInt64);

@@ -17,9 +17,11 @@ from torch import nn
 from ...ops import gmm_nt
 from ...ops._ext import has_ext
 
-# Opt-in: the NT kernel beats rocBLAS ~2x at the projection shapes in
-# isolation, but per-call host overhead (group-offset build + H2D) cancels
-# the win end-to-end at bench scale; measured A/B kept rocBLAS as default.
+# Still opt-in after round 2: with the E==1 offset tables cached device-
+# side (gmm.hip) the per-call host overhead is gone, but the measured
+# end-to-end A/B is 116.1k (on) vs 116.9k (off) tokens/s — rocBLAS keeps
+# the dense projections. The NT kernel's isolated-shape win does not
+# survive the step's cache/occupancy interleaving.
 _ENABLED = bool(int(os.environ.get("D9D_KERNEL_LINEAR", "0")))
 
 
