@@ -48,9 +48,14 @@ union ushort2_t {  // 4 bf16 lanes packed for one 8-byte LDS store
 // (128-byte rows, ((d&7)<<4) swizzle) using 4B LDS reads + 8B LDS writes.
 template <int D, int BLOCK = 256>
 D9D_DEVICE void transpose_lds_tile(bf16_t* dst, const bf16_t* src_rm, int tid) {
+  // Work mapping: consecutive threads take consecutive ROW-quads of the
+  // same d-pair, so the two b64 writes of a 16-thread group land on one
+  // [d][64] row contiguously (128 B span, conflict-free). The b32 source
+  // reads pay a 4-way conflict instead of the old mapping's 8-way b64
+  // write conflict (rows 256 B apart XOR-spread only 4 ways per group).
   for (int idx = tid; idx < (64 / 4) * (D / 2); idx += BLOCK) {
-    const int d0 = (idx % (D / 2)) * 2;
-    const int rb = (idx / (D / 2)) * 4;
+    const int d0 = (idx / 16) * 2;
+    const int rb = (idx % 16) * 4;
     ushort2_t c0, c1;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {

@@ -420,3 +420,34 @@ def test_adamw_multi_matches_single():
         assert torch.equal(ps[i], ps2[i]), i  # bitwise: same per-slot RNG
         torch.testing.assert_close(ms[i], ms2[i])
         torch.testing.assert_close(vs[i], vs2[i])
+
+
+@pytest.mark.gpu
+def test_cce_lse_grad_kernel_path_gpu():
+    """The fused dlogits kernel's dlse scale (bf16 GPU path) matches the
+    eager fp32 reference for a mixed loss+lse objective."""
+    import torch
+
+    from d9d_amd.ops.cce import linear_cross_entropy
+
+    torch.manual_seed(12)
+    T, H, V = 256, 128, 512
+    e = (torch.randn(T, H, device="cuda") * 0.5).bfloat16().requires_grad_(True)
+    c = (torch.randn(V, H, device="cuda") * 0.1).bfloat16().requires_grad_(True)
+    tg = torch.randint(0, V, (T,), device="cuda")
+    tg[0] = -100
+
+    loss, lse = linear_cross_entropy(e, c, tg, return_lse=True, filter_eps=None)
+    (loss.sum() + 0.3 * (lse ** 2).sum()).backward()
+
+    e32 = e.detach().float().requires_grad_(True)
+    c32 = c.detach().float().requires_grad_(True)
+    logits = e32 @ c32.t()
+    lse32 = torch.logsumexp(logits, -1)
+    nll = torch.nn.functional.cross_entropy(
+        logits, tg.clamp(min=0), reduction="none"
+    ) * (tg != -100)
+    (nll.sum() + 0.3 * (lse32 ** 2).sum()).backward()
+
+    torch.testing.assert_close(e.grad.float(), e32.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(c.grad.float(), c32.grad, rtol=5e-2, atol=5e-2)
