@@ -23,6 +23,7 @@ namespace d9d {
 
 typedef __bf16 bf16_t;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 D9D_DEVICE f32x4 mfma16g(bf16x8 a, bf16x8 b, f32x4 c) {
@@ -808,6 +809,249 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_8phase_kernel(
   }
 }
 
+// 8-phase counted-vmcnt gmm for the NN layout (dgrad: out = g @ w[e] with
+// w stored (E, K, N), n-contiguous rows). Identical pipeline to
+// gmm_nt_8phase_kernel; the ONE difference is the B operand: its MFMA
+// fragment needs 8 reduction (k) elements at fixed n — a COLUMN of the
+// row-major weight. glds cannot transpose, so B stages ROW-major into a
+// [K/4][BN/16][4][16] blocked image (16-byte units are 8 consecutive n at
+// one k: still lane-linear, still coalesced) and fragments are read with
+// ds_read_b64_tr_b16: each 16-lane group reads one 128-B [4 k][16 n]
+// block transposed, two reads per fragment (probe: scratch/tr16_probe.py,
+// map in gpurun_out/tr16_map.txt).
+template <int BN8>
+__global__ __launch_bounds__(512, 1) void gmm_nn_8phase_kernel(
+    const bf16_t* __restrict__ a,    // (T, K)
+    const bf16_t* __restrict__ w,    // (E, K, N)
+    bf16_t* __restrict__ out,        // (T, N)
+    const int* __restrict__ row_off,
+    const int* __restrict__ mtile_pref,
+    int E, int K, int N, int n_tiles) {
+  constexpr int NJ = BN8 / 64;
+  constexpr int SLOT = (256 + BN8) * 128;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  const int nwg = gridDim.x;
+  const int xcd = blockIdx.x & 7;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int wid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                  + (blockIdx.x >> 3);
+  const int mt_global = wid / n_tiles;
+  int lo = 0, hi = E - 1;
+  while (lo < hi) {
+    const int mid = (lo + hi + 1) >> 1;
+    if (mtile_pref[mid] <= mt_global) lo = mid; else hi = mid - 1;
+  }
+  const int e = lo;
+  const int m_tile = mt_global - mtile_pref[e];
+  const int row0 = row_off[e] + m_tile * kBM;
+  const int row_end = row_off[e + 1];
+  const int n0 = (wid % n_tiles) * BN8;
+
+  const bf16_t* w_e = w + (int64_t)e * K * N;
+
+  f32x4 acc[8][NJ];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int n_ktiles = (K + kBK - 1) / kBK;
+  const bool k_tail = (K % kBK) != 0;   // K % 32 == 0 guaranteed by host
+  const int last_kt = n_ktiles - 1;
+
+  auto ktile_k0 = [&](int kt) {
+    return (k_tail && kt == last_kt) ? K - kBK : kt * kBK;
+  };
+
+  // A half (128 rows x 64 k = 16 KB): 2 glds per thread (same as NT).
+  auto stage_a = [&](int kt, int half) {
+    const int k0 = ktile_k0(kt);
+    char* base = smem + (size_t)(kt & 1) * SLOT + half * (16 * 1024);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int chunk = wave + i * 8;
+      const int l = chunk * 1024 + lane * 16;
+      const int arow = half * 128 + (l >> 7);
+      const int colb = (l & 127) ^ ((arow & 7) << 4);
+      const int64_t sr = min(row0 + arow, row_end - 1);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const uint32_t*>(
+              reinterpret_cast<const char*>(a) + (sr * (int64_t)K + k0) * 2 + colb),
+          reinterpret_cast<uint32_t*>(base + l), 16, 0, 0);
+    }
+  };
+  // B chunk (8 KB of the blocked [64/4][BN8/16][4][16] k-tile image):
+  // 1 glds per thread. Image element e: kb = e / (BN8*4), then
+  // nb = (e % (BN8*4)) / 64, krow = (e % 64) / 16, ncol = e % 16.
+  auto stage_b = [&](int kt, int chunk) {
+    const int k0 = ktile_k0(kt);
+    char* base = smem + (size_t)(kt & 1) * SLOT + (32 * 1024);
+    const int l = chunk * 8192 + wave * 1024 + lane * 16;  // byte offset
+    const int el = l >> 1;                                  // bf16 element
+    const int kb = el / (BN8 * 4);
+    const int rem = el % (BN8 * 4);
+    const int nb = rem >> 6;
+    const int krow = (rem & 63) >> 4;
+    const int ncol = rem & 15;   // 0 or 8 at 16-B granularity
+    const int brow = k0 + kb * 4 + krow;
+    const int64_t bcol = min((int64_t)(n0 + nb * 16 + ncol), (int64_t)N - 8);
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const uint32_t*>(
+            reinterpret_cast<const char*>(w_e) + ((int64_t)brow * N + bcol) * 2),
+        reinterpret_cast<uint32_t*>(base + l), 16, 0, 0);
+  };
+
+#pragma unroll
+  for (int c = 0; c < NJ; ++c) stage_b(0, c);
+  stage_a(0, 0); stage_a(0, 1);
+  if (n_ktiles > 1) {
+#pragma unroll
+    for (int c = 0; c < NJ; ++c) stage_b(1, c);
+  }
+
+  const int n_full = k_tail ? n_ktiles - 1 : n_ktiles;
+  for (int kt = 0; kt < n_full; ++kt) {
+    const char* a_base = smem + (size_t)(kt & 1) * SLOT;
+    const char* b_base = a_base + 32 * 1024;
+
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NJ) : "memory");
+    __builtin_amdgcn_s_barrier();
+
+    bf16x8 b_frag[NJ][2];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      bf16x8 a_frag[2][2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int kk = ks * 32 + (lane >> 4) * 8;
+#pragma unroll
+        for (int il = 0; il < 2; ++il) {
+          const int row = wm * 128 + (2 * p + il) * 16 + (lane & 15);
+          const int byte = (kk * 2) ^ ((row & 7) << 4);
+          a_frag[il][ks] = *reinterpret_cast<const bf16x8*>(
+              a_base + row * (kBK * 2) + byte);
+        }
+        if (p == 0) {
+#pragma unroll
+          for (int j = 0; j < NJ; ++j) {
+            const int nb = (wn * (BN8 / 4) + j * 16) >> 4;
+            // fragment k-range kk..kk+8 spans kb blocks 2*(kk/8), +1
+            const int kb0 = kk >> 2;  // = (kk/4); kk multiple of 8
+            const char* blk0 = b_base + ((size_t)kb0 * (BN8 / 16) + nb) * 128
+                               + (lane & 15) * 8;
+            const char* blk1 = b_base + ((size_t)(kb0 + 1) * (BN8 / 16) + nb) * 128
+                               + (lane & 15) * 8;
+            bf16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (__attribute__((address_space(3))) bf16x4*)(blk0));
+            bf16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (__attribute__((address_space(3))) bf16x4*)(blk1));
+            bf16x8 f;
+#pragma unroll
+            for (int t = 0; t < 4; ++t) { f[t] = lo4[t]; f[4 + t] = hi4[t]; }
+            b_frag[j][ks] = f;
+          }
+        }
+      }
+      if (p == 0) {
+        if (kt + 1 < n_ktiles) stage_a(kt + 1, 0);
+      } else if (p == 1) {
+        if (kt + 1 < n_ktiles) stage_a(kt + 1, 1);
+        if (kt + 2 < n_ktiles) {
+          stage_b(kt + 2, 0);
+          if (NJ == 4) stage_b(kt + 2, 1);
+        }
+      } else if (p == 2) {
+        if (kt + 2 < n_ktiles) {
+          if (NJ == 4) { stage_b(kt + 2, 2); stage_b(kt + 2, 3); }
+          else stage_b(kt + 2, 1);
+        }
+      } else {
+        if (NJ == 3 && kt + 2 < n_ktiles) stage_b(kt + 2, 2);
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int il = 0; il < 2; ++il)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[2 * p + il][j] =
+                mfma16g(a_frag[il][ks], b_frag[j][ks], acc[2 * p + il][j]);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  if (k_tail) {
+    const int kt = last_kt;
+    const char* a_base = smem + (size_t)(kt & 1) * SLOT;
+    const char* b_base = a_base + 32 * 1024;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    bf16x8 b_frag_t[NJ];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      bf16x8 a_frag_t[2];
+      const int kk = 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int il = 0; il < 2; ++il) {
+        const int row = wm * 128 + (2 * p + il) * 16 + (lane & 15);
+        const int byte = (kk * 2) ^ ((row & 7) << 4);
+        a_frag_t[il] = *reinterpret_cast<const bf16x8*>(
+            a_base + row * (kBK * 2) + byte);
+      }
+      if (p == 0) {
+#pragma unroll
+        for (int j = 0; j < NJ; ++j) {
+          const int nb = (wn * (BN8 / 4) + j * 16) >> 4;
+          const int kb0 = kk >> 2;
+          const char* blk0 = b_base + ((size_t)kb0 * (BN8 / 16) + nb) * 128
+                             + (lane & 15) * 8;
+          const char* blk1 = b_base + ((size_t)(kb0 + 1) * (BN8 / 16) + nb) * 128
+                             + (lane & 15) * 8;
+          bf16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4*)(blk0));
+          bf16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4*)(blk1));
+          bf16x8 f;
+#pragma unroll
+          for (int t = 0; t < 4; ++t) { f[t] = lo4[t]; f[4 + t] = hi4[t]; }
+          b_frag_t[j] = f;
+        }
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int il = 0; il < 2; ++il)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j)
+          acc[2 * p + il][j] =
+              mfma16g(a_frag_t[il], b_frag_t[j], acc[2 * p + il][j]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) {
+#pragma unroll
+      for (int r2 = 0; r2 < 4; ++r2) {
+        const int row = row0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r2;
+        const int col = n0 + wn * (BN8 / 4) + j * 16 + (lane & 15);
+        if (row < row_end && col < N) {
+          out[(int64_t)row * N + col] = (bf16_t)acc[i][j][r2];
+        }
+      }
+    }
+  }
+}
+
 // db[e] = a[rows_e]^T @ g[rows_e]: out (E, K, N).
 // Grid (ceil(N/256), ceil(K/256), E): 256x256 tiles per workgroup -- 8 waves
 // as 2 (k-halves of 128) x 4 (n-quarters of 64) -- looping the expert's rows
@@ -1027,11 +1271,55 @@ torch::Tensor gmm(torch::Tensor a, torch::Tensor b, torch::Tensor batch_sizes) {
   auto [row_off, mtile_pref, total_mtiles] =
       build_offsets(batch_sizes, a.device(), d9d::kBM);
   if (total_mtiles == 0) return out;
+  auto stream = at::hip::getCurrentHIPStream();
+
+  // 8-phase + tr16-fragment path (B stays row-major, staged by glds into
+  // the [K/4][BN/16][4][16] blocked image). D9D_GMM_NN_8PHASE=0 opts out.
+  static const bool use_nn8 = []() {
+    const char* v = getenv("D9D_GMM_NN_8PHASE");
+    return v == nullptr || v[0] != '0';
+  }();
+  if (use_nn8 && K % 32 == 0 && K >= 64 && N % 16 == 0) {
+    const int pad256 = ((N + 255) / 256) * 256 - N;
+    const int pad192 = ((N + 191) / 192) * 192 - N;
+    const bool use256 = pad256 <= pad192;
+    const int bn = use256 ? 256 : 192;
+    const int n_tiles8 = (N + bn - 1) / bn;
+    const dim3 grid8(n_tiles8 * total_mtiles);
+    const size_t smem8 = (size_t)2 * (256 + bn) * 128;
+    static bool attr_set_nn = false;
+    if (!attr_set_nn) {
+      hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&d9d::gmm_nn_8phase_kernel<256>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 128 * 1024);
+      hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&d9d::gmm_nn_8phase_kernel<192>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 112 * 1024);
+      attr_set_nn = true;
+    }
+    if (use256) {
+      hipLaunchKernelGGL(d9d::gmm_nn_8phase_kernel<256>, grid8, dim3(512),
+                         smem8, stream,
+                         reinterpret_cast<const __bf16*>(a.data_ptr()),
+                         reinterpret_cast<const __bf16*>(b.data_ptr()),
+                         reinterpret_cast<__bf16*>(out.data_ptr()),
+                         row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
+                         E, K, N, n_tiles8);
+    } else {
+      hipLaunchKernelGGL(d9d::gmm_nn_8phase_kernel<192>, grid8, dim3(512),
+                         smem8, stream,
+                         reinterpret_cast<const __bf16*>(a.data_ptr()),
+                         reinterpret_cast<const __bf16*>(b.data_ptr()),
+                         reinterpret_cast<__bf16*>(out.data_ptr()),
+                         row_off.data_ptr<int>(), mtile_pref.data_ptr<int>(),
+                         E, K, N, n_tiles8);
+    }
+    return out;
+  }
 
   const int n_tiles = (N + d9d::kBN - 1) / d9d::kBN;
   const dim3 grid(n_tiles * total_mtiles);
   const size_t smem = (d9d::kBM * d9d::kBK + d9d::kBN * d9d::kBK) * sizeof(__bf16);
-  auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(d9d::gmm_kernel, grid, dim3(512), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(b.data_ptr()),
