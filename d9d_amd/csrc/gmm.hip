@@ -692,8 +692,16 @@ __global__ __launch_bounds__(512, 1) void gmm_nt_8phase_kernel(
     const char* a_base = smem + (size_t)(kt & 1) * SLOT;
     const char* b_base = a_base + 32 * 1024;
 
-    // k-tile boundary: this tile's staging landed; <= NJ newer in flight.
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NJ) : "memory");
+    // k-tile boundary: this tile's staging landed. In steady state the NJ
+    // newer loads are B(kt+1); when B(kt+1) was NOT staged (last tile) the
+    // newest outstanding loads are THIS tile's A halves — drain fully
+    // (vmcnt(NJ) there let the final tile read in-flight A: rare flake
+    // caught by the 12x race screen).
+    if (kt + 1 < n_ktiles) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NJ) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
 
     bf16x8 b_frag[NJ][2];
@@ -919,7 +927,11 @@ __global__ __launch_bounds__(512, 1) void gmm_nn_8phase_kernel(
     const char* a_base = smem + (size_t)(kt & 1) * SLOT;
     const char* b_base = a_base + 32 * 1024;
 
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NJ) : "memory");
+    if (kt + 1 < n_ktiles) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NJ) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
 
     bf16x8 b_frag[NJ][2];
@@ -1046,6 +1058,172 @@ __global__ __launch_bounds__(512, 1) void gmm_nn_8phase_kernel(
         const int col = n0 + wn * (BN8 / 4) + j * 16 + (lane & 15);
         if (row < row_end && col < N) {
           out[(int64_t)row * N + col] = (bf16_t)acc[i][j][r2];
+        }
+      }
+    }
+  }
+}
+
+// 8-phase glds + tr16 wgrad: db[e] = a[rows_e]^T @ g[rows_e], out (E, K, N).
+// Same block mapping as gmm_db_kernel (expert-order round-robin over
+// (e, k-tile, n-tile), 256x256 C tiles, ragged row loop in 64-row chunks)
+// but the staging is the deep pipeline: BOTH operand chunks stage ROW-major
+// via global_load_lds into [rows/4][cols/16][4][16] blocked images and BOTH
+// MFMA fragments (reduction = rows) come from ds_read_b64_tr_b16 pairs.
+// Rows past the expert's end read from a 16-byte zero buffer (glds cannot
+// mask; a per-lane address select costs one v_cndmask, no branch), so pad
+// rows contribute exact zeros to the row-sum.
+__global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
+    const bf16_t* __restrict__ a,   // (T, K)
+    const bf16_t* __restrict__ g,   // (T, N)
+    bf16_t* __restrict__ db,        // (E, K, N)
+    const int* __restrict__ row_off,
+    const int* __restrict__ expert_order,
+    const bf16_t* __restrict__ zero16,  // >= 8 zero bf16
+    int E, int K, int N, int kt_tiles, int nt_tiles) {
+  constexpr int SLOT = 64 * 1024;  // A image 32 KB + B image 32 KB
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;  // k-half (128 rows of db)
+  const int wn = wave & 3;   // n-quarter (64 cols)
+
+  const int wid = blockIdx.x;
+  const int tiles_per_e = kt_tiles * nt_tiles;
+  const int e = expert_order[wid / tiles_per_e];
+  const int k0 = ((wid % tiles_per_e) / nt_tiles) * 256;
+  const int n0 = (wid % nt_tiles) * 256;
+  const int r_start = row_off[e];
+  const int r_end = row_off[e + 1];
+  if (r_start >= r_end) return;  // empty expert: db stays zero (host zeros it)
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int n_rtiles = (r_end - r_start + 63) / 64;
+
+  // Stage one 8 KB chunk (16 rows x 256 cols) of operand A or B for row
+  // tile rt: 1 glds per thread. Image element e (bf16):
+  // rb = e/1024, cb = (e%1024)/64, rrow = (e%64)/16, col8 = e%16.
+  auto stage = [&](const bf16_t* src, int64_t stride, int c0, int cmax,
+                   int rt, int chunk, int img_off) {
+    char* base = smem + (size_t)(rt & 1) * SLOT + img_off;
+    const int l = chunk * 8192 + wave * 1024 + lane * 16;  // byte offset
+    const int el = l >> 1;
+    const int rb = el >> 10;
+    const int rem = el & 1023;
+    const int cb = rem >> 6;
+    const int rrow = (rem & 63) >> 4;
+    const int ccol = rem & 15;  // 0 or 8
+    const int row = r_start + rt * 64 + rb * 4 + rrow;
+    const int64_t col = min((int64_t)(c0 + cb * 16 + ccol), (int64_t)cmax - 8);
+    const char* sp =
+        (row < r_end)
+            ? reinterpret_cast<const char*>(src) + (row * stride + col) * 2
+            : reinterpret_cast<const char*>(zero16);
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const uint32_t*>(sp),
+        reinterpret_cast<uint32_t*>(base + l), 16, 0, 0);
+  };
+  auto stage_a = [&](int rt, int chunk) { stage(a, K, k0, K, rt, chunk, 0); };
+  auto stage_g = [&](int rt, int chunk) {
+    stage(g, N, n0, N, rt, chunk, 32 * 1024);
+  };
+
+  // tr16 fragment: reduction slice red0..red0+8 at 16 cols [ct*16, ct*16+16)
+  auto frag = [&](const char* img, int red0, int ct) -> bf16x8 {
+    const int rb0 = red0 >> 2;
+    const char* blk0 = img + ((size_t)rb0 * 16 + ct) * 128 + (lane & 15) * 8;
+    const char* blk1 = img + ((size_t)(rb0 + 1) * 16 + ct) * 128 + (lane & 15) * 8;
+    bf16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4*)(blk0));
+    bf16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4*)(blk1));
+    bf16x8 f;
+#pragma unroll
+    for (int t = 0; t < 4; ++t) { f[t] = lo4[t]; f[4 + t] = hi4[t]; }
+    return f;
+  };
+
+  // Prologue mirrors the steady-state issue order.
+#pragma unroll
+  for (int c = 0; c < 4; ++c) stage_g(0, c);
+  stage_a(0, 0); stage_a(0, 1); stage_a(0, 2); stage_a(0, 3);
+  if (n_rtiles > 1) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) stage_g(1, c);
+  }
+
+  for (int rt = 0; rt < n_rtiles; ++rt) {
+    const char* a_img = smem + (size_t)(rt & 1) * SLOT;
+    const char* g_img = a_img + 32 * 1024;
+
+    if (rt + 1 < n_rtiles) {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      // no B(rt+1) staged: the newest outstanding loads are THIS row
+      // tile's A chunks — drain fully
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    bf16x8 b_frag[4][2];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      bf16x8 a_frag[2][2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int red0 = ks * 32 + (lane >> 4) * 8;
+#pragma unroll
+        for (int il = 0; il < 2; ++il) {
+          const int kct = (wm * 128 + (2 * p + il) * 16) >> 4;
+          a_frag[il][ks] = frag(a_img, red0, kct);
+        }
+        if (p == 0) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const int nct = (wn * 64 + j * 16) >> 4;
+            b_frag[j][ks] = frag(g_img, red0, nct);
+          }
+        }
+      }
+      if (p == 0) {
+        if (rt + 1 < n_rtiles) { stage_a(rt + 1, 0); stage_a(rt + 1, 1); }
+      } else if (p == 1) {
+        if (rt + 1 < n_rtiles) { stage_a(rt + 1, 2); stage_a(rt + 1, 3); }
+        if (rt + 2 < n_rtiles) { stage_g(rt + 2, 0); stage_g(rt + 2, 1); }
+      } else if (p == 2) {
+        if (rt + 2 < n_rtiles) { stage_g(rt + 2, 2); stage_g(rt + 2, 3); }
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int il = 0; il < 2; ++il)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[2 * p + il][j] =
+                mfma16g(a_frag[il][ks], b_frag[j][ks], acc[2 * p + il][j]);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  bf16_t* db_e = db + (int64_t)e * K * N;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r2 = 0; r2 < 4; ++r2) {
+        const int krow = k0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r2;
+        const int ncol = n0 + wn * 64 + j * 16 + (lane & 15);
+        if (krow < K && ncol < N) {
+          db_e[(int64_t)krow * N + ncol] = (bf16_t)acc[i][j][r2];
         }
       }
     }
@@ -1449,8 +1627,50 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
 
   const int nt_tiles = (N + 255) / 256, kt_tiles = (K + 255) / 256;
   const dim3 grid(nt_tiles * kt_tiles * E);
-  const size_t smem = (2 * 256 * 64) * sizeof(__bf16);
   auto stream = at::hip::getCurrentHIPStream();
+
+  // 8-phase glds + tr16 pipeline (both operands row-major staged).
+  // D9D_GMM_DB_8PHASE=0 opts back into the scatter-staged kernel.
+  static const bool use_db8 = []() {
+    const char* v = getenv("D9D_GMM_DB_8PHASE");
+    return v == nullptr || v[0] != '0';
+  }();
+  // K/N must be multiples of 8: a 16-byte staging unit must never straddle
+  // the tensor edge (the clamp would shift VALID columns' data — caught by
+  // the ragged K=100 parity case).
+  if (use_db8 && K >= 8 && N >= 8 && K % 8 == 0 && N % 8 == 0) {
+    // cached 16-byte zero source for ragged row tails (glds cannot mask)
+    static std::map<int, torch::Tensor> zeros_by_dev;
+    torch::Tensor z;
+    {
+      std::lock_guard<std::mutex> lock(g_e1_offsets_mu);
+      auto it = zeros_by_dev.find((int)a.device().index());
+      if (it == zeros_by_dev.end()) {
+        z = torch::zeros({8}, a.options());
+        zeros_by_dev.emplace((int)a.device().index(), z);
+      } else {
+        z = it->second;
+      }
+    }
+    static bool attr_set_db = false;
+    if (!attr_set_db) {
+      hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&d9d::gmm_db_8phase_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 128 * 1024);
+      attr_set_db = true;
+    }
+    hipLaunchKernelGGL(d9d::gmm_db_8phase_kernel, grid, dim3(512),
+                       128 * 1024, stream,
+                       reinterpret_cast<const __bf16*>(a.data_ptr()),
+                       reinterpret_cast<const __bf16*>(g.data_ptr()),
+                       reinterpret_cast<__bf16*>(db.data_ptr()),
+                       row_off.data_ptr<int>(), expert_order.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(z.data_ptr()),
+                       E, K, N, kt_tiles, nt_tiles);
+    return db;
+  }
+
+  const size_t smem = (2 * 256 * 64) * sizeof(__bf16);
   hipLaunchKernelGGL(d9d::gmm_db_kernel, grid, dim3(512), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
