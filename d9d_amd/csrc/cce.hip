@@ -104,11 +104,13 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
   const int my_row_local = m_tile * 16 + (lane & 15);
 
   float m_run[4], l_run[4], tgt[4];
+  int tg_row[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     m_run[r] = -1e30f;
     l_run[r] = 0.f;
     tgt[r] = -1e30f;
+    tg_row[r] = (int)targets[min(t0 + m_tile * 16 + (lane >> 4) * 4 + r, T - 1)];
   }
 
   f32x4 acc[4];
@@ -140,10 +142,10 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     if (kt == n_ktiles - 1) {
-      // logits tile complete: online lse + target capture, reset acc.
-      float m_new[4];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) m_new[r] = m_run[r];
+      // Logits tile complete. PER-LANE online lse over this lane's own 4
+      // columns — the 16-lane merge happens ONCE after the sweep, so the
+      // hot epilogue has NO cross-lane shuffles (they were ~2/3 of the
+      // per-tile VALU work: 96 shfl_xor per vocab tile).
       float vals[4][4];
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -153,43 +155,25 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
           vals[nt][r] = (col < V) ? acc[nt][r] : -1e30f;
         }
       }
-      // target capture: row = t0 + m_tile*16 + (lane>>4)*4 + r
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row_g = t0 + m_tile * 16 + (lane >> 4) * 4 + r;
-        const int64_t tg = targets[min(row_g, T - 1)];
 #pragma unroll
         for (int nt = 0; nt < 4; ++nt) {
           const int col = vt * 128 + vhalf * 64 + nt * 16 + (lane & 15);
-          float hit = (col == (int)tg) ? vals[nt][r] : -1e30f;
-#pragma unroll
-          for (int off = 1; off < 16; off <<= 1) {
-            hit = fmaxf(hit, __shfl_xor(hit, off, 64));
-          }
-          tgt[r] = fmaxf(tgt[r], hit);
+          if (col == (int)tg_row[r]) tgt[r] = fmaxf(tgt[r], vals[nt][r]);
         }
-      }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float mx = fmaxf(fmaxf(vals[0][r], vals[1][r]),
-                         fmaxf(vals[2][r], vals[3][r]));
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1) {
-          mx = fmaxf(mx, __shfl_xor(mx, off, 64));
-        }
-        m_new[r] = fmaxf(m_new[r], mx);
+        const float mx = fmaxf(fmaxf(vals[0][r], vals[1][r]),
+                               fmaxf(vals[2][r], vals[3][r]));
+        const float m_new = fmaxf(m_run[r], mx);
         float add = 0.f;
 #pragma unroll
         for (int nt = 0; nt < 4; ++nt) {
-          add += __builtin_amdgcn_exp2f((vals[nt][r] - m_new[r]) * kLog2eC);
+          add += __builtin_amdgcn_exp2f((vals[nt][r] - m_new) * kLog2eC);
         }
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1) {
-          add += __shfl_xor(add, off, 64);
-        }
-        const float alpha = __builtin_amdgcn_exp2f((m_run[r] - m_new[r]) * kLog2eC);
-        l_run[r] = l_run[r] * alpha + add;
-        m_run[r] = m_new[r];
+        l_run[r] = l_run[r] *
+                       __builtin_amdgcn_exp2f((m_run[r] - m_new) * kLog2eC) +
+                   add;
+        m_run[r] = m_new;
       }
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
@@ -202,7 +186,21 @@ __global__ __launch_bounds__(512, 2) void cce_fwd_kernel(
     }
   }
 
-  // ---- merge the two vocab halves, write lse + target logit ---------------
+  // ---- merge this wave's 16 col-lanes (once per kernel), then the two
+  // vocab halves; write lse + target logit ----------------------------------
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1) {
+      const float mo = __shfl_xor(m_run[r], off, 64);
+      const float lo = __shfl_xor(l_run[r], off, 64);
+      const float m2 = fmaxf(m_run[r], mo);
+      l_run[r] = l_run[r] * __builtin_amdgcn_exp2f((m_run[r] - m2) * kLog2eC) +
+                 lo * __builtin_amdgcn_exp2f((mo - m2) * kLog2eC);
+      m_run[r] = m2;
+      tgt[r] = fmaxf(tgt[r], __shfl_xor(tgt[r], off, 64));
+    }
+  }
   // scratch (reuses c_lds): per half, per row: lse and tgt.
   float* half_lse = reinterpret_cast<float*>(c_lds);          // [2][64]
   float* half_tgt = half_lse + 2 * 64;                        // [2][64]
