@@ -206,7 +206,9 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
   // sinks[h] and no value row: seed the running max/sum with it and the
   // online softmax (and the LSE the backward reads) absorbs it for free.
   const float m_init = sinks ? sinks[h] : -1e30f;
-  const float l_init = sinks ? 1.f : 0.f;
+  // l is tracked per lane and merged at the epilogue: seed the sink's
+  // softmax column on ONE lane of each row group only
+  const float l_init = (sinks && (threadIdx.x & 15) == 0) ? 1.f : 0.f;
   float m_run[2][4], l_run[2][4];
 #pragma unroll
   for (int m = 0; m < 2; ++m)
@@ -358,6 +360,10 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
         }
         m_new[r] = fmaxf(m_new[r], mx);
       }
+      // l accumulates PER LANE (each lane's own 4 columns): the row-wide
+      // sum only matters at the epilogue, where one 16-lane merge replaces
+      // a 4-step shuffle cascade per row per KV tile (the fwd kernel's
+      // PMC showed a 10:1 VALU:MFMA instruction ratio).
       float l_add[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -367,10 +373,6 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
           const float p = __builtin_amdgcn_exp2f((p_val[nt][r] - m_new[r]) * kLog2e);
           p_val[nt][r] = p;
           acc += p;
-        }
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1) {
-          acc += __shfl_xor(acc, off, 64);
         }
         l_add[r] = acc;
       }
@@ -433,7 +435,15 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     }
   }
 
-  // ---- epilogue: O /= l; stage in LDS; coalesced 16B stores; LSE -----------
+  // ---- epilogue: merge per-lane l across the 16 column lanes (once),
+  // then O /= l; stage in LDS; coalesced 16B stores; LSE --------------------
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        l_run[m][r] += __shfl_xor(l_run[m][r], off, 64);
   {
     bf16_t* o_lds = k_lds;  // reuse: [64][D] staged twice (two 64-row halves)
 #pragma unroll
