@@ -119,13 +119,21 @@ __global__ void adamw_sr_bf16_multi_kernel(
 
   const int64_t tid = blockIdx.x * static_cast<int64_t>(BLOCK) + threadIdx.x;
   const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
-  for (int64_t slot = tid; slot < total_slots; slot += stride) {
+  // ONE binary search to place the thread's first slot; the grid-stride
+  // walk advances t monotonically after that (a 9-step dependent-load
+  // search per 4-element slot was ~2x the whole kernel's useful time at
+  // 1.6e9 parameters).
+  int t = 0;
+  {
     int lo = 0, hi = n_tensors - 1;
     while (lo < hi) {
       const int mid = (lo + hi + 1) >> 1;
-      if (prefix[mid] <= slot) lo = mid; else hi = mid - 1;
+      if (prefix[mid] <= min(tid, total_slots - 1)) lo = mid; else hi = mid - 1;
     }
-    const int t = lo;
+    t = lo;
+  }
+  for (int64_t slot = tid; slot < total_slots; slot += stride) {
+    while (t + 1 < n_tensors && prefix[t + 1] <= slot) ++t;
     const int64_t vi = slot - prefix[t];
     const int64_t n = count[t];
     const int64_t base = vi * 4;
@@ -255,7 +263,7 @@ void adamw_stochastic_bf16_multi_(
   auto meta = meta_cpu.to(params[0].device(), /*non_blocking=*/true);
   auto bc = bc_cpu.to(params[0].device(), /*non_blocking=*/true);
   constexpr int kBlock = 256;
-  const int grid = (int)std::min<int64_t>((slots + kBlock - 1) / kBlock, 4096);
+  const int grid = (int)std::min<int64_t>((slots + kBlock - 1) / kBlock, 16384);
   const size_t smem = (7 * (size_t)n + 1) * 8 + 2 * (size_t)n * 4;
   TORCH_CHECK(smem <= 64 * 1024, "adamw multi: too many tensors for LDS");
   auto stream = at::hip::getCurrentHIPStream();
