@@ -134,22 +134,29 @@ __global__ void causal_conv_silu_bwd_kernel(
 constexpr int kGdnC = 64;   // chunk rows
 constexpr int kGdnD = 64;   // Dk = Dv
 
+// DVT: this block's Dv slice (64 = whole head, 32 = half). The state
+// columns are independent, so gridDim.y splits Dv across blocks — K-side
+// work (KK^T, M, N, the decay scan) is duplicated per split, but B*H
+// workgroups only fill half the 256 CUs at the bench shape, so a 2-way
+// split still wins well below chip saturation.
+template <int DVT>
 struct GdnLds {
-  float state[kGdnD][kGdnD];      // S (k, v) fp32
-  float M[kGdnC][kGdnC + 1];      // solve matrix / N scratch (+1 pad)
-  float R[kGdnC][kGdnD];          // solve result fp32
-  float rhs[kGdnC][kGdnD];        // rhs / partials scratch
+  float state[kGdnD][DVT];        // S (k, v-slice) fp32
+  float M[kGdnC][kGdnC + 1];      // solve matrix (+1 pad)
+  float R[kGdnC][DVT];            // solve result fp32
+  float rhs[kGdnC][DVT];          // rhs / partials scratch
   bf16_t kc[kGdnC][kGdnD];        // K chunk
   bf16_t qc[kGdnC][kGdnD];        // Q chunk
-  bf16_t vc[kGdnC][kGdnD];        // V chunk
+  bf16_t vc[kGdnC][DVT];          // V chunk slice
   bf16_t ks[kGdnC][kGdnD];        // scaled K / scaled Q / scaled-K2 scratch
-  bf16_t sbT[kGdnD][kGdnD];       // state^T cast to bf16: sbT[v][k]
+  bf16_t sbT[DVT][kGdnD];         // state^T cast to bf16: sbT[v][k]
   bf16_t nb[kGdnC][kGdnC];        // N cast to bf16
-  bf16_t rbT[kGdnD][kGdnC];       // R^T cast to bf16: rbT[v][t]
+  bf16_t rbT[DVT][kGdnC];         // R^T cast to bf16: rbT[v][t]
   float gc[kGdnC];
   float beta[kGdnC];
 };
 
+template <int DVT>
 __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
     const bf16_t* __restrict__ q,   // (B, H, S, D)
     const bf16_t* __restrict__ k,
@@ -159,12 +166,14 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
     bf16_t* __restrict__ out,            // (B, H, S, D)
     float* __restrict__ final_state,     // (B, H, D, D) or nullptr
     int64_t BH, int64_t S) {
+  constexpr int NTV = DVT / 16;        // v tiles per wave quarter
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  GdnLds& L = *reinterpret_cast<GdnLds*>(smem_raw);
+  GdnLds<DVT>& L = *reinterpret_cast<GdnLds<DVT>*>(smem_raw);
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;   // 4 waves; wave w owns m-rows 16w..16w+15
   const int64_t bh = blockIdx.x;
+  const int v0 = blockIdx.y * DVT;     // this block's Dv slice
   if (bh >= BH) return;
 
   const bf16_t* qp = q + bh * S * kGdnD;
@@ -174,7 +183,7 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
   const float* gp = decay_log + bh * S;
   bf16_t* op = out + bh * S * kGdnD;
 
-  for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
+  for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
     (&L.state[0][0])[i] = 0.f;
   }
   __syncthreads();
@@ -195,16 +204,25 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
             *reinterpret_cast<const bf16x8*>(kp + off);
         *reinterpret_cast<bf16x8*>(&L.qc[row][col]) =
             *reinterpret_cast<const bf16x8*>(qp + off);
-        *reinterpret_cast<bf16x8*>(&L.vc[row][col]) =
-            *reinterpret_cast<const bf16x8*>(vp + off);
       } else {
         bf16x8 z = {};
         *reinterpret_cast<bf16x8*>(&L.kc[row][col]) = z;
         *reinterpret_cast<bf16x8*>(&L.qc[row][col]) = z;
+      }
+    }
+    for (int i = threadIdx.x; i < kGdnC * DVT / 8; i += 256) {
+      const int row = (i * 8) / DVT;
+      const int col = (i * 8) % DVT;
+      if (row < c_rows) {
+        *reinterpret_cast<bf16x8*>(&L.vc[row][col]) =
+            *reinterpret_cast<const bf16x8*>(
+                vp + (int64_t)(s0 + row) * kGdnD + v0 + col);
+      } else {
+        bf16x8 z = {};
         *reinterpret_cast<bf16x8*>(&L.vc[row][col]) = z;
       }
     }
-    for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
+    for (int i = threadIdx.x; i < DVT * kGdnD; i += 256) {
       const int vcol = i / kGdnD;
       const int dk = i % kGdnD;
       L.sbT[vcol][dk] = (bf16_t)L.state[dk][vcol];
@@ -265,23 +283,23 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
     }
     __syncthreads();
     {
-      f32x4 acc[4];
+      f32x4 acc[NTV];
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      for (int nt = 0; nt < NTV; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
       const int arow = wave * 16 + (lane & 15);
 #pragma unroll
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         const int d0 = ks2 * 32 + (lane >> 4) * 8;
         const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.ks[arow][d0]);
 #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
+        for (int nt = 0; nt < NTV; ++nt) {
           const int vcol = nt * 16 + (lane & 15);
           const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.sbT[vcol][d0]);
           acc[nt] = mfma16gdn(a, b, acc[nt]);
         }
       }
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NTV; ++nt) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int t = wave * 16 + (lane >> 4) * 4 + r;
@@ -299,18 +317,25 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
     // j-quarter of M[t,:]*R[:,col], a 2-step xor-shuffle combines the 4
     // partials, and the qj==0 lane writes R[t][col] (+ bf16 R^T copy).
     {
-      const int c = lane & 15;
-      const int qj = lane >> 4;
-      const int col = wave * 16 + c;
+      // wave owns DVT/4 columns; its 64 lanes split as (j-group, col):
+      // DVT=64: 16 cols x 4 j-quarters; DVT=32: 8 cols x 8 j-octants.
+      constexpr int CPW = DVT / 4;           // cols per wave
+      constexpr int JG = 64 / CPW;           // j groups
+      const int c = lane % CPW;
+      const int qj = lane / CPW;
+      const int col = wave * CPW + c;
+      const int jspan = kGdnC / JG;
       for (int t = 0; t < kGdnC; ++t) {
         float part = 0.f;
-        const int jend = min(t, (qj + 1) * 16);
-        for (int j = qj * 16; j < jend; ++j) {
+        const int jend = min(t, (qj + 1) * jspan);
+        for (int j = qj * jspan; j < jend; ++j) {
           part += L.M[t][j] * L.R[j][col];
         }
         // all 64 lanes active and convergent: xor-shuffles are safe
-        part += __shfl_xor(part, 16, 64);
-        part += __shfl_xor(part, 32, 64);
+#pragma unroll
+        for (int off = CPW; off < 64; off <<= 1) {
+          part += __shfl_xor(part, off, 64);
+        }
         const float r_t = L.rhs[t][col] - part;
         if (qj == 0) {
           L.R[t][col] = r_t;
@@ -367,9 +392,9 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
     }
     __syncthreads();
     {
-      f32x4 acc[4];
+      f32x4 acc[NTV];
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      for (int nt = 0; nt < NTV; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
       const int arow = wave * 16 + (lane & 15);
       // (e^{gc} Q) @ S
 #pragma unroll
@@ -377,7 +402,7 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
         const int d0 = ks2 * 32 + (lane >> 4) * 8;
         const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.ks[arow][d0]);
 #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
+        for (int nt = 0; nt < NTV; ++nt) {
           const int vcol = nt * 16 + (lane & 15);
           const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.sbT[vcol][d0]);
           acc[nt] = mfma16gdn(a, b, acc[nt]);
@@ -389,20 +414,20 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
         const int j0 = ks2 * 32 + (lane >> 4) * 8;
         const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.nb[arow][j0]);
 #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
+        for (int nt = 0; nt < NTV; ++nt) {
           const int vcol = nt * 16 + (lane & 15);
           const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.rbT[vcol][j0]);
           acc[nt] = mfma16gdn(a, b, acc[nt]);
         }
       }
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < NTV; ++nt) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int t = wave * 16 + (lane >> 4) * 4 + r;
           const int vcol = nt * 16 + (lane & 15);
           if (t < c_rows) {
-            op[(int64_t)(s0 + t) * kGdnD + vcol] = (bf16_t)acc[nt][r];
+            op[(int64_t)(s0 + t) * kGdnD + v0 + vcol] = (bf16_t)acc[nt][r];
           }
         }
       }
@@ -426,9 +451,9 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
       }
       __syncthreads();
       const float e_tot = __builtin_amdgcn_exp2f(g_tot * kLog2eG);
-      for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
-        const int dk = i / kGdnD;
-        const int dv = i % kGdnD;
+      for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
+        const int dk = i / DVT;
+        const int dv = i % DVT;
         float acc = L.state[dk][dv] * e_tot;
         for (int t = 0; t < c_rows; ++t) {
           acc += (float)L.ks[t][dk] * L.R[t][dv];
@@ -441,8 +466,10 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
 
   if (final_state != nullptr) {
     float* fs = final_state + bh * kGdnD * kGdnD;
-    for (int i = threadIdx.x; i < kGdnD * kGdnD; i += 256) {
-      fs[i] = (&L.state[0][0])[i];
+    for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
+      const int dk = i / DVT;
+      const int dv = i % DVT;
+      fs[dk * kGdnD + v0 + dv] = L.state[dk][dv];
     }
   }
 }
@@ -515,23 +542,42 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
                       torch::dtype(torch::kFloat32).device(q.device()));
     fs_ptr = fs.data_ptr<float>();
   }
-  const size_t smem = sizeof(d9d::GdnLds);
+  auto stream = at::hip::getCurrentHIPStream();
   static bool attr_set = false;
   if (!attr_set) {
     hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&d9d::gdn_chunk_fwd_kernel),
-        hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);
+        reinterpret_cast<const void*>(&d9d::gdn_chunk_fwd_kernel<64>),
+        hipFuncAttributeMaxDynamicSharedMemorySize,
+        (int)sizeof(d9d::GdnLds<64>));
+    hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&d9d::gdn_chunk_fwd_kernel<32>),
+        hipFuncAttributeMaxDynamicSharedMemorySize,
+        (int)sizeof(d9d::GdnLds<32>));
     attr_set = true;
   }
-  auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(d9d::gdn_chunk_fwd_kernel, dim3((unsigned)(B * H)),
-                     dim3(256), smem, stream,
-                     reinterpret_cast<const __bf16*>(qc.data_ptr()),
-                     reinterpret_cast<const __bf16*>(kc.data_ptr()),
-                     reinterpret_cast<const __bf16*>(vc.data_ptr()),
-                     bc.data_ptr<float>(), gc.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(out.data_ptr()),
-                     fs_ptr, B * H, S);
+  // Dv-split: below ~1.5 workgroups/CU the duplicated K-side work is
+  // cheaper than idle CUs.
+  if (B * H * 2 <= 384) {
+    hipLaunchKernelGGL(d9d::gdn_chunk_fwd_kernel<32>,
+                       dim3((unsigned)(B * H), 2), dim3(256),
+                       sizeof(d9d::GdnLds<32>), stream,
+                       reinterpret_cast<const __bf16*>(qc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(kc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(vc.data_ptr()),
+                       bc.data_ptr<float>(), gc.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(out.data_ptr()),
+                       fs_ptr, B * H, S);
+  } else {
+    hipLaunchKernelGGL(d9d::gdn_chunk_fwd_kernel<64>,
+                       dim3((unsigned)(B * H), 1), dim3(256),
+                       sizeof(d9d::GdnLds<64>), stream,
+                       reinterpret_cast<const __bf16*>(qc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(kc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(vc.data_ptr()),
+                       bc.data_ptr<float>(), gc.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(out.data_ptr()),
+                       fs_ptr, B * H, S);
+  }
   if (return_state) return {out, fs};
   return {out};
 }
