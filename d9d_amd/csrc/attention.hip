@@ -44,6 +44,15 @@ union ushort2_t {  // 4 bf16 lanes packed for one 8-byte LDS store
   ushort s[4];
 };
 
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_attn;
+#define bf16x4 bf16x4_attn
+
+D9D_DEVICE ushort bf16_bits(bf16_t v) {
+  union { bf16_t b; ushort u; } cv;
+  cv.b = v;
+  return cv.u;
+}
+
 // Transpose a swizzled row-major [64][D] LDS image into a [D][64] LDS image
 // (128-byte rows, ((d&7)<<4) swizzle) using 4B LDS reads + 8B LDS writes.
 template <int D, int BLOCK = 256>
@@ -558,9 +567,16 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
   bf16_t* tdo_lds = do_lds + kQBlk * D;              // [D][64]: dO^T
   bf16_t* tq_lds = tdo_lds + D * kQBlk;              // [D][64]: Q^T
   bf16_t* kt_lds = tq_lds + D * kQBlk;               // [D][kBwdKv]
-  bf16_t* x1_lds = kt_lds + D * kBwdKv;              // [kBwdKv][64+8]: P^T
-  bf16_t* x2_lds = x1_lds + kBwdKv * (kQBlk + 8);    // [kBwdKv][64+8]: dS^T
-  bf16_t* x3_lds = x2_lds + kBwdKv * (kQBlk + 8);    // [64][kBwdKv+8]: dS
+  // P and dS live in ONE q-major blocked image each: [4 q][16 kv] 128-B
+  // tr16 blocks at (q>>2, kv>>4), qb stride padded +32 B so the packed
+  // b64 stores of 4 kv values per (q, nt) are bank-spread. dV/dK read
+  // their kv-major A fragments via ds_read_b64_tr_b16 pairs; dQ reads its
+  // q-major fragments as plain 16-B rows of the SAME dS image — the old
+  // third (q-major) copy and 48 scalar stores per wave are gone.
+  constexpr int kXkvB = kBwdKv / 16;                 // kv blocks per q-row
+  constexpr int kXqStride = kXkvB * 128 + 32;        // bytes per q-block row
+  char* x1_lds = reinterpret_cast<char*>(kt_lds + D * kBwdKv);  // P image
+  char* x2_lds = x1_lds + (kQBlk / 4) * kXqStride;              // dS image
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;   // 8 waves x 16 kv rows = 128 kv/block
@@ -783,34 +799,51 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
       }
     }
 
-    // ---- stage P^T, dS^T (kv-major) and dS (q-major), ONE barrier ----------
+    // ---- stage P and dS into the blocked images, ONE barrier --------------
+    // value (q, kv) at qb = q>>2 row, block kv>>4, inner (q&3)*32 + (kv&15)*2;
+    // the 4 r-values (4 consecutive kv at fixed q) pack into one b64 store.
     {
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
+        const int q_l = nt * 16 + (lane & 15);
+        const int kv_l = wave * 16 + (lane >> 4) * 4;
+        const size_t byte = (size_t)(q_l >> 2) * kXqStride +
+                            (kv_l >> 4) * 128 + (q_l & 3) * 32 +
+                            (kv_l & 15) * 2;
+        ushort2_t p4, d4;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int kv_l = wave * 16 + (lane >> 4) * 4 + r;
-          const int q_l = nt * 16 + (lane & 15);
-          x1_lds[kv_l * (kQBlk + 8) + q_l] = (bf16_t)pt_val[nt][r];
-          x2_lds[kv_l * (kQBlk + 8) + q_l] = (bf16_t)dst_val[nt][r];
-          x3_lds[q_l * (kBwdKv + 8) + kv_l] = (bf16_t)dst_val[nt][r];
+          p4.s[r] = bf16_bits((bf16_t)pt_val[nt][r]);
+          d4.s[r] = bf16_bits((bf16_t)dst_val[nt][r]);
         }
+        *reinterpret_cast<uint64_t*>(x1_lds + byte) = p4.u;
+        *reinterpret_cast<uint64_t*>(x2_lds + byte) = d4.u;
       }
       __syncthreads();
     }
 
     // ---- dV += P^T @ dO; dK += dS^T @ Q: back-to-back, no barrier ----------
     {
-      const bf16_t* x1r = x1_lds + (wave * 16) * (kQBlk + 8);
-      const bf16_t* x2r = x2_lds + (wave * 16) * (kQBlk + 8);
+      // A fragment (m = kv row wave*16 + lane&15, k = 8 q at q_off) = two
+      // ds_read_b64_tr_b16 of the [4 q][16 kv] blocks (qb, wave)
+      auto xfrag = [&](const char* img, int q_off) -> bf16x8 {
+        const size_t b0 = (size_t)(q_off >> 2) * kXqStride + wave * 128 +
+                          (lane & 15) * 8;
+        bf16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (__attribute__((address_space(3))) bf16x4*)(img + b0));
+        bf16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (__attribute__((address_space(3))) bf16x4*)(img + b0 + kXqStride));
+        bf16x8 f;
+#pragma unroll
+        for (int t2 = 0; t2 < 4; ++t2) { f[t2] = lo4[t2]; f[4 + t2] = hi4[t2]; }
+        return f;
+      };
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         const int q_off = ks2 * 32 + (lane >> 4) * 8;
-        const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-            x1r + (lane & 15) * (kQBlk + 8) + q_off);
-        const bf16x8 da = *reinterpret_cast<const bf16x8*>(
-            x2r + (lane & 15) * (kQBlk + 8) + q_off);
+        const bf16x8 pa = xfrag(x1_lds, q_off);
+        const bf16x8 da = xfrag(x2_lds, q_off);
 #pragma unroll
         for (int nt = 0; nt < kNT; ++nt) {
           const int d = nt * 16 + (lane & 15);
@@ -838,13 +871,14 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
         f32x4 dq_acc[kNTH];
 #pragma unroll
         for (int nt = 0; nt < kNTH; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
-        const bf16_t* xr = x3_lds + (q_mt * 16) * (kBwdKv + 8);
+        const int q_row = q_mt * 16 + (lane & 15);
+        const size_t qbase = (size_t)(q_row >> 2) * kXqStride + (q_row & 3) * 32;
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks2 = 0; ks2 < kBwdKv / 32; ++ks2) {
           const int kv_off = ks2 * 32 + (lane >> 4) * 8;
           const bf16x8 da = *reinterpret_cast<const bf16x8*>(
-              xr + (lane & 15) * (kBwdKv + 8) + kv_off);
+              x2_lds + qbase + (kv_off >> 4) * 128 + (kv_off & 15) * 2);
 #pragma unroll
           for (int nt = 0; nt < kNTH; ++nt) {
             const int d = (nt0 + nt) * 16 + (lane & 15);
@@ -1108,11 +1142,11 @@ std::vector<torch::Tensor> flash_attn_bwd(
 
   const dim3 grid(varlen ? n_kvtiles : (Skv + d9d::kBwdKv - 1) / d9d::kBwdKv,
                   varlen ? Hq : B * Hq);
+  // 4 q/do/t-images + K^T + the two blocked P/dS images (padded q-block
+  // rows: (kBwdKv/16)*128 + 32 bytes per 4 q rows)
   const size_t smem =
-      (size_t)(4 * d9d::kQBlk * D_pad + D_pad * d9d::kBwdKv +
-               2 * d9d::kBwdKv * (d9d::kQBlk + 8) +
-               d9d::kQBlk * (d9d::kBwdKv + 8)) *
-      sizeof(__bf16);
+      (size_t)(4 * d9d::kQBlk * D_pad + D_pad * d9d::kBwdKv) * sizeof(__bf16)
+      + 2 * (size_t)(d9d::kQBlk / 4) * ((d9d::kBwdKv / 16) * 128 + 32);
 
 #define LAUNCH_BWD(DP)                                                        \
   hipLaunchKernelGGL((d9d::flash_bwd_kernel<DP>), grid, dim3(512), smem,      \
