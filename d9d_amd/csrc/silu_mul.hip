@@ -10,7 +10,7 @@
 namespace d9d {
 
 D9D_DEVICE float silu_f(float x) {
-  const float sig = 1.f / (1.f + __expf(-x));
+  const float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-x * 1.44269504f));
   return x * sig;
 }
 
@@ -60,7 +60,7 @@ __global__ void silu_mul_bwd_kernel(
       const float af = bf16_bits_to_f32(av.s[j]);
       const float bf = bf16_bits_to_f32(bv.s[j]);
       const float gf = bf16_bits_to_f32(gv.s[j]);
-      const float sig = 1.f / (1.f + __expf(-af));
+      const float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-af * 1.44269504f));
       const float silu = af * sig;
       dav.s[j] = f32_to_bf16_rne(gf * bf * sig * (1.f + af * (1.f - sig)));
       dbv.s[j] = f32_to_bf16_rne(gf * silu);
@@ -72,7 +72,7 @@ __global__ void silu_mul_bwd_kernel(
     const float af = bf16_bits_to_f32(a[i]);
     const float bf = bf16_bits_to_f32(b[i]);
     const float gf = bf16_bits_to_f32(g[i]);
-    const float sig = 1.f / (1.f + __expf(-af));
+    const float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-af * 1.44269504f));
     da[i] = f32_to_bf16_rne(gf * bf * sig * (1.f + af * (1.f - sig)));
     db[i] = f32_to_bf16_rne(gf * af * sig);
   }
@@ -92,8 +92,10 @@ __global__ void silu_mul_packed_fwd_kernel(
   const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
   const int iv = inter / 8;
   for (int64_t v = tid; v < n_vec; v += stride) {
-    const int64_t t = v / iv;
-    const int i = (int)(v % iv) * 8;
+    // u32 div: a 64-bit software divide per element was ~the same cost
+    // as the whole silu math chain
+    const uint32_t t = (uint32_t)v / (uint32_t)iv;
+    const int i = (int)((uint32_t)v - t * (uint32_t)iv) * 8;
     const ushort* xp = x + t * (int64_t)(2 * inter);
     Bf16x8 av, bv, ov;
     av.u = *reinterpret_cast<const ushort8v*>(xp + i);
@@ -118,8 +120,10 @@ __global__ void silu_mul_packed_bwd_kernel(
   const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
   const int iv = inter / 8;
   for (int64_t v = tid; v < n_vec; v += stride) {
-    const int64_t t = v / iv;
-    const int i = (int)(v % iv) * 8;
+    // u32 div: a 64-bit software divide per element was ~the same cost
+    // as the whole silu math chain
+    const uint32_t t = (uint32_t)v / (uint32_t)iv;
+    const int i = (int)((uint32_t)v - t * (uint32_t)iv) * 8;
     const ushort* xp = x + t * (int64_t)(2 * inter);
     ushort* dxp = dx + t * (int64_t)(2 * inter);
     Bf16x8 av, bv, gv, dav, dbv;
@@ -131,7 +135,7 @@ __global__ void silu_mul_packed_bwd_kernel(
       const float af = bf16_bits_to_f32(av.s[j]);
       const float bf = bf16_bits_to_f32(bv.s[j]);
       const float gf = bf16_bits_to_f32(gv.s[j]);
-      const float sig = 1.f / (1.f + __expf(-af));
+      const float sig = 1.f / (1.f + __builtin_amdgcn_exp2f(-af * 1.44269504f));
       dav.s[j] = f32_to_bf16_rne(gf * bf * sig * (1.f + af * (1.f - sig)));
       dbv.s[j] = f32_to_bf16_rne(gf * af * sig);
     }
@@ -194,6 +198,8 @@ torch::Tensor silu_mul_packed_fwd(torch::Tensor x) {
   if (out.numel() == 0) return out;
   constexpr int kBlock = 256;
   auto stream = at::hip::getCurrentHIPStream();
+  TORCH_CHECK(rows * (int64_t)(inter / 8) < (int64_t)1 << 31,
+              "silu_mul_packed: activation too large for u32 indexing");
   hipLaunchKernelGGL(
       (d9d::silu_mul_packed_fwd_kernel<kBlock>),
       dim3(silu_grid(rows * (inter / 8), kBlock)), dim3(kBlock), 0, stream,
@@ -211,6 +217,8 @@ torch::Tensor silu_mul_packed_bwd(torch::Tensor x, torch::Tensor g) {
   if (x.numel() == 0) return dx;
   constexpr int kBlock = 256;
   auto stream = at::hip::getCurrentHIPStream();
+  TORCH_CHECK(rows * (int64_t)(inter / 8) < (int64_t)1 << 31,
+              "silu_mul_packed: activation too large for u32 indexing");
   hipLaunchKernelGGL(
       (d9d::silu_mul_packed_bwd_kernel<kBlock>),
       dim3(silu_grid(rows * (inter / 8), kBlock)), dim3(kBlock), 0, stream,

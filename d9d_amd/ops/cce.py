@@ -140,8 +140,6 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
         )
         chunk = _ROW_CHUNK
         single = chunk >= T
-        if not single:
-            dc = torch.zeros(c.shape, dtype=torch.float32, device=c.device)
         for s in range(0, T, chunk):
             sl = slice(s, min(s + _ROW_CHUNK, T))
             e_chunk = e[sl]
@@ -185,7 +183,13 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
                     # reduce_e_grad, d9d/kernel/cce/cce.py:190-198).
                     dist.all_reduce(de, group=vp_group)
                 return de, dc, None, None, None, None, None, None
-            dc += torch.matmul(pb.t(), e_chunk.to(c.dtype)).float()
+            part = torch.matmul(pb.t(), e_chunk.to(c.dtype))
+            if dc is None:
+                # first chunk INITIALIZES the fp32 accumulator (a zeros
+                # prefill is a 4*V*H-byte fill per backward call)
+                dc = part.float()
+            else:
+                dc += part.float()
 
         if vp_group is not None:
             dist.all_reduce(de, group=vp_group)
