@@ -197,10 +197,21 @@ def _shard_linear(lin: nn.Linear, mesh: DeviceMesh, tp_dim: int, shard_dim: int)
     lin._parameters["weight"] = nn.Parameter(dt, requires_grad=w.requires_grad)
     if shard_dim == 0:
         lin.out_features = local.shape[0]
+        b = lin._parameters.get("bias")
+        if b is not None:
+            tp_rank = mesh.get_coordinate()[tp_dim]
+            local_b = b.data.chunk(tp_size, dim=0)[tp_rank].contiguous()
+            bp: list[Placement] = [Replicate()] * mesh.ndim
+            bp[tp_dim] = Shard(0)
+            lin._parameters["bias"] = nn.Parameter(
+                DTensor.from_local(local_b, mesh, tuple(bp), run_check=False),
+                requires_grad=b.requires_grad,
+            )
     else:
         lin.in_features = local.shape[1]
     if not getattr(lin, "_d9d_to_local_params", None):
-        lin.__class__ = _to_local_class(type(lin), ("weight",))
+        names = ("weight", "bias") if lin._parameters.get("bias") is not None else ("weight",)
+        lin.__class__ = _to_local_class(type(lin), names)
 
 
 def parallelize_tp_attention(
@@ -507,14 +518,102 @@ def parallelize_tp_lm_head(
     head.logits = tp_logits
 
 
+def _wrap_tp_boundary(module: nn.Module, group, sequence_parallel: bool) -> None:
+    """Entry copy-to-tp (SP: sequence all-gather), exit all-reduce (SP:
+    reduce-scatter) around a block whose internals produce partial sums."""
+    orig_forward = module.forward
+
+    def forward(hidden_states, *args, **kwargs):
+        if sequence_parallel:
+            hidden_states = _AllGatherSeq.apply(hidden_states, group, 1)
+        else:
+            hidden_states = _CopyToTP.apply(hidden_states, group)
+        out = orig_forward(hidden_states, *args, **kwargs)
+        if sequence_parallel:
+            return _ReduceScatterSeq.apply(out, group, 1)
+        return _ReduceFromTP.apply(out, group)
+
+    module.forward = forward
+
+
+def parallelize_tp_mla(
+    mla, mesh: DeviceMesh, tp_dim_name: str = "tp",
+    sequence_parallel: bool = False,
+) -> None:
+    """Head-parallel TP for MultiHeadLatentAttention: per-head projections
+    (q up / kv up) shard colwise, o_proj rowwise; the shared latent path
+    (kv_down, kv_norm, low-rank q down/norm) stays replicated inside the
+    region with autograd-level tp-sum on its weights."""
+    tp_dim = mesh.mesh_dim_names.index(tp_dim_name)
+    tp_size = mesh.shape[tp_dim]
+    if tp_size == 1:
+        return
+    group = mesh.get_group(tp_dim)
+    assert mla.num_heads % tp_size == 0, "MLA heads must divide tp"
+
+    q = mla.q_proj
+    if isinstance(q, nn.Linear):
+        _shard_linear(q, mesh, tp_dim, 0)
+    else:  # LowRankProjection: shard the up matrix, replicate down + norm
+        _shard_linear(q.up, mesh, tp_dim, 0)
+        _install_tp_sumgrad(q.down, ("weight",), group)
+        _install_tp_sumgrad(q.norm, ("weight",), group)
+    _shard_linear(mla.kv_up, mesh, tp_dim, 0)
+    _shard_linear(mla.o_proj, mesh, tp_dim, 1)
+    _install_tp_sumgrad(mla.kv_down, ("weight",), group)
+    _install_tp_sumgrad(mla.kv_norm, ("weight",), group)
+    mla.num_heads //= tp_size
+    _wrap_tp_boundary(mla, group, sequence_parallel)
+
+
+def parallelize_tp_gdn(
+    gdn, mesh: DeviceMesh, tp_dim_name: str = "tp",
+    sequence_parallel: bool = False,
+) -> None:
+    """Head-parallel TP for GatedDeltaNet: q/k/v/beta/decay/out-gate
+    projections and the causal-conv channels shard with the heads, o_proj
+    rowwise; the per-head-dim output RMSNorm weight is shared across heads
+    (replicated inside the region, tp-sum grads)."""
+    tp_dim = mesh.mesh_dim_names.index(tp_dim_name)
+    tp_size = mesh.shape[tp_dim]
+    if tp_size == 1:
+        return
+    group = mesh.get_group(tp_dim)
+    rank = mesh.get_coordinate()[tp_dim]
+    assert gdn.num_heads % tp_size == 0 and gdn.num_kv_heads % tp_size == 0
+
+    def shard_conv(conv):
+        w = conv._parameters["weight"]
+        local = w.data.chunk(tp_size, dim=0)[rank].contiguous()
+        placements: list[Placement] = [Replicate()] * mesh.ndim
+        placements[tp_dim] = Shard(0)
+        dt = DTensor.from_local(local, mesh, tuple(placements), run_check=False)
+        conv._parameters["weight"] = nn.Parameter(dt, requires_grad=w.requires_grad)
+        if not getattr(conv, "_d9d_to_local_params", None):
+            conv.__class__ = _to_local_class(type(conv), ("weight",))
+
+    for lin in (gdn.q_proj, gdn.k_proj, gdn.v_proj, gdn.beta_proj,
+                gdn.decay_gate.proj, gdn.out_gate):
+        _shard_linear(lin, mesh, tp_dim, 0)
+    for conv in (gdn.q_conv, gdn.k_conv, gdn.v_conv):
+        shard_conv(conv)
+    _shard_linear(gdn.o_proj, mesh, tp_dim, 1)
+    _install_tp_sumgrad(gdn.out_norm, ("weight",), group)
+    gdn.num_heads //= tp_size
+    gdn.num_kv_heads //= tp_size
+    _wrap_tp_boundary(gdn, group, sequence_parallel)
+
+
 def parallelize_tensor_parallel(
     module: nn.Module,
     mesh: DeviceMesh,
     tp_dim_name: str = "tp",
     sequence_parallel: bool = False,
 ) -> nn.Module:
-    """Apply TP (optionally SP) to every GQA attention, SwiGLU FFN, MoE
-    layer, split token embedding and split LM head in the module tree."""
+    """Apply TP (optionally SP) to every GQA/MLA/GDN attention, SwiGLU FFN,
+    MoE layer, split token embedding and split LM head in the module tree."""
+    from ..module.block.attention import MultiHeadLatentAttention
+    from ..module.block.attention.linear import GatedDeltaNet
     from ..module.block.embedding import SplitTokenEmbeddings
     from ..module.block.head import SplitLanguageModellingHead
     from ..module.block.moe import MoELayer
@@ -532,6 +631,10 @@ def parallelize_tensor_parallel(
     for sub in module.modules():
         if isinstance(sub, GroupedQueryAttention):
             parallelize_tp_attention(sub, mesh, tp_dim_name, sequence_parallel)
+        elif isinstance(sub, MultiHeadLatentAttention):
+            parallelize_tp_mla(sub, mesh, tp_dim_name, sequence_parallel)
+        elif isinstance(sub, GatedDeltaNet):
+            parallelize_tp_gdn(sub, mesh, tp_dim_name, sequence_parallel)
         elif isinstance(sub, MoELayer):
             parallelize_tp_moe(sub, mesh, tp_dim_name, sequence_parallel)
         elif isinstance(sub, SwiGLU):

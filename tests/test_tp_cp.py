@@ -432,3 +432,113 @@ def _cp2_whole_model_golden(rank, world_size):
 @pytest.mark.distributed
 def test_cp2_whole_model_golden():
     assert all(run_distributed(_cp2_whole_model_golden, world_size=2))
+
+
+# -- TP over the other attention families (MLA, GatedDeltaNet) ---------------
+
+
+def _tp2_mla_exact(rank, world_size):
+    from torch.distributed.device_mesh import init_device_mesh
+
+    from d9d_amd.module.block.attention import MultiHeadLatentAttention
+    from d9d_amd.module.block.positional import RotaryEmbeddingProvider
+    from d9d_amd.parallel.tensor import parallelize_tp_mla
+
+    def build():
+        torch.manual_seed(21)
+        m = MultiHeadLatentAttention(
+            64, 4, qk_nope_head_dim=16, qk_rope_head_dim=8, v_head_dim=16,
+            kv_lora_rank=32, q_lora_rank=24,
+        )
+        m.reset_parameters()
+        return m
+
+    ref = build()
+    mla = build()
+    mesh = init_device_mesh("cpu", (2,), mesh_dim_names=("tp",))
+    parallelize_tp_mla(mla, mesh)
+
+    prov = RotaryEmbeddingProvider(rope_dim=8)
+    pos = torch.arange(8).unsqueeze(0).expand(2, 8)
+    cos_sin = prov(pos)
+    torch.manual_seed(33)
+    x = torch.randn(2, 8, 64, requires_grad=True)
+    out = mla(x, cos_sin)
+    out.sum().backward()
+
+    x_ref = x.detach().clone().requires_grad_(True)
+    ref_out = ref(x_ref, cos_sin)
+    ref_out.sum().backward()
+
+    torch.testing.assert_close(out, ref_out, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(x.grad, x_ref.grad, rtol=1e-4, atol=1e-5)
+    # replicated-inside-region param: full grad everywhere
+    torch.testing.assert_close(
+        mla.kv_down._parameters["weight"].grad, ref.kv_down.weight.grad,
+        rtol=1e-4, atol=1e-5,
+    )
+    # head-sharded kv_up grad slice
+    from torch.distributed.tensor import DTensor
+
+    g = mla.kv_up._parameters["weight"].grad
+    assert isinstance(g, DTensor)
+    torch.testing.assert_close(
+        g.to_local(), ref.kv_up.weight.grad.chunk(2, 0)[rank], rtol=1e-4, atol=1e-5
+    )
+    return True
+
+
+@pytest.mark.distributed
+def test_tp2_mla_exact():
+    assert all(run_distributed(_tp2_mla_exact, world_size=2))
+
+
+def _tp2_gdn_exact(rank, world_size):
+    from torch.distributed.device_mesh import init_device_mesh
+
+    from d9d_amd.module.block.attention.linear import GatedDeltaNet
+    from d9d_amd.parallel.tensor import parallelize_tp_gdn
+
+    def build():
+        torch.manual_seed(23)
+        m = GatedDeltaNet(64, num_heads=4, num_kv_heads=2,
+                          head_k_dim=16, head_v_dim=16)
+        m.reset_parameters()
+        return m
+
+    ref = build()
+    gdn = build()
+    mesh = init_device_mesh("cpu", (2,), mesh_dim_names=("tp",))
+    parallelize_tp_gdn(gdn, mesh)
+
+    torch.manual_seed(35)
+    x = torch.randn(2, 32, 64, requires_grad=True)
+    out = gdn(x)
+    out.sum().backward()
+
+    x_ref = x.detach().clone().requires_grad_(True)
+    ref_out = ref(x_ref)
+    ref_out.sum().backward()
+
+    torch.testing.assert_close(out, ref_out, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(x.grad, x_ref.grad, rtol=1e-4, atol=1e-4)
+    # decay-gate bias is head-sharded (the one biased projection)
+    from torch.distributed.tensor import DTensor
+
+    b = gdn.decay_gate.proj._parameters["bias"]
+    assert isinstance(b, DTensor)
+    torch.testing.assert_close(
+        b.grad.to_local(), ref.decay_gate.proj.bias.grad.chunk(2, 0)[rank],
+        rtol=1e-4, atol=1e-4,
+    )
+    # shared out_norm weight: full grad everywhere (tp-summed)
+    torch.testing.assert_close(
+        gdn.out_norm._parameters["weight"].grad, ref.out_norm.weight.grad,
+        rtol=1e-4, atol=1e-4,
+    )
+    return True
+
+
+@pytest.mark.distributed
+def test_tp2_gdn_exact():
+    assert all(run_distributed(_tp2_gdn_exact, world_size=2))
