@@ -931,6 +931,336 @@ __global__ __launch_bounds__(512, 1) void flash_bwd_kernel(
   }
 }
 
+// flash bwd v2 — the occupancy-first geometry (opt-in: D9D_FLASH_BWD_V2=1).
+// MEASURED RESULT: 176 TF/s vs the default kernel's 326 at B8 S4096 H16/4
+// D128 causal — occupancy does NOT pay here. The default kernel is pinned
+// at 2 waves/SIMD (212 VGPR, 127 KB LDS) and this variant reaches 3
+// waves/SIMD x 3 blocks/CU, but the price of fitting 168 VGPRs / 51 KB is
+// 4x K/V fragment re-reads (streamed from L2 per n-tile), a 16-KB K-tile
+// re-stage per q-iteration, double the Q/dO traffic (kv-64 tiles instead
+// of kv-128) and atomic dQ — memory work grows faster than the extra
+// latency hiding recovers. Kept as a correct (15/15 parity) reference
+// point for the occupancy/traffic trade-off; see docs/perf_notes.md.
+// Original design notes:
+//   * kv tile 64, ONE 4-wave (256-thread) workgroup per tile, ~49 KB LDS
+//     and a <=168-VGPR budget -> 3 workgroups per CU, 3 waves per SIMD.
+//   * Q and dO live in ONE [4 q][16 d] 128-B-blocked image each: the
+//     S^T/dP^T B fragments are plain 16-B row reads, the dV/dK B
+//     fragments are ds_read_b64_tr_b16 pairs of the same blocks — the
+//     LDS->LDS transposed copies (32 KB + a pass per tile) are gone.
+//   * dQ's K^T fragments read straight from global K (the 16-KB tile is
+//     L2-resident; each fragment row is a 32-B coalesced lane group) —
+//     the 16-KB K^T image is gone.
+//   * P/dS stage once into q-major tr16 block images (as v1).
+template <int D>
+__global__ __launch_bounds__(256, 3) void flash_bwd_kernel_v2(
+    const bf16_t* __restrict__ q,
+    const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v,
+    const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse,    // (B,Hq,Sq)
+    const float* __restrict__ delta,  // (B,Hq,Sq)
+    float* __restrict__ dq,           // (B,Sq,Hq,D) fp32 accum
+    float* __restrict__ dk,           // (B,Skv,Hkv,D) fp32 accum
+    float* __restrict__ dv,           // (B,Skv,Hkv,D) fp32 accum
+    const int* __restrict__ cu_q,     // packed-seq bounds, or nullptr
+    const int* __restrict__ cu_k,
+    const int* __restrict__ kvtile_pref,  // prefix of ceil(len_k/64)
+    int nseq,
+    int B, int Sq, int Skv, int Hq, int Hkv,
+    float scale, int causal, int window_left, int q_offset) {
+  constexpr int kNT = D / 16;
+  constexpr int kKS = D / 32;
+  constexpr int kKv = 64;                   // kv rows per workgroup
+  constexpr int kDB = D / 16;               // d blocks per q-block row
+  constexpr int kQStride = kDB * 128 + 32;  // q/do image: bytes per 4-q row
+  constexpr int kXkvB = kKv / 16;           // x image: kv blocks per row
+  constexpr int kXStride = kXkvB * 128 + 32;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* q_img = smem;                                  // (kQBlk/4)*kQStride
+  char* do_img = q_img + (kQBlk / 4) * kQStride;
+  char* x1_img = do_img + (kQBlk / 4) * kQStride;      // P
+  char* x2_img = x1_img + (kQBlk / 4) * kXStride;      // dS
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;   // 4 waves x 16 kv rows = 64 kv/block
+
+  int b, h, kv_tile, q_lo, Sq_loc, kv_lo, Skv_loc, causal_off;
+  if (cu_q != nullptr) {
+    h = blockIdx.y;
+    int lo = 0, hi = nseq - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (kvtile_pref[mid] <= (int)blockIdx.x) lo = mid; else hi = mid - 1;
+    }
+    b = 0;
+    kv_tile = blockIdx.x - kvtile_pref[lo];
+    q_lo = cu_q[lo];
+    Sq_loc = cu_q[lo + 1] - q_lo;
+    kv_lo = cu_k[lo];
+    Skv_loc = cu_k[lo + 1] - kv_lo;
+    causal_off = Skv_loc - Sq_loc + q_offset;
+    if (Skv_loc <= 0 || Sq_loc <= 0 || kv_tile * kKv >= Skv_loc) return;
+  } else {
+    const int nwg = gridDim.x * gridDim.y;
+    const int lin = blockIdx.y * gridDim.x + blockIdx.x;
+    const int xcd = lin & 7;
+    const int qd = nwg >> 3, rd = nwg & 7;
+    const int wid = (xcd < rd ? xcd * (qd + 1) : rd * (qd + 1) + (xcd - rd) * qd)
+                    + (lin >> 3);
+    const int bh = wid / gridDim.x;
+    b = bh / Hq;
+    h = bh % Hq;
+    kv_tile = wid % gridDim.x;
+    q_lo = 0; Sq_loc = Sq; kv_lo = 0; Skv_loc = Skv;
+    causal_off = q_offset;
+  }
+  const int hkv = h / (Hq / Hkv);
+
+  const int64_t q_base = ((int64_t)b * Sq * Hq + h) * D;
+  const int64_t kv_base = ((int64_t)b * Skv * Hkv + hkv) * D;
+  const int64_t q_row_stride = (int64_t)Hq * D;
+  const int64_t kv_row_stride = (int64_t)Hkv * D;
+  const float* lse_row = lse + ((int64_t)b * Hq + h) * Sq + q_lo;
+  const float* delta_row = delta + ((int64_t)b * Hq + h) * Sq + q_lo;
+
+  const int kv0 = kv_tile * kKv;
+
+  // This wave's 16 kv rows. K/V A-fragments are NOT register-resident in
+  // v2 (the 64 VGPRs go to occupancy instead): the 16-KB K/V tiles are
+  // L2-hot, and each fragment re-load is one coalesced 16-B lane read.
+  const int kv_row_global =
+      kv_lo + min(kv0 + wave * 16 + (lane & 15), Skv_loc - 1);
+  const bf16_t* kp_row = k + kv_base + (int64_t)kv_row_global * kv_row_stride;
+  const bf16_t* vp_row = v + kv_base + (int64_t)kv_row_global * kv_row_stride;
+
+  f32x4 dk_acc[kNT], dv_acc[kNT];
+#pragma unroll
+  for (int nt = 0; nt < kNT; ++nt) {
+    dk_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+    dv_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  int q_start = 0;
+  if (causal) {
+    q_start = (max(kv0 - causal_off, 0) / kQBlk) * kQBlk;
+  }
+  if (q_start >= Sq_loc) return;
+
+  // Q/dO staging straight into the blocked images: thread chunk = one row's
+  // 8 consecutive d (16 B), landing contiguously inside a [4 q][16 d] block.
+  auto store_qdo = [&](int qt) {
+#pragma unroll
+    for (int it = 0; it < (kQBlk * D) / (256 * 8); ++it) {
+      const int idx = (threadIdx.x + it * 256) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int g_row = q_lo + min(qt + row, Sq_loc - 1);
+      const size_t byte = (size_t)(row >> 2) * kQStride + (col >> 4) * 128 +
+                          (row & 3) * 32 + (col & 15) * 2;
+      *reinterpret_cast<bf16x8*>(q_img + byte) =
+          *reinterpret_cast<const bf16x8*>(
+              q + q_base + (int64_t)g_row * q_row_stride + col);
+      *reinterpret_cast<bf16x8*>(do_img + byte) =
+          *reinterpret_cast<const bf16x8*>(
+              dout + q_base + (int64_t)g_row * q_row_stride + col);
+    }
+  };
+
+  // B fragment (k = 8 d at fixed q row): one 16-B row read of the image.
+  auto row_frag = [&](const char* img, int q_row, int d0) -> bf16x8 {
+    return *reinterpret_cast<const bf16x8*>(
+        img + (size_t)(q_row >> 2) * kQStride + (d0 >> 4) * 128 +
+        (q_row & 3) * 32 + (d0 & 15) * 2);
+  };
+  // B fragment (k = 8 q at fixed d col): tr16 pair of [4 q][16 d] blocks.
+  auto tr_frag = [&](const char* img, int q0, int dt) -> bf16x8 {
+    const size_t b0 = (size_t)(q0 >> 2) * kQStride + dt * 128 + (lane & 15) * 8;
+    bf16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4*)(img + b0));
+    bf16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4*)(img + b0 + kQStride));
+    bf16x8 f;
+#pragma unroll
+    for (int t2 = 0; t2 < 4; ++t2) { f[t2] = lo4[t2]; f[4 + t2] = hi4[t2]; }
+    return f;
+  };
+
+  store_qdo(q_start);
+  __syncthreads();
+
+  for (int qt = q_start; qt < Sq_loc; qt += kQBlk) {
+    // ---- S^T, dP^T, softmax — fused PER n-tile so only ONE st/dpt f32x4
+    // pair is live at a time (the [4]-array version costs 24 more VGPRs and
+    // pushes the kernel past the 168-reg / 3-wave budget). K/V fragment
+    // loads repeat per nt; the 16-KB tiles are L2-hot.
+    const bool any_mask =
+        (kv0 + kKv > Skv_loc) || (qt + kQBlk > Sq_loc) ||
+        (causal && kv0 + kKv - 1 > qt + causal_off) ||
+        (window_left >= 0 && kv0 < qt + kQBlk - 1 + causal_off - window_left);
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int q_row = nt * 16 + (lane & 15);
+      f32x4 st_nt = {0.f, 0.f, 0.f, 0.f};
+      f32x4 dpt_nt = {0.f, 0.f, 0.f, 0.f};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < kKS; ++ks) {
+        const int d0 = ks * 32 + (lane >> 4) * 8;
+        const bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp_row + d0);
+        st_nt = mfma16(kf, row_frag(q_img, q_row, d0), st_nt);
+        const bf16x8 vf = *reinterpret_cast<const bf16x8*>(vp_row + d0);
+        dpt_nt = mfma16(vf, row_frag(do_img, q_row, d0), dpt_nt);
+      }
+      __builtin_amdgcn_s_setprio(0);
+      const int q_g = qt + nt * 16 + (lane & 15);
+      const float l = (q_g < Sq_loc) ? lse_row[min(q_g, Sq_loc - 1)] : 1e30f;
+      const float dlt = (q_g < Sq_loc) ? delta_row[min(q_g, Sq_loc - 1)] : 0.f;
+      float pt4[4], dst4[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kv_g = kv0 + wave * 16 + (lane >> 4) * 4 + r;
+        float s = st_nt[r] * scale;
+        float p;
+        if (any_mask) {
+          bool masked = (kv_g >= Skv_loc) || (q_g >= Sq_loc);
+          if (causal) masked |= kv_g > q_g + causal_off;
+          if (window_left >= 0) masked |= kv_g < q_g + causal_off - window_left;
+          p = masked ? 0.f : __builtin_amdgcn_exp2f((s - l) * kLog2e);
+        } else {
+          p = __builtin_amdgcn_exp2f((s - l) * kLog2e);
+        }
+        pt4[r] = p;
+        dst4[r] = p * (dpt_nt[r] - dlt) * scale;
+      }
+      const int q_l = nt * 16 + (lane & 15);
+      const int kv_l = wave * 16 + (lane >> 4) * 4;
+      const size_t byte = (size_t)(q_l >> 2) * kXStride + (kv_l >> 4) * 128 +
+                          (q_l & 3) * 32 + (kv_l & 15) * 2;
+      ushort2_t p4, d4;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        p4.s[r] = bf16_bits((bf16_t)pt4[r]);
+        d4.s[r] = bf16_bits((bf16_t)dst4[r]);
+      }
+      *reinterpret_cast<uint64_t*>(x1_img + byte) = p4.u;
+      *reinterpret_cast<uint64_t*>(x2_img + byte) = d4.u;
+    }
+    __syncthreads();
+
+    // ---- dV += P^T @ dO; dK += dS^T @ Q ----------------------------------
+    {
+      auto xtr = [&](const char* img, int q0) -> bf16x8 {
+        const size_t b0 =
+            (size_t)(q0 >> 2) * kXStride + wave * 128 + (lane & 15) * 8;
+        bf16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (__attribute__((address_space(3))) bf16x4*)(img + b0));
+        bf16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (__attribute__((address_space(3))) bf16x4*)(img + b0 + kXStride));
+        bf16x8 f;
+#pragma unroll
+        for (int t2 = 0; t2 < 4; ++t2) { f[t2] = lo4[t2]; f[4 + t2] = hi4[t2]; }
+        return f;
+      };
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int q_off = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 pa = xtr(x1_img, q_off);
+        const bf16x8 da = xtr(x2_img, q_off);
+#pragma unroll
+        for (int nt = 0; nt < kNT; ++nt) {
+          const bf16x8 dob = tr_frag(do_img, q_off, nt);
+          const bf16x8 qb = tr_frag(q_img, q_off, nt);
+          dv_acc[nt] = mfma16(pa, dob, dv_acc[nt]);
+          dk_acc[nt] = mfma16(da, qb, dk_acc[nt]);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+
+    // ---- dQ partial = dS @ K ---------------------------------------------
+    // do_img is dead once the dV mfma above has consumed it: re-stage the
+    // 16-KB K tile THERE as the same [4 kv][16 d] blocked image (tr_frag
+    // pairs give the B[k=kv][n=d] fragments). A global-gather alternative
+    // (8 ushort loads + perms per fragment) measured 38 scratch spill ops:
+    // the hoisted gathers blew the 168-reg budget.
+    __syncthreads();
+    {
+#pragma unroll
+      for (int it = 0; it < (kKv * D) / (256 * 8); ++it) {
+        const int idx = (threadIdx.x + it * 256) * 8;
+        const int row = idx / D;
+        const int col = idx % D;
+        const int g_row = kv_lo + min(kv0 + row, Skv_loc - 1);
+        const size_t byte = (size_t)(row >> 2) * kQStride + (col >> 4) * 128 +
+                            (row & 3) * 32 + (col & 15) * 2;
+        *reinterpret_cast<bf16x8*>(do_img + byte) =
+            *reinterpret_cast<const bf16x8*>(
+                k + kv_base + (int64_t)g_row * kv_row_stride + col);
+      }
+    }
+    __syncthreads();
+    {
+      // 4 waves x (q m-tile = wave); d handled in full by each wave
+      const int q_mt = wave;
+      const int q_row = q_mt * 16 + (lane & 15);
+      const size_t qbase = (size_t)(q_row >> 2) * kXStride + (q_row & 3) * 32;
+      f32x4 dq_acc[kNT];
+#pragma unroll
+      for (int nt = 0; nt < kNT; ++nt) dq_acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int kv_off = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 da = *reinterpret_cast<const bf16x8*>(
+            x2_img + qbase + (kv_off >> 4) * 128 + (kv_off & 15) * 2);
+#pragma unroll
+        for (int nt = 0; nt < kNT; ++nt) {
+          dq_acc[nt] = mfma16(da, tr_frag(do_img, kv_off, nt), dq_acc[nt]);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q_g = qt + q_mt * 16 + (lane >> 4) * 4 + r;
+        if (q_g < Sq_loc) {
+#pragma unroll
+          for (int nt = 0; nt < kNT; ++nt) {
+            atomicAdd(dq + q_base +
+                          (int64_t)(q_lo + q_g) * q_row_stride + nt * 16 +
+                          (lane & 15),
+                      dq_acc[nt][r]);
+          }
+        }
+      }
+    }
+    // next tile's Q/dO overwrite the images: wait for all readers
+    __syncthreads();
+    if (qt + kQBlk < Sq_loc) {
+      store_qdo(qt + kQBlk);
+      __syncthreads();
+    }
+  }
+
+  // ---- flush dK, dV (atomicAdd: GQA groups and padded tiles overlap) -------
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kv_g = kv0 + wave * 16 + (lane >> 4) * 4 + r;
+    if (kv_g < Skv_loc) {
+#pragma unroll
+      for (int nt = 0; nt < kNT; ++nt) {
+        const int d = nt * 16 + (lane & 15);
+        atomicAdd(dk + kv_base + (int64_t)(kv_lo + kv_g) * kv_row_stride + d,
+                  dk_acc[nt][r]);
+        atomicAdd(dv + kv_base + (int64_t)(kv_lo + kv_g) * kv_row_stride + d,
+                  dv_acc[nt][r]);
+      }
+    }
+  }
+}
+
 // MFMA fragment-map self-check: C = A@B for one 16x32 @ 32x16 tile.
 __global__ void mfma_selfcheck_kernel(
     const bf16_t* __restrict__ a,  // (16, 32) row-major
@@ -1140,6 +1470,39 @@ std::vector<torch::Tensor> flash_attn_bwd(
                        delta.data_ptr<float>(), B, Sq, Hq, D_pad);
   }
 
+  // v2 geometry experiment (kv-64 tiles, 3 workgroups/CU): opt-in via
+  // D9D_FLASH_BWD_V2=1, non-varlen only.
+  static const bool use_bwd_v2 = []() {
+    const char* e = getenv("D9D_FLASH_BWD_V2");
+    return e != nullptr && e[0] == '1';
+  }();
+  const bool run_v2 = use_bwd_v2 && !varlen;
+  if (run_v2) {
+    const dim3 grid2((Skv + 63) / 64, B * Hq);
+    const size_t qs = (size_t)(d9d::kQBlk / 4) * ((D_pad / 16) * 128 + 32);
+    const size_t xs = (size_t)(d9d::kQBlk / 4) * ((64 / 16) * 128 + 32);
+    const size_t smem2 = 2 * qs + 2 * xs;
+#define LAUNCH_BWD2(DP)                                                       \
+  hipLaunchKernelGGL((d9d::flash_bwd_kernel_v2<DP>), grid2, dim3(256), smem2, \
+                     stream,                                                  \
+                     reinterpret_cast<const __bf16*>(qp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(kp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(vp.data_ptr()),          \
+                     reinterpret_cast<const __bf16*>(dop.data_ptr()),         \
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),          \
+                     dq32.data_ptr<float>(), dk32.data_ptr<float>(),          \
+                     dv32.data_ptr<float>(), nullptr, nullptr, nullptr, 0,    \
+                     B, Sq, Skv, Hq, Hkv, (float)softmax_scale,               \
+                     causal ? 1 : 0, (int)window_left, (int)q_offset)
+    switch (D_pad) {
+      case 32: LAUNCH_BWD2(32); break;
+      case 64: LAUNCH_BWD2(64); break;
+      case 96: LAUNCH_BWD2(96); break;
+      case 128: LAUNCH_BWD2(128); break;
+    }
+#undef LAUNCH_BWD2
+  }
+
   const dim3 grid(varlen ? n_kvtiles : (Skv + d9d::kBwdKv - 1) / d9d::kBwdKv,
                   varlen ? Hq : B * Hq);
   // 4 q/do/t-images + K^T + the two blocked P/dS images (padded q-block
@@ -1161,11 +1524,13 @@ std::vector<torch::Tensor> flash_attn_bwd(
                      nseq, B, Sq, Skv, Hq, Hkv,                               \
                      (float)softmax_scale, causal ? 1 : 0, (int)window_left,   \
                      (int)q_offset)
-  switch (D_pad) {
-    case 32: LAUNCH_BWD(32); break;
-    case 64: LAUNCH_BWD(64); break;
-    case 96: LAUNCH_BWD(96); break;
-    case 128: LAUNCH_BWD(128); break;
+  if (!run_v2) {
+    switch (D_pad) {
+      case 32: LAUNCH_BWD(32); break;
+      case 64: LAUNCH_BWD(64); break;
+      case 96: LAUNCH_BWD(96); break;
+      case 128: LAUNCH_BWD(128); break;
+    }
   }
 #undef LAUNCH_BWD
 
