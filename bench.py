@@ -293,6 +293,9 @@ def main():
     else:
         device = torch.device("cuda", local_rank)
         torch.cuda.set_device(device)
+        from d9d_amd.ops.tunable import load_tuned_gemm_table
+
+        load_tuned_gemm_table()  # pre-tuned hipBLASLt algo table (+2.7%)
 
     if distributed:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")

@@ -115,6 +115,10 @@ class TrainingConfigurator:
         return d["dp"]
 
     def _build(self, ctx: DistributedContext, bus: EventBus, timeout) -> "Trainer":
+        from ..ops.tunable import load_tuned_gemm_table
+
+        load_tuned_gemm_table()  # pre-tuned hipBLASLt algo table for gfx950
+
         cfg = self.config
         dp = self._dp_degree(ctx)
         pp = self.mesh.pipeline_parallel
