@@ -91,9 +91,10 @@ __global__ void adamw_sr_bf16_kernel(
 
 
 // Multi-tensor fused AdamW: one launch for the whole parameter list (the
-// per-tensor variant costs ~360 launches per optimizer step). Slots of 4
-// elements; each slot binary-searches its tensor in the prefix table.
-// Per-slot RNG matches the single-tensor kernel (seed[t] ^ local_slot).
+// per-tensor variant costs ~360 launches per optimizer step). Slots of 8
+// elements, software-pipelined one slot ahead; each thread binary-searches
+// its first tensor in the prefix table once. Per-element SR noise is
+// bitwise identical to the single-tensor kernel (4-vector salt = elem/4).
 template <int BLOCK>
 __global__ void adamw_sr_bf16_multi_kernel(
     const int64_t* __restrict__ meta_g,  // prefix(N+1) | count(N) | p|g|m|v ptrs (N each) | seed(N)
@@ -121,8 +122,8 @@ __global__ void adamw_sr_bf16_multi_kernel(
   const int64_t stride = static_cast<int64_t>(gridDim.x) * BLOCK;
   // ONE binary search to place the thread's first slot; the grid-stride
   // walk advances t monotonically after that (a 9-step dependent-load
-  // search per 4-element slot was ~2x the whole kernel's useful time at
-  // 1.6e9 parameters).
+  // search per slot was ~2x the whole kernel's useful time at 1.6e9
+  // parameters).
   int t = 0;
   {
     int lo = 0, hi = n_tensors - 1;
@@ -132,41 +133,110 @@ __global__ void adamw_sr_bf16_multi_kernel(
     }
     t = lo;
   }
-  for (int64_t slot = tid; slot < total_slots; slot += stride) {
-    while (t + 1 < n_tensors && prefix[t + 1] <= slot) ++t;
-    const int64_t vi = slot - prefix[t];
-    const int64_t n = count[t];
-    const int64_t base = vi * 4;
-    ushort* p = reinterpret_cast<ushort*>(p_ptrs[t]);
-    const ushort* g = reinterpret_cast<const ushort*>(g_ptrs[t]);
-    float* m = reinterpret_cast<float*>(m_ptrs[t]);
-    float* v = reinterpret_cast<float*>(v_ptrs[t]);
-    const float inv_bc1 = 1.f / bc[t * 2];
-    const float inv_sqrt_bc2 = rsqrtf(bc[t * 2 + 1]);
-    const uint64_t r = splitmix64((uint64_t)seeds[t] ^ (uint64_t)vi);
 
-    if (base + 4 <= n) {
-      ushort4v pv = *reinterpret_cast<const ushort4v*>(p + base);
-      const ushort4v gv = *reinterpret_cast<const ushort4v*>(g + base);
-      float4v mv = *reinterpret_cast<const float4v*>(m + base);
-      float4v vv = *reinterpret_cast<const float4v*>(v + base);
+  // 8-element slots, software-pipelined one slot ahead: vmcnt completion is
+  // IN-ORDER on CDNA, so a store-then-load iteration shape makes every
+  // iteration's load wait drain the previous stores too. Issuing the next
+  // slot's loads BEFORE the current slot's stores breaks that chain (the
+  // 4-element store-then-load version measured ~1.9 TB/s effective).
+  struct SlotData {
+    ushort4v pv0, pv1, gv0, gv1;
+    float4v mv0, mv1, vv0, vv1;
+    int64_t base, n;
+    int t;
+    bool full;
+  };
+  auto load_slot = [&](int64_t slot, int& tt, SlotData& s) {
+    while (tt + 1 < n_tensors && prefix[tt + 1] <= slot) ++tt;
+    const int64_t vi = slot - prefix[tt];
+    s.t = tt;
+    s.n = count[tt];
+    s.base = vi * 8;
+    s.full = s.base + 8 <= s.n;
+    if (s.full) {
+      const ushort* p = reinterpret_cast<const ushort*>(p_ptrs[tt]);
+      const ushort* g = reinterpret_cast<const ushort*>(g_ptrs[tt]);
+      const float* m = reinterpret_cast<const float*>(m_ptrs[tt]);
+      const float* v = reinterpret_cast<const float*>(v_ptrs[tt]);
+      s.pv0 = *reinterpret_cast<const ushort4v*>(p + s.base);
+      s.pv1 = *reinterpret_cast<const ushort4v*>(p + s.base + 4);
+      s.gv0 = *reinterpret_cast<const ushort4v*>(g + s.base);
+      s.gv1 = *reinterpret_cast<const ushort4v*>(g + s.base + 4);
+      s.mv0 = *reinterpret_cast<const float4v*>(m + s.base);
+      s.mv1 = *reinterpret_cast<const float4v*>(m + s.base + 4);
+      s.vv0 = *reinterpret_cast<const float4v*>(v + s.base);
+      s.vv1 = *reinterpret_cast<const float4v*>(v + s.base + 4);
+    }
+  };
+  auto compute_store = [&](SlotData& s) {
+    const float inv_bc1 = 1.f / bc[s.t * 2];
+    const float inv_sqrt_bc2 = rsqrtf(bc[s.t * 2 + 1]);
+    if (s.full) {
+      const int64_t vi = s.base >> 3;
+      const uint64_t r0 = splitmix64((uint64_t)seeds[s.t] ^ (uint64_t)(vi << 1));
+      const uint64_t r1 =
+          splitmix64((uint64_t)seeds[s.t] ^ (uint64_t)((vi << 1) | 1));
+      ushort4v* pvs[2] = {&s.pv0, &s.pv1};
+      const ushort4v* gvs[2] = {&s.gv0, &s.gv1};
+      float4v* mvs[2] = {&s.mv0, &s.mv1};
+      float4v* vvs[2] = {&s.vv0, &s.vv1};
+      const uint64_t rs[2] = {r0, r1};
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        float pf = bf16_bits_to_f32(pv[j]) * decay_mul;
-        const float gf = bf16_bits_to_f32(gv[j]);
-        float mf = beta1 * mv[j] + (1.f - beta1) * gf;
-        float vf = beta2 * vv[j] + (1.f - beta2) * gf * gf;
-        pf -= lr * (mf * inv_bc1) / (sqrtf(vf) * inv_sqrt_bc2 + eps);
-        mv[j] = mf;
-        vv[j] = vf;
-        pv[j] = f32_to_bf16_stochastic(pf, static_cast<uint32_t>(r >> (16 * j)));
+      for (int h = 0; h < 2; ++h) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float pf = bf16_bits_to_f32((*pvs[h])[j]) * decay_mul;
+          const float gf = bf16_bits_to_f32((*gvs[h])[j]);
+          float mf = beta1 * (*mvs[h])[j] + (1.f - beta1) * gf;
+          float vf = beta2 * (*vvs[h])[j] + (1.f - beta2) * gf * gf;
+          pf -= lr * (mf * inv_bc1) / (sqrtf(vf) * inv_sqrt_bc2 + eps);
+          (*mvs[h])[j] = mf;
+          (*vvs[h])[j] = vf;
+          (*pvs[h])[j] =
+              f32_to_bf16_stochastic(pf, static_cast<uint32_t>(rs[h] >> (16 * j)));
+        }
       }
-      *reinterpret_cast<ushort4v*>(p + base) = pv;
-      *reinterpret_cast<float4v*>(m + base) = mv;
-      *reinterpret_cast<float4v*>(v + base) = vv;
+      ushort* p = reinterpret_cast<ushort*>(p_ptrs[s.t]);
+      float* m = reinterpret_cast<float*>(m_ptrs[s.t]);
+      float* v = reinterpret_cast<float*>(v_ptrs[s.t]);
+      *reinterpret_cast<ushort4v*>(p + s.base) = s.pv0;
+      *reinterpret_cast<ushort4v*>(p + s.base + 4) = s.pv1;
+      *reinterpret_cast<float4v*>(m + s.base) = s.mv0;
+      *reinterpret_cast<float4v*>(m + s.base + 4) = s.mv1;
+      *reinterpret_cast<float4v*>(v + s.base) = s.vv0;
+      *reinterpret_cast<float4v*>(v + s.base + 4) = s.vv1;
     } else {
-      // tail slot: per-element RNG exactly as the single-tensor kernel
-      for (int64_t i = base; i < n; ++i) {
+      // tail slot, bitwise-matching the single-tensor kernel: any full
+      // 4-group left in the window uses the 4-vector RNG (salt = i/4),
+      // the rest the per-element RNG.
+      ushort* p = reinterpret_cast<ushort*>(p_ptrs[s.t]);
+      const ushort* g = reinterpret_cast<const ushort*>(g_ptrs[s.t]);
+      float* m = reinterpret_cast<float*>(m_ptrs[s.t]);
+      float* v = reinterpret_cast<float*>(v_ptrs[s.t]);
+      int64_t i4 = s.base;
+      for (; i4 + 4 <= s.n; i4 += 4) {
+        ushort4v pv = *reinterpret_cast<const ushort4v*>(p + i4);
+        const ushort4v gv = *reinterpret_cast<const ushort4v*>(g + i4);
+        float4v mv = *reinterpret_cast<const float4v*>(m + i4);
+        float4v vv = *reinterpret_cast<const float4v*>(v + i4);
+        const uint64_t r =
+            splitmix64((uint64_t)seeds[s.t] ^ (uint64_t)(i4 >> 2));
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float pf = bf16_bits_to_f32(pv[j]) * decay_mul;
+          const float gf = bf16_bits_to_f32(gv[j]);
+          float mf = beta1 * mv[j] + (1.f - beta1) * gf;
+          float vf = beta2 * vv[j] + (1.f - beta2) * gf * gf;
+          pf -= lr * (mf * inv_bc1) / (sqrtf(vf) * inv_sqrt_bc2 + eps);
+          mv[j] = mf;
+          vv[j] = vf;
+          pv[j] = f32_to_bf16_stochastic(pf, static_cast<uint32_t>(r >> (16 * j)));
+        }
+        *reinterpret_cast<ushort4v*>(p + i4) = pv;
+        *reinterpret_cast<float4v*>(m + i4) = mv;
+        *reinterpret_cast<float4v*>(v + i4) = vv;
+      }
+      for (int64_t i = i4; i < s.n; ++i) {
         float pf = bf16_bits_to_f32(p[i]) * decay_mul;
         const float gf = bf16_bits_to_f32(g[i]);
         float mf = beta1 * m[i] + (1.f - beta1) * gf;
@@ -175,10 +245,27 @@ __global__ void adamw_sr_bf16_multi_kernel(
         m[i] = mf;
         v[i] = vf;
         const uint64_t rt = splitmix64(
-            (uint64_t)seeds[t] ^ (0x8000000000000000ull | (uint64_t)i));
+            (uint64_t)seeds[s.t] ^ (0x8000000000000000ull | (uint64_t)i));
         p[i] = f32_to_bf16_stochastic(pf, static_cast<uint32_t>(rt));
       }
     }
+  };
+
+  int64_t slot = tid;
+  if (slot >= total_slots) return;
+  SlotData cur;
+  load_slot(slot, t, cur);
+  while (true) {
+    const int64_t nslot = slot + stride;
+    if (nslot >= total_slots) {
+      compute_store(cur);
+      break;
+    }
+    SlotData nxt;
+    load_slot(nslot, t, nxt);  // next loads issue BEFORE cur's stores
+    compute_store(cur);
+    cur = nxt;
+    slot = nslot;
   }
 }
 
@@ -257,7 +344,7 @@ void adamw_stochastic_bf16_multi_(
     mp[6 * n + 1 + i] = seeds[i];
     bp[i * 2] = 1.f - powf((float)beta1, (float)steps[i]);
     bp[i * 2 + 1] = 1.f - powf((float)beta2, (float)steps[i]);
-    slots += (cnt + 3) / 4;
+    slots += (cnt + 7) / 8;
   }
   mp[n] = slots;
   auto meta = meta_cpu.to(params[0].device(), /*non_blocking=*/true);

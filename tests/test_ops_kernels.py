@@ -399,7 +399,8 @@ def test_adamw_multi_matches_single():
 
     torch.manual_seed(7)
     ext = get_ext()
-    shapes = [(1000,), (257,), (64, 64), (3,)]
+    # 1004 % 8 == 4: exercises the multi kernel's partial-slot 4-vector RNG
+    shapes = [(1000,), (257,), (64, 64), (3,), (1004,)]
     ps = [torch.randn(s, dtype=torch.bfloat16, device="cuda") for s in shapes]
     gs = [torch.randn(s, dtype=torch.bfloat16, device="cuda") for s in shapes]
     ms = [torch.randn(s, dtype=torch.float32, device="cuda").abs() for s in shapes]
@@ -407,16 +408,16 @@ def test_adamw_multi_matches_single():
     ps2 = [p.clone() for p in ps]
     ms2 = [m.clone() for m in ms]
     vs2 = [v.clone() for v in vs]
-    seeds = [11, 22, 33, 44]
-    for i in range(4):
+    seeds = [11, 22, 33, 44, 55]
+    for i in range(len(shapes)):
         ext.adamw_stochastic_bf16_(
             ps[i].view(-1), gs[i].contiguous().view(-1), ms[i].view(-1),
             vs[i].view(-1), 1e-3, 0.9, 0.95, 1e-8, 0.01, 3, seeds[i])
     ext.adamw_stochastic_bf16_multi_(
         [p.view(-1) for p in ps2], [g.contiguous().view(-1) for g in gs],
         [m.view(-1) for m in ms2], [v.view(-1) for v in vs2],
-        1e-3, 0.9, 0.95, 1e-8, 0.01, [3, 3, 3, 3], seeds)
-    for i in range(4):
+        1e-3, 0.9, 0.95, 1e-8, 0.01, [3] * len(shapes), seeds)
+    for i in range(len(shapes)):
         assert torch.equal(ps[i], ps2[i]), i  # bitwise: same per-slot RNG
         torch.testing.assert_close(ms[i], ms2[i])
         torch.testing.assert_close(vs[i], vs2[i])
