@@ -1457,11 +1457,17 @@ std::vector<torch::Tensor> flash_attn_bwd(
 
   auto f32opt = q.options().dtype(torch::kFloat32);
   auto delta = torch::empty({B, Hq, Sq}, f32opt);
-  auto dq32 = torch::zeros({B, Sq, Hq, D_pad}, f32opt);
-  auto dk32 = torch::zeros({B, Skv, Hkv, D_pad}, f32opt);
-  auto dv32 = torch::zeros({B, Skv, Hkv, D_pad}, f32opt);
+  // atomicAdd accumulators: cleared via the copy/DMA engine (hipMemsetAsync)
+  // instead of torch::zeros — the fill kernels measured ~1.7 TB/s in-step
+  // and 400 MB/call of clears showed up at ~1.3% of the whole bench step.
+  auto dq32 = torch::empty({B, Sq, Hq, D_pad}, f32opt);
+  auto dk32 = torch::empty({B, Skv, Hkv, D_pad}, f32opt);
+  auto dv32 = torch::empty({B, Skv, Hkv, D_pad}, f32opt);
 
   auto stream = at::hip::getCurrentHIPStream();
+  hipMemsetAsync(dq32.data_ptr(), 0, dq32.numel() * sizeof(float), stream);
+  hipMemsetAsync(dk32.data_ptr(), 0, dk32.numel() * sizeof(float), stream);
+  hipMemsetAsync(dv32.data_ptr(), 0, dv32.numel() * sizeof(float), stream);
   {
     const int64_t rows = (int64_t)B * Sq * Hq;
     hipLaunchKernelGGL(d9d::attn_delta_kernel, dim3(rows), dim3(64), 0, stream,
