@@ -46,3 +46,25 @@ for _ in range(10):
     ts.append(time.perf_counter() - t0)
 t = statistics.median(ts)
 print(f"causal_conv_silu_fwd: {by/t/1e12:.2f} TB/s")
+
+# backward: derived closed-form vs autograd through the torch WY graph
+from d9d_amd.module.block.attention.linear.gated_deltanet import _chunk_gdn_backward
+
+qf, kf, vf = q.float(), k.float(), v.float()
+do = torch.randn(B, H, S, D, device="cuda")
+
+def bwd_derived():
+    return _chunk_gdn_backward(qf, kf, vf, beta, g, do)
+
+def bwd_autograd():
+    with torch.enable_grad():
+        qs = qf.detach().requires_grad_(True)
+        ks = kf.detach().requires_grad_(True)
+        vs = vf.detach().requires_grad_(True)
+        bs = beta.detach().requires_grad_(True)
+        gs = g.detach().requires_grad_(True)
+        ref = _chunk_gated_delta_rule_torch(qs, ks, vs, bs, gs)
+        return torch.autograd.grad(ref, (qs, ks, vs, bs, gs), do)
+
+bench(bwd_derived, "gdn bwd (derived)", 3 * fl)
+bench(bwd_autograd, "gdn bwd (autograd-through-graph)", 3 * fl)

@@ -110,42 +110,60 @@ def chunk_gated_delta_rule(
     GEMM lands on MFMA; backward recomputes through the torch WY graph
     (a native bwd kernel is the follow-up). Other shapes / CPU: the
     chunked-parallel WY torch path below."""
-    if (
+    use_kernel = (
         _use_gdn_kernels(q)
         and q.shape[-1] == 64
         and v.shape[-1] == 64
         and chunk_size == 64
+    )
+    if torch.is_grad_enabled() and any(
+        t.requires_grad for t in (q, k, v, beta, decay_log)
     ):
-        return _GdnChunkFunction.apply(q, k, v, beta, decay_log)
+        return _GdnChunkFunction.apply(q, k, v, beta, decay_log, chunk_size, use_kernel)
+    if use_kernel:
+        return _gdn_kernel_fwd(q, k, v, beta, decay_log)
     return _chunk_gated_delta_rule_torch(q, k, v, beta, decay_log, chunk_size)
 
 
+def _gdn_kernel_fwd(q, k, v, beta, decay_log):
+    out = get_ext().gdn_chunk_fwd(
+        q.to(torch.bfloat16).contiguous(),
+        k.to(torch.bfloat16).contiguous(),
+        v.to(torch.bfloat16).contiguous(),
+        beta.contiguous(),
+        decay_log.contiguous(),
+        False,
+    )[0]
+    return out.to(v.dtype)
+
+
 class _GdnChunkFunction(torch.autograd.Function):
+    """Forward on the CDNA4 chunk kernel (or the torch WY path); backward
+    via the hand-derived `_chunk_gdn_backward` — one light forward state
+    scan + one reverse dState scan, everything else batched over chunks
+    (the previous backward re-built and backpropped the whole sequential
+    autograd graph)."""
+
     @staticmethod
-    def forward(ctx, q, k, v, beta, decay_log):
-        out = get_ext().gdn_chunk_fwd(
-            q.to(torch.bfloat16).contiguous(),
-            k.to(torch.bfloat16).contiguous(),
-            v.to(torch.bfloat16).contiguous(),
-            beta.contiguous(),
-            decay_log.contiguous(),
-            False,
-        )[0]
+    def forward(ctx, q, k, v, beta, decay_log, chunk_size, use_kernel):
+        if use_kernel:
+            out = _gdn_kernel_fwd(q, k, v, beta, decay_log)
+        else:
+            with torch.no_grad():
+                out = _chunk_gated_delta_rule_torch(
+                    q, k, v, beta, decay_log, chunk_size
+                )
         ctx.save_for_backward(q, k, v, beta, decay_log)
-        return out.to(v.dtype)
+        ctx.chunk_size = chunk_size
+        return out
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, beta, decay_log = ctx.saved_tensors
-        with torch.enable_grad():
-            qs = q.detach().requires_grad_(True)
-            ks = k.detach().requires_grad_(True)
-            vs = v.detach().requires_grad_(True)
-            bs = beta.detach().requires_grad_(True)
-            gs = decay_log.detach().requires_grad_(True)
-            ref = _chunk_gated_delta_rule_torch(qs, ks, vs, bs, gs)
-            grads = torch.autograd.grad(ref, (qs, ks, vs, bs, gs), dout)
-        return grads
+        dq, dk, dv, dbeta, dg = _chunk_gdn_backward(
+            q, k, v, beta, decay_log, dout, ctx.chunk_size
+        )
+        return dq, dk, dv, dbeta, dg, None, None
 
 
 def _chunk_gated_delta_rule_torch(
@@ -214,6 +232,180 @@ def _chunk_gated_delta_rule_torch(
             k_scaled.transpose(-1, -2), R
         )
     return torch.cat(outs, dim=2).to(v.dtype)
+
+
+def _chunk_gdn_backward(q, k, v, beta, decay_log, dout, chunk_size=64):
+    """Explicit backward of `_chunk_gated_delta_rule_torch` (same WY chunk
+    formulation), derived by hand so the backward does NOT rebuild and
+    backprop the sequential autograd graph. Only the inter-chunk dState
+    scan is sequential; every per-chunk GEMM before/after the scan runs
+    batched over (B, H, n_chunks).
+
+    Forward per chunk (c x Dk matrices Q,K; c x Dv V; vectors b, gc):
+        E = exp(gc);  ratio_tj = E_t / E_j  (t >= j)
+        A = I + tril_{-1}(b_row * K K^T * ratio)
+        P = V - (E*K) S0;   rhs = b_row * P;   R = A^{-1} rhs
+        N = tril_0(Q K^T * ratio)
+        O = (E*Q) S0 + N R
+        S1 = Eend S0 + (w*K)^T R,  w = Eend / E
+    Reverse per chunk (dS1 from the next chunk, zero at the end):
+        dR   = (w*K) dS1 + N^T dO
+        drhs = A^{-T} dR;          dT = -(b_row * drhs)
+        dS0  = Eend dS1 + (E*K)^T dT + (E*Q)^T dO
+    and all remaining leaf grads are local chunk expressions of
+    (dO, dS1, dR, drhs) — see the body. Returns (dq, dk, dv, dbeta, dg)."""
+    B, H, S, Dk = q.shape
+    Dv = v.shape[-1]
+    C = min(chunk_size, S)
+    dtype_v = v.dtype
+    q32, k32, v32 = q.float(), k.float(), v.float()
+    b32, g32 = beta.float(), decay_log.float()
+    do32 = dout.float()
+
+    # pad S to a chunk multiple (padded tail: beta = 0, g = 0, q/k/v/do = 0
+    # => the padded rows write nothing into the state and produce no grads)
+    pad = (C - S % C) % C
+    if pad:
+        zq = q32.new_zeros(B, H, pad, Dk)
+        q32 = torch.cat([q32, zq], 2)
+        k32 = torch.cat([k32, zq], 2)
+        v32 = torch.cat([v32, v32.new_zeros(B, H, pad, Dv)], 2)
+        do32 = torch.cat([do32, do32.new_zeros(B, H, pad, Dv)], 2)
+        b32 = torch.cat([b32, b32.new_zeros(B, H, pad)], 2)
+        g32 = torch.cat([g32, g32.new_zeros(B, H, pad)], 2)
+    Sp = S + pad
+    nc = Sp // C
+
+    # (B, H, nc, C, D) chunked views
+    Qc = q32.view(B, H, nc, C, Dk)
+    Kc = k32.view(B, H, nc, C, Dk)
+    Vc = v32.view(B, H, nc, C, Dv)
+    dOc = do32.view(B, H, nc, C, Dv)
+    bc = b32.view(B, H, nc, C)
+    gc = g32.view(B, H, nc, C).cumsum(dim=-1)
+    E = gc.exp()                                  # (B,H,nc,C)
+    g_tot = gc[..., -1:]                          # (B,H,nc,1)
+    Eend = g_tot.exp()
+    w = (g_tot - gc).exp()                        # Eend / E
+
+    ratio = torch.exp(gc.unsqueeze(-1) - gc.unsqueeze(-2)).tril(0)
+    kk = torch.matmul(Kc, Kc.transpose(-1, -2))
+    M = (bc.unsqueeze(-1) * kk * ratio).tril(-1)
+    eye = torch.eye(C, dtype=torch.float32, device=q.device)
+    A = M + eye
+    qk = torch.matmul(Qc, Kc.transpose(-1, -2))
+    N = (qk * ratio).tril(0)
+
+    EK = E.unsqueeze(-1) * Kc
+    EQ = E.unsqueeze(-1) * Qc
+    wK = w.unsqueeze(-1) * Kc
+
+    # ---- forward state scan (recompute S0 per chunk + R) --------------------
+    S0s = torch.empty(B, H, nc, Dk, Dv, dtype=torch.float32, device=q.device)
+    Rs = torch.empty(B, H, nc, C, Dv, dtype=torch.float32, device=q.device)
+    state = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+    for i in range(nc):
+        S0s[:, :, i] = state
+        rhs = bc[:, :, i].unsqueeze(-1) * (
+            Vc[:, :, i] - torch.matmul(EK[:, :, i], state)
+        )
+        R = torch.linalg.solve_triangular(
+            A[:, :, i], rhs, upper=False, unitriangular=False
+        )
+        Rs[:, :, i] = R
+        state = Eend[:, :, i].unsqueeze(-1) * state + torch.matmul(
+            wK[:, :, i].transpose(-1, -2), R
+        )
+
+    # ---- reverse dState scan ------------------------------------------------
+    dRs = torch.empty_like(Rs)
+    drhss = torch.empty_like(Rs)
+    dS0s = torch.empty_like(S0s)
+    AT = A.transpose(-1, -2)
+    dS = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+    for i in range(nc - 1, -1, -1):
+        dR = torch.matmul(wK[:, :, i], dS) + torch.matmul(
+            N[:, :, i].transpose(-1, -2), dOc[:, :, i]
+        )
+        drhs = torch.linalg.solve_triangular(
+            AT[:, :, i], dR, upper=True, unitriangular=False
+        )
+        dT = -(bc[:, :, i].unsqueeze(-1) * drhs)
+        dS0 = (
+            Eend[:, :, i].unsqueeze(-1) * dS
+            + torch.matmul(EK[:, :, i].transpose(-1, -2), dT)
+            + torch.matmul(EQ[:, :, i].transpose(-1, -2), dOc[:, :, i])
+        )
+        dRs[:, :, i] = dR
+        drhss[:, :, i] = drhs
+        dS0s[:, :, i] = dS0  # dL/dS0 of THIS chunk == dS1 of the previous
+        dS = dS0
+
+    dS1s = torch.cat(
+        [dS0s[:, :, 1:], torch.zeros_like(dS0s[:, :, :1])], dim=2
+    )  # per-chunk dL/d(S1)
+
+    # ---- batched leaf gradients --------------------------------------------
+    dgc = torch.zeros_like(gc)
+
+    # S1 = Eend S0 + (w*K)^T R
+    dwK = torch.matmul(Rs, dS1s.transpose(-1, -2))           # (B,H,nc,C,Dk)
+    dK = dwK * w.unsqueeze(-1)
+    dw = (dwK * Kc).sum(-1)                                  # (B,H,nc,C)
+    dg_tot = (dS1s * S0s).sum((-1, -2)) * Eend.squeeze(-1)   # (B,H,nc)
+    dg_tot = dg_tot + (dw * w).sum(-1)
+    dgc = dgc - dw * w
+    dgc[..., -1] += dg_tot
+    # NOTE: dR from this path entered the scan already (wK dS term)
+
+    # O = (E*Q) S0 + N R
+    dEQ = torch.matmul(dOc, S0s.transpose(-1, -2))           # (B,H,nc,C,Dk)
+    dQ = dEQ * E.unsqueeze(-1)
+    dgc = dgc + (dEQ * Qc).sum(-1) * E
+    dN = torch.matmul(dOc, Rs.transpose(-1, -2)).tril(0)
+    # (dR's N^T dO term entered the scan)
+
+    # N = tril0(QK^T * ratio)
+    dqk = dN * ratio
+    dQ = dQ + torch.matmul(dqk, Kc)
+    dK = dK + torch.matmul(dqk.transpose(-1, -2), Qc)
+    dratio = dN * qk
+
+    # R = A^{-1} rhs: dA = -drhs R^T (restricted to M's strictly-lower part)
+    dM = -torch.matmul(drhss, Rs.transpose(-1, -2)).tril(-1)
+    # M = tril_{-1}(b_row KK^T ratio)
+    db = (dM * kk * ratio).sum(-1)
+    dkkr = dM * bc.unsqueeze(-1)
+    dkk = dkkr * ratio
+    dK = dK + torch.matmul(dkk, Kc) + torch.matmul(dkk.transpose(-1, -2), Kc)
+    dratio = dratio + dkkr * kk
+
+    # ratio_tj = exp(gc_t - gc_j), masked t >= j
+    rr = dratio * ratio
+    dgc = dgc + rr.sum(-1) - rr.sum(-2)
+
+    # rhs = b_row * P, P = V - (E*K) S0
+    P = Vc - torch.matmul(EK, S0s)
+    db = db + (drhss * P).sum(-1)
+    dT_all = -(bc.unsqueeze(-1) * drhss)
+    dV = bc.unsqueeze(-1) * drhss
+    dEK = torch.matmul(dT_all, S0s.transpose(-1, -2))
+    dK = dK + dEK * E.unsqueeze(-1)
+    dgc = dgc + (dEK * Kc).sum(-1) * E
+    # (dS0 contributions entered the scan)
+
+    # gc = cumsum(g): dg_t = sum_{tau >= t} dgc_tau  (within the chunk)
+    dg = dgc.flip(-1).cumsum(-1).flip(-1)
+
+    dq = dQ.reshape(B, H, Sp, Dk)[:, :, :S]
+    dk = dK.reshape(B, H, Sp, Dk)[:, :, :S]
+    dv = dV.reshape(B, H, Sp, Dv)[:, :, :S]
+    dbeta = db.reshape(B, H, Sp)[:, :, :S]
+    dg = dg.reshape(B, H, Sp)[:, :, :S]
+    return (
+        dq.to(q.dtype), dk.to(k.dtype), dv.to(dtype_v),
+        dbeta.to(beta.dtype), dg.to(decay_log.dtype),
+    )
 
 
 class LogSigmoidDecayGate(nn.Module):

@@ -318,3 +318,35 @@ def test_gdn_module_uses_kernel_and_trains_gpu():
     assert x.grad is not None and torch.isfinite(x.grad.float()).all()
     for n, p in m.named_parameters():
         assert p.grad is None or torch.isfinite(p.grad.float()).all(), n
+
+
+def test_gdn_derived_backward_matches_autograd():
+    """The hand-derived chunk backward (_chunk_gdn_backward) matches
+    autograd through the torch WY graph, including a padded tail chunk."""
+    import torch
+
+    from d9d_amd.module.block.attention.linear.gated_deltanet import (
+        _chunk_gated_delta_rule_torch,
+        _chunk_gdn_backward,
+    )
+
+    torch.manual_seed(3)
+    for B, H, S, Dk, Dv, C in [(2, 2, 96, 32, 48, 64), (1, 3, 64, 16, 16, 32)]:
+        q = torch.randn(B, H, S, Dk, requires_grad=True)
+        k = torch.nn.functional.normalize(
+            torch.randn(B, H, S, Dk), dim=-1
+        ).requires_grad_(True)
+        v = torch.randn(B, H, S, Dv, requires_grad=True)
+        b = torch.rand(B, H, S, requires_grad=True)
+        g = (-torch.rand(B, H, S) * 0.2).requires_grad_(True)
+        out = _chunk_gated_delta_rule_torch(q, k, v, b, g, C)
+        dout = torch.randn_like(out)
+        ref = torch.autograd.grad(out, (q, k, v, b, g), dout)
+        mine = _chunk_gdn_backward(
+            q.detach(), k.detach(), v.detach(), b.detach(), g.detach(), dout, C
+        )
+        for name, a, m in zip("qkvbg", ref, mine):
+            torch.testing.assert_close(
+                a, m, rtol=2e-4, atol=2e-4,
+                msg=lambda s, n=name: f"d{n}: {s}",
+            )
