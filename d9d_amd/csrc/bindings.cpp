@@ -44,7 +44,11 @@ std::vector<torch::Tensor> causal_conv_silu_bwd(torch::Tensor x, torch::Tensor w
 std::vector<torch::Tensor> gdn_chunk_fwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor beta,
                                          torch::Tensor decay_log,
-                                         bool return_state);
+                                         bool return_state, bool return_aux);
+std::vector<torch::Tensor> gdn_chunk_bwd_scan(torch::Tensor q, torch::Tensor k,
+                                              torch::Tensor dout,
+                                              torch::Tensor beta,
+                                              torch::Tensor decay_log);
 
 torch::Tensor cce_dlogits_(torch::Tensor logits, torch::Tensor lse, torch::Tensor targets,
                            torch::Tensor dl, c10::optional<torch::Tensor> dlse,
@@ -82,6 +86,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("causal_conv_silu_fwd", &causal_conv_silu_fwd, "fused causal depthwise conv + SiLU fwd");
   m.def("causal_conv_silu_bwd", &causal_conv_silu_bwd, "fused causal depthwise conv + SiLU bwd");
   m.def("gdn_chunk_fwd", &gdn_chunk_fwd, "chunked gated delta rule forward (GDN)");
+  m.def("gdn_chunk_bwd_scan", &gdn_chunk_bwd_scan,
+        "chunked gated delta rule backward reverse scan (GDN)");
   m.def("adamw_stochastic_bf16_multi_", &adamw_stochastic_bf16_multi_, "multi-tensor fused SR-AdamW");
   m.def("gmm", &gmm, "CDNA4 grouped GEMM (MoE experts)");
   m.def("gmm_nt", &gmm_nt, "CDNA4 grouped GEMM, weight (E,N,K)");

@@ -165,6 +165,8 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
     const float* __restrict__ decay_log, // (B, H, S)
     bf16_t* __restrict__ out,            // (B, H, S, D)
     float* __restrict__ final_state,     // (B, H, D, D) or nullptr
+    float* __restrict__ r_out,           // (B, H, S, D) fp32 or nullptr
+    float* __restrict__ s0_out,          // (B, H, nc, D, D) fp32 or nullptr
     int64_t BH, int64_t S) {
   constexpr int NTV = DVT / 16;        // v tiles per wave quarter
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
@@ -236,6 +238,15 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
       }
     }
     __syncthreads();
+
+    if (s0_out != nullptr) {
+      float* sp = s0_out + (bh * n_chunks + ch) * (int64_t)kGdnD * kGdnD;
+      for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
+        const int dk = i / DVT;
+        const int dv = i % DVT;
+        sp[dk * kGdnD + v0 + dv] = L.state[dk][dv];
+      }
+    }
 
     // ---- M = tril(beta * KK^T * ratio, -1) -----------------------------
     {
@@ -346,6 +357,16 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
       }
     }
     __syncthreads();
+
+    if (r_out != nullptr) {
+      for (int i = threadIdx.x; i < kGdnC * DVT; i += 256) {
+        const int t = i / DVT;
+        const int dv = i % DVT;
+        if (t < c_rows) {
+          r_out[(bh * S + s0 + t) * kGdnD + v0 + dv] = L.R[t][dv];
+        }
+      }
+    }
 
     // ---- N (bf16) and O = (e^{gc} Q) @ S + N @ R -----------------------
     for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
@@ -474,6 +495,304 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// chunked gated delta rule backward REVERSE SCAN, Dk = Dv = 64
+// ---------------------------------------------------------------------------
+//
+// Mirror of the forward chunk kernel for the sequential half of the
+// hand-derived backward (gated_deltanet.py _chunk_gdn_backward): one
+// workgroup per (b, h), chunks walked BACKWARD with dL/dState fp32 in LDS.
+// Per chunk (incoming dS = dL/d(S1) of this chunk):
+//   M, N      recomputed exactly as the forward kernel
+//   dR        = (w K) @ dS + N^T @ dO,     w_t = e^{gc_last - gc_t}
+//   drhs      : (I + M)^T drhs = dR        backward substitution, in place
+//   dS        = e^{gc_last} dS + (e^{gc} K)^T @ (-beta drhs)
+//                                + (e^{gc} Q)^T @ dO
+// Outputs drhs (B,H,S,D fp32) and the per-chunk dL/dS0 (B,H,nc,D,D fp32);
+// the non-sequential leaf gradients stay batched torch GEMMs.
+template <int DVT>
+struct GdnBwdLds {
+  float dS[kGdnD][DVT];           // dL/dState fp32 (v slice)
+  float M[kGdnC][kGdnC + 1];      // solve matrix (+1 pad)
+  float dx[kGdnC][DVT + 1];       // dR -> drhs in place (+1 pad)
+  bf16_t kc[kGdnC][kGdnD];        // K chunk
+  bf16_t qc[kGdnC][kGdnD];        // Q chunk
+  bf16_t doc[kGdnC][DVT];         // dO chunk slice
+  bf16_t doT[DVT][kGdnC];         // dO^T (B fragments of N^T @ dO)
+  bf16_t ks[kGdnC][kGdnD];        // scaled K / scaled Q scratch
+  bf16_t dsT[DVT][kGdnD];         // dS^T cast to bf16 (B fragments)
+  bf16_t nbT[kGdnC][kGdnC];       // N^T cast to bf16 (A fragments)
+  float gc[kGdnC];
+  float beta[kGdnC];
+};
+
+template <int DVT>
+__global__ __launch_bounds__(256, 1) void gdn_chunk_bwd_scan_kernel(
+    const bf16_t* __restrict__ q,    // (B, H, S, D)
+    const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ dout, // (B, H, S, D)
+    const float* __restrict__ beta,
+    const float* __restrict__ decay_log,
+    float* __restrict__ drhs_out,    // (B, H, S, D)
+    float* __restrict__ ds0_out,     // (B, H, nc, D, D)
+    int64_t BH, int64_t S) {
+  constexpr int NTV = DVT / 16;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  GdnBwdLds<DVT>& L = *reinterpret_cast<GdnBwdLds<DVT>*>(smem_raw);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t bh = blockIdx.x;
+  const int v0 = blockIdx.y * DVT;
+  if (bh >= BH) return;
+
+  const bf16_t* qp = q + bh * S * kGdnD;
+  const bf16_t* kp = k + bh * S * kGdnD;
+  const bf16_t* dop = dout + bh * S * kGdnD;
+  const float* bp = beta + bh * S;
+  const float* gp = decay_log + bh * S;
+
+  for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
+    (&L.dS[0][0])[i] = 0.f;
+  }
+  __syncthreads();
+
+  const float kLog2eG = 1.44269504f;
+  const int n_chunks = (int)((S + kGdnC - 1) / kGdnC);
+  for (int ch = n_chunks - 1; ch >= 0; --ch) {
+    const int s0 = ch * kGdnC;
+    const int c_rows = min((int)(S - (int64_t)s0), kGdnC);
+
+    // ---- stage chunk tiles (zero-pad), dS^T, decay scan ----------------
+    for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+      const int row = (i * 8) / kGdnD;
+      const int col = (i * 8) % kGdnD;
+      if (row < c_rows) {
+        const int64_t off = (int64_t)(s0 + row) * kGdnD + col;
+        *reinterpret_cast<bf16x8*>(&L.kc[row][col]) =
+            *reinterpret_cast<const bf16x8*>(kp + off);
+        *reinterpret_cast<bf16x8*>(&L.qc[row][col]) =
+            *reinterpret_cast<const bf16x8*>(qp + off);
+      } else {
+        bf16x8 z = {};
+        *reinterpret_cast<bf16x8*>(&L.kc[row][col]) = z;
+        *reinterpret_cast<bf16x8*>(&L.qc[row][col]) = z;
+      }
+    }
+    for (int i = threadIdx.x; i < kGdnC * DVT / 8; i += 256) {
+      const int row = (i * 8) / DVT;
+      const int col = (i * 8) % DVT;
+      if (row < c_rows) {
+        *reinterpret_cast<bf16x8*>(&L.doc[row][col]) =
+            *reinterpret_cast<const bf16x8*>(
+                dop + (int64_t)(s0 + row) * kGdnD + v0 + col);
+      } else {
+        bf16x8 z = {};
+        *reinterpret_cast<bf16x8*>(&L.doc[row][col]) = z;
+      }
+    }
+    if (threadIdx.x == 0) {
+      float run = 0.f;
+      for (int t = 0; t < kGdnC; ++t) {
+        if (t < c_rows) run += gp[s0 + t];
+        L.gc[t] = run;
+        L.beta[t] = (t < c_rows) ? bp[s0 + t] : 0.f;
+      }
+    }
+    for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
+      const int a = i / DVT;
+      const int b = i % DVT;
+      L.dsT[b][a] = (bf16_t)L.dS[a][b];
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < kGdnC * DVT; i += 256) {
+      const int t = i / DVT;
+      const int d = i % DVT;
+      L.doT[d][t] = L.doc[t][d];
+    }
+
+    // ---- M (fp32, solve matrix) and N^T (bf16) -------------------------
+    {
+      f32x4 accm[4], accn[4];
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        accm[nt] = {0.f, 0.f, 0.f, 0.f};
+        accn[nt] = {0.f, 0.f, 0.f, 0.f};
+      }
+      const int arow = wave * 16 + (lane & 15);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int d0 = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 ak = *reinterpret_cast<const bf16x8*>(&L.kc[arow][d0]);
+        const bf16x8 aq = *reinterpret_cast<const bf16x8*>(&L.qc[arow][d0]);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int brow = nt * 16 + (lane & 15);
+          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.kc[brow][d0]);
+          accm[nt] = mfma16gdn(ak, b, accm[nt]);
+          accn[nt] = mfma16gdn(aq, b, accn[nt]);
+        }
+      }
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int t = wave * 16 + (lane >> 4) * 4 + r;
+          const int j = nt * 16 + (lane & 15);
+          const float ratio =
+              __builtin_amdgcn_exp2f((L.gc[t] - L.gc[j]) * kLog2eG);
+          L.M[t][j] = (j < t) ? L.beta[t] * accm[nt][r] * ratio : 0.f;
+          L.nbT[j][t] = (bf16_t)((j <= t) ? accn[nt][r] * ratio : 0.f);
+        }
+      }
+    }
+    // wK scratch: ks[t][dk] = K * e^{gc_last - gc_t}
+    {
+      const float g_tot = L.gc[c_rows - 1];
+      for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+        const int row = (i * 8) / kGdnD;
+        const int col = (i * 8) % kGdnD;
+        const float sc =
+            (row < c_rows)
+                ? __builtin_amdgcn_exp2f((g_tot - L.gc[row]) * kLog2eG)
+                : 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          L.ks[row][col + j] = (bf16_t)((float)L.kc[row][col + j] * sc);
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- dR = wK @ dS + N^T @ dO  (fp32 into L.dx) ---------------------
+    {
+      f32x4 acc[NTV];
+#pragma unroll
+      for (int nt = 0; nt < NTV; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+      const int arow = wave * 16 + (lane & 15);
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        const int d0 = ks2 * 32 + (lane >> 4) * 8;
+        const bf16x8 aw = *reinterpret_cast<const bf16x8*>(&L.ks[arow][d0]);
+        const bf16x8 an = *reinterpret_cast<const bf16x8*>(&L.nbT[arow][d0]);
+#pragma unroll
+        for (int nt = 0; nt < NTV; ++nt) {
+          const int dv = nt * 16 + (lane & 15);
+          const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(&L.dsT[dv][d0]);
+          const bf16x8 b2 = *reinterpret_cast<const bf16x8*>(&L.doT[dv][d0]);
+          acc[nt] = mfma16gdn(aw, b1, acc[nt]);
+          acc[nt] = mfma16gdn(an, b2, acc[nt]);
+        }
+      }
+      // wait: nbT rows are A fragments over k = t (the q-row index) —
+      // dimensionally N^T is (c x c) with reduction over t' = dO rows; the
+      // A fragment of row j needs nbT[j][t' slice] which is exactly
+      // L.nbT[arow][d0] with arow = j. Correct as written.
+#pragma unroll
+      for (int nt = 0; nt < NTV; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int t = wave * 16 + (lane >> 4) * 4 + r;
+          const int dv = nt * 16 + (lane & 15);
+          L.dx[t][dv] = acc[nt][r];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- backward substitution: (I + M)^T drhs = dR, in place ----------
+    // drhs[t] = dR[t] - sum_{j > t} M[j][t] * drhs[j]; same 4-way j-split
+    // as the forward solve, t walked downward.
+    {
+      constexpr int CPW = DVT / 4;           // cols per wave
+      constexpr int JG = 64 / CPW;           // j groups
+      const int c = lane % CPW;
+      const int qj = lane / CPW;
+      const int col = wave * CPW + c;
+      const int jspan = kGdnC / JG;
+      for (int t = kGdnC - 2; t >= 0; --t) {
+        float part = 0.f;
+        const int jlo = max(t + 1, qj * jspan);
+        const int jhi = (qj + 1) * jspan;
+        for (int j = jlo; j < jhi; ++j) {
+          part += L.M[j][t] * L.dx[j][col];
+        }
+#pragma unroll
+        for (int off = CPW; off < 64; off <<= 1) {
+          part += __shfl_xor(part, off, 64);
+        }
+        if (qj == 0) {
+          L.dx[t][col] -= part;
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+      }
+    }
+    __syncthreads();
+
+    // drhs out (fp32)
+    for (int i = threadIdx.x; i < kGdnC * DVT; i += 256) {
+      const int t = i / DVT;
+      const int dv = i % DVT;
+      if (t < c_rows) {
+        drhs_out[(bh * S + s0 + t) * kGdnD + v0 + dv] = L.dx[t][dv];
+      }
+    }
+
+    // ---- dS = e^{g_tot} dS + EK^T @ (-beta drhs) + EQ^T @ dO -----------
+    {
+      const float g_tot = L.gc[c_rows - 1];
+      // ks = e^{gc} K
+      for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+        const int row = (i * 8) / kGdnD;
+        const int col = (i * 8) % kGdnD;
+        const float sc =
+            (row < c_rows) ? __builtin_amdgcn_exp2f(L.gc[row] * kLog2eG) : 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          L.ks[row][col + j] = (bf16_t)((float)L.kc[row][col + j] * sc);
+        }
+      }
+      __syncthreads();
+      const float e_tot = __builtin_amdgcn_exp2f(g_tot * kLog2eG);
+      for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
+        const int dk = i / DVT;
+        const int dv = i % DVT;
+        float acc = L.dS[dk][dv] * e_tot;
+        for (int t = 0; t < c_rows; ++t) {
+          acc -= (float)L.ks[t][dk] * L.beta[t] * L.dx[t][dv];
+        }
+        L.dS[dk][dv] = acc;
+      }
+      __syncthreads();
+      // ks = e^{gc} Q
+      for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+        const int row = (i * 8) / kGdnD;
+        const int col = (i * 8) % kGdnD;
+        const float sc =
+            (row < c_rows) ? __builtin_amdgcn_exp2f(L.gc[row] * kLog2eG) : 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          L.ks[row][col + j] = (bf16_t)((float)L.qc[row][col + j] * sc);
+        }
+      }
+      __syncthreads();
+      for (int i = threadIdx.x; i < kGdnD * DVT; i += 256) {
+        const int dk = i / DVT;
+        const int dv = i % DVT;
+        float acc = L.dS[dk][dv];
+        for (int t = 0; t < c_rows; ++t) {
+          acc += (float)L.ks[t][dk] * (float)L.doc[t][dv];
+        }
+        L.dS[dk][dv] = acc;
+        ds0_out[(bh * n_chunks + ch) * (int64_t)kGdnD * kGdnD +
+                dk * kGdnD + v0 + dv] = acc;
+      }
+    }
+    __syncthreads();
+  }
+}
+
 }  // namespace d9d
 
 // ---------------------------------------------------------------------------
@@ -522,9 +841,65 @@ std::vector<torch::Tensor> causal_conv_silu_bwd(
   return {dx, dw.to(w.scalar_type())};
 }
 
+std::vector<torch::Tensor> gdn_chunk_bwd_scan(
+    torch::Tensor q, torch::Tensor k, torch::Tensor dout,
+    torch::Tensor beta, torch::Tensor decay_log) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(dout.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.size(-1) == 64, "gdn_chunk_bwd_scan supports D = 64");
+  auto qc = q.contiguous();
+  auto kc = k.contiguous();
+  auto dc = dout.contiguous();
+  auto bc = beta.to(torch::kFloat32).contiguous();
+  auto gc = decay_log.to(torch::kFloat32).contiguous();
+  const int64_t B = qc.size(0), H = qc.size(1), S = qc.size(2);
+  const int64_t nc = (S + 63) / 64;
+  auto f32 = torch::dtype(torch::kFloat32).device(q.device());
+  auto drhs = torch::empty({B, H, S, 64}, f32);
+  auto ds0 = torch::empty({B, H, nc, 64, 64}, f32);
+  auto stream = at::hip::getCurrentHIPStream();
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&d9d::gdn_chunk_bwd_scan_kernel<64>),
+        hipFuncAttributeMaxDynamicSharedMemorySize,
+        (int)sizeof(d9d::GdnBwdLds<64>));
+    hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&d9d::gdn_chunk_bwd_scan_kernel<32>),
+        hipFuncAttributeMaxDynamicSharedMemorySize,
+        (int)sizeof(d9d::GdnBwdLds<32>));
+    attr_set = true;
+  }
+  // same Dv-split heuristic as the forward kernel: duplicate the K-side
+  // work when B*H alone would leave half the CUs idle
+  if (B * H * 2 <= 384) {
+    hipLaunchKernelGGL(d9d::gdn_chunk_bwd_scan_kernel<32>,
+                       dim3((unsigned)(B * H), 2), dim3(256),
+                       sizeof(d9d::GdnBwdLds<32>), stream,
+                       reinterpret_cast<const __bf16*>(qc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(kc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(dc.data_ptr()),
+                       bc.data_ptr<float>(), gc.data_ptr<float>(),
+                       drhs.data_ptr<float>(), ds0.data_ptr<float>(),
+                       B * H, S);
+  } else {
+    hipLaunchKernelGGL(d9d::gdn_chunk_bwd_scan_kernel<64>,
+                       dim3((unsigned)(B * H), 1), dim3(256),
+                       sizeof(d9d::GdnBwdLds<64>), stream,
+                       reinterpret_cast<const __bf16*>(qc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(kc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(dc.data_ptr()),
+                       bc.data_ptr<float>(), gc.data_ptr<float>(),
+                       drhs.data_ptr<float>(), ds0.data_ptr<float>(),
+                       B * H, S);
+  }
+  return {drhs, ds0};
+}
+
 std::vector<torch::Tensor> gdn_chunk_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
-    torch::Tensor beta, torch::Tensor decay_log, bool return_state) {
+    torch::Tensor beta, torch::Tensor decay_log, bool return_state,
+    bool return_aux) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.size(-1) == 64 && v.size(-1) == 64,
               "gdn_chunk_fwd supports Dk = Dv = 64");
@@ -541,6 +916,19 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
     fs = torch::empty({B, H, 64, 64},
                       torch::dtype(torch::kFloat32).device(q.device()));
     fs_ptr = fs.data_ptr<float>();
+  }
+  // aux outputs for the hand-derived backward: per-row solve results R and
+  // the per-chunk entry states S0
+  torch::Tensor r_aux, s0_aux;
+  float* r_ptr = nullptr;
+  float* s0_ptr = nullptr;
+  const int64_t nc = (S + 63) / 64;
+  if (return_aux) {
+    auto f32 = torch::dtype(torch::kFloat32).device(q.device());
+    r_aux = torch::empty({B, H, S, 64}, f32);
+    s0_aux = torch::empty({B, H, nc, 64, 64}, f32);
+    r_ptr = r_aux.data_ptr<float>();
+    s0_ptr = s0_aux.data_ptr<float>();
   }
   auto stream = at::hip::getCurrentHIPStream();
   static bool attr_set = false;
@@ -566,7 +954,7 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
                        reinterpret_cast<const __bf16*>(vc.data_ptr()),
                        bc.data_ptr<float>(), gc.data_ptr<float>(),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
-                       fs_ptr, B * H, S);
+                       fs_ptr, r_ptr, s0_ptr, B * H, S);
   } else {
     hipLaunchKernelGGL(d9d::gdn_chunk_fwd_kernel<64>,
                        dim3((unsigned)(B * H), 1), dim3(256),
@@ -576,8 +964,13 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
                        reinterpret_cast<const __bf16*>(vc.data_ptr()),
                        bc.data_ptr<float>(), gc.data_ptr<float>(),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
-                       fs_ptr, B * H, S);
+                       fs_ptr, r_ptr, s0_ptr, B * H, S);
   }
-  if (return_state) return {out, fs};
-  return {out};
+  std::vector<torch::Tensor> outs = {out};
+  if (return_state) outs.push_back(fs);
+  if (return_aux) {
+    outs.push_back(r_aux);
+    outs.push_back(s0_aux);
+  }
+  return outs;
 }

@@ -133,6 +133,7 @@ def _gdn_kernel_fwd(q, k, v, beta, decay_log):
         beta.contiguous(),
         decay_log.contiguous(),
         False,
+        False,
     )[0]
     return out.to(v.dtype)
 
@@ -290,56 +291,76 @@ def _chunk_gdn_backward(q, k, v, beta, decay_log, dout, chunk_size=64):
 
     ratio = torch.exp(gc.unsqueeze(-1) - gc.unsqueeze(-2)).tril(0)
     kk = torch.matmul(Kc, Kc.transpose(-1, -2))
-    M = (bc.unsqueeze(-1) * kk * ratio).tril(-1)
-    eye = torch.eye(C, dtype=torch.float32, device=q.device)
-    A = M + eye
     qk = torch.matmul(Qc, Kc.transpose(-1, -2))
-    N = (qk * ratio).tril(0)
 
     EK = E.unsqueeze(-1) * Kc
     EQ = E.unsqueeze(-1) * Qc
     wK = w.unsqueeze(-1) * Kc
 
-    # ---- forward state scan (recompute S0 per chunk + R) --------------------
-    S0s = torch.empty(B, H, nc, Dk, Dv, dtype=torch.float32, device=q.device)
-    Rs = torch.empty(B, H, nc, C, Dv, dtype=torch.float32, device=q.device)
-    state = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
-    for i in range(nc):
-        S0s[:, :, i] = state
-        rhs = bc[:, :, i].unsqueeze(-1) * (
-            Vc[:, :, i] - torch.matmul(EK[:, :, i], state)
-        )
-        R = torch.linalg.solve_triangular(
-            A[:, :, i], rhs, upper=False, unitriangular=False
-        )
-        Rs[:, :, i] = R
-        state = Eend[:, :, i].unsqueeze(-1) * state + torch.matmul(
-            wK[:, :, i].transpose(-1, -2), R
-        )
+    # ---- the two sequential scans: CDNA4 kernels when they apply, else
+    # torch loops (the scans are 92% of the backward at bench shape —
+    # profiles/gdn_kernels.md)
+    import os
 
-    # ---- reverse dState scan ------------------------------------------------
-    dRs = torch.empty_like(Rs)
-    drhss = torch.empty_like(Rs)
-    dS0s = torch.empty_like(S0s)
-    AT = A.transpose(-1, -2)
-    dS = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
-    for i in range(nc - 1, -1, -1):
-        dR = torch.matmul(wK[:, :, i], dS) + torch.matmul(
-            N[:, :, i].transpose(-1, -2), dOc[:, :, i]
+    use_scan_kernels = (
+        _use_gdn_kernels(q) and Dk == 64 and Dv == 64 and C == 64 and pad == 0
+        and os.environ.get("D9D_GDN_BWD_SCAN", "1") != "0"
+    )
+    if use_scan_kernels:
+        ext = get_ext()
+        q16 = q32.to(torch.bfloat16).contiguous()
+        k16 = k32.to(torch.bfloat16).contiguous()
+        _, r_flat, S0s = ext.gdn_chunk_fwd(
+            q16, k16, v32.to(torch.bfloat16).contiguous(),
+            b32.contiguous(), g32.contiguous(), False, True,
         )
-        drhs = torch.linalg.solve_triangular(
-            AT[:, :, i], dR, upper=True, unitriangular=False
+        Rs = r_flat.view(B, H, nc, C, Dv)
+        drhs_flat, dS0s = ext.gdn_chunk_bwd_scan(
+            q16, k16, do32.to(torch.bfloat16).contiguous(),
+            b32.contiguous(), g32.contiguous(),
         )
-        dT = -(bc[:, :, i].unsqueeze(-1) * drhs)
-        dS0 = (
-            Eend[:, :, i].unsqueeze(-1) * dS
-            + torch.matmul(EK[:, :, i].transpose(-1, -2), dT)
-            + torch.matmul(EQ[:, :, i].transpose(-1, -2), dOc[:, :, i])
-        )
-        dRs[:, :, i] = dR
-        drhss[:, :, i] = drhs
-        dS0s[:, :, i] = dS0  # dL/dS0 of THIS chunk == dS1 of the previous
-        dS = dS0
+        drhss = drhs_flat.view(B, H, nc, C, Dv)
+    else:
+        M = (bc.unsqueeze(-1) * kk * ratio).tril(-1)
+        eye = torch.eye(C, dtype=torch.float32, device=q.device)
+        A = M + eye
+        N = (qk * ratio).tril(0)
+        S0s = torch.empty(B, H, nc, Dk, Dv, dtype=torch.float32, device=q.device)
+        Rs = torch.empty(B, H, nc, C, Dv, dtype=torch.float32, device=q.device)
+        state = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+        for i in range(nc):
+            S0s[:, :, i] = state
+            rhs = bc[:, :, i].unsqueeze(-1) * (
+                Vc[:, :, i] - torch.matmul(EK[:, :, i], state)
+            )
+            R = torch.linalg.solve_triangular(
+                A[:, :, i], rhs, upper=False, unitriangular=False
+            )
+            Rs[:, :, i] = R
+            state = Eend[:, :, i].unsqueeze(-1) * state + torch.matmul(
+                wK[:, :, i].transpose(-1, -2), R
+            )
+
+        drhss = torch.empty_like(Rs)
+        dS0s = torch.empty_like(S0s)
+        AT = A.transpose(-1, -2)
+        dS = torch.zeros(B, H, Dk, Dv, dtype=torch.float32, device=q.device)
+        for i in range(nc - 1, -1, -1):
+            dR = torch.matmul(wK[:, :, i], dS) + torch.matmul(
+                N[:, :, i].transpose(-1, -2), dOc[:, :, i]
+            )
+            drhs = torch.linalg.solve_triangular(
+                AT[:, :, i], dR, upper=True, unitriangular=False
+            )
+            dT = -(bc[:, :, i].unsqueeze(-1) * drhs)
+            dS0 = (
+                Eend[:, :, i].unsqueeze(-1) * dS
+                + torch.matmul(EK[:, :, i].transpose(-1, -2), dT)
+                + torch.matmul(EQ[:, :, i].transpose(-1, -2), dOc[:, :, i])
+            )
+            drhss[:, :, i] = drhs
+            dS0s[:, :, i] = dS0  # dL/dS0 of THIS chunk == dS1 of the previous
+            dS = dS0
 
     dS1s = torch.cat(
         [dS0s[:, :, 1:], torch.zeros_like(dS0s[:, :, :1])], dim=2
