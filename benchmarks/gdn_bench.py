@@ -29,7 +29,7 @@ def bench(fn, name, flops):
 # + solve ~64^3/2 -> ~5.5 * 2*64^3 FLOP per chunk per (B,H)
 chunks = S // 64
 fl = B * H * chunks * 5.5 * 2 * 64**3
-bench(lambda: ext.gdn_chunk_fwd(q, k, v, beta, g, False, False), "gdn_chunk_fwd(kernel)", fl)
+bench(lambda: ext.gdn_chunk_fwd(q, k, v, beta, g, False, False, False), "gdn_chunk_fwd(kernel)", fl)
 bench(lambda: _chunk_gated_delta_rule_torch(q.float(), k.float(), v.float(), beta, g),
       "gdn_chunk_fwd(torch WY)", fl)
 

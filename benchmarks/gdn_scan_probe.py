@@ -24,6 +24,6 @@ def timed(fn, n=10):
     return (time.perf_counter() - t0) / n * 1e3
 
 
-print("fwd plain : %.2f ms" % timed(lambda: ext.gdn_chunk_fwd(q, k, v, beta, g, False, False)))
-print("fwd + aux : %.2f ms" % timed(lambda: ext.gdn_chunk_fwd(q, k, v, beta, g, False, True)))
+print("fwd plain : %.2f ms" % timed(lambda: ext.gdn_chunk_fwd(q, k, v, beta, g, False, False, False)))
+print("fwd + aux : %.2f ms" % timed(lambda: ext.gdn_chunk_fwd(q, k, v, beta, g, False, True, True)))
 print("bwd scan  : %.2f ms" % timed(lambda: ext.gdn_chunk_bwd_scan(q, k, do, beta, g)))

@@ -44,7 +44,8 @@ std::vector<torch::Tensor> causal_conv_silu_bwd(torch::Tensor x, torch::Tensor w
 std::vector<torch::Tensor> gdn_chunk_fwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor beta,
                                          torch::Tensor decay_log,
-                                         bool return_state, bool return_aux);
+                                         bool return_state, bool return_aux,
+                                         bool skip_out);
 std::vector<torch::Tensor> gdn_chunk_bwd_scan(torch::Tensor q, torch::Tensor k,
                                               torch::Tensor dout,
                                               torch::Tensor beta,

@@ -369,91 +369,96 @@ __global__ __launch_bounds__(256, 1) void gdn_chunk_fwd_kernel(
     }
 
     // ---- N (bf16) and O = (e^{gc} Q) @ S + N @ R -----------------------
-    for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
-      const int row = (i * 8) / kGdnD;
-      const int col = (i * 8) % kGdnD;
-      const float sc = __builtin_amdgcn_exp2f(L.gc[row] * kLog2eG);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        L.ks[row][col + j] = (bf16_t)((float)L.qc[row][col + j] * sc);
+    // (skipped when out == nullptr: the backward recompute only needs the
+    // solve results and chunk states)
+    if (out != nullptr) {
+      for (int i = threadIdx.x; i < kGdnC * kGdnD / 8; i += 256) {
+        const int row = (i * 8) / kGdnD;
+        const int col = (i * 8) % kGdnD;
+        const float sc = __builtin_amdgcn_exp2f(L.gc[row] * kLog2eG);
+  #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          L.ks[row][col + j] = (bf16_t)((float)L.qc[row][col + j] * sc);
+        }
       }
-    }
-    __syncthreads();
-    {
-      // N = tril(QK^T * ratio, 0) -> nb (bf16)
-      f32x4 acc[4];
-#pragma unroll
-      for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
-      const int arow = wave * 16 + (lane & 15);
-#pragma unroll
-      for (int ks2 = 0; ks2 < 2; ++ks2) {
-        const int d0 = ks2 * 32 + (lane >> 4) * 8;
-        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.qc[arow][d0]);
-#pragma unroll
+      __syncthreads();
+      {
+        // N = tril(QK^T * ratio, 0) -> nb (bf16)
+        f32x4 acc[4];
+  #pragma unroll
+        for (int nt = 0; nt < 4; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+        const int arow = wave * 16 + (lane & 15);
+  #pragma unroll
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
+          const int d0 = ks2 * 32 + (lane >> 4) * 8;
+          const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.qc[arow][d0]);
+  #pragma unroll
+          for (int nt = 0; nt < 4; ++nt) {
+            const int brow = nt * 16 + (lane & 15);
+            const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.kc[brow][d0]);
+            acc[nt] = mfma16gdn(a, b, acc[nt]);
+          }
+        }
+  #pragma unroll
         for (int nt = 0; nt < 4; ++nt) {
-          const int brow = nt * 16 + (lane & 15);
-          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.kc[brow][d0]);
-          acc[nt] = mfma16gdn(a, b, acc[nt]);
-        }
-      }
-#pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int t = wave * 16 + (lane >> 4) * 4 + r;
-          const int j = nt * 16 + (lane & 15);
-          float n = 0.f;
-          if (j <= t) {
-            n = acc[nt][r] *
-                __builtin_amdgcn_exp2f((L.gc[t] - L.gc[j]) * kLog2eG);
-          }
-          L.nb[t][j] = (bf16_t)n;
-        }
-      }
-    }
-    __syncthreads();
-    {
-      f32x4 acc[NTV];
-#pragma unroll
-      for (int nt = 0; nt < NTV; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
-      const int arow = wave * 16 + (lane & 15);
-      // (e^{gc} Q) @ S
-#pragma unroll
-      for (int ks2 = 0; ks2 < 2; ++ks2) {
-        const int d0 = ks2 * 32 + (lane >> 4) * 8;
-        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.ks[arow][d0]);
-#pragma unroll
-        for (int nt = 0; nt < NTV; ++nt) {
-          const int vcol = nt * 16 + (lane & 15);
-          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.sbT[vcol][d0]);
-          acc[nt] = mfma16gdn(a, b, acc[nt]);
-        }
-      }
-      // + N @ R  (k dim = chunk rows j, 64 wide)
-#pragma unroll
-      for (int ks2 = 0; ks2 < 2; ++ks2) {
-        const int j0 = ks2 * 32 + (lane >> 4) * 8;
-        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.nb[arow][j0]);
-#pragma unroll
-        for (int nt = 0; nt < NTV; ++nt) {
-          const int vcol = nt * 16 + (lane & 15);
-          const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.rbT[vcol][j0]);
-          acc[nt] = mfma16gdn(a, b, acc[nt]);
-        }
-      }
-#pragma unroll
-      for (int nt = 0; nt < NTV; ++nt) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int t = wave * 16 + (lane >> 4) * 4 + r;
-          const int vcol = nt * 16 + (lane & 15);
-          if (t < c_rows) {
-            op[(int64_t)(s0 + t) * kGdnD + v0 + vcol] = (bf16_t)acc[nt][r];
+  #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int t = wave * 16 + (lane >> 4) * 4 + r;
+            const int j = nt * 16 + (lane & 15);
+            float n = 0.f;
+            if (j <= t) {
+              n = acc[nt][r] *
+                  __builtin_amdgcn_exp2f((L.gc[t] - L.gc[j]) * kLog2eG);
+            }
+            L.nb[t][j] = (bf16_t)n;
           }
         }
       }
+      __syncthreads();
+      {
+        f32x4 acc[NTV];
+  #pragma unroll
+        for (int nt = 0; nt < NTV; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+        const int arow = wave * 16 + (lane & 15);
+        // (e^{gc} Q) @ S
+  #pragma unroll
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
+          const int d0 = ks2 * 32 + (lane >> 4) * 8;
+          const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.ks[arow][d0]);
+  #pragma unroll
+          for (int nt = 0; nt < NTV; ++nt) {
+            const int vcol = nt * 16 + (lane & 15);
+            const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.sbT[vcol][d0]);
+            acc[nt] = mfma16gdn(a, b, acc[nt]);
+          }
+        }
+        // + N @ R  (k dim = chunk rows j, 64 wide)
+  #pragma unroll
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
+          const int j0 = ks2 * 32 + (lane >> 4) * 8;
+          const bf16x8 a = *reinterpret_cast<const bf16x8*>(&L.nb[arow][j0]);
+  #pragma unroll
+          for (int nt = 0; nt < NTV; ++nt) {
+            const int vcol = nt * 16 + (lane & 15);
+            const bf16x8 b = *reinterpret_cast<const bf16x8*>(&L.rbT[vcol][j0]);
+            acc[nt] = mfma16gdn(a, b, acc[nt]);
+          }
+        }
+  #pragma unroll
+        for (int nt = 0; nt < NTV; ++nt) {
+  #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int t = wave * 16 + (lane >> 4) * 4 + r;
+            const int vcol = nt * 16 + (lane & 15);
+            if (t < c_rows) {
+              op[(int64_t)(s0 + t) * kGdnD + v0 + vcol] = (bf16_t)acc[nt][r];
+            }
+          }
+        }
+      }
+      __syncthreads();
+
     }
-    __syncthreads();
 
     // ---- state = e^{gc_last} S + (K e^{gc_last - gc})^T @ R ------------
     {
@@ -899,7 +904,7 @@ std::vector<torch::Tensor> gdn_chunk_bwd_scan(
 std::vector<torch::Tensor> gdn_chunk_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor beta, torch::Tensor decay_log, bool return_state,
-    bool return_aux) {
+    bool return_aux, bool skip_out) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.size(-1) == 64 && v.size(-1) == 64,
               "gdn_chunk_fwd supports Dk = Dv = 64");
@@ -909,7 +914,9 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
   auto bc = beta.to(torch::kFloat32).contiguous();
   auto gc = decay_log.to(torch::kFloat32).contiguous();
   const int64_t B = qc.size(0), H = qc.size(1), S = qc.size(2);
-  auto out = torch::empty_like(vc);
+  auto out = skip_out ? torch::Tensor() : torch::empty_like(vc);
+  __bf16* out_ptr =
+      skip_out ? nullptr : reinterpret_cast<__bf16*>(out.data_ptr());
   torch::Tensor fs;
   float* fs_ptr = nullptr;
   if (return_state) {
@@ -953,8 +960,7 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
                        reinterpret_cast<const __bf16*>(kc.data_ptr()),
                        reinterpret_cast<const __bf16*>(vc.data_ptr()),
                        bc.data_ptr<float>(), gc.data_ptr<float>(),
-                       reinterpret_cast<__bf16*>(out.data_ptr()),
-                       fs_ptr, r_ptr, s0_ptr, B * H, S);
+                       out_ptr, fs_ptr, r_ptr, s0_ptr, B * H, S);
   } else {
     hipLaunchKernelGGL(d9d::gdn_chunk_fwd_kernel<64>,
                        dim3((unsigned)(B * H), 1), dim3(256),
@@ -963,8 +969,7 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
                        reinterpret_cast<const __bf16*>(kc.data_ptr()),
                        reinterpret_cast<const __bf16*>(vc.data_ptr()),
                        bc.data_ptr<float>(), gc.data_ptr<float>(),
-                       reinterpret_cast<__bf16*>(out.data_ptr()),
-                       fs_ptr, r_ptr, s0_ptr, B * H, S);
+                       out_ptr, fs_ptr, r_ptr, s0_ptr, B * H, S);
   }
   std::vector<torch::Tensor> outs = {out};
   if (return_state) outs.push_back(fs);

@@ -134,6 +134,7 @@ def _gdn_kernel_fwd(q, k, v, beta, decay_log):
         decay_log.contiguous(),
         False,
         False,
+        False,
     )[0]
     return out.to(v.dtype)
 
@@ -310,9 +311,9 @@ def _chunk_gdn_backward(q, k, v, beta, decay_log, dout, chunk_size=64):
         ext = get_ext()
         q16 = q32.to(torch.bfloat16).contiguous()
         k16 = k32.to(torch.bfloat16).contiguous()
-        _, r_flat, S0s = ext.gdn_chunk_fwd(
+        r_flat, S0s = ext.gdn_chunk_fwd(
             q16, k16, v32.to(torch.bfloat16).contiguous(),
-            b32.contiguous(), g32.contiguous(), False, True,
+            b32.contiguous(), g32.contiguous(), False, True, True,
         )
         Rs = r_flat.view(B, H, nc, C, Dv)
         drhs_flat, dS0s = ext.gdn_chunk_bwd_scan(

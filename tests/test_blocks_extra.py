@@ -239,7 +239,7 @@ def test_gdn_chunk_kernel_vs_oracle_gpu():
     beta = torch.rand(B, H, S, device="cuda")
     g = -torch.rand(B, H, S, device="cuda") * 0.2
 
-    out = get_ext().gdn_chunk_fwd(q, k, v, beta, g, False, False)[0]
+    out = get_ext().gdn_chunk_fwd(q, k, v, beta, g, False, False, False)[0]
     oracle = step_gated_delta_rule(q.float(), k.float(), v.float(), beta, g)
     wy = _chunk_gated_delta_rule_torch(q.float(), k.float(), v.float(), beta, g)
     torch.testing.assert_close(wy, oracle.to(wy.dtype), rtol=1e-3, atol=1e-3)
@@ -261,7 +261,7 @@ def test_gdn_chunk_kernel_final_state_gpu():
     beta = torch.rand(B, H, S, device="cuda")
     g = -torch.rand(B, H, S, device="cuda") * 0.1
 
-    out, fs = get_ext().gdn_chunk_fwd(q, k, v, beta, g, True, False)
+    out, fs = get_ext().gdn_chunk_fwd(q, k, v, beta, g, True, False, False)
     assert fs.shape == (B, H, D, D)
     assert torch.isfinite(fs).all()
 
