@@ -1066,13 +1066,22 @@ __global__ __launch_bounds__(512, 1) void gmm_nn_8phase_kernel(
 
 // 8-phase glds + tr16 wgrad: db[e] = a[rows_e]^T @ g[rows_e], out (E, K, N).
 // Same block mapping as gmm_db_kernel (expert-order round-robin over
-// (e, k-tile, n-tile), 256x256 C tiles, ragged row loop in 64-row chunks)
-// but the staging is the deep pipeline: BOTH operand chunks stage ROW-major
-// via global_load_lds into [rows/4][cols/16][4][16] blocked images and BOTH
+// (e, k-tile, n-tile), ragged row loop in 64-row chunks) but the staging is
+// the deep pipeline: BOTH operand chunks stage ROW-major via
+// global_load_lds into [rows/4][cols/16][4][16] blocked images and BOTH
 // MFMA fragments (reduction = rows) come from ds_read_b64_tr_b16 pairs.
 // Rows past the expert's end read from a 16-byte zero buffer (glds cannot
 // mask; a per-lane address select costs one v_cndmask, no branch), so pad
 // rows contribute exact zeros to the row-sum.
+//
+// Templated on BOTH output tile dims (256 or 192): the wgrad's output is
+// (K x N) and both can quantize badly — the bench's down-projection is
+// K = 576 (2.25 tiles of 256 -> 33% wasted MFMA), its gate_up N = 1152
+// (4.5 tiles). 192 divides both exactly. Phase count = BK8/64 (k subtiles
+// walked two per phase), chunk counts scale with the image sizes, and the
+// staging issue order (all a(rt+1) before any g(rt+2)) keeps the counted
+// vmcnt waits exact.
+template <int BK8, int BN8>
 __global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
     const bf16_t* __restrict__ a,   // (T, K)
     const bf16_t* __restrict__ g,   // (T, N)
@@ -1081,26 +1090,33 @@ __global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
     const int* __restrict__ expert_order,
     const bf16_t* __restrict__ zero16,  // >= 8 zero bf16
     int E, int K, int N, int kt_tiles, int nt_tiles) {
-  constexpr int SLOT = 64 * 1024;  // A image 32 KB + B image 32 KB
+  constexpr int KI = BK8 / 32;          // 16-wide k subtiles per wave
+  constexpr int NJ = BN8 / 64;          // 16-wide n subtiles per wave
+  constexpr int P = KI / 2;             // phases (2 k subtiles each)
+  constexpr int ACH = BK8 / 64;         // 8 KB staging chunks per a tile
+  constexpr int GCH = BN8 / 64;         // 8 KB staging chunks per g tile
+  constexpr int ABYTES = 64 * BK8 * 2;  // blocked-image sizes
+  constexpr int GBYTES = 64 * BN8 * 2;
+  constexpr int SLOT = ABYTES + GBYTES;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = wave >> 2;  // k-half (128 rows of db)
-  const int wn = wave & 3;   // n-quarter (64 cols)
+  const int wm = wave >> 2;  // k-half (BK8/2 rows of db)
+  const int wn = wave & 3;   // n-quarter (BN8/4 cols)
 
   const int wid = blockIdx.x;
   const int tiles_per_e = kt_tiles * nt_tiles;
   const int e = expert_order[wid / tiles_per_e];
-  const int k0 = ((wid % tiles_per_e) / nt_tiles) * 256;
-  const int n0 = (wid % nt_tiles) * 256;
+  const int k0 = ((wid % tiles_per_e) / nt_tiles) * BK8;
+  const int n0 = (wid % nt_tiles) * BN8;
   const int r_start = row_off[e];
   const int r_end = row_off[e + 1];
 
-  f32x4 acc[8][4];
+  f32x4 acc[KI][NJ];
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
+  for (int i = 0; i < KI; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   if (r_start >= r_end) {
     // empty expert: this block zero-fills its tile so the host can hand us
@@ -1108,13 +1124,13 @@ __global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
     // — ~0.9 GB per wgrad at the bench shape, ~1.5% of the whole step)
     bf16_t* dbe = db + (int64_t)e * K * N;
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
+    for (int i = 0; i < KI; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
+      for (int j = 0; j < NJ; ++j)
 #pragma unroll
         for (int r2 = 0; r2 < 4; ++r2) {
-          const int krow = k0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r2;
-          const int ncol = n0 + wn * 64 + j * 16 + (lane & 15);
+          const int krow = k0 + wm * (BK8 / 2) + i * 16 + (lane >> 4) * 4 + r2;
+          const int ncol = n0 + wn * (BN8 / 4) + j * 16 + (lane & 15);
           if (krow < K && ncol < N) dbe[(int64_t)krow * N + ncol] = (bf16_t)0;
         }
     return;
@@ -1122,21 +1138,20 @@ __global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
 
   const int n_rtiles = (r_end - r_start + 63) / 64;
 
-  // Stage one 8 KB chunk (16 rows x 256 cols) of operand A or B for row
-  // tile rt: 1 glds per thread. Image element e (bf16):
-  // rb = e/1024, cb = (e%1024)/64, rrow = (e%64)/16, col8 = e%16.
+  // Stage one 8 KB chunk of an operand tile (width W) for row tile rt:
+  // 1 glds per thread. l = byte offset in the blocked image; the inverse
+  // mapping recovers (row, col) for the source address.
   auto stage = [&](const bf16_t* src, int64_t stride, int c0, int cmax,
-                   int rt, int chunk, int img_off) {
+                   int rt, int chunk, int img_off, int W) {
     char* base = smem + (size_t)(rt & 1) * SLOT + img_off;
     const int l = chunk * 8192 + wave * 1024 + lane * 16;  // byte offset
-    const int el = l >> 1;
-    const int rb = el >> 10;
-    const int rem = el & 1023;
-    const int cb = rem >> 6;
-    const int rrow = (rem & 63) >> 4;
-    const int ccol = rem & 15;  // 0 or 8
-    const int row = r_start + rt * 64 + rb * 4 + rrow;
-    const int64_t col = min((int64_t)(c0 + cb * 16 + ccol), (int64_t)cmax - 8);
+    const int blk = l >> 7;                // 128-B [4][16] block index
+    const int inb = l & 127;
+    const int rb = blk / (W / 16);
+    const int cb = blk % (W / 16);
+    const int row = r_start + rt * 64 + rb * 4 + (inb >> 5);
+    const int col8 = cb * 16 + (((inb >> 4) & 1) * 8);
+    const int64_t col = min((int64_t)(c0 + col8), (int64_t)cmax - 8);
     const char* sp =
         (row < r_end)
             ? reinterpret_cast<const char*>(src) + (row * stride + col) * 2
@@ -1145,16 +1160,20 @@ __global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
         reinterpret_cast<const uint32_t*>(sp),
         reinterpret_cast<uint32_t*>(base + l), 16, 0, 0);
   };
-  auto stage_a = [&](int rt, int chunk) { stage(a, K, k0, K, rt, chunk, 0); };
+  auto stage_a = [&](int rt, int chunk) {
+    stage(a, K, k0, K, rt, chunk, 0, BK8);
+  };
   auto stage_g = [&](int rt, int chunk) {
-    stage(g, N, n0, N, rt, chunk, 32 * 1024);
+    stage(g, N, n0, N, rt, chunk, ABYTES, BN8);
   };
 
   // tr16 fragment: reduction slice red0..red0+8 at 16 cols [ct*16, ct*16+16)
-  auto frag = [&](const char* img, int red0, int ct) -> bf16x8 {
+  auto frag = [&](const char* img, int W, int red0, int ct) -> bf16x8 {
     const int rb0 = red0 >> 2;
-    const char* blk0 = img + ((size_t)rb0 * 16 + ct) * 128 + (lane & 15) * 8;
-    const char* blk1 = img + ((size_t)(rb0 + 1) * 16 + ct) * 128 + (lane & 15) * 8;
+    const char* blk0 =
+        img + ((size_t)rb0 * (W / 16) + ct) * 128 + (lane & 15) * 8;
+    const char* blk1 =
+        img + ((size_t)(rb0 + 1) * (W / 16) + ct) * 128 + (lane & 15) * 8;
     bf16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
         (__attribute__((address_space(3))) bf16x4*)(blk0));
     bf16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
@@ -1165,62 +1184,69 @@ __global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
     return f;
   };
 
-  // Prologue mirrors the steady-state issue order.
+  // Prologue mirrors the steady-state issue order (a before g).
 #pragma unroll
-  for (int c = 0; c < 4; ++c) stage_g(0, c);
-  stage_a(0, 0); stage_a(0, 1); stage_a(0, 2); stage_a(0, 3);
+  for (int c = 0; c < GCH; ++c) stage_g(0, c);
+#pragma unroll
+  for (int c = 0; c < ACH; ++c) stage_a(0, c);
   if (n_rtiles > 1) {
 #pragma unroll
-    for (int c = 0; c < 4; ++c) stage_g(1, c);
+    for (int c = 0; c < GCH; ++c) stage_g(1, c);
   }
 
   for (int rt = 0; rt < n_rtiles; ++rt) {
     const char* a_img = smem + (size_t)(rt & 1) * SLOT;
-    const char* g_img = a_img + 32 * 1024;
+    const char* g_img = a_img + ABYTES;
 
     if (rt + 1 < n_rtiles) {
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GCH) : "memory");
     } else {
-      // no B(rt+1) staged: the newest outstanding loads are THIS row
-      // tile's A chunks — drain fully
+      // no g(rt+1) staged: the newest outstanding loads are THIS row
+      // tile's a chunks — drain fully
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
 
-    bf16x8 b_frag[4][2];
+    bf16x8 b_frag[NJ][2];
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < P; ++p) {
       bf16x8 a_frag[2][2];
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         const int red0 = ks * 32 + (lane >> 4) * 8;
 #pragma unroll
         for (int il = 0; il < 2; ++il) {
-          const int kct = (wm * 128 + (2 * p + il) * 16) >> 4;
-          a_frag[il][ks] = frag(a_img, red0, kct);
+          const int kct = (wm * (BK8 / 2) + (2 * p + il) * 16) >> 4;
+          a_frag[il][ks] = frag(a_img, BK8, red0, kct);
         }
         if (p == 0) {
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            const int nct = (wn * 64 + j * 16) >> 4;
-            b_frag[j][ks] = frag(g_img, red0, nct);
+          for (int j = 0; j < NJ; ++j) {
+            const int nct = (wn * (BN8 / 4) + j * 16) >> 4;
+            b_frag[j][ks] = frag(g_img, BN8, red0, nct);
           }
         }
       }
-      if (p == 0) {
-        if (rt + 1 < n_rtiles) { stage_a(rt + 1, 0); stage_a(rt + 1, 1); }
-      } else if (p == 1) {
-        if (rt + 1 < n_rtiles) { stage_a(rt + 1, 2); stage_a(rt + 1, 3); }
-        if (rt + 2 < n_rtiles) { stage_g(rt + 2, 0); stage_g(rt + 2, 1); }
-      } else if (p == 2) {
-        if (rt + 2 < n_rtiles) { stage_g(rt + 2, 2); stage_g(rt + 2, 3); }
+      // Stage a(rt+1) then g(rt+2), spread evenly over the phases; a fully
+      // before g keeps the top-of-iteration vmcnt(GCH) count exact.
+      {
+        constexpr int TOT = ACH + GCH;
+        constexpr int PER = (TOT + P - 1) / P;
+#pragma unroll
+        for (int s = p * PER; s < (p + 1) * PER && s < TOT; ++s) {
+          if (s < ACH) {
+            if (rt + 1 < n_rtiles) stage_a(rt + 1, s);
+          } else {
+            if (rt + 2 < n_rtiles) stage_g(rt + 2, s - ACH);
+          }
+        }
       }
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int il = 0; il < 2; ++il)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NJ; ++j)
 #pragma unroll
           for (int ks = 0; ks < 2; ++ks)
             acc[2 * p + il][j] =
@@ -1232,13 +1258,13 @@ __global__ __launch_bounds__(512, 1) void gmm_db_8phase_kernel(
 
   bf16_t* db_e = db + (int64_t)e * K * N;
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < KI; ++i) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < NJ; ++j) {
 #pragma unroll
       for (int r2 = 0; r2 < 4; ++r2) {
-        const int krow = k0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r2;
-        const int ncol = n0 + wn * 64 + j * 16 + (lane & 15);
+        const int krow = k0 + wm * (BK8 / 2) + i * 16 + (lane >> 4) * 4 + r2;
+        const int ncol = n0 + wn * (BN8 / 4) + j * 16 + (lane & 15);
         if (krow < K && ncol < N) {
           db_e[(int64_t)krow * N + ncol] = (bf16_t)acc[i][j][r2];
         }
@@ -1645,7 +1671,12 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
                                   /*descending=*/true).to(torch::kInt32);
   auto expert_order = order_cpu.to(a.device(), /*non_blocking=*/true);
 
-  const int nt_tiles = (N + 255) / 256, kt_tiles = (K + 255) / 256;
+  // wgrad tile choice: prefer the width that divides the dim exactly
+  // (the bench's down-projection K = 576 loses 33% of its MFMA to a
+  // 256-tile ceiling; 192 divides it exactly — likewise N = 1152)
+  const int BK = (K % 256 != 0 && K % 192 == 0) ? 192 : 256;
+  const int BN = (N % 256 != 0 && N % 192 == 0) ? 192 : 256;
+  const int nt_tiles = (N + BN - 1) / BN, kt_tiles = (K + BK - 1) / BK;
   const dim3 grid(nt_tiles * kt_tiles * E);
   auto stream = at::hip::getCurrentHIPStream();
 
@@ -1674,29 +1705,42 @@ torch::Tensor gmm_db(torch::Tensor a, torch::Tensor g, torch::Tensor batch_sizes
     }
     static bool attr_set_db = false;
     if (!attr_set_db) {
-      hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&d9d::gmm_db_8phase_kernel),
-          hipFuncAttributeMaxDynamicSharedMemorySize, 128 * 1024);
+#define DB_ATTR(BK8, BN8)                                                     \
+      hipFuncSetAttribute(                                                    \
+          reinterpret_cast<const void*>(                                      \
+              &d9d::gmm_db_8phase_kernel<BK8, BN8>),                          \
+          hipFuncAttributeMaxDynamicSharedMemorySize,                         \
+          2 * 64 * (BK8 + BN8) * 2)
+      DB_ATTR(256, 256); DB_ATTR(256, 192); DB_ATTR(192, 256); DB_ATTR(192, 192);
+#undef DB_ATTR
       attr_set_db = true;
     }
-    hipLaunchKernelGGL(d9d::gmm_db_8phase_kernel, grid, dim3(512),
-                       128 * 1024, stream,
-                       reinterpret_cast<const __bf16*>(a.data_ptr()),
-                       reinterpret_cast<const __bf16*>(g.data_ptr()),
-                       reinterpret_cast<__bf16*>(db.data_ptr()),
-                       row_off.data_ptr<int>(), expert_order.data_ptr<int>(),
-                       reinterpret_cast<const __bf16*>(z.data_ptr()),
-                       E, K, N, kt_tiles, nt_tiles);
+#define DB_LAUNCH(BK8, BN8)                                                   \
+    hipLaunchKernelGGL((d9d::gmm_db_8phase_kernel<BK8, BN8>), grid,           \
+                       dim3(512), 2 * 64 * (BK8 + BN8) * 2, stream,           \
+                       reinterpret_cast<const __bf16*>(a.data_ptr()),         \
+                       reinterpret_cast<const __bf16*>(g.data_ptr()),         \
+                       reinterpret_cast<__bf16*>(db.data_ptr()),              \
+                       row_off.data_ptr<int>(), expert_order.data_ptr<int>(), \
+                       reinterpret_cast<const __bf16*>(z.data_ptr()),         \
+                       E, K, N, kt_tiles, nt_tiles)
+    if (BK == 256 && BN == 256) DB_LAUNCH(256, 256);
+    else if (BK == 256) DB_LAUNCH(256, 192);
+    else if (BN == 256) DB_LAUNCH(192, 256);
+    else DB_LAUNCH(192, 192);
+#undef DB_LAUNCH
     return db;
   }
 
   db.zero_();  // legacy kernel early-returns on empty experts
+  const int nt256 = (N + 255) / 256, kt256 = (K + 255) / 256;
+  const dim3 grid256(nt256 * kt256 * E);
   const size_t smem = (2 * 256 * 64) * sizeof(__bf16);
-  hipLaunchKernelGGL(d9d::gmm_db_kernel, grid, dim3(512), smem, stream,
+  hipLaunchKernelGGL(d9d::gmm_db_kernel, grid256, dim3(512), smem, stream,
                      reinterpret_cast<const __bf16*>(a.data_ptr()),
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
                      reinterpret_cast<__bf16*>(db.data_ptr()),
                      row_off.data_ptr<int>(), expert_order.data_ptr<int>(),
-                     E, K, N, kt_tiles, nt_tiles);
+                     E, K, N, kt256, nt256);
   return db;
 }
