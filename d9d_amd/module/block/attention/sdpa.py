@@ -40,12 +40,20 @@ class EagerSdpaConfig(BaseModel):
     backend: Literal["eager"] = "eager"
 
 
+class TorchSdpaConfig(BaseModel):
+    """torch.nn.functional.scaled_dot_product_attention (reference
+    TorchSdpa, sdpa/impl/torch_sdpa.py): no sink or q_offset support, but
+    exercises torch's own fused path (AOTriton/MIOpen on ROCm)."""
+
+    backend: Literal["torch"] = "torch"
+
+
 class AutoSdpaConfig(BaseModel):
     backend: Literal["auto"] = "auto"
 
 
 SdpaBackendConfig = Annotated[
-    Union[Cdna4FlashSdpaConfig, EagerSdpaConfig, AutoSdpaConfig],
+    Union[Cdna4FlashSdpaConfig, EagerSdpaConfig, TorchSdpaConfig, AutoSdpaConfig],
     Field(discriminator="backend"),
 ]
 
@@ -69,6 +77,37 @@ def _eager(q, k, v, *, causal=True, softmax_scale=None,
     return out
 
 
+def _torch_sdpa(q, k, v, *, causal=True, softmax_scale=None,
+                window_size=(-1, -1), sinks=None, q_offset=0):
+    if sinks is not None or q_offset != 0:
+        raise ValueError("torch sdpa backend does not support sinks/q_offset")
+    import torch.nn.functional as F
+
+    # (B, S, H, D) -> (B, H, S, D)
+    qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))
+    kw = {}
+    if window_size != (-1, -1):
+        # explicit mask: sliding window + causal
+        S, Skv = q.shape[1], k.shape[1]
+        pos_q = torch.arange(S, device=q.device).unsqueeze(1)
+        pos_k = torch.arange(Skv, device=q.device).unsqueeze(0)
+        off = Skv - S
+        mask = torch.ones(S, Skv, dtype=torch.bool, device=q.device)
+        if causal:
+            mask &= pos_k <= pos_q + off
+        if window_size[0] >= 0:
+            mask &= pos_k >= pos_q + off - window_size[0]
+        out = F.scaled_dot_product_attention(
+            qt, kt, vt, attn_mask=mask, scale=softmax_scale, enable_gqa=True
+        )
+    else:
+        out = F.scaled_dot_product_attention(
+            qt, kt, vt, is_causal=causal, scale=softmax_scale, enable_gqa=True,
+            **kw,
+        )
+    return out.transpose(1, 2)
+
+
 def build_sdpa_backend(config: SdpaBackendConfig | None = None) -> SdpaBackend:
     name = None
     if config is not None and config.backend != "auto":
@@ -77,6 +116,8 @@ def build_sdpa_backend(config: SdpaBackendConfig | None = None) -> SdpaBackend:
         name = os.environ[SDPA_ENV_VAR]
     if name == "eager":
         return _eager
+    if name == "torch":
+        return _torch_sdpa
     # auto and cdna4_flash both resolve to the flash op (which itself keeps
     # the eager path for CPU and for features the kernel lacks).
     return _cdna4_flash

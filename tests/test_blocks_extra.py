@@ -350,3 +350,27 @@ def test_gdn_derived_backward_matches_autograd():
                 a, m, rtol=2e-4, atol=2e-4,
                 msg=lambda s, n=name: f"d{n}: {s}",
             )
+
+
+def test_torch_sdpa_backend_matches_eager():
+    """The torch-SDPA backend (reference TorchSdpa analog) matches the
+    eager reference, including GQA and sliding window."""
+    import torch
+
+    from d9d_amd.module.block.attention.sdpa import (
+        TorchSdpaConfig,
+        build_sdpa_backend,
+        _eager,
+    )
+
+    torch.manual_seed(5)
+    be = build_sdpa_backend(TorchSdpaConfig())
+    B, S, Hq, Hkv, D = 2, 33, 4, 2, 16
+    q = torch.randn(B, S, Hq, D)
+    k = torch.randn(B, S, Hkv, D)
+    v = torch.randn(B, S, Hkv, D)
+    for kw in [dict(causal=True), dict(causal=False),
+               dict(causal=True, window_size=(7, 0))]:
+        out = be(q, k, v, **kw)
+        ref = _eager(q, k, v, **kw)
+        torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-4)
