@@ -107,6 +107,21 @@ class CastDType(_SingleTensor):
         return t.to(self.dtype)
 
 
+class SelectChild(ModelStateMapper):
+    """Select one tensor out of a stacked/batched input along dim 0
+    (reference leaf SelectChild: e.g. pick expert i's slice from an
+    (E, ...) tensor into its own key)."""
+
+    def __init__(self, src: str, dst: str, index: int, dim: int = 0) -> None:
+        self.src, self.dst, self.index, self.dim = src, dst, index, dim
+
+    def state_dependency_groups(self) -> list[StateGroup]:
+        return [StateGroup.of([self.src], [self.dst])]
+
+    def apply_group(self, group, tensors):
+        return {self.dst: tensors[self.src].select(self.dim, self.index).contiguous()}
+
+
 class StackTensors(ModelStateMapper):
     def __init__(self, srcs: list[str], dst: str, dim: int = 0) -> None:
         self.srcs, self.dst, self.dim = list(srcs), dst, dim

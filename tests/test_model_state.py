@@ -262,3 +262,17 @@ def test_trainer_export_reimport_ws2():
 
     with tempfile.TemporaryDirectory() as d:
         assert all(run_distributed(_trainer_export_hf_reimport, world_size=2, args=(d,)))
+
+
+def test_select_child_leaf():
+    """SelectChild picks one dim-0 slice into its own key (reference leaf)."""
+    import torch
+
+    from d9d_amd.model_state import SelectChild, Parallel
+
+    t = torch.arange(24, dtype=torch.float32).view(3, 8)
+    m = Parallel(*[SelectChild("experts", f"expert_{i}.w", i) for i in range(3)])
+    out = m.apply({"experts": t})
+    assert set(out) == {"expert_0.w", "expert_1.w", "expert_2.w"}
+    for i in range(3):
+        torch.testing.assert_close(out[f"expert_{i}.w"], t[i])
