@@ -1281,6 +1281,42 @@ __global__ void mfma_selfcheck_kernel(
   }
 }
 
+// Fused fp32 -> bf16 cast of the three backward accumulators in ONE launch
+// (the three at::native casts measured ~2.6 TB/s and ~0.7 ms per backward at
+// bench shape). Sizes are multiples of 8 (B*S*H*D), so 8-float groups never
+// straddle a buffer boundary.
+__global__ void cast3_f32_bf16_kernel(
+    const float* __restrict__ a, const float* __restrict__ b,
+    const float* __restrict__ c,
+    ushort* __restrict__ oa, ushort* __restrict__ ob, ushort* __restrict__ oc,
+    int64_t na, int64_t nb, int64_t nc) {
+  const int64_t total = na + nb + nc;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+       i < total; i += stride) {
+    const float* src;
+    ushort* dst;
+    int64_t off = i;
+    if (off < na) {
+      src = a; dst = oa;
+    } else if (off < na + nb) {
+      src = b; dst = ob; off -= na;
+    } else {
+      src = c; dst = oc; off -= na + nb;
+    }
+    f32x4 lo = *reinterpret_cast<const f32x4*>(src + off);
+    f32x4 hi = *reinterpret_cast<const f32x4*>(src + off + 4);
+    ushort2_t p0, p1;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      p0.s[j] = bf16_bits((bf16_t)lo[j]);
+      p1.s[j] = bf16_bits((bf16_t)hi[j]);
+    }
+    *reinterpret_cast<uint64_t*>(dst + off) = p0.u;
+    *reinterpret_cast<uint64_t*>(dst + off + 4) = p1.u;
+  }
+}
+
 }  // namespace d9d
 
 // ---------------------------------------------------------------------------
@@ -1540,9 +1576,28 @@ std::vector<torch::Tensor> flash_attn_bwd(
   }
 #undef LAUNCH_BWD
 
-  auto dq = dq32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
-  auto dk = dk32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
-  auto dv = dv32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
+  torch::Tensor dq, dk, dv;
+  if (D == D_pad) {
+    // one fused cast launch for all three accumulators
+    auto bf = q.options().dtype(torch::kBFloat16);
+    dq = torch::empty(dq32.sizes(), bf);
+    dk = torch::empty(dk32.sizes(), bf);
+    dv = torch::empty(dv32.sizes(), bf);
+    const int64_t na = dq32.numel(), nb = dk32.numel(), nc = dv32.numel();
+    const int64_t groups = (na + nb + nc) / 8;
+    const int blocks = (int)std::min<int64_t>((groups + 255) / 256, 16384);
+    hipLaunchKernelGGL(d9d::cast3_f32_bf16_kernel, dim3(blocks), dim3(256), 0,
+                       stream, dq32.data_ptr<float>(), dk32.data_ptr<float>(),
+                       dv32.data_ptr<float>(),
+                       reinterpret_cast<ushort*>(dq.data_ptr()),
+                       reinterpret_cast<ushort*>(dk.data_ptr()),
+                       reinterpret_cast<ushort*>(dv.data_ptr()),
+                       na, nb, nc);
+  } else {
+    dq = dq32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
+    dk = dk32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
+    dv = dv32.narrow(-1, 0, D).to(torch::kBFloat16).contiguous();
+  }
   if (varlen) {
     dq = dq.squeeze(0);
     dk = dk.squeeze(0);
