@@ -13,6 +13,7 @@ std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor a, torch::Tensor b, torch:
 
 // stochastic.hip
 void copy_fp32_to_bf16_stochastic_(torch::Tensor dst, torch::Tensor src, int64_t seed);
+void add_bf16_into_f32_(torch::Tensor acc, torch::Tensor x);
 void adamw_stochastic_bf16_(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
                             double lr, double beta1, double beta2, double eps, double weight_decay,
                             int64_t step, int64_t seed);
@@ -107,6 +108,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_mul_packed_fwd", &silu_mul_packed_fwd, "packed SwiGLU fwd");
   m.def("silu_mul_packed_bwd", &silu_mul_packed_bwd, "packed SwiGLU bwd");
   m.def("silu_mul_bwd", &silu_mul_bwd, "fused silu(a)*b backward");
+  m.def("add_bf16_into_f32_", &add_bf16_into_f32_,
+        "acc(fp32) += x(bf16), single pass");
   m.def("copy_fp32_to_bf16_stochastic_", &copy_fp32_to_bf16_stochastic_,
         "stochastic-rounding fp32->bf16 copy");
   m.def("adamw_stochastic_bf16_", &adamw_stochastic_bf16_,
