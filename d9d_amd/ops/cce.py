@@ -17,6 +17,7 @@ differentiate through it. `softcap` applies z = cap * tanh(logit / cap)
 before the softmax (reference kernel/cce/main.py:59).
 """
 
+import os
 from dataclasses import dataclass
 
 from ._ext import get_ext, has_ext
@@ -26,11 +27,16 @@ import torch.distributed as dist
 
 LM_IGNORE_INDEX = -100
 # Backward row-chunking bounds the transient (rows, V) dlogits buffer.
-# 2048 rows (~0.6 GB at V=152k) measured fastest on MI355X: larger chunks
-# (1.2 GB+) push the transient past the caching allocator's reuse size and
-# the resulting hipMalloc/hipFree per microbatch serializes the device
-# (measured 2x END-TO-END regression at 4096 rows).
-_ROW_CHUNK = 2048
+# With the buffer ALLOCATED ONCE per backward and matmul'd into via out=
+# (per-chunk torch.matmul allocations past the caching allocator's reuse
+# size caused a hipMalloc/hipFree storm per microbatch — a measured 2x
+# end-to-end regression at 4096 rows in round 1), taller chunks mean
+# fewer, fatter GEMMs over the vocab. In-box sweep on the bench model
+# (V=152k): 2048 -> 130.5k tok/s, 4096 -> 132.8k, 8192 -> 134.9k,
+# 16384 -> 136.0k, 32768 (one 10-GB chunk) -> 130.6k. Default 16384
+# (~5 GB transient at V=152k); D9D_CCE_ROW_CHUNK overrides for
+# memory-tight configurations.
+_ROW_CHUNK = int(os.environ.get("D9D_CCE_ROW_CHUNK", "16384"))
 
 
 @dataclass
@@ -140,10 +146,19 @@ class _LinearCrossEntropyFunction(torch.autograd.Function):
         )
         chunk = _ROW_CHUNK
         single = chunk >= T
+        logits_buf = (
+            torch.empty(min(chunk, T), V, dtype=e.dtype, device=e.device)
+            if bf16_fast and T > 0
+            else None
+        )
         for s in range(0, T, chunk):
             sl = slice(s, min(s + _ROW_CHUNK, T))
             e_chunk = e[sl]
-            logits = torch.matmul(e_chunk, c.t().to(e_chunk.dtype))  # (Tc, V)
+            if logits_buf is not None:
+                rows = e_chunk.shape[0]
+                logits = torch.mm(e_chunk, c.t(), out=logits_buf[:rows])
+            else:
+                logits = torch.matmul(e_chunk, c.t().to(e_chunk.dtype))  # (Tc, V)
             local_targets = targets[sl] - vocab_start
             in_shard = (local_targets >= 0) & (local_targets < V) & (~ignored[sl])
             safe = local_targets.clamp(0, V - 1)
