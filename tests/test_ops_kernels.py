@@ -452,3 +452,38 @@ def test_cce_lse_grad_kernel_path_gpu():
 
     torch.testing.assert_close(e.grad.float(), e32.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(c.grad.float(), c32.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.gpu
+def test_cce_backward_multi_chunk_gpu():
+    """The chunked backward path (preallocated logits buffer + fused
+    fp32 classifier-grad accumulation) matches the single-chunk result."""
+    import torch
+
+    import d9d_amd.ops.cce as cce_mod
+    from d9d_amd.ops.cce import linear_cross_entropy
+
+    torch.manual_seed(21)
+    T, H, V = 700, 128, 1024
+    e0 = (torch.randn(T, H, device="cuda") * 0.5).bfloat16()
+    c0 = (torch.randn(V, H, device="cuda") * 0.1).bfloat16()
+    tg = torch.randint(0, V, (T,), device="cuda")
+    tg[::13] = -100
+
+    def run(chunk):
+        old = cce_mod._ROW_CHUNK
+        cce_mod._ROW_CHUNK = chunk
+        try:
+            e = e0.clone().requires_grad_(True)
+            c = c0.clone().requires_grad_(True)
+            loss = linear_cross_entropy(e, c, tg, reduction="mean", filter_eps=None)
+            loss.backward()
+            return loss.detach(), e.grad.clone(), c.grad.clone()
+        finally:
+            cce_mod._ROW_CHUNK = old
+
+    loss_1, de_1, dc_1 = run(4096)   # single chunk (T < chunk)
+    loss_m, de_m, dc_m = run(256)    # 3 chunks: buffer reuse + fused acc
+    torch.testing.assert_close(loss_1, loss_m, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(de_1, de_m, rtol=2e-2, atol=2e-3)
+    torch.testing.assert_close(dc_1, dc_m, rtol=2e-2, atol=2e-3)
