@@ -206,3 +206,18 @@ def test_profiling_wrapper_smoke(tmp_path):
         prof.step()
     prof.close()
     assert any(tmp_path.glob("trace_r0_*.json.tar.gz")), "no trace written"
+
+
+def test_tuned_gemm_table_loader_cpu_noop():
+    """The TunableOp table loader is a safe no-op without a GPU and the
+    shipped table exists with validator headers."""
+    import os
+
+    from d9d_amd.ops.tunable import load_tuned_gemm_table, _TABLE
+
+    assert load_tuned_gemm_table() is False  # CPU host: no-op
+    path = os.path.abspath(_TABLE)
+    assert os.path.exists(path)
+    with open(path) as f:
+        head = f.read(200)
+    assert head.startswith("Validator")
