@@ -40,7 +40,12 @@ def test_gmm_forward_parity(E, total, K, N):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("E,total,K,N", [(8, 4096, 768, 576), (16, 999, 100, 72)])
+@pytest.mark.parametrize("E,total,K,N", [
+    (8, 4096, 768, 576),    # BK 256, BN 192
+    (8, 4096, 576, 768),    # BK 192, BN 256 (the bench's down-projection wgrad)
+    (4, 2048, 1152, 576),   # BK 192, BN 192
+    (16, 999, 100, 72),     # legacy fallback (K % 8 != 0)
+])
 def test_gmm_db_parity(E, total, K, N):
     device = torch.device("cuda")
     sizes = _rand_sizes(E, total)
