@@ -971,7 +971,8 @@ std::vector<torch::Tensor> gdn_chunk_fwd(
                        bc.data_ptr<float>(), gc.data_ptr<float>(),
                        out_ptr, fs_ptr, r_ptr, s0_ptr, B * H, S);
   }
-  std::vector<torch::Tensor> outs = {out};
+  std::vector<torch::Tensor> outs;
+  if (!skip_out) outs.push_back(out);
   if (return_state) outs.push_back(fs);
   if (return_aux) {
     outs.push_back(r_aux);
